@@ -1,0 +1,147 @@
+"""Command-line driver, argument-compatible with the reference
+(``gmm num_clusters infile outfile [target_num_clusters]``,
+gaussian.cu:1111-1178) plus runtime flags for every compile-time knob.
+
+Multi-GPU: run under ``torchrun --nproc-per-node N`` (one process per GPU,
+RCCL over xGMI) or with ``--gpus N`` which self-launches torchrun-style
+workers on one node.
+"""
+from __future__ import annotations
+
+import argparse
+import sys
+
+import numpy as np
+import torch
+
+from .engine import build_engine
+from .parallel import dist as pdist
+from .utils import io as gio
+from .utils.config import GmmConfig, MAX_CLUSTERS
+from .utils.timers import Profile
+
+
+def make_parser() -> argparse.ArgumentParser:
+    p = argparse.ArgumentParser(
+        prog="gmm",
+        description="MI355X-native GMM EM clustering with MDL order reduction",
+    )
+    p.add_argument("num_clusters", type=int,
+                   help="The number of starting clusters")
+    p.add_argument("infile", help="FCS data file (.bin binary or CSV)")
+    p.add_argument("outfile", help="Clustering results output file stem")
+    p.add_argument("target_num_clusters", type=int, nargs="?", default=0,
+                   help="A desired number of clusters. Must be less than or "
+                        "equal to num_clusters")
+    p.add_argument("--min-iters", type=int, default=100)
+    p.add_argument("--max-iters", type=int, default=100)
+    p.add_argument("--covariance-dynamic-range", type=float, default=1e3)
+    p.add_argument("--diag-only", action="store_true")
+    p.add_argument("--no-bug-compat", dest="bug_compat", action="store_false",
+                   help="use corrected ln determinant in the merge path "
+                        "instead of the reference's log10 quirk")
+    p.add_argument("--no-output", dest="enable_output", action="store_false",
+                   help="skip .summary/.results content (files still created "
+                        "like the reference)")
+    p.add_argument("--no-results", dest="write_results", action="store_false",
+                   help="write only the .summary file")
+    p.add_argument("--print", dest="enable_print", action="store_true")
+    p.add_argument("--estep-dtype", choices=["fp32", "bf16"], default="fp32")
+    p.add_argument("--no-center", dest="center_data", action="store_false")
+    p.add_argument("--device", default=None,
+                   help="cpu | cuda (default: cuda when available)")
+    p.add_argument("--profile", action="store_true",
+                   help="print the per-GPU timing report (gaussian.cu:967)")
+    return p
+
+
+def config_from_args(args) -> GmmConfig:
+    cfg = GmmConfig(
+        num_clusters=args.num_clusters,
+        target_num_clusters=args.target_num_clusters,
+        min_iters=args.min_iters, max_iters=args.max_iters,
+        covariance_dynamic_range=args.covariance_dynamic_range,
+        diag_only=args.diag_only, bug_compat=args.bug_compat,
+        enable_print=args.enable_print, enable_output=args.enable_output,
+        estep_dtype=args.estep_dtype, center_data=args.center_data,
+    )
+    cfg.validate()
+    return cfg
+
+
+def run_clustering(data: np.ndarray, cfg: GmmConfig, outfile: str,
+                   device: str, write_results: bool = True,
+                   profile_report: bool = False) -> dict:
+    """Full pipeline on already-initialized process group. Returns a result
+    dict (rank 0) with num_clusters / rissanen / likelihood."""
+    rank, local_rank, world = pdist.rank(), 0, pdist.world_size()
+    if device == "cuda":
+        local_rank = torch.cuda.current_device()
+    prof = Profile(device)
+    engine = build_engine(data, cfg, device=device, profile=prof)
+    result = engine.sweep()
+
+    # de-center the saved model for output
+    out_state = result.state.to("cpu")
+    out_state.means += engine.center.cpu().unsqueeze(0)
+
+    w_shard = engine.recompute_memberships(result.state)
+    memberships = engine.gather_memberships(w_shard)
+
+    if rank == 0:
+        gio.write_summary(outfile + ".summary", out_state, cfg.enable_output)
+        if cfg.enable_output and write_results and memberships is not None:
+            gio.write_results(outfile + ".results", data, memberships)
+
+    if profile_report:
+        print(engine.profile.report(rank, local_rank))
+
+    return {
+        "num_clusters": result.num_clusters,
+        "rissanen": result.min_rissanen,
+        "likelihood": result.likelihood,
+        "rissanen_by_k": result.rissanen_by_k,
+        "state": out_state,
+        "memberships": memberships,
+    }
+
+
+def main(argv=None) -> int:
+    args = make_parser().parse_args(argv)
+    if not (1 <= args.num_clusters <= MAX_CLUSTERS):
+        print("Invalid number of starting clusters\n")
+        return 1
+    if args.target_num_clusters > args.num_clusters:
+        print("target_num_clusters must be less than equal to num_clusters\n")
+        return 4
+    try:
+        cfg = config_from_args(args)
+    except ValueError as e:
+        print(str(e))
+        return 1
+
+    device = args.device
+    if device is None:
+        device = "cuda" if torch.cuda.is_available() else "cpu"
+
+    rank, local_rank, world = pdist.init_process_group()
+    try:
+        try:
+            data = gio.read_data(args.infile)
+        except (OSError, ValueError) as e:
+            print(f"Invalid infile. ({e})\n")
+            return 2
+        result = run_clustering(
+            data, cfg, args.outfile, device,
+            write_results=args.write_results, profile_report=args.profile,
+        )
+        if rank == 0 and args.enable_print:
+            print(f"Ideal clusters: {result['num_clusters']} "
+                  f"(rissanen {result['rissanen']:.4f})")
+    finally:
+        pdist.destroy()
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

@@ -1,0 +1,332 @@
+"""Device-resident EM + MDL order-reduction engine.
+
+MI355X-native redesign of the reference driver (gaussian.cu:128-1106):
+
+ - one process per GPU; events sharded across ranks (remainder handled
+   correctly — fixes SURVEY §2.6 #4);
+ - all sufficient statistics live on the device; ONE fused in-place RCCL
+   all-reduce of {N, mean numerators, second moments} per EM iteration
+   (+ one scalar likelihood reduce) replaces the reference's 7 host
+   round-trips and 4 staged MPI_Allreduce calls;
+ - covariance is finalized as (S - N mu mu^T + G*avgvar*I)/N, algebraically
+   identical to the reference's centered per-event sums because S is
+   mean-independent — this is what makes the single fused reduce possible;
+ - the data is internally centered by the global mean (translation
+   invariant math; output means get the center added back), eliminating
+   the fp32 cancellation that the uncentered second-moment would suffer;
+ - MDL outer loop: rank 0 merges on host with reference-faithful fp32
+   math (including the log10/ln determinant quirk under bug_compat),
+   broadcasts one fused parameter vector (vs 7 MPI_Bcasts).
+"""
+from __future__ import annotations
+
+import dataclasses
+import math
+
+import numpy as np
+import torch
+
+from .models.merge import HostClusters, reduce_order
+from .models.seed import seed_means_host, seed_state
+from .models.state import GmmState
+from .ops import functional as F
+from .parallel import dist as pdist
+from .utils.config import GmmConfig, em_epsilon, rissanen_score
+from .utils.timers import Profile
+
+
+@dataclasses.dataclass
+class SweepResult:
+    """Best-MDL model of the sweep (rank 0) plus bookkeeping."""
+    state: GmmState                  # saved best params (host copies OK)
+    num_clusters: int
+    min_rissanen: float
+    likelihood: float
+    rissanen_by_k: dict[int, float]
+
+
+class EmEngine:
+    """EM on one event shard; collectives across the world."""
+
+    def __init__(self, data_shard_by_event: torch.Tensor, config: GmmConfig,
+                 num_events_total: int, center: torch.Tensor,
+                 seed_means: torch.Tensor, var_per_dim: torch.Tensor,
+                 device: torch.device | str = "cpu",
+                 profile: Profile | None = None):
+        self.cfg = config
+        self.device = torch.device(device)
+        self.rank = pdist.rank()
+        self.world = pdist.world_size()
+        self.n_total = num_events_total
+        self.profile = profile or Profile(self.device)
+
+        shard = data_shard_by_event.to(self.device, torch.float32)
+        self.n_shard = int(shard.shape[0])
+        self.d = int(shard.shape[1])
+        self.center = center.to(self.device, torch.float32)       # [D]
+        shard = shard - self.center.unsqueeze(0)
+        # dimension-major resident copy for the quadratic-form kernels
+        self.x = shard.T.contiguous()                              # [D, n]
+        # augmented transpose for the fused N+means GEMM
+        ones = torch.ones(self.n_shard, 1, dtype=torch.float32, device=self.device)
+        self.x_aug_t = torch.cat([shard, ones], dim=1).contiguous()  # [n, D+1]
+        if config.estep_dtype == "bf16" and self.device.type == "cuda":
+            self.x_estep: torch.Tensor = self.x.to(torch.bfloat16)
+        else:
+            self.x_estep = self.x
+
+        k0 = config.num_clusters
+        self.state = GmmState.empty(k0, self.d, self.device)
+        seed_state(
+            self.state, seed_means - center.unsqueeze(0), var_per_dim,
+            num_events_total, config.covariance_dynamic_range,
+        )
+        # constants for the seeded R=I state (constants_kernel after seeding,
+        # gaussian.cu:404)
+        with self.profile.time("constants"):
+            rinv, const = F.constants(self.state.R, config.diag_only)
+            self.state.Rinv.copy_(rinv)
+            self.state.constant.copy_(const)
+        self.profile.count("constants")
+
+        # membership / logw buffer, cluster-major [K, n_shard]
+        self.w = torch.empty(k0, self.n_shard, dtype=torch.float32,
+                             device=self.device)
+        # fused reduce buffer: [K + K*D + K*D*D]
+        self._stats = torch.empty(k0 * (1 + self.d + self.d * self.d),
+                                  dtype=torch.float32, device=self.device)
+        self.epsilon = em_epsilon(self.d, num_events_total)
+        self.likelihood = 0.0
+
+    # ------------------------------------------------------------------ EM
+
+    def _estep(self, k: int) -> torch.Tensor:
+        """E-step into self.w[:k]; returns shard-partial likelihood tensor."""
+        st = self.state.shrink(k)
+        with self.profile.time("e_step"):
+            logw = F.estep_logw(
+                self.x_estep, st.means, st.Rinv, st.constant, st.pi,
+                self.cfg.diag_only, out=self.w[:k],
+            )
+            w, lik = F.estep_posteriors(logw)
+        self.profile.count("regroup")
+        return lik
+
+    def _reduce_likelihood(self, lik_part: torch.Tensor) -> float:
+        with self.profile.time("comm"):
+            t = lik_part.reshape(1)
+            pdist.all_reduce_(t)
+        return float(t.item())
+
+    def _mstep(self, k: int) -> None:
+        """M-step: stats, fused all-reduce, finalize params + constants."""
+        d = self.d
+        st = self.state.shrink(k)
+        buf = self._stats[: k * (1 + d + d * d)]
+        n_v = buf[:k]
+        mn_v = buf[k: k + k * d].view(k, d)
+        s_v = buf[k + k * d:].view(k, d, d)
+        with self.profile.time("m_step"):
+            n_c, mean_num = F.mstep_n_means(self.x_aug_t, self.w[:k])
+            n_v.copy_(n_c)
+            mn_v.copy_(mean_num)
+            F.mstep_covariance_s(self.x, self.w[:k], out=s_v)
+        with self.profile.time("comm"):
+            pdist.all_reduce_(buf)
+        with self.profile.time("m_step"):
+            st.N.copy_(n_v)
+            st.means.copy_(F.finalize_means(n_v, mn_v))
+            st.R.copy_(F.finalize_covariance(
+                n_v, st.means, s_v, st.avgvar, self.world, self.cfg.diag_only,
+            ))
+        self.profile.count("params")
+        with self.profile.time("constants"):
+            rinv, const = F.constants(st.R, self.cfg.diag_only)
+            st.Rinv.copy_(rinv)
+            st.constant.copy_(const)
+            st.pi.copy_(F.compute_pi(st.N))
+        self.profile.count("constants")
+
+    def run_em(self, k: int) -> float:
+        """Full EM at fixed K (the inner loop of gaussian.cu:479-755).
+
+        Returns the final global log-likelihood.
+        """
+        cfg = self.cfg
+        lik = self._reduce_likelihood(self._estep(k))
+        iters = 0
+        change = self.epsilon * 2
+        while iters < cfg.min_iters or (
+            abs(change) > self.epsilon and iters < cfg.max_iters
+        ):
+            old_lik = lik
+            self._mstep(k)
+            lik = self._reduce_likelihood(self._estep(k))
+            change = lik - old_lik
+            iters += 1
+        self.likelihood = lik
+        return lik
+
+    def em_iteration(self, k: int) -> None:
+        """One benchmark step: M-step + constants + E-step + reduces.
+
+        (Requires a prior _estep so self.w holds posteriors.)
+        """
+        self._mstep(k)
+        self._reduce_likelihood(self._estep(k))
+
+    # ------------------------------------------------- MDL sweep / merging
+
+    def _host_clusters(self, k: int) -> HostClusters:
+        st = self.state.shrink(k)
+        return HostClusters(
+            N=st.N.cpu().numpy().copy(),
+            pi=st.pi.cpu().numpy().copy(),
+            constant=st.constant.cpu().numpy().copy(),
+            avgvar=st.avgvar.cpu().numpy().copy(),
+            means=st.means.cpu().numpy().copy(),
+            R=st.R.cpu().numpy().copy(),
+            Rinv=st.Rinv.cpu().numpy().copy(),
+        )
+
+    def _load_host_clusters(self, hc: HostClusters, k: int) -> None:
+        st = self.state.shrink(k)
+        st.N.copy_(torch.from_numpy(np.ascontiguousarray(hc.N[:k])))
+        st.pi.copy_(torch.from_numpy(np.ascontiguousarray(hc.pi[:k])))
+        st.constant.copy_(torch.from_numpy(np.ascontiguousarray(hc.constant[:k])))
+        st.avgvar.copy_(torch.from_numpy(np.ascontiguousarray(hc.avgvar[:k])))
+        st.means.copy_(torch.from_numpy(np.ascontiguousarray(hc.means[:k])))
+        st.R.copy_(torch.from_numpy(np.ascontiguousarray(hc.R[:k])))
+        st.Rinv.copy_(torch.from_numpy(np.ascontiguousarray(hc.Rinv[:k])))
+
+    def sweep(self) -> SweepResult:
+        """MDL model-order sweep K0 -> stop (gaussian.cu:479-960)."""
+        cfg = self.cfg
+        k = cfg.num_clusters
+        stop = cfg.stop_number
+        best_state: GmmState | None = None
+        best_k = k
+        min_rissanen = float("inf")
+        best_lik = 0.0
+        riss_by_k: dict[int, float] = {}
+
+        while k >= stop:
+            lik = self.run_em(k)
+            riss = rissanen_score(lik, k, self.d, self.n_total)
+            riss_by_k[k] = riss
+            if cfg.enable_print and self.rank == 0:
+                print(f"\nRissanen Score: {riss:e}")
+
+            save = (
+                k == cfg.num_clusters
+                or (riss < min_rissanen and cfg.target_num_clusters == 0)
+                or k == cfg.target_num_clusters
+            )
+            if save:
+                min_rissanen = riss
+                best_k = k
+                best_lik = lik
+                best_state = self.state.shrink(k).clone(with_memberships=False)
+
+            if k <= stop:
+                break
+
+            # ---- order reduction on rank 0, fused param broadcast
+            with self.profile.time("reduce"):
+                new_k = k
+                if self.rank == 0:
+                    hc = self._host_clusters(k)
+                    new_k, _, _ = reduce_order(hc, bug_compat=cfg.bug_compat)
+                if self.world > 1:
+                    nk = torch.tensor([new_k], dtype=torch.int64)
+                    pdist.broadcast_(nk, src=0)
+                    new_k = int(nk.item())
+                if self.rank == 0:
+                    self._load_host_clusters(hc.truncated(new_k), new_k)
+                if self.world > 1:
+                    with self.profile.time("comm"):
+                        vec = self.state.shrink(new_k).param_vector().contiguous()
+                        pdist.broadcast_(vec, src=0)
+                    if self.rank != 0:
+                        self.state.shrink(new_k).load_param_vector(vec)
+            self.profile.count("reduce")
+            k = new_k
+
+        assert best_state is not None
+        if cfg.enable_print and self.rank == 0:
+            print(f"\nFinal rissanen score was: {min_rissanen:f}, "
+                  f"with {best_k} clusters.")
+        return SweepResult(
+            state=best_state, num_clusters=best_k,
+            min_rissanen=min_rissanen, likelihood=best_lik,
+            rissanen_by_k=riss_by_k,
+        )
+
+    # ------------------------------------------------------------- output
+
+    def recompute_memberships(self, saved: GmmState) -> torch.Tensor:
+        """Shard posteriors [K, n_shard] for the saved best model.
+
+        Memberships never leave their shard during the sweep; for the
+        .results file they are regenerated from the saved parameters (the
+        final E-step of the best K is a pure function of those parameters)
+        and gathered once.
+        """
+        k = saved.num_clusters
+        st = saved.to(self.device)
+        logw = F.estep_logw(
+            self.x_estep, st.means, st.Rinv, st.constant, st.pi,
+            self.cfg.diag_only,
+        )
+        w, _ = F.estep_posteriors(logw)
+        return w
+
+    def gather_memberships(self, w_shard: torch.Tensor) -> np.ndarray | None:
+        """Gather per-event posteriors to rank 0 (event-major contiguous
+        reassembly). Returns [K, N_total] on rank 0, None elsewhere."""
+        k = int(w_shard.shape[0])
+        if self.world == 1:
+            return w_shard.cpu().numpy()
+        per = self.n_total // self.world
+        maxn = self.n_total - per * (self.world - 1)
+        pad = torch.zeros(k, maxn, dtype=torch.float32, device=w_shard.device)
+        pad[:, : self.n_shard] = w_shard
+        parts = [torch.empty_like(pad) for _ in range(self.world)]
+        torch.distributed.all_gather(parts, pad)
+        if self.rank != 0:
+            return None
+        out = np.empty((k, self.n_total), dtype=np.float32)
+        for r in range(self.world):
+            s, e = pdist.shard_bounds(self.n_total, self.world, r)
+            out[:, s:e] = parts[r][:, : e - s].cpu().numpy()
+        return out
+
+
+def build_engine(data_by_event: np.ndarray | torch.Tensor, config: GmmConfig,
+                 device: torch.device | str = "cpu",
+                 profile: Profile | None = None) -> EmEngine:
+    """Build an engine from rank-consistent full data.
+
+    Every rank passes the same full dataset (CLI: rank 0 broadcasts it
+    first — see cli.run_clustering); each rank keeps only its shard.
+    Seeding statistics (strided means, per-dim variance, global center)
+    are computed from the full data so results are world-size independent.
+    """
+    if isinstance(data_by_event, np.ndarray):
+        data = torch.from_numpy(np.ascontiguousarray(data_by_event, np.float32))
+    else:
+        data = data_by_event.to(torch.float32).cpu()
+    n, d = data.shape
+    config.validate()
+
+    mean = data.double().mean(dim=0)
+    var = data.double().pow(2).mean(dim=0) - mean * mean
+    center = mean.float() if config.center_data else torch.zeros(d)
+    seed_means = seed_means_host(data, config.num_clusters)
+
+    r, w = pdist.rank(), pdist.world_size()
+    s, e = pdist.shard_bounds(n, w, r)
+    return EmEngine(
+        data[s:e], config, n, center, seed_means, var.float(),
+        device=device, profile=profile,
+    )

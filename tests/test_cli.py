@@ -1,0 +1,82 @@
+import numpy as np
+import pytest
+
+from cuda_gmm_mpi_amd.cli import main
+from cuda_gmm_mpi_amd.utils import io as gio
+from cuda_gmm_mpi_amd.utils.synthetic import make_blobs
+
+
+@pytest.fixture
+def csv_file(tmp_path):
+    data, _ = make_blobs(600, 2, 3, seed=4)
+    path = tmp_path / "data.csv"
+    with open(path, "w") as f:
+        f.write("c0,c1\n")
+        for row in data:
+            f.write(f"{row[0]:.6f},{row[1]:.6f}\n")
+    return str(path), data
+
+
+def test_cli_end_to_end_csv(tmp_path, csv_file):
+    path, data = csv_file
+    out = str(tmp_path / "out")
+    rc = main([
+        "3", path, out, "3", "--min-iters", "5", "--max-iters", "5",
+        "--device", "cpu",
+    ])
+    assert rc == 0
+    summary = open(out + ".summary").read()
+    assert summary.count("Cluster #") == 3
+    results = open(out + ".results").read().splitlines()
+    assert len(results) == 600
+    first = results[0].split("\t")
+    assert len(first) == 2
+    assert len(first[0].split(",")) == 2
+    memb = np.array([float(v) for v in first[1].split(",")])
+    assert len(memb) == 3
+    assert memb.sum() == pytest.approx(1.0, abs=1e-4)
+
+
+def test_cli_bin_input(tmp_path):
+    data, _ = make_blobs(500, 3, 2, seed=6)
+    binpath = str(tmp_path / "d.bin")
+    gio.write_bin(binpath, data)
+    out = str(tmp_path / "o")
+    rc = main(["2", binpath, out, "2", "--min-iters", "4", "--max-iters", "4",
+               "--device", "cpu", "--no-results"])
+    assert rc == 0
+    assert open(out + ".summary").read().count("Cluster #") == 2
+
+
+def test_cli_sweep_with_mdl(tmp_path, csv_file):
+    path, _ = csv_file
+    out = str(tmp_path / "sweep")
+    rc = main(["5", path, out, "2", "--min-iters", "3", "--max-iters", "3",
+               "--device", "cpu", "--no-results"])
+    assert rc == 0
+    assert open(out + ".summary").read().count("Cluster #") == 2
+
+
+def test_cli_invalid_cluster_count(tmp_path, csv_file):
+    path, _ = csv_file
+    assert main(["0", path, str(tmp_path / "x")]) == 1
+    assert main(["513", path, str(tmp_path / "x")]) == 1
+
+
+def test_cli_target_exceeds_start(tmp_path, csv_file):
+    path, _ = csv_file
+    assert main(["3", path, str(tmp_path / "x"), "5"]) == 4
+
+
+def test_cli_missing_infile(tmp_path):
+    assert main(["3", str(tmp_path / "nope.csv"), str(tmp_path / "x"),
+                 "--device", "cpu"]) == 2
+
+
+def test_cli_no_output_creates_empty_summary(tmp_path, csv_file):
+    path, _ = csv_file
+    out = str(tmp_path / "noout")
+    rc = main(["2", path, out, "2", "--min-iters", "2", "--max-iters", "2",
+               "--device", "cpu", "--no-output"])
+    assert rc == 0
+    assert open(out + ".summary").read() == ""

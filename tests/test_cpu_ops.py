@@ -1,0 +1,166 @@
+"""Unit tests of the torch-CPU reference ops against independent numpy math."""
+import math
+
+import numpy as np
+import pytest
+import torch
+
+from cuda_gmm_mpi_amd.ops import cpu_reference as cpu
+
+
+def random_model(rng, k, d):
+    means = rng.standard_normal((k, d)).astype(np.float32) * 3
+    rs = []
+    for _ in range(k):
+        a = rng.standard_normal((d, d))
+        rs.append((a @ a.T + d * np.eye(d)).astype(np.float32))
+    r = np.stack(rs)
+    pi = rng.dirichlet(np.ones(k)).astype(np.float32)
+    return means, r, pi
+
+
+def test_estep_logw_matches_multivariate_logpdf(rng):
+    k, d, n = 3, 4, 50
+    means, r, pi = random_model(rng, k, d)
+    x = rng.standard_normal((d, n)).astype(np.float32)
+    rinv = np.stack([np.linalg.inv(r[c]) for c in range(k)]).astype(np.float32)
+    const = np.array(
+        [-d / 2 * math.log(2 * math.pi)
+         - 0.5 * np.linalg.slogdet(r[c].astype(np.float64))[1]
+         for c in range(k)], dtype=np.float32)
+
+    logw = cpu.estep_logw(
+        torch.from_numpy(x), torch.from_numpy(means), torch.from_numpy(rinv),
+        torch.from_numpy(const), torch.from_numpy(pi),
+    ).numpy()
+
+    for c in range(k):
+        for e in range(0, n, 7):
+            diff = x[:, e] - means[c]
+            q = diff @ rinv[c] @ diff
+            expect = -0.5 * q + const[c] + math.log(pi[c])
+            assert logw[c, e] == pytest.approx(expect, rel=1e-4, abs=1e-4)
+
+
+def test_estep_posteriors_sum_to_one_and_likelihood(rng):
+    k, n = 5, 200
+    logw = torch.from_numpy(rng.standard_normal((k, n)).astype(np.float32) * 5)
+    w, lik = cpu.estep_posteriors(logw.clone())
+    np.testing.assert_allclose(w.sum(dim=0).numpy(), np.ones(n), rtol=1e-5)
+    # independent logsumexp
+    ref = torch.logsumexp(logw, dim=0).sum()
+    assert float(lik) == pytest.approx(float(ref), rel=1e-5)
+
+
+def test_mstep_stats_match_manual(rng):
+    k, d, n = 3, 4, 64
+    x = torch.from_numpy(rng.standard_normal((d, n)).astype(np.float32))
+    w = torch.from_numpy(rng.uniform(0, 1, (k, n)).astype(np.float32))
+    n_c, mean_num, s = cpu.mstep_sufficient_stats(x, w)
+    np.testing.assert_allclose(n_c.numpy(), w.sum(dim=1).numpy(), rtol=1e-5)
+    ref_mn = np.einsum("cn,dn->cd", w.numpy(), x.numpy())
+    np.testing.assert_allclose(mean_num.numpy(), ref_mn, rtol=1e-4)
+    ref_s = np.einsum("cn,dn,en->cde", w.numpy(), x.numpy(), x.numpy())
+    np.testing.assert_allclose(s.numpy(), ref_s, rtol=1e-3, atol=1e-3)
+
+
+def test_finalize_covariance_equals_centered_sum(rng):
+    """S - N mu mu^T must equal the reference's centered per-event sum
+    (gaussian_kernel.cu:644-646) when mu is the exact weighted mean."""
+    k, d, n = 2, 3, 500
+    x = torch.from_numpy(rng.standard_normal((d, n)).astype(np.float32))
+    w = torch.from_numpy(rng.uniform(0, 1, (k, n)).astype(np.float32))
+    n_c, mean_num, s = cpu.mstep_sufficient_stats(x.double(), w.double())
+    means = mean_num / n_c.unsqueeze(1)
+    avgvar = torch.full((k,), 0.123, dtype=torch.float64)
+    r = cpu.finalize_covariance(n_c, means, s, avgvar, world_size=1)
+    for c in range(k):
+        diff = x.double() - means[c].unsqueeze(1)
+        ref = (diff * w[c].double()) @ diff.T
+        ref += 0.123 * torch.eye(d, dtype=torch.float64)
+        ref /= n_c[c]
+        np.testing.assert_allclose(r[c].numpy(), ref.numpy(), rtol=1e-8)
+
+
+def test_finalize_covariance_empty_cluster_rules():
+    k, d = 3, 2
+    n_c = torch.tensor([10.0, 0.7, 0.1])
+    means = torch.zeros(k, d)
+    s = torch.ones(k, d, d) * 5.0
+    avgvar = torch.full((k,), 2.0)
+    r = cpu.finalize_covariance(n_c, means, s, avgvar, world_size=4)
+    # N >= 1: (S + G*avgvar*I)/N
+    np.testing.assert_allclose(
+        r[0].numpy(), (np.full((d, d), 5.0) + 8.0 * np.eye(d)) / 10.0
+    )
+    # 0.5 < N < 1: kernel zeroed cov_num, diag = G*avgvar, divided by N
+    np.testing.assert_allclose(r[1].numpy(), (8.0 * np.eye(d)) / 0.7,
+                               rtol=1e-6)
+    # N <= 0.5: identity reset (gaussian.cu:669-678)
+    np.testing.assert_allclose(r[2].numpy(), np.eye(d))
+
+
+def test_finalize_means_zeroes_empty():
+    n_c = torch.tensor([4.0, 0.2])
+    mean_num = torch.tensor([[8.0, 2.0], [5.0, 5.0]])
+    m = cpu.finalize_means(n_c, mean_num)
+    np.testing.assert_allclose(m.numpy(), [[2.0, 0.5], [0.0, 0.0]])
+
+
+def test_compute_pi_floor():
+    n_c = torch.tensor([99.0, 0.3, 1.0])
+    pi = cpu.compute_pi(n_c)
+    total = 99.0 + 0.3 + 1.0
+    assert pi[0] == pytest.approx(99.0 / total)
+    assert pi[1] == pytest.approx(1e-10)
+    assert pi[2] == pytest.approx(1.0 / total)
+
+
+@pytest.mark.parametrize("d", [1, 2, 5, 24])
+def test_lu_invert_nopivot_batched(rng, d):
+    k = 4
+    rs = []
+    for _ in range(k):
+        a = rng.standard_normal((d, d))
+        rs.append((a @ a.T + d * np.eye(d)).astype(np.float32))
+    r = torch.from_numpy(np.stack(rs))
+    inv, logdet = cpu.lu_invert_nopivot(r)
+    for c in range(k):
+        np.testing.assert_allclose(
+            inv[c].numpy(), np.linalg.inv(r[c].numpy()), rtol=5e-3, atol=5e-3
+        )
+        ref_ld = np.linalg.slogdet(r[c].numpy().astype(np.float64))[1]
+        assert float(logdet[c]) == pytest.approx(ref_ld, rel=1e-3, abs=1e-3)
+
+
+def test_compute_constants(rng):
+    d = 6
+    a = rng.standard_normal((d, d))
+    r = torch.from_numpy((a @ a.T + d * np.eye(d)).astype(np.float32))[None]
+    rinv, const = cpu.compute_constants(r)
+    ref_ld = np.linalg.slogdet(r[0].numpy().astype(np.float64))[1]
+    expect = -d / 2 * math.log(2 * math.pi) - 0.5 * ref_ld
+    assert float(const[0]) == pytest.approx(expect, rel=1e-4)
+
+
+def test_diag_only_paths(rng):
+    k, d, n = 2, 3, 40
+    means, r, pi = random_model(rng, k, d)
+    # make R diagonal for a clean check
+    r = np.stack([np.diag(np.diag(r[c])) for c in range(k)]).astype(np.float32)
+    rt = torch.from_numpy(r)
+    rinv, const = cpu.compute_constants(rt, diag_only=True)
+    for c in range(k):
+        np.testing.assert_allclose(
+            rinv[c].numpy(), np.diag(1.0 / np.diag(r[c])), rtol=1e-6)
+    x = torch.from_numpy(rng.standard_normal((d, n)).astype(np.float32))
+    logw = cpu.estep_logw(
+        x, torch.from_numpy(means), rinv, const, torch.from_numpy(pi),
+        diag_only=True,
+    )
+    full = cpu.estep_logw(
+        x, torch.from_numpy(means), rinv, const, torch.from_numpy(pi),
+        diag_only=False,
+    )
+    np.testing.assert_allclose(logw.numpy(), full.numpy(), rtol=1e-4,
+                               atol=1e-4)

@@ -1,0 +1,143 @@
+"""End-to-end engine tests on the CPU golden path (BASELINE config 1)."""
+import numpy as np
+import pytest
+import torch
+
+from cuda_gmm_mpi_amd.engine import build_engine
+from cuda_gmm_mpi_amd.utils.config import GmmConfig, em_epsilon, rissanen_score
+from cuda_gmm_mpi_amd.utils.synthetic import make_blobs
+
+
+def test_epsilon_formula():
+    # eps = (1 + D + 0.5*(D+1)*D) * ln(N*D) * 0.01 (gaussian.cu:458)
+    assert em_epsilon(24, 1_000_000) == pytest.approx(
+        (1 + 24 + 0.5 * 25 * 24) * np.log(24e6) * 0.01)
+
+
+def test_rissanen_formula():
+    # gaussian.cu:826
+    lik = -1234.5
+    assert rissanen_score(lik, 64, 24, 10_000) == pytest.approx(
+        -lik + 0.5 * (64 * (1 + 24 + 0.5 * 25 * 24) - 1) * np.log(240_000.0))
+
+
+def test_em_recovers_planted_blobs():
+    """Config 1: K=4, D=2, N=10k synthetic blobs on the CPU path."""
+    data, labels = make_blobs(10_000, 2, 4, seed=5)
+    cfg = GmmConfig(num_clusters=4, target_num_clusters=4,
+                    min_iters=40, max_iters=40)
+    eng = build_engine(data, cfg, device="cpu")
+    lik = eng.run_em(4)
+    assert np.isfinite(lik)
+    means = (eng.state.means + eng.center.unsqueeze(0)).numpy()
+    true_means = np.stack([data[labels == c].mean(axis=0) for c in range(4)])
+    # each true mean matched by some learned mean within a few units
+    d2 = ((means[None] - true_means[:, None]) ** 2).sum(-1) ** 0.5
+    assert d2.min(axis=1).max() < 5.0
+    # mixture weights sane
+    pi = eng.state.pi.numpy()
+    assert pi.sum() == pytest.approx(1.0, abs=1e-3)
+    assert (pi > 0.01).all()
+
+
+def test_likelihood_monotone_on_well_conditioned_data():
+    data, _ = make_blobs(4000, 3, 3, seed=11)
+    cfg = GmmConfig(num_clusters=3, target_num_clusters=3,
+                    min_iters=1, max_iters=1)
+    eng = build_engine(data, cfg, device="cpu")
+    liks = [eng._reduce_likelihood(eng._estep(3))]
+    for _ in range(15):
+        eng._mstep(3)
+        liks.append(eng._reduce_likelihood(eng._estep(3)))
+    liks = np.array(liks)
+    # monotone non-decreasing up to fp32 noise
+    assert (np.diff(liks) > -abs(liks[-1]) * 1e-5).all()
+
+
+def test_posteriors_sum_to_one_after_em():
+    data, _ = make_blobs(2000, 2, 3, seed=5)
+    cfg = GmmConfig(num_clusters=3, target_num_clusters=3,
+                    min_iters=5, max_iters=5)
+    eng = build_engine(data, cfg, device="cpu")
+    eng.run_em(3)
+    sums = eng.w[:3].sum(dim=0).numpy()
+    np.testing.assert_allclose(sums, np.ones_like(sums), rtol=1e-4)
+
+
+def test_sweep_reduces_and_saves_best():
+    data, _ = make_blobs(3000, 2, 3, seed=9)
+    cfg = GmmConfig(num_clusters=6, target_num_clusters=2,
+                    min_iters=8, max_iters=8)
+    eng = build_engine(data, cfg, device="cpu")
+    res = eng.sweep()
+    assert res.num_clusters == 2
+    assert set(res.rissanen_by_k) <= set(range(2, 7))
+    assert 6 in res.rissanen_by_k and 2 in res.rissanen_by_k
+    assert np.isfinite(res.min_rissanen)
+
+
+def test_sweep_best_mdl_selection():
+    """With no target, the saved model is the min-rissanen K
+    (gaussian.cu:839)."""
+    data, _ = make_blobs(4000, 2, 4, seed=13)
+    cfg = GmmConfig(num_clusters=7, target_num_clusters=0,
+                    min_iters=6, max_iters=6)
+    eng = build_engine(data, cfg, device="cpu")
+    res = eng.sweep()
+    ks = [k for k in res.rissanen_by_k]
+    best_k = min(res.rissanen_by_k, key=res.rissanen_by_k.get)
+    assert res.num_clusters == best_k
+    assert res.min_rissanen == pytest.approx(res.rissanen_by_k[best_k])
+    assert min(ks) == 1  # swept down to 1
+
+
+def test_centering_is_transparent():
+    """center_data must not change the fitted model (beyond fp noise)."""
+    data, _ = make_blobs(2000, 3, 3, seed=21)
+    data = data + 500.0  # large offset to stress the uncentered path
+    results = []
+    for center in (True, False):
+        cfg = GmmConfig(num_clusters=3, target_num_clusters=3,
+                        min_iters=10, max_iters=10, center_data=center)
+        eng = build_engine(data, cfg, device="cpu")
+        eng.run_em(3)
+        means = (eng.state.means + eng.center.unsqueeze(0)).numpy().copy()
+        results.append((means, eng.state.R.numpy().copy()))
+    m_c, r_c = results[0]
+    m_u, r_u = results[1]
+    # sort clusters by first mean coordinate for comparison
+    oc, ou = np.argsort(m_c[:, 0]), np.argsort(m_u[:, 0])
+    np.testing.assert_allclose(m_c[oc], m_u[ou], rtol=2e-3, atol=1e-2)
+    np.testing.assert_allclose(r_c[oc], r_u[ou], rtol=0.3, atol=2.0)
+
+
+def test_seed_matches_reference_formula():
+    data, _ = make_blobs(1000, 3, 4, seed=2)
+    cfg = GmmConfig(num_clusters=4, target_num_clusters=4, center_data=False)
+    eng = build_engine(data, cfg, device="cpu")
+    seed = (1000 - 1.0) / (4 - 1.0)
+    for c in range(4):
+        np.testing.assert_allclose(
+            eng.state.means[c].numpy(), data[int(c * seed)], rtol=1e-6)
+    assert eng.state.N[0] == pytest.approx(1000 // 4)
+    assert eng.state.pi[0] == pytest.approx(0.25)
+    # R seeded to identity, constants computed for it
+    np.testing.assert_allclose(eng.state.R[0].numpy(), np.eye(3))
+    expect_const = -3 * 0.5 * np.log(2 * np.pi)
+    assert eng.state.constant[0] == pytest.approx(expect_const, rel=1e-5)
+    # avgvar = mean(var)/COVARIANCE_DYNAMIC_RANGE
+    var = data.astype(np.float64).var(axis=0).mean()
+    assert eng.state.avgvar[0] == pytest.approx(var / 1e3, rel=1e-3)
+
+
+def test_diag_only_engine_runs():
+    data, _ = make_blobs(1500, 3, 2, seed=8)
+    cfg = GmmConfig(num_clusters=2, target_num_clusters=2,
+                    min_iters=5, max_iters=5, diag_only=True)
+    eng = build_engine(data, cfg, device="cpu")
+    lik = eng.run_em(2)
+    assert np.isfinite(lik)
+    r = eng.state.R.numpy()
+    for c in range(2):
+        off = r[c] - np.diag(np.diag(r[c]))
+        assert np.abs(off).max() == 0.0
