@@ -60,20 +60,51 @@ def mstep_n_means(x_aug_t: torch.Tensor, w: torch.Tensor
     return nm[:, -1].contiguous(), nm[:, :-1].contiguous()
 
 
+_tri_cache: dict = {}
+
+
+def _tri_unpack_index(d: int, device) -> torch.Tensor:
+    """[D*D] gather index from packed lower-triangle to full matrix."""
+    key = (d, str(device))
+    idx = _tri_cache.get(key)
+    if idx is None:
+        full = torch.empty(d, d, dtype=torch.long)
+        for i in range(d):
+            for j in range(d):
+                r, c = (i, j) if i >= j else (j, i)
+                full[i, j] = r * (r + 1) // 2 + c
+        idx = full.reshape(-1).to(device)
+        _tri_cache[key] = idx
+    return idx
+
+
 def mstep_covariance_s(x: torch.Tensor, w: torch.Tensor,
-                       out: torch.Tensor | None = None) -> torch.Tensor:
+                       out: torch.Tensor | None = None,
+                       nchunk: int = 64) -> torch.Tensor:
     """Uncentered weighted second moments S [K, D, D].
 
-    CUDA: custom LDS-tiled kernel (the covariance showpiece); CPU: batched
-    torch matmuls.
+    CUDA: custom LDS-tiled kernel (the covariance showpiece) producing
+    deterministic per-chunk packed partials, summed + unpacked here; CPU:
+    batched torch matmuls.
     """
     if x.is_cuda:
         k = w.shape[0]
         d = x.shape[0]
-        if out is None:
-            out = torch.empty((k, d, d), dtype=torch.float32, device=x.device)
-        hip_ext().mstep_covariance(x, w, out)
-        return out
+        n = x.shape[1]
+        p = d * (d + 1) // 2
+        te = 256 if d <= 64 else 128
+        tiles = (n + te - 1) // te
+        nchunk = int(min(nchunk, tiles))
+        partials = torch.empty((nchunk, k, p), dtype=torch.float32,
+                               device=x.device)
+        hip_ext().mstep_covariance_partials(x, w, partials)
+        packed = partials.sum(dim=0)                      # [K, P]
+        idx = _tri_unpack_index(d, x.device)
+        s = packed[:, idx].view(k, d, d)
+        if out is not None:
+            out.copy_(s)
+            return out
+        return s
     k, n = w.shape
     d = x.shape[0]
     xf = x.float()

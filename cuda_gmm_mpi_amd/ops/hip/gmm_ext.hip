@@ -1,0 +1,202 @@
+// Host-side launchers + PyTorch bindings for the gfx950 GMM kernels.
+#include <ATen/hip/HIPContext.h>
+#include <torch/extension.h>
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+#define HIP_CHECK(expr)                                                     \
+  do {                                                                      \
+    hipError_t _e = (expr);                                                 \
+    TORCH_CHECK(_e == hipSuccess, "HIP error: ", hipGetErrorString(_e));    \
+  } while (0)
+
+#include "gmm_kernels.hip"
+
+namespace {
+
+constexpr int kNT = 256;
+
+hipStream_t stream() { return at::hip::getCurrentHIPStream().stream(); }
+
+void check_f32(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda() && t.is_contiguous() &&
+                  t.scalar_type() == torch::kFloat32,
+              name, " must be contiguous fp32 on device");
+}
+
+int grid_x_for(int64_t n) {
+  // >> 256 workgroups to fill 8 XCDs x 32 CUs; cap so partial buffers and
+  // tail-effect stay reasonable
+  int64_t tiles = (n + kNT - 1) / kNT;
+  return (int)std::min<int64_t>(tiles, 16384);
+}
+
+template <typename T>
+const T* data_as(const torch::Tensor& t) {
+  return reinterpret_cast<const T*>(t.data_ptr());
+}
+
+template <typename T>
+void estep_logw_impl(const torch::Tensor& x, const torch::Tensor& means,
+                     const torch::Tensor& rinv, const torch::Tensor& constant,
+                     const torch::Tensor& logpi, torch::Tensor& logw,
+                     bool diag_only) {
+  const int d = (int)x.size(0);
+  const int64_t n = x.size(1);
+  const int k = (int)means.size(0);
+  dim3 grid(grid_x_for(n), k);
+  const int p = d * (d + 1) / 2;
+  const size_t lds = sizeof(float) * (d + p);
+  auto s = stream();
+  const T* xp = data_as<T>(x);
+  const float* mp = means.data_ptr<float>();
+  const float* rp = rinv.data_ptr<float>();
+  const float* cp = constant.data_ptr<float>();
+  const float* lp = logpi.data_ptr<float>();
+  float* op = logw.data_ptr<float>();
+
+  if (diag_only) {
+    const size_t lds_d = sizeof(float) * 2 * d;
+    hipLaunchKernelGGL(gmm::estep_logw_diag_kernel<T>, grid, dim3(kNT), lds_d,
+                       s, xp, mp, rp, cp, lp, op, d, n);
+  } else if (d <= 8) {
+    hipLaunchKernelGGL((gmm::estep_logw_reg_kernel<8, T>), grid, dim3(kNT),
+                       lds, s, xp, mp, rp, cp, lp, op, d, n);
+  } else if (d <= 16) {
+    hipLaunchKernelGGL((gmm::estep_logw_reg_kernel<16, T>), grid, dim3(kNT),
+                       lds, s, xp, mp, rp, cp, lp, op, d, n);
+  } else if (d <= 24) {
+    hipLaunchKernelGGL((gmm::estep_logw_reg_kernel<24, T>), grid, dim3(kNT),
+                       lds, s, xp, mp, rp, cp, lp, op, d, n);
+  } else if (d <= 32) {
+    hipLaunchKernelGGL((gmm::estep_logw_reg_kernel<32, T>), grid, dim3(kNT),
+                       lds, s, xp, mp, rp, cp, lp, op, d, n);
+  } else {
+    hipLaunchKernelGGL(gmm::estep_logw_gen_kernel<T>, grid, dim3(kNT), lds, s,
+                       xp, mp, rp, cp, lp, op, d, n);
+  }
+  HIP_CHECK(hipGetLastError());
+}
+
+void estep_logw(torch::Tensor x, torch::Tensor means, torch::Tensor rinv,
+                torch::Tensor constant, torch::Tensor logpi,
+                torch::Tensor logw, bool diag_only) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous(), "x must be contiguous device");
+  check_f32(means, "means");
+  check_f32(rinv, "rinv");
+  check_f32(constant, "constant");
+  check_f32(logpi, "logpi");
+  check_f32(logw, "logw");
+  TORCH_CHECK(logw.size(0) == means.size(0) && logw.size(1) == x.size(1),
+              "logw shape mismatch");
+  if (x.scalar_type() == torch::kFloat32) {
+    estep_logw_impl<float>(x, means, rinv, constant, logpi, logw, diag_only);
+  } else if (x.scalar_type() == torch::kBFloat16) {
+    estep_logw_impl<__hip_bfloat16>(x, means, rinv, constant, logpi, logw,
+                                    diag_only);
+  } else {
+    TORCH_CHECK(false, "x must be fp32 or bf16");
+  }
+}
+
+void estep_posteriors(torch::Tensor logw, torch::Tensor partial) {
+  check_f32(logw, "logw");
+  check_f32(partial, "partial");
+  const int k = (int)logw.size(0);
+  const int64_t n = logw.size(1);
+  int grid = (int)std::min<int64_t>((n + kNT - 1) / kNT, partial.size(0));
+  TORCH_CHECK(grid >= 1, "empty logw");
+  // zero the tail of the partial buffer (grid may be < partial size)
+  hipLaunchKernelGGL(gmm::estep_posteriors_kernel, dim3(grid), dim3(kNT), 0,
+                     stream(), logw.data_ptr<float>(),
+                     partial.data_ptr<float>(), k, n);
+  HIP_CHECK(hipGetLastError());
+}
+
+template <typename T>
+void mstep_cov_impl(const torch::Tensor& x, const torch::Tensor& w,
+                    torch::Tensor& partials) {
+  const int d = (int)x.size(0);
+  const int64_t n = x.size(1);
+  const int k = (int)w.size(0);
+  const int nchunk = (int)partials.size(0);
+  const int p = d * (d + 1) / 2;
+  TORCH_CHECK(partials.size(1) == k && partials.size(2) == p,
+              "partials must be [nchunk, K, D*(D+1)/2]");
+  const int te = (d <= 64) ? 256 : 128;
+  const size_t lds = sizeof(float) * ((size_t)d * (te + 4) + te);
+  TORCH_CHECK(lds <= 64 * 1024, "covariance tile exceeds LDS budget");
+  dim3 grid(k, nchunk);
+  auto s = stream();
+  const int ppt = (p + kNT - 1) / kNT;
+  const T* xp = data_as<T>(x);
+  const float* wp = w.data_ptr<float>();
+  float* pp = partials.data_ptr<float>();
+#define LAUNCH_COV(PPT)                                                     \
+  hipLaunchKernelGGL((gmm::mstep_cov_kernel<PPT, T>), grid, dim3(kNT), lds, \
+                     s, xp, wp, pp, d, k, n, te, nchunk)
+  if (ppt <= 1) LAUNCH_COV(1);
+  else if (ppt <= 2) LAUNCH_COV(2);
+  else if (ppt <= 4) LAUNCH_COV(4);
+  else if (ppt <= 8) LAUNCH_COV(8);
+  else if (ppt <= 17) LAUNCH_COV(17);
+  else if (ppt <= 33) LAUNCH_COV(33);
+  else TORCH_CHECK(false, "D too large for covariance kernel (max 128)");
+#undef LAUNCH_COV
+  HIP_CHECK(hipGetLastError());
+}
+
+void mstep_covariance_partials(torch::Tensor x, torch::Tensor w,
+                               torch::Tensor partials) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous(), "x must be contiguous device");
+  check_f32(w, "w");
+  check_f32(partials, "partials");
+  if (x.scalar_type() == torch::kFloat32) {
+    mstep_cov_impl<float>(x, w, partials);
+  } else if (x.scalar_type() == torch::kBFloat16) {
+    mstep_cov_impl<__hip_bfloat16>(x, w, partials);
+  } else {
+    TORCH_CHECK(false, "x must be fp32 or bf16");
+  }
+}
+
+void constants(torch::Tensor r, torch::Tensor rinv, torch::Tensor logdet,
+               bool diag_only) {
+  check_f32(r, "r");
+  check_f32(rinv, "rinv");
+  check_f32(logdet, "logdet");
+  const int k = (int)r.size(0);
+  const int d = (int)r.size(1);
+  auto s = stream();
+  if (diag_only) {
+    hipLaunchKernelGGL(gmm::constants_diag_kernel, dim3(k), dim3(kNT), 0, s,
+                       r.data_ptr<float>(), rinv.data_ptr<float>(),
+                       logdet.data_ptr<float>(), d);
+  } else {
+    const size_t lds = sizeof(float) * (size_t)d * d;
+    if (lds > 64 * 1024) {
+      static bool raised = false;
+      HIP_CHECK(hipFuncSetAttribute(
+          reinterpret_cast<const void*>(&gmm::constants_lu_kernel),
+          hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds));
+      (void)raised;
+    }
+    hipLaunchKernelGGL(gmm::constants_lu_kernel, dim3(k), dim3(kNT), lds, s,
+                       r.data_ptr<float>(), rinv.data_ptr<float>(),
+                       logdet.data_ptr<float>(), d);
+  }
+  HIP_CHECK(hipGetLastError());
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("estep_logw", &estep_logw,
+        "log weighted likelihoods [K,N] (gfx950 kernel)");
+  m.def("estep_posteriors", &estep_posteriors,
+        "in-place posteriors + per-block likelihood partials");
+  m.def("mstep_covariance_partials", &mstep_covariance_partials,
+        "packed weighted second-moment partials [nchunk,K,P]");
+  m.def("constants", &constants, "batched no-pivot LU inverse + ln|det|");
+}
