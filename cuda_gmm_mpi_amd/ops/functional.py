@@ -92,7 +92,7 @@ def mstep_covariance_s(x: torch.Tensor, w: torch.Tensor,
         d = x.shape[0]
         n = x.shape[1]
         p = d * (d + 1) // 2
-        te = 256 if d <= 64 else 128
+        te = min((16384 - 4 * d) // (d + 1) & ~3, 256)  # matches gmm_ext.hip
         tiles = (n + te - 1) // te
         nchunk = int(min(nchunk, tiles))
         partials = torch.empty((nchunk, k, p), dtype=torch.float32,

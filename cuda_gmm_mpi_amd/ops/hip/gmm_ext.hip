@@ -124,9 +124,12 @@ void mstep_cov_impl(const torch::Tensor& x, const torch::Tensor& w,
   const int p = d * (d + 1) / 2;
   TORCH_CHECK(partials.size(1) == k && partials.size(2) == p,
               "partials must be [nchunk, K, D*(D+1)/2]");
-  const int te = (d <= 64) ? 256 : 128;
+  // largest event tile (multiple of 4) fitting the 64 KiB LDS budget:
+  // d*(te+4) + te floats  <=  16384
+  int te = (16384 - 4 * d) / (d + 1);
+  te = std::min(te & ~3, 256);
+  TORCH_CHECK(te >= 32, "D too large for covariance tile");
   const size_t lds = sizeof(float) * ((size_t)d * (te + 4) + te);
-  TORCH_CHECK(lds <= 64 * 1024, "covariance tile exceeds LDS budget");
   dim3 grid(k, nchunk);
   auto s = stream();
   const int ppt = (p + kNT - 1) / kNT;
@@ -174,13 +177,12 @@ void constants(torch::Tensor r, torch::Tensor rinv, torch::Tensor logdet,
                        r.data_ptr<float>(), rinv.data_ptr<float>(),
                        logdet.data_ptr<float>(), d);
   } else {
-    const size_t lds = sizeof(float) * (size_t)d * d;
-    if (lds > 64 * 1024) {
-      static bool raised = false;
+    // working buffer + read-only LU snapshot
+    const size_t lds = sizeof(float) * 2 * (size_t)d * d;
+    if (lds > 64 * 1024) {  // gfx950: 160 KiB LDS/CU; opt in past 64 KiB
       HIP_CHECK(hipFuncSetAttribute(
           reinterpret_cast<const void*>(&gmm::constants_lu_kernel),
           hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds));
-      (void)raised;
     }
     hipLaunchKernelGGL(gmm::constants_lu_kernel, dim3(k), dim3(kNT), lds, s,
                        r.data_ptr<float>(), rinv.data_ptr<float>(),

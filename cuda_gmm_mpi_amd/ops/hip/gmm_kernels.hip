@@ -286,9 +286,15 @@ mstep_cov_kernel(const T* __restrict__ x, const float* __restrict__ w,
 __global__ void __launch_bounds__(NT)
 constants_lu_kernel(const float* __restrict__ r, float* __restrict__ rinv,
                     float* __restrict__ logdet, int d) {
-  extern __shared__ float a[];  // [d][d], row-major (cold kernel; conflicts OK)
+  // lds: a[d*d] working buffer, o[d*d] read-only snapshot of the LU factor.
+  // The reference's in-place triangular inversion reads a mix of original
+  // and already-inverted entries in a serial order; parallelized across
+  // columns/rows that order only survives if cross-thread reads come from
+  // a snapshot (o) and each thread's own running values stay private (a).
+  extern __shared__ float a[];
   const int c = blockIdx.x;
   const int tid = threadIdx.x;
+  float* o = a + d * d;
   const float* rc = r + (int64_t)c * d * d;
   float* oc = rinv + (int64_t)c * d * d;
 
@@ -342,28 +348,34 @@ constants_lu_kernel(const float* __restrict__ r, float* __restrict__ rinv,
     }
   }
   __syncthreads();
+  // snapshot the LU factor: the inversion below reads original L/U values
+  // that the in-place writes would otherwise clobber across threads
+  for (int t = tid; t < d * d; t += NT) o[t] = a[t];
+  __syncthreads();
 
-  // invert L in place: column i independent per thread, serial down rows
-  // (gaussian_kernel.cu:142-151)
+  // invert L: column i per thread, serial down rows; cross-column reads and
+  // the diagonal divisor come from the snapshot (gaussian_kernel.cu:142-151:
+  // data[j,k] for k>i and data[j,j] are pre-inversion values there)
   for (int i = tid; i < d; i += NT) {
     for (int j = i; j < d; ++j) {
       float xv = 1.0f;
       if (i != j) {
         xv = 0.0f;
         for (int kk = i; kk < j; ++kk)
-          xv -= a[j * d + kk] * a[kk * d + i];
+          xv -= o[j * d + kk] * a[kk * d + i];
       }
-      a[j * d + i] = xv / a[j * d + j];
+      a[j * d + i] = xv / o[j * d + j];
     }
   }
-  __syncthreads();
-  // invert U in place: row i independent per thread, serial across cols
-  // (gaussian_kernel.cu:152-159)
+  // invert U: row i per thread, serial across cols; column reads from the
+  // snapshot (gaussian_kernel.cu:152-159: data[k,j] for k>i pre-inversion),
+  // row reads from this thread's own inverted values. Disjoint from the L
+  // writes (strict upper vs lower+diag) so no barrier is needed between.
   for (int i = tid; i < d; i += NT) {
     for (int j = i + 1; j < d; ++j) {
       float s = 0.0f;
       for (int kk = i; kk < j; ++kk)
-        s += a[kk * d + j] * ((i == kk) ? 1.0f : a[i * d + kk]);
+        s += o[kk * d + j] * ((i == kk) ? 1.0f : a[i * d + kk]);
       a[i * d + j] = -s;
     }
   }
