@@ -67,15 +67,22 @@ class EmEngine:
         shard = shard - self.center.unsqueeze(0)
         # dimension-major resident copy for the quadratic-form kernels
         self.x = shard.T.contiguous()                              # [D, n]
-        # augmented transpose for the fused N+means GEMM
-        ones = torch.ones(self.n_shard, 1, dtype=torch.float32, device=self.device)
-        self.x_aug_t = torch.cat([shard, ones], dim=1).contiguous()  # [n, D+1]
         if config.estep_dtype == "bf16" and self.device.type == "cuda":
             self.x_estep: torch.Tensor = self.x.to(torch.bfloat16)
         else:
             self.x_estep = self.x
 
         k0 = config.num_clusters
+        # fused bf16-MFMA E-step path (D <= 31, LDS-bounded K)
+        self.use_fused_estep = F.estep_fused_available(
+            self.device, config.estep_dtype, self.d, k0,
+        ) and not config.diag_only
+        self.mfac = (
+            torch.empty(k0, 2, 32, 32, dtype=torch.bfloat16,
+                        device=self.device)
+            if self.use_fused_estep else None
+        )
+
         self.state = GmmState.empty(k0, self.d, self.device)
         seed_state(
             self.state, seed_means - center.unsqueeze(0), var_per_dim,
@@ -83,20 +90,23 @@ class EmEngine:
         )
         # constants for the seeded R=I state (constants_kernel after seeding,
         # gaussian.cu:404)
-        with self.profile.time("constants"):
-            rinv, const = F.constants(self.state.R, config.diag_only)
-            self.state.Rinv.copy_(rinv)
-            self.state.constant.copy_(const)
-        self.profile.count("constants")
+        self._update_constants(self.state)
 
         # membership / logw buffer, cluster-major [K, n_shard]
         self.w = torch.empty(k0, self.n_shard, dtype=torch.float32,
                              device=self.device)
-        # fused reduce buffer: [K + K*D + K*D*D]
-        self._stats = torch.empty(k0 * (1 + self.d + self.d * self.d),
-                                  dtype=torch.float32, device=self.device)
         self.epsilon = em_epsilon(self.d, num_events_total)
         self.likelihood = 0.0
+
+    def _update_constants(self, st: GmmState) -> None:
+        k = st.num_clusters
+        with self.profile.time("constants"):
+            mfac = self.mfac[:k] if self.mfac is not None else None
+            rinv, const = F.constants(st.R, st.means, self.cfg.diag_only,
+                                      mfac)
+            st.Rinv.copy_(rinv)
+            st.constant.copy_(const)
+        self.profile.count("constants")
 
     # ------------------------------------------------------------------ EM
 
@@ -104,11 +114,16 @@ class EmEngine:
         """E-step into self.w[:k]; returns shard-partial likelihood tensor."""
         st = self.state.shrink(k)
         with self.profile.time("e_step"):
-            logw = F.estep_logw(
-                self.x_estep, st.means, st.Rinv, st.constant, st.pi,
-                self.cfg.diag_only, out=self.w[:k],
-            )
-            w, lik = F.estep_posteriors(logw)
+            if self.use_fused_estep:
+                add = st.constant + torch.log(st.pi)
+                w, lik = F.estep_fused(self.x_estep, self.mfac[:k], add,
+                                       self.w[:k])
+            else:
+                logw = F.estep_logw(
+                    self.x_estep, st.means, st.Rinv, st.constant, st.pi,
+                    self.cfg.diag_only, out=self.w[:k],
+                )
+                w, lik = F.estep_posteriors(logw)
         self.profile.count("regroup")
         return lik
 
@@ -119,33 +134,28 @@ class EmEngine:
         return float(t.item())
 
     def _mstep(self, k: int) -> None:
-        """M-step: stats, fused all-reduce, finalize params + constants."""
-        d = self.d
+        """M-step: fused moments, ONE all-reduce, finalize params + constants.
+
+        The packed buffer [K, (D+1)(D+2)/2] carries S, the mean numerators
+        and N in one contiguous payload (functional.mstep_moments layout) —
+        a single in-place RCCL all-reduce replaces the reference's staged
+        N / means / R reductions (gaussian.cu:545-686).
+        """
         st = self.state.shrink(k)
-        buf = self._stats[: k * (1 + d + d * d)]
-        n_v = buf[:k]
-        mn_v = buf[k: k + k * d].view(k, d)
-        s_v = buf[k + k * d:].view(k, d, d)
         with self.profile.time("m_step"):
-            n_c, mean_num = F.mstep_n_means(self.x_aug_t, self.w[:k])
-            n_v.copy_(n_c)
-            mn_v.copy_(mean_num)
-            F.mstep_covariance_s(self.x, self.w[:k], out=s_v)
+            packed = F.mstep_moments(self.x, self.w[:k])
         with self.profile.time("comm"):
-            pdist.all_reduce_(buf)
+            pdist.all_reduce_(packed)
         with self.profile.time("m_step"):
-            st.N.copy_(n_v)
-            st.means.copy_(F.finalize_means(n_v, mn_v))
+            n_c, mean_num, s = F.moments_views(packed, self.d)
+            st.N.copy_(n_c)
+            st.means.copy_(F.finalize_means(n_c, mean_num))
             st.R.copy_(F.finalize_covariance(
-                n_v, st.means, s_v, st.avgvar, self.world, self.cfg.diag_only,
+                n_c, st.means, s, st.avgvar, self.world, self.cfg.diag_only,
             ))
         self.profile.count("params")
-        with self.profile.time("constants"):
-            rinv, const = F.constants(st.R, self.cfg.diag_only)
-            st.Rinv.copy_(rinv)
-            st.constant.copy_(const)
-            st.pi.copy_(F.compute_pi(st.N))
-        self.profile.count("constants")
+        self._update_constants(st)
+        st.pi.copy_(F.compute_pi(st.N))
 
     def run_em(self, k: int) -> float:
         """Full EM at fixed K (the inner loop of gaussian.cu:479-755).

@@ -114,17 +114,96 @@ def mstep_covariance_s(x: torch.Tensor, w: torch.Tensor,
     return s
 
 
-def constants(r: torch.Tensor, diag_only: bool = False
+def constants(r: torch.Tensor, means: torch.Tensor | None = None,
+              diag_only: bool = False, mfac: torch.Tensor | None = None
               ) -> tuple[torch.Tensor, torch.Tensor]:
-    """(Rinv [K,D,D], constant [K]) via no-pivot LU + ln|det|."""
+    """(Rinv [K,D,D], constant [K]) via no-pivot LU + ln|det|.
+
+    On CUDA, when ``mfac`` (bf16 [K,2,32,32]) is given, the kernel also
+    emits the fused-E-step factor M = [U | -U mu] (U^T U = Rinv) as bf16
+    hi/lo fragment pairs — requires ``means`` and D <= 31.
+    """
     if r.is_cuda:
         k, d, _ = r.shape
         rinv = torch.empty_like(r)
         logdet = torch.empty(k, dtype=torch.float32, device=r.device)
-        hip_ext().constants(r, rinv, logdet, bool(diag_only))
+        if means is None:
+            means = torch.zeros(k, d, dtype=torch.float32, device=r.device)
+        if mfac is None:
+            mfac = torch.empty(0, dtype=torch.bfloat16, device=r.device)
+        hip_ext().constants(r, means, rinv, logdet, mfac, bool(diag_only))
         const = -d * 0.5 * cpu.LOG_2PI - 0.5 * logdet
         return rinv, const
     return cpu.compute_constants(r, diag_only)
+
+
+def estep_fused(z: torch.Tensor, mfac: torch.Tensor, add: torch.Tensor,
+                w_out: torch.Tensor) -> tuple[torch.Tensor, torch.Tensor]:
+    """Fused bf16-MFMA E-step (CUDA only): posteriors into w_out [K,N] and
+    the shard-partial likelihood. One pass over the data; logw never
+    touches HBM."""
+    n = z.shape[1]
+    nblk = (n + 127) // 128
+    partial = torch.zeros(nblk, dtype=torch.float32, device=z.device)
+    hip_ext().estep_fused(z, mfac, add, w_out, partial)
+    return w_out, partial.sum()
+
+
+def estep_fused_available(device: torch.device, dtype: str, d: int,
+                          k: int) -> bool:
+    """Fused path gate (mirrors the LDS check in gmm_ext.hip)."""
+    if device.type != "cuda" or dtype != "bf16" or d > 31:
+        return False
+    zbytes = (d * 136 * 2 + 3) & ~3
+    return zbytes + 4 * k * 132 <= 64 * 1024
+
+
+def mstep_moments(x: torch.Tensor, w: torch.Tensor,
+                  nchunk: int = 64) -> torch.Tensor:
+    """Fused augmented sufficient statistics, packed lower triangle of
+    T_c = sum_e w_ce [x;1][x;1]^T per cluster: [K, Dp*(Dp+1)/2] with layout
+    [S_tri (D rows) | mean numerators (row D) | N (corner)] — the single
+    all-reduce payload of the M-step.
+
+    CUDA fast path (D <= 31): one f32-MFMA kernel (exact fp32 fmaf chain).
+    Otherwise composed from the covariance kernel + rocBLAS/torch ops.
+    """
+    k, n = w.shape
+    d = x.shape[0]
+    dp = d + 1
+    pp = dp * (dp + 1) // 2
+    p = d * (d + 1) // 2
+    if x.is_cuda and d <= 31:
+        tiles = (n + 127) // 128
+        nchunk = int(min(nchunk, tiles))
+        partials = torch.empty((nchunk, k, pp), dtype=torch.float32,
+                               device=x.device)
+        hip_ext().mstep_moments(x, w, partials)
+        return partials.sum(dim=0)
+    packed = torch.empty((k, pp), dtype=torch.float32, device=x.device)
+    if x.is_cuda:
+        s = mstep_covariance_s(x, w)        # custom kernel, [K, D, D]
+        mean_num = w @ x.T                  # rocBLAS
+        n_c = w.sum(dim=1)
+    else:
+        n_c, mean_num, s = cpu.mstep_sufficient_stats(x.float(), w)
+    tri = torch.tril_indices(d, d, device=x.device)
+    packed[:, :p] = s[:, tri[0], tri[1]]
+    packed[:, p:p + d] = mean_num
+    packed[:, -1] = n_c
+    return packed
+
+
+def moments_views(packed: torch.Tensor, d: int
+                  ) -> tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
+    """(N [K], mean numerators [K,D], S [K,D,D]) from the packed layout."""
+    k = packed.shape[0]
+    p = d * (d + 1) // 2
+    n_c = packed[:, -1]
+    mean_num = packed[:, p:p + d]
+    idx = _tri_unpack_index(d, packed.device)
+    s = packed[:, :p][:, idx].view(k, d, d)
+    return n_c, mean_num, s
 
 
 finalize_means = cpu.finalize_means

@@ -164,30 +164,113 @@ void mstep_covariance_partials(torch::Tensor x, torch::Tensor w,
   }
 }
 
-void constants(torch::Tensor r, torch::Tensor rinv, torch::Tensor logdet,
-               bool diag_only) {
+void constants(torch::Tensor r, torch::Tensor means, torch::Tensor rinv,
+               torch::Tensor logdet, torch::Tensor mfac, bool diag_only) {
   check_f32(r, "r");
+  check_f32(means, "means");
   check_f32(rinv, "rinv");
   check_f32(logdet, "logdet");
   const int k = (int)r.size(0);
   const int d = (int)r.size(1);
+  const bool make_mfac = mfac.numel() > 0;
+  __hip_bfloat16* mp = nullptr;
+  if (make_mfac) {
+    TORCH_CHECK(mfac.is_cuda() && mfac.is_contiguous() &&
+                    mfac.scalar_type() == torch::kBFloat16 &&
+                    mfac.numel() == (int64_t)k * 2 * 32 * 32 && d <= 31,
+                "mfac must be bf16 [K,2,32,32] with D <= 31");
+    mp = reinterpret_cast<__hip_bfloat16*>(mfac.data_ptr());
+  }
   auto s = stream();
   if (diag_only) {
     hipLaunchKernelGGL(gmm::constants_diag_kernel, dim3(k), dim3(kNT), 0, s,
                        r.data_ptr<float>(), rinv.data_ptr<float>(),
                        logdet.data_ptr<float>(), d);
+    if (make_mfac) {
+      // diag Rinv is a valid input to the generic factor path: run the LU
+      // kernel's emit on the diagonal inverse via a tiny dedicated kernel
+      hipLaunchKernelGGL(gmm::emit_mfac_from_rinv_kernel, dim3(k), dim3(kNT),
+                         sizeof(float) * (2 * (size_t)d * d + d), s,
+                         rinv.data_ptr<float>(), means.data_ptr<float>(), mp,
+                         d);
+    }
   } else {
-    // working buffer + read-only LU snapshot
-    const size_t lds = sizeof(float) * 2 * (size_t)d * d;
+    // working buffer + read-only LU snapshot (+ u0 scratch for the factor)
+    const size_t lds = sizeof(float) * (2 * (size_t)d * d + d);
     if (lds > 64 * 1024) {  // gfx950: 160 KiB LDS/CU; opt in past 64 KiB
       HIP_CHECK(hipFuncSetAttribute(
           reinterpret_cast<const void*>(&gmm::constants_lu_kernel),
           hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds));
     }
     hipLaunchKernelGGL(gmm::constants_lu_kernel, dim3(k), dim3(kNT), lds, s,
-                       r.data_ptr<float>(), rinv.data_ptr<float>(),
-                       logdet.data_ptr<float>(), d);
+                       r.data_ptr<float>(), means.data_ptr<float>(),
+                       rinv.data_ptr<float>(), logdet.data_ptr<float>(), mp,
+                       d);
   }
+  HIP_CHECK(hipGetLastError());
+}
+
+void mstep_moments(torch::Tensor x, torch::Tensor w, torch::Tensor partials) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() &&
+                  x.scalar_type() == torch::kFloat32,
+              "x must be contiguous fp32 (exact M-step)");
+  check_f32(w, "w");
+  check_f32(partials, "partials");
+  const int d = (int)x.size(0);
+  const int64_t n = x.size(1);
+  const int k = (int)w.size(0);
+  const int nchunk = (int)partials.size(0);
+  const int dp = d + 1;
+  TORCH_CHECK(d <= 31, "mstep_moments fast path needs D <= 31");
+  TORCH_CHECK(partials.size(1) == k &&
+                  partials.size(2) == dp * (dp + 1) / 2,
+              "partials must be [nchunk, K, Dp*(Dp+1)/2]");
+  const size_t lds = sizeof(float) * ((size_t)d * (128 + 4) + 4 * 128);
+  dim3 grid((k + 3) / 4, nchunk);
+  hipLaunchKernelGGL((gmm::mstep_moments_kernel<float>), grid, dim3(kNT), lds,
+                     stream(), x.data_ptr<float>(), w.data_ptr<float>(),
+                     partials.data_ptr<float>(), d, k, n, nchunk);
+  HIP_CHECK(hipGetLastError());
+}
+
+void estep_fused(torch::Tensor z, torch::Tensor mfac, torch::Tensor add,
+                 torch::Tensor w_out, torch::Tensor partial) {
+  TORCH_CHECK(z.is_cuda() && z.is_contiguous() &&
+                  z.scalar_type() == torch::kBFloat16,
+              "z must be contiguous bf16 [D,N]");
+  TORCH_CHECK(mfac.is_contiguous() && mfac.scalar_type() == torch::kBFloat16,
+              "mfac must be bf16");
+  check_f32(add, "add");
+  check_f32(w_out, "w_out");
+  check_f32(partial, "partial");
+  const int d = (int)z.size(0);
+  const int64_t n = z.size(1);
+  const int k = (int)add.size(0);
+  TORCH_CHECK(d <= 31, "estep_fused needs D <= 31");
+  TORCH_CHECK(mfac.numel() >= (int64_t)k * 2 * 32 * 32, "mfac too small");
+  TORCH_CHECK(w_out.size(0) == k && w_out.size(1) == n, "w_out shape");
+  const int64_t nblk = (n + 128 - 1) / 128;
+  TORCH_CHECK(partial.size(0) >= nblk, "partial buffer too small");
+  const size_t zbytes = ((size_t)d * (128 + 8) * 2 + 3) & ~(size_t)3;
+  const size_t lds = zbytes + sizeof(float) * (size_t)k * (128 + 4);
+  TORCH_CHECK(lds <= 64 * 1024, "estep_fused LDS budget exceeded (K too big)");
+  hipLaunchKernelGGL(gmm::estep_fused_kernel, dim3((uint32_t)nblk), dim3(kNT),
+                     lds, stream(),
+                     reinterpret_cast<const __hip_bfloat16*>(z.data_ptr()),
+                     reinterpret_cast<const __hip_bfloat16*>(mfac.data_ptr()),
+                     add.data_ptr<float>(), w_out.data_ptr<float>(),
+                     partial.data_ptr<float>(), d, k, n);
+  HIP_CHECK(hipGetLastError());
+}
+
+void mfma_probe(torch::Tensor a, torch::Tensor b, torch::Tensor c) {
+  TORCH_CHECK(a.scalar_type() == torch::kBFloat16 && a.numel() == 16 * 32);
+  TORCH_CHECK(b.scalar_type() == torch::kBFloat16 && b.numel() == 32 * 16);
+  check_f32(c, "c");
+  hipLaunchKernelGGL(gmm::mfma_probe_kernel, dim3(1), dim3(64), 0, stream(),
+                     reinterpret_cast<const __hip_bfloat16*>(a.data_ptr()),
+                     reinterpret_cast<const __hip_bfloat16*>(b.data_ptr()),
+                     c.data_ptr<float>());
   HIP_CHECK(hipGetLastError());
 }
 
@@ -200,5 +283,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "in-place posteriors + per-block likelihood partials");
   m.def("mstep_covariance_partials", &mstep_covariance_partials,
         "packed weighted second-moment partials [nchunk,K,P]");
-  m.def("constants", &constants, "batched no-pivot LU inverse + ln|det|");
+  m.def("constants", &constants,
+        "batched no-pivot LU inverse + ln|det| + bf16 Cholesky factors");
+  m.def("mstep_moments", &mstep_moments,
+        "fused augmented moments [S|mean_num|N] via f32 MFMA");
+  m.def("estep_fused", &estep_fused,
+        "fused bf16-MFMA E-step: posteriors + likelihood partials");
+  m.def("mfma_probe", &mfma_probe, "bf16 MFMA fragment-layout probe");
 }

@@ -39,6 +39,12 @@ __device__ inline void tri_row_col(int t, int* i, int* j) {
   *j = t - r * (r + 1) / 2;
 }
 
+// k index held by lane group (l>>4) element u for mfma_f32_16x16x32_bf16
+// A/B fragments. Verified on hardware by the mfma_probe test
+// (tests/test_gpu_kernels.py::test_mfma_bf16_probe); flip to the split-K
+// variant (4*(l>>4) + u%4 + 16*(u/4)) if the probe ever disagrees.
+__device__ inline int mfma_b16_k(int group, int u) { return 8 * group + u; }
+
 // Stage cluster c's means + packed/pre-symmetrized Rinv into LDS.
 // lds layout: [0, d) means, [d, d + d(d+1)/2) packed rinv.
 __device__ inline void stage_cluster_params(
@@ -283,9 +289,59 @@ mstep_cov_kernel(const T* __restrict__ x, const float* __restrict__ w,
 // sized for D up to 128 — no NUM_DIMENSIONS cap).
 // One workgroup per cluster. Crout-style: L keeps the diagonal, U unit.
 // ---------------------------------------------------------------------------
+// Emit the fused-E-step factor M = [U | -U mu] (U^T U = Rinv, upper
+// Cholesky) as bf16 hi/lo pairs into mfac[c][2][32][32], stored in MFMA
+// A-fragment k-order (mfma_b16_k). `u` is scratch LDS holding Rinv [d*d]
+// (upper triangle becomes U in place) + u0 [d].
+__device__ inline void emit_mfac(float* u, const float* __restrict__ means,
+                                 __hip_bfloat16* __restrict__ mfac, int c,
+                                 int d) {
+  const int tid = threadIdx.x;
+  float* u0 = u + d * d;
+  // in-place upper Cholesky of Rinv (SPD up to rounding; clamped pivots)
+  for (int j = 0; j < d; ++j) {
+    if (tid == 0) {
+      float s = u[j * d + j];
+      for (int kk = 0; kk < j; ++kk) s -= u[kk * d + j] * u[kk * d + j];
+      u[j * d + j] = sqrtf(fmaxf(s, 1e-30f));
+    }
+    __syncthreads();
+    const float piv = u[j * d + j];
+    for (int i = j + 1 + tid; i < d; i += NT) {
+      float s = u[j * d + i];
+      for (int kk = 0; kk < j; ++kk) s -= u[kk * d + j] * u[kk * d + i];
+      u[j * d + i] = s / piv;
+    }
+    __syncthreads();
+  }
+  // u0 = -U mu (mu = centered cluster means)
+  for (int i = tid; i < d; i += NT) {
+    float s = 0.0f;
+    for (int j = i; j < d; ++j) s += u[i * d + j] * means[c * d + j];
+    u0[i] = -s;
+  }
+  __syncthreads();
+  // write hi/lo bf16 fragments, rows padded to 32, k-order per mfma_b16_k
+  __hip_bfloat16* out = mfac + (int64_t)c * 2 * 32 * 32;
+  for (int t = tid; t < 32 * 32; t += NT) {
+    const int i = t / 32, slot = t % 32;
+    const int kx = mfma_b16_k(slot / 8, slot % 8);
+    float v = 0.0f;
+    if (i < d) {
+      if (kx < d) v = (kx >= i) ? u[i * d + kx] : 0.0f;
+      else if (kx == d) v = u0[i];
+    }
+    const __hip_bfloat16 hi = __float2bfloat16(v);
+    out[t] = hi;
+    out[32 * 32 + t] = __float2bfloat16(v - __bfloat162float(hi));
+  }
+}
+
 __global__ void __launch_bounds__(NT)
-constants_lu_kernel(const float* __restrict__ r, float* __restrict__ rinv,
-                    float* __restrict__ logdet, int d) {
+constants_lu_kernel(const float* __restrict__ r,
+                    const float* __restrict__ means, float* __restrict__ rinv,
+                    float* __restrict__ logdet,
+                    __hip_bfloat16* __restrict__ mfac, int d) {
   // lds: a[d*d] working buffer, o[d*d] read-only snapshot of the LU factor.
   // The reference's in-place triangular inversion reads a mix of original
   // and already-inverted entries in a serial order; parallelized across
@@ -305,7 +361,10 @@ constants_lu_kernel(const float* __restrict__ r, float* __restrict__ rinv,
     if (tid == 0) {
       logdet[c] = __logf(a[0]);
       oc[0] = 1.0f / a[0];
+      o[0] = oc[0];
     }
+    __syncthreads();
+    if (mfac != nullptr) emit_mfac(o, means, mfac, c, 1);
     return;
   }
 
@@ -390,7 +449,277 @@ constants_lu_kernel(const float* __restrict__ r, float* __restrict__ rinv,
     for (int kk = (i > j ? i : j); kk < d; ++kk)
       s = fmaf((j == kk) ? 1.0f : a[j * d + kk], a[kk * d + i], s);
     oc[j * d + i] = s;
+    o[j * d + i] = s;  // LDS copy for the Cholesky factor below
   }
+  if (mfac != nullptr) {
+    __syncthreads();
+    emit_mfac(o, means, mfac, c, d);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// MFMA M-step: fused augmented moments  T_c = sum_e w_ce [x;1][x;1]^T
+// via v_mfma_f32_16x16x4_f32 (exact f32 fmaf chain — bitwise an f32 VALU
+// loop, guide §3) — one kernel produces S, the mean numerators AND N
+// (replaces mstep_N + mstep_means + mstep_covariance1,
+// gaussian_kernel.cu:522-677, and the separate rocBLAS GEMM).
+//
+// Packed output layout per cluster (lower triangle of (D+1)x(D+1)):
+//   [0, P)        S rows 0..D-1   (P = D(D+1)/2)
+//   [P, P+D)      mean numerators (row D, cols 0..D-1)
+//   [P+D]         N               (row D, col D)
+// Grid (ceil(K/4), nchunk): 4 waves per block, one cluster per wave, all
+// sharing the LDS event tile. D <= 31 (Dp = D+1 <= 32: 2 row-tiles).
+// f32 MFMA fragment maps (guide §3): A lane l -> A[i=l&15][k=l>>4],
+// B lane l -> B[k=l>>4][j=l&15]; C/D: col=l&15, row=(l>>4)*4+reg.
+// ---------------------------------------------------------------------------
+#define MOM_BK 128
+
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+template <typename T>
+__global__ void __launch_bounds__(NT)
+mstep_moments_kernel(const T* __restrict__ x, const float* __restrict__ w,
+                     float* __restrict__ partials, int d, int k, int64_t n,
+                     int nchunk) {
+  extern __shared__ float lds[];
+  const int row = MOM_BK + 4;
+  float* xs = lds;                       // [d][MOM_BK+4]
+  float* wt = lds + (int64_t)d * row;    // [4][MOM_BK]
+
+  const int wave = threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int c = blockIdx.x * 4 + wave;
+  const int chunk = blockIdx.y;
+  const int dp = d + 1;
+  const int rt2 = dp > 16;               // second row-tile present?
+
+  f32x4 acc00 = {0, 0, 0, 0}, acc10 = {0, 0, 0, 0}, acc11 = {0, 0, 0, 0};
+
+  const int i_loc = lane & 15;
+  const int kk = lane >> 4;              // event sub-index 0..3
+
+  const int64_t tiles = (n + MOM_BK - 1) / MOM_BK;
+  for (int64_t tile = chunk; tile < tiles; tile += nchunk) {
+    const int64_t e0 = tile * MOM_BK;
+    const int cnt = (int)min((int64_t)MOM_BK, n - e0);
+    __syncthreads();
+    for (int idx = threadIdx.x; idx < d * MOM_BK; idx += NT) {
+      const int di = idx / MOM_BK, ei = idx % MOM_BK;
+      xs[di * row + ei] =
+          (ei < cnt) ? load_x(x, (int64_t)di * n + e0 + ei) : 0.0f;
+    }
+    for (int idx = threadIdx.x; idx < 4 * MOM_BK; idx += NT) {
+      const int wv = idx / MOM_BK, ei = idx % MOM_BK;
+      const int cw = blockIdx.x * 4 + wv;
+      wt[wv * MOM_BK + ei] =
+          (cw < k && ei < cnt) ? w[(int64_t)cw * n + e0 + ei] : 0.0f;
+    }
+    __syncthreads();
+
+    for (int ks = 0; ks < MOM_BK / 4; ++ks) {
+      const int e = ks * 4 + kk;
+      const float we = wt[wave * MOM_BK + e];
+      // z values for this lane's row position in each row-tile
+      const int g0 = i_loc;             // row-tile 0 rows 0..15
+      const int g1 = 16 + i_loc;        // row-tile 1 rows 16..31
+      const float z0 = (g0 < d) ? xs[g0 * row + e] : (g0 == d ? 1.0f : 0.0f);
+      const float z1 = (g1 < d) ? xs[g1 * row + e] : (g1 == d ? 1.0f : 0.0f);
+      const float a0 = we * z0;
+      const float b0 = z0;
+      acc00 = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b0, acc00, 0, 0, 0);
+      if (rt2) {
+        const float a1 = we * z1;
+        acc10 = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b0, acc10, 0, 0, 0);
+        acc11 = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, z1, acc11, 0, 0, 0);
+      }
+    }
+  }
+
+  if (c >= k) return;
+  const int p_aug = dp * (dp + 1) / 2;
+  float* out = partials + ((int64_t)chunk * k + c) * p_aug;
+  // C/D layout: col = lane&15, row = (lane>>4)*4 + reg
+  const int col = lane & 15;
+  const int row0 = (lane >> 4) * 4;
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    {  // tile (0,0): keep lower triangle
+      const int gi = row0 + r, gj = col;
+      if (gi < dp && gj <= gi) out[gi * (gi + 1) / 2 + gj] = acc00[r];
+    }
+    if (rt2) {
+      const int gi = 16 + row0 + r;
+      {  // tile (1,0)
+        const int gj = col;
+        if (gi < dp) out[gi * (gi + 1) / 2 + gj] = acc10[r];
+      }
+      {  // tile (1,1)
+        const int gj = 16 + col;
+        if (gi < dp && gj <= gi) out[gi * (gi + 1) / 2 + gj] = acc11[r];
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// MFMA fused E-step (replaces estep1 + estep2 in one pass for bf16, D<=31,
+// moderate K): logw via  q = ||M_c z||^2  with M_c = [U_c | -U_c mu_c],
+// U_c^T U_c = Rinv_c (Cholesky, emitted by the constants kernel), split into
+// bf16 hi+lo parts so accuracy matches a bf16-data fp32-VALU quadratic form.
+// Per block: stage z tile once, loop clusters (B fragments built once and
+// reused across all K!), logw kept in LDS, then in-block posteriors +
+// likelihood partial — memberships are written ONCE and logw never touches
+// HBM (vs 3 reads + 2 writes of [K,N] in the two-kernel path).
+// v_mfma_f32_16x16x32_bf16: A lane l -> A[i=l&15][k=8*(l>>4)+u];
+// B lane l -> B[k=8*(l>>4)+u][j=l&15]; C/D col=l&15, row=(l>>4)*4+reg.
+// ---------------------------------------------------------------------------
+#define EST_BE 128
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+
+
+// Layout probe: D = A(16x32) @ B(32x16) in bf16 via one MFMA.
+__global__ void mfma_probe_kernel(const __hip_bfloat16* __restrict__ a,
+                                  const __hip_bfloat16* __restrict__ b,
+                                  float* __restrict__ c) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  bf16x8 av, bv;
+#pragma unroll
+  for (int u = 0; u < 8; ++u) {
+    const int kx = mfma_b16_k(lane >> 4, u);
+    av[u] = (__bf16)__bfloat162float(a[(lane & 15) * 32 + kx]);
+    bv[u] = (__bf16)__bfloat162float(b[kx * 16 + (lane & 15)]);
+  }
+  f32x4 acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+      av, bv, (f32x4){0, 0, 0, 0}, 0, 0, 0);
+#pragma unroll
+  for (int r = 0; r < 4; ++r)
+    c[((lane >> 4) * 4 + r) * 16 + (lane & 15)] = acc[r];
+}
+
+__global__ void __launch_bounds__(NT)
+estep_fused_kernel(const __hip_bfloat16* __restrict__ z,
+                   const __hip_bfloat16* __restrict__ mfac,  // [K][2][32][32]
+                   const float* __restrict__ add,            // const + ln pi
+                   float* __restrict__ w_out, float* __restrict__ partial,
+                   int d, int k, int64_t n) {
+  extern __shared__ float lds[];
+  const int zrow = EST_BE + 8;   // bf16 elements per z row (pad)
+  const int lrow = EST_BE + 4;   // f32 elements per logw row
+  __hip_bfloat16* zs = (__hip_bfloat16*)lds;            // [d][zrow]
+  float* lw = lds + ((int64_t)d * zrow + 1) / 2;        // [k][lrow]
+
+  const int wave = threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int64_t e0 = (int64_t)blockIdx.x * EST_BE;
+  const int cnt = (int)min((int64_t)EST_BE, n - e0);
+
+  for (int idx = threadIdx.x; idx < d * EST_BE; idx += NT) {
+    const int di = idx / EST_BE, ei = idx % EST_BE;
+    zs[di * zrow + ei] = (ei < cnt)
+        ? z[(int64_t)di * n + e0 + ei]
+        : __hip_bfloat16(0.0f);
+  }
+  __syncthreads();
+
+  // B fragments for this wave's two 16-event tiles (reused across clusters)
+  const int j_loc = lane & 15;
+  const int kbase = 8 * (lane >> 4);
+  bf16x8 bfrag[2];
+#pragma unroll
+  for (int t = 0; t < 2; ++t) {
+    const int e = wave * 32 + t * 16 + j_loc;
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+      const int krow = mfma_b16_k(lane >> 4, u);
+      float v;
+      if (krow < d) v = __bfloat162float(zs[krow * zrow + e]);
+      else if (krow == d) v = 1.0f;
+      else v = 0.0f;
+      bfrag[t][u] = (__bf16)v;
+    }
+  }
+
+  const int dp = d + 1;
+  const int rt2 = dp > 16;
+  const bf16x8* mf = (const bf16x8*)mfac;  // rows of 32 bf16 = 4 frags each
+
+  for (int c = 0; c < k; ++c) {
+    // A fragments: M rows for this lane, hi and lo parts, both row-tiles
+    // mfac[c][h][i][kbase..kbase+7], row stride 32 bf16 = 4 bf16x8
+    const int64_t base = ((int64_t)c * 2) * 32 * 4;  // in bf16x8 units
+    const int i_loc = lane & 15;
+    const int fq = kbase / 8;
+    const bf16x8 a_hi0 = mf[base + i_loc * 4 + fq];
+    const bf16x8 a_lo0 = mf[base + 32 * 4 + i_loc * 4 + fq];
+    bf16x8 a_hi1, a_lo1;
+    if (rt2) {
+      a_hi1 = mf[base + (16 + i_loc) * 4 + fq];
+      a_lo1 = mf[base + 32 * 4 + (16 + i_loc) * 4 + fq];
+    }
+    const float addc = add[c];
+#pragma unroll
+    for (int t = 0; t < 2; ++t) {
+      f32x4 y0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          a_hi0, bfrag[t], (f32x4){0, 0, 0, 0}, 0, 0, 0);
+      y0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_lo0, bfrag[t], y0, 0, 0, 0);
+      float s = y0[0] * y0[0] + y0[1] * y0[1] + y0[2] * y0[2] + y0[3] * y0[3];
+      if (rt2) {
+        f32x4 y1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a_hi1, bfrag[t], (f32x4){0, 0, 0, 0}, 0, 0, 0);
+        y1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_lo1, bfrag[t], y1,
+                                                     0, 0, 0);
+        s += y1[0] * y1[0] + y1[1] * y1[1] + y1[2] * y1[2] + y1[3] * y1[3];
+      }
+      // rows of the padded Y are spread over lane groups: sum across them
+      s += __shfl_xor(s, 16, WAVE);
+      s += __shfl_xor(s, 32, WAVE);
+      if (lane < 16) {
+        const int e = wave * 32 + t * 16 + lane;
+        lw[c * lrow + e] = -0.5f * s + addc;
+      }
+    }
+  }
+  __syncthreads();
+
+  // pass 2: posteriors + likelihood over this block's events
+  float acc = 0.0f;
+  if (threadIdx.x < EST_BE && threadIdx.x < cnt) {
+    const int t = threadIdx.x;
+    float m = lw[t];
+    for (int c = 1; c < k; ++c) m = fmaxf(m, lw[c * lrow + t]);
+    float s = 0.0f;
+    for (int c = 0; c < k; ++c) s += __expf(lw[c * lrow + t] - m);
+    const float denom = m + __logf(s);
+    for (int c = 0; c < k; ++c)
+      w_out[(int64_t)c * n + e0 + t] = __expf(lw[c * lrow + t] - denom);
+    acc = denom;
+  }
+  __shared__ float wsum[NT / WAVE];
+  for (int off = WAVE / 2; off > 0; off >>= 1)
+    acc += __shfl_down(acc, off, WAVE);
+  if (lane == 0) wsum[wave] = acc;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float total = 0.0f;
+    for (int wv = 0; wv < NT / WAVE; ++wv) total += wsum[wv];
+    partial[blockIdx.x] = total;
+  }
+}
+
+// Factor emission from an existing Rinv (diag-only path): stage Rinv into
+// LDS and run the same Cholesky/emit as the LU kernel.
+__global__ void __launch_bounds__(NT)
+emit_mfac_from_rinv_kernel(const float* __restrict__ rinv,
+                           const float* __restrict__ means,
+                           __hip_bfloat16* __restrict__ mfac, int d) {
+  extern __shared__ float buf[];
+  const int c = blockIdx.x;
+  for (int t = threadIdx.x; t < d * d; t += NT)
+    buf[t] = rinv[(int64_t)c * d * d + t];
+  __syncthreads();
+  emit_mfac(buf, means, mfac, c, d);
 }
 
 // DIAG_ONLY constants (gaussian_kernel.cu:215-223)

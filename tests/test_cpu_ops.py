@@ -164,3 +164,21 @@ def test_diag_only_paths(rng):
     )
     np.testing.assert_allclose(logw.numpy(), full.numpy(), rtol=1e-4,
                                atol=1e-4)
+
+
+def test_mstep_moments_packed_layout(rng):
+    """functional.mstep_moments packed layout == [S_tri | mean_num | N]."""
+    import torch as t
+    from cuda_gmm_mpi_amd.ops import functional as F
+    k, d, n = 3, 5, 200
+    x = t.from_numpy(rng.standard_normal((d, n)).astype(np.float32))
+    w = t.from_numpy(rng.uniform(0, 1, (k, n)).astype(np.float32))
+    packed = F.mstep_moments(x, w)
+    assert packed.shape == (k, (d + 1) * (d + 2) // 2)
+    n_c, mean_num, s = F.moments_views(packed, d)
+    rn, rm, rs = cpu.mstep_sufficient_stats(x, w)
+    np.testing.assert_allclose(n_c.numpy(), rn.numpy(), rtol=1e-5)
+    np.testing.assert_allclose(mean_num.numpy(), rm.numpy(), rtol=1e-4)
+    np.testing.assert_allclose(s.numpy(), rs.numpy(), rtol=1e-3, atol=1e-3)
+    # unpacked S symmetric by construction
+    assert float((s - s.transpose(1, 2)).abs().max()) == 0.0

@@ -172,3 +172,120 @@ def test_missing_extension_raises_loudly(monkeypatch, device):
                      torch.zeros(1, 2, 2, device=device),
                      torch.zeros(1, device=device),
                      torch.ones(1, device=device))
+
+
+def test_mfma_bf16_probe(device):
+    """Empirically verify the assumed bf16 MFMA fragment layout (guide G9:
+    asymmetric B catches operand/output transposes)."""
+    from cuda_gmm_mpi_amd.ops.backend import hip_ext
+    rng = np.random.default_rng(77)
+    a = rng.standard_normal((16, 32)).astype(np.float32)
+    b = (rng.standard_normal((32, 16)) + np.arange(16)[None, :]).astype(
+        np.float32)  # asymmetric
+    at = torch.from_numpy(a).to(device).to(torch.bfloat16)
+    bt = torch.from_numpy(b).to(device).to(torch.bfloat16)
+    c = torch.zeros(16, 16, dtype=torch.float32, device=device)
+    hip_ext().mfma_probe(at.contiguous(), bt.contiguous(), c)
+    ref = at.float() @ bt.float()
+    np.testing.assert_allclose(c.cpu().numpy(), ref.cpu().numpy(),
+                               rtol=2e-2, atol=2e-2)
+
+
+@pytest.mark.parametrize("d,n", [(2, 3001), (15, 5000), (16, 5000),
+                                 (24, 100000), (31, 4097)])
+def test_mstep_moments_matches_cpu(device, d, n):
+    from cuda_gmm_mpi_amd.ops import functional as F
+    rng = np.random.default_rng(d * 1000 + 7)
+    k = 6
+    x = rng.standard_normal((d, n)).astype(np.float32)
+    w = rng.uniform(0, 1, (k, n)).astype(np.float32)
+    packed = F.mstep_moments(torch.from_numpy(x).to(device),
+                             torch.from_numpy(w).to(device))
+    n_c, mean_num, s = F.moments_views(packed, d)
+    rn, rm, rs = cpu.mstep_sufficient_stats(torch.from_numpy(x),
+                                            torch.from_numpy(w))
+    np.testing.assert_allclose(n_c.cpu().numpy(), rn.numpy(), rtol=1e-4)
+    np.testing.assert_allclose(mean_num.cpu().numpy(), rm.numpy(),
+                               rtol=1e-3, atol=1e-2)
+    np.testing.assert_allclose(s.cpu().numpy(), rs.numpy(),
+                               rtol=2e-3, atol=5e-2)
+
+
+def test_mstep_moments_determinism(device):
+    from cuda_gmm_mpi_amd.ops import functional as F
+    rng = np.random.default_rng(3)
+    d, n, k = 24, 200000, 16
+    x = torch.from_numpy(rng.standard_normal((d, n)).astype(np.float32)).to(device)
+    w = torch.from_numpy(rng.uniform(0, 1, (k, n)).astype(np.float32)).to(device)
+    assert torch.equal(F.mstep_moments(x, w), F.mstep_moments(x, w))
+
+
+def test_estep_fused_matches_cpu(device):
+    """Fused bf16 MFMA E-step vs plain fp32 torch reference."""
+    from cuda_gmm_mpi_amd.ops import functional as F
+    rng = np.random.default_rng(11)
+    k, d, n = 8, 24, 20000
+    means, r, pi = random_model(rng, k, d, device)
+    rinv, const = cpu.compute_constants(r.cpu())
+    x = rng.standard_normal((d, n)).astype(np.float32) * 2
+    xt = torch.from_numpy(x).to(device)
+    # emit mfac via the constants kernel
+    mfac = torch.empty(k, 2, 32, 32, dtype=torch.bfloat16, device=device)
+    rinv_g, const_g = F.constants(r, means, False, mfac)
+    add = const_g + torch.log(pi)
+    w_out = torch.empty(k, n, dtype=torch.float32, device=device)
+    w, lik = F.estep_fused(xt.to(torch.bfloat16), mfac, add, w_out)
+    # reference: logw + posteriors in fp32
+    ref_logw = cpu.estep_logw(torch.from_numpy(x), means.cpu(), rinv, const,
+                              pi.cpu())
+    ref_w, ref_lik = cpu.estep_posteriors(ref_logw)
+    # bf16 data + bf16 hi/lo factors: posteriors within bf16-class tolerance
+    np.testing.assert_allclose(w.cpu().numpy(), ref_w.numpy(),
+                               rtol=5e-2, atol=2e-2)
+    assert float(lik) == pytest.approx(float(ref_lik), rel=2e-3)
+    np.testing.assert_allclose(w.sum(dim=0).cpu().numpy(), np.ones(n),
+                               rtol=1e-3)
+
+
+def test_estep_fused_tail_events(device):
+    """n not a multiple of the 128-event block."""
+    from cuda_gmm_mpi_amd.ops import functional as F
+    rng = np.random.default_rng(13)
+    k, d, n = 4, 8, 1000 + 37
+    means, r, pi = random_model(rng, k, d, device)
+    mfac = torch.empty(k, 2, 32, 32, dtype=torch.bfloat16, device=device)
+    rinv_g, const_g = F.constants(r, means, False, mfac)
+    add = const_g + torch.log(pi)
+    x = rng.standard_normal((d, n)).astype(np.float32)
+    w_out = torch.empty(k, n, dtype=torch.float32, device=device)
+    w, lik = F.estep_fused(
+        torch.from_numpy(x).to(device).to(torch.bfloat16), mfac, add, w_out)
+    rinv, const = cpu.compute_constants(r.cpu())
+    ref_logw = cpu.estep_logw(torch.from_numpy(x), means.cpu(), rinv, const,
+                              pi.cpu())
+    ref_w, ref_lik = cpu.estep_posteriors(ref_logw)
+    np.testing.assert_allclose(w.cpu().numpy(), ref_w.numpy(),
+                               rtol=5e-2, atol=2e-2)
+    assert float(lik) == pytest.approx(float(ref_lik), rel=5e-3)
+
+
+def test_constants_emits_valid_cholesky(device):
+    """mfac hi+lo must reconstruct M with U^T U = Rinv and u0 = -U mu."""
+    from cuda_gmm_mpi_amd.ops import functional as F
+    rng = np.random.default_rng(19)
+    k, d = 5, 24
+    means, r, pi = random_model(rng, k, d, device)
+    mfac = torch.empty(k, 2, 32, 32, dtype=torch.bfloat16, device=device)
+    rinv, const = F.constants(r, means, False, mfac)
+    m = mfac.float()
+    mm = (m[:, 0] + m[:, 1]).cpu().numpy()  # [K, 32, 32] fragment k-order
+    # identity k-map: column j of the stored fragment == M column j
+    for c in range(k):
+        u = mm[c][:d, :d]
+        np.testing.assert_allclose(u.T @ u, rinv[c].cpu().numpy(),
+                                   rtol=2e-2, atol=2e-2)
+        u0 = mm[c][:d, d]
+        np.testing.assert_allclose(
+            u0, -(u @ means[c].cpu().numpy()), rtol=2e-2, atol=2e-2)
+        assert np.abs(mm[c][:, d + 1:]).max() == 0.0
+        assert np.abs(mm[c][d:, :]).max() == 0.0
