@@ -97,6 +97,12 @@ class EmEngine:
                              device=self.device)
         self.epsilon = em_epsilon(self.d, num_events_total)
         self.likelihood = 0.0
+        # hipGraph capture of the EM iteration (launch-overhead elimination);
+        # per-K graphs, lazily captured. Disable with use_graphs=False or
+        # GMM_NO_GRAPHS=1 (e.g. when per-bucket profiling is wanted).
+        self.use_graphs = True
+        self._graphs: dict[int, object] = {}
+        self._lik_dev = torch.zeros(1, dtype=torch.float32, device=self.device)
 
     def _update_constants(self, st: GmmState) -> None:
         k = st.num_clusters
@@ -127,11 +133,34 @@ class EmEngine:
         self.profile.count("regroup")
         return lik
 
-    def _reduce_likelihood(self, lik_part: torch.Tensor) -> float:
+    def _finish_likelihood(self, lik_part: torch.Tensor) -> None:
+        """All-reduce the shard likelihood into the persistent device
+        scalar (graph-capturable: no host sync)."""
         with self.profile.time("comm"):
             t = lik_part.reshape(1)
             pdist.all_reduce_(t)
-        return float(t.item())
+            self._lik_dev.copy_(t)
+
+    def _reduce_likelihood(self, lik_part: torch.Tensor) -> float:
+        self._finish_likelihood(lik_part)
+        return float(self._lik_dev.item())
+
+    def _iteration_body(self, k: int) -> None:
+        """One EM iteration ending with the reduced likelihood on device."""
+        self._mstep(k)
+        self._finish_likelihood(self._estep(k))
+
+    def _can_graph(self) -> bool:
+        import os
+        if not self.use_graphs or os.environ.get("GMM_NO_GRAPHS"):
+            return False
+        if self.device.type != "cuda":
+            return False
+        if self.world > 1:
+            import torch.distributed as dist
+            if dist.get_backend() != "nccl":
+                return False
+        return True
 
     def _mstep(self, k: int) -> None:
         """M-step: fused moments, ONE all-reduce, finalize params + constants.
@@ -170,20 +199,47 @@ class EmEngine:
             abs(change) > self.epsilon and iters < cfg.max_iters
         ):
             old_lik = lik
-            self._mstep(k)
-            lik = self._reduce_likelihood(self._estep(k))
+            self.em_iteration(k)
+            lik = float(self._lik_dev.item())
             change = lik - old_lik
             iters += 1
         self.likelihood = lik
         return lik
 
     def em_iteration(self, k: int) -> None:
-        """One benchmark step: M-step + constants + E-step + reduces.
+        """One EM iteration: M-step + all-reduce + constants + E-step +
+        likelihood reduce. On GPU the whole sequence replays as one
+        hipGraph (captured lazily per K; the capturing call runs the
+        iteration eagerly so iteration counts stay exact).
 
         (Requires a prior _estep so self.w holds posteriors.)
         """
-        self._mstep(k)
-        self._reduce_likelihood(self._estep(k))
+        if self._can_graph():
+            g = self._graphs.get(k)
+            if g is None:
+                g = self._capture_iteration(k)
+                if g is not None:
+                    return  # the capture's eager warmup WAS this iteration
+            if g:
+                g.replay()
+                return
+        self._iteration_body(k)
+
+    def _capture_iteration(self, k: int):
+        """Run one eager iteration (counts), then capture the graph."""
+        self._iteration_body(k)
+        try:
+            self.profile.paused = True
+            torch.cuda.synchronize()
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                self._iteration_body(k)
+            self._graphs[k] = g
+        except Exception:  # noqa: BLE001 — graphs are an optimization only
+            self._graphs[k] = False
+        finally:
+            self.profile.paused = False
+        return None
 
     # ------------------------------------------------- MDL sweep / merging
 

@@ -159,7 +159,7 @@ def estep_fused_available(device: torch.device, dtype: str, d: int,
 
 
 def mstep_moments(x: torch.Tensor, w: torch.Tensor,
-                  nchunk: int = 64) -> torch.Tensor:
+                  nchunk: int | None = None) -> torch.Tensor:
     """Fused augmented sufficient statistics, packed lower triangle of
     T_c = sum_e w_ce [x;1][x;1]^T per cluster: [K, Dp*(Dp+1)/2] with layout
     [S_tri (D rows) | mean numerators (row D) | N (corner)] — the single
@@ -175,6 +175,11 @@ def mstep_moments(x: torch.Tensor, w: torch.Tensor,
     p = d * (d + 1) // 2
     if x.is_cuda and d <= 31:
         tiles = (n + 127) // 128
+        if nchunk is None:
+            # enough chunk-parallelism to fill 8 XCDs x 32 CUs (measured:
+            # 512 chunks beat 64 by ~25% at K=64), capped so the partial
+            # buffer stays <= 64 MB
+            nchunk = max(1, min(512, (64 << 20) // (4 * k * pp)))
         nchunk = int(min(nchunk, tiles))
         partials = torch.empty((nchunk, k, pp), dtype=torch.float32,
                                device=x.device)

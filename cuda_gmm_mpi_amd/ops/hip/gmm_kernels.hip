@@ -517,6 +517,7 @@ mstep_moments_kernel(const T* __restrict__ x, const float* __restrict__ w,
     }
     __syncthreads();
 
+#pragma unroll 4
     for (int ks = 0; ks < MOM_BK / 4; ++ks) {
       const int e = ks * 4 + kk;
       const float we = wt[wave * MOM_BK + e];
@@ -644,41 +645,70 @@ estep_fused_kernel(const __hip_bfloat16* __restrict__ z,
   const int dp = d + 1;
   const int rt2 = dp > 16;
   const bf16x8* mf = (const bf16x8*)mfac;  // rows of 32 bf16 = 4 frags each
+  const int i_loc = lane & 15;
+  const int fq = kbase / 8;
+
+  // software-pipelined A-fragment loads: cluster c+1's factors stream from
+  // L2 while cluster c's MFMAs run (the per-cluster load latency otherwise
+  // serializes the K loop)
+  bf16x8 nx_hi0, nx_lo0, nx_hi1, nx_lo1;
+  float nx_add;
+  auto load_a = [&](int c) {
+    const int64_t base = ((int64_t)c * 2) * 32 * 4;  // in bf16x8 units
+    nx_hi0 = mf[base + i_loc * 4 + fq];
+    nx_lo0 = mf[base + 32 * 4 + i_loc * 4 + fq];
+    if (rt2) {
+      nx_hi1 = mf[base + (16 + i_loc) * 4 + fq];
+      nx_lo1 = mf[base + 32 * 4 + (16 + i_loc) * 4 + fq];
+    }
+    nx_add = add[c];
+  };
+  load_a(0);
 
   for (int c = 0; c < k; ++c) {
-    // A fragments: M rows for this lane, hi and lo parts, both row-tiles
-    // mfac[c][h][i][kbase..kbase+7], row stride 32 bf16 = 4 bf16x8
-    const int64_t base = ((int64_t)c * 2) * 32 * 4;  // in bf16x8 units
-    const int i_loc = lane & 15;
-    const int fq = kbase / 8;
-    const bf16x8 a_hi0 = mf[base + i_loc * 4 + fq];
-    const bf16x8 a_lo0 = mf[base + 32 * 4 + i_loc * 4 + fq];
-    bf16x8 a_hi1, a_lo1;
+    const bf16x8 a_hi0 = nx_hi0, a_lo0 = nx_lo0;
+    const bf16x8 a_hi1 = nx_hi1, a_lo1 = nx_lo1;
+    const float addc = nx_add;
+    if (c + 1 < k) load_a(c + 1);
+    // issue all four independent MFMA chains (2 event-tiles x 2 row-tiles)
+    // before any epilogue math: the dependent-accumulator latency and the
+    // square/shuffle epilogue overlap across chains instead of serializing
+    f32x4 y00 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+        a_hi0, bfrag[0], (f32x4){0, 0, 0, 0}, 0, 0, 0);
+    f32x4 y01 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+        a_hi0, bfrag[1], (f32x4){0, 0, 0, 0}, 0, 0, 0);
+    f32x4 y10, y11;
     if (rt2) {
-      a_hi1 = mf[base + (16 + i_loc) * 4 + fq];
-      a_lo1 = mf[base + 32 * 4 + (16 + i_loc) * 4 + fq];
+      y10 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          a_hi1, bfrag[0], (f32x4){0, 0, 0, 0}, 0, 0, 0);
+      y11 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          a_hi1, bfrag[1], (f32x4){0, 0, 0, 0}, 0, 0, 0);
     }
-    const float addc = add[c];
-#pragma unroll
-    for (int t = 0; t < 2; ++t) {
-      f32x4 y0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-          a_hi0, bfrag[t], (f32x4){0, 0, 0, 0}, 0, 0, 0);
-      y0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_lo0, bfrag[t], y0, 0, 0, 0);
-      float s = y0[0] * y0[0] + y0[1] * y0[1] + y0[2] * y0[2] + y0[3] * y0[3];
-      if (rt2) {
-        f32x4 y1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            a_hi1, bfrag[t], (f32x4){0, 0, 0, 0}, 0, 0, 0);
-        y1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_lo1, bfrag[t], y1,
-                                                     0, 0, 0);
-        s += y1[0] * y1[0] + y1[1] * y1[1] + y1[2] * y1[2] + y1[3] * y1[3];
-      }
-      // rows of the padded Y are spread over lane groups: sum across them
-      s += __shfl_xor(s, 16, WAVE);
-      s += __shfl_xor(s, 32, WAVE);
-      if (lane < 16) {
-        const int e = wave * 32 + t * 16 + lane;
-        lw[c * lrow + e] = -0.5f * s + addc;
-      }
+    y00 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_lo0, bfrag[0], y00, 0, 0, 0);
+    y01 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_lo0, bfrag[1], y01, 0, 0, 0);
+    if (rt2) {
+      y10 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_lo1, bfrag[0], y10, 0, 0, 0);
+      y11 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_lo1, bfrag[1], y11, 0, 0, 0);
+    }
+    float s0 = y00[0] * y00[0] + y00[1] * y00[1] + y00[2] * y00[2] +
+               y00[3] * y00[3];
+    float s1 = y01[0] * y01[0] + y01[1] * y01[1] + y01[2] * y01[2] +
+               y01[3] * y01[3];
+    if (rt2) {
+      s0 += y10[0] * y10[0] + y10[1] * y10[1] + y10[2] * y10[2] +
+            y10[3] * y10[3];
+      s1 += y11[0] * y11[0] + y11[1] * y11[1] + y11[2] * y11[2] +
+            y11[3] * y11[3];
+    }
+    // rows of the padded Y are spread over lane groups: sum across them
+    s0 += __shfl_xor(s0, 16, WAVE);
+    s0 += __shfl_xor(s0, 32, WAVE);
+    s1 += __shfl_xor(s1, 16, WAVE);
+    s1 += __shfl_xor(s1, 32, WAVE);
+    if (lane < 16) {
+      const int e = wave * 32 + lane;
+      lw[c * lrow + e] = -0.5f * s0 + addc;
+      lw[c * lrow + e + 16] = -0.5f * s1 + addc;
     }
   }
   __syncthreads();

@@ -60,9 +60,13 @@ class Profile:
         use_events = dev.type == "cuda" and torch.cuda.is_available()
         self.timers = {b: _Timer(use_events) for b in BUCKETS}
         self.iterations = {"regroup": 0, "params": 0, "constants": 0, "reduce": 0}
+        self.paused = False  # hipEvents cannot be recorded inside graph capture
 
     @contextmanager
     def time(self, bucket: str):
+        if self.paused:
+            yield
+            return
         t = self.timers[bucket]
         t.start()
         try:
@@ -71,7 +75,8 @@ class Profile:
             t.stop()
 
     def count(self, which: str, n: int = 1):
-        self.iterations[which] += n
+        if not self.paused:
+            self.iterations[which] += n
 
     def report(self, rank: int = 0, gpu: int = 0) -> str:
         """Per-GPU report in the reference's shape (gaussian.cu:967)."""
