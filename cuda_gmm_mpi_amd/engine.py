@@ -217,9 +217,8 @@ class EmEngine:
         if self._can_graph():
             g = self._graphs.get(k)
             if g is None:
-                g = self._capture_iteration(k)
-                if g is not None:
-                    return  # the capture's eager warmup WAS this iteration
+                self._capture_iteration(k)
+                return  # the capture's eager warmup WAS this iteration
             if g:
                 g.replay()
                 return
