@@ -154,8 +154,7 @@ def estep_fused_available(device: torch.device, dtype: str, d: int,
     """Fused path gate (mirrors the LDS check in gmm_ext.hip)."""
     if device.type != "cuda" or dtype != "bf16" or d > 31:
         return False
-    zbytes = (d * 136 * 2 + 3) & ~3
-    return zbytes + 4 * k * 132 <= 64 * 1024
+    return 128 * 40 * 2 + 4 * k * 132 <= 64 * 1024
 
 
 def mstep_moments(x: torch.Tensor, w: torch.Tensor,

@@ -251,7 +251,7 @@ void estep_fused(torch::Tensor z, torch::Tensor mfac, torch::Tensor add,
   TORCH_CHECK(w_out.size(0) == k && w_out.size(1) == n, "w_out shape");
   const int64_t nblk = (n + 128 - 1) / 128;
   TORCH_CHECK(partial.size(0) >= nblk, "partial buffer too small");
-  const size_t zbytes = ((size_t)d * (128 + 8) * 2 + 3) & ~(size_t)3;
+  const size_t zbytes = (size_t)128 * 40 * 2;  // transposed z tile
   const size_t lds = zbytes + sizeof(float) * (size_t)k * (128 + 4);
   TORCH_CHECK(lds <= 64 * 1024, "estep_fused LDS budget exceeded (K too big)");
   hipLaunchKernelGGL(gmm::estep_fused_kernel, dim3((uint32_t)nblk), dim3(kNT),
@@ -260,6 +260,17 @@ void estep_fused(torch::Tensor z, torch::Tensor mfac, torch::Tensor add,
                      reinterpret_cast<const __hip_bfloat16*>(mfac.data_ptr()),
                      add.data_ptr<float>(), w_out.data_ptr<float>(),
                      partial.data_ptr<float>(), d, k, n);
+  HIP_CHECK(hipGetLastError());
+}
+
+void mfma_probe32(torch::Tensor a, torch::Tensor b, torch::Tensor c) {
+  TORCH_CHECK(a.scalar_type() == torch::kBFloat16 && a.numel() == 32 * 16);
+  TORCH_CHECK(b.scalar_type() == torch::kBFloat16 && b.numel() == 16 * 32);
+  check_f32(c, "c");
+  hipLaunchKernelGGL(gmm::mfma_probe32_kernel, dim3(1), dim3(64), 0, stream(),
+                     reinterpret_cast<const __hip_bfloat16*>(a.data_ptr()),
+                     reinterpret_cast<const __hip_bfloat16*>(b.data_ptr()),
+                     c.data_ptr<float>());
   HIP_CHECK(hipGetLastError());
 }
 
@@ -290,4 +301,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("estep_fused", &estep_fused,
         "fused bf16-MFMA E-step: posteriors + likelihood partials");
   m.def("mfma_probe", &mfma_probe, "bf16 MFMA fragment-layout probe");
+  m.def("mfma_probe32", &mfma_probe32, "32x32x16 bf16 layout probe");
 }

@@ -599,121 +599,115 @@ __global__ void mfma_probe_kernel(const __hip_bfloat16* __restrict__ a,
     c[((lane >> 4) * 4 + r) * 16 + (lane & 15)] = acc[r];
 }
 
+// Layout probe for the 32x32x16 shape: D = A(32x16) @ B(16x32).
+__global__ void mfma_probe32_kernel(const __hip_bfloat16* __restrict__ a,
+                                    const __hip_bfloat16* __restrict__ b,
+                                    float* __restrict__ c) {
+  typedef __attribute__((ext_vector_type(16))) float f32x16;
+  const int lane = threadIdx.x & (WAVE - 1);
+  bf16x8 av, bv;
+#pragma unroll
+  for (int u = 0; u < 8; ++u) {
+    const int kx = 8 * (lane >> 5) + u;
+    av[u] = (__bf16)__bfloat162float(a[(lane & 31) * 16 + kx]);
+    bv[u] = (__bf16)__bfloat162float(b[kx * 32 + (lane & 31)]);
+  }
+  f32x16 acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+      av, bv, (f32x16)(0.0f), 0, 0, 0);
+#pragma unroll
+  for (int r = 0; r < 16; ++r)
+    c[((r & 3) + 8 * (r >> 2) + 4 * (lane >> 5)) * 32 + (lane & 31)] = acc[r];
+}
+
+#define EST_ZROW 40  // bf16 per transposed-z row (32 k-slots + pad)
+
 __global__ void __launch_bounds__(NT)
 estep_fused_kernel(const __hip_bfloat16* __restrict__ z,
                    const __hip_bfloat16* __restrict__ mfac,  // [K][2][32][32]
                    const float* __restrict__ add,            // const + ln pi
                    float* __restrict__ w_out, float* __restrict__ partial,
                    int d, int k, int64_t n) {
+  // LDS: zs_t [EST_BE][EST_ZROW] bf16 — z staged TRANSPOSED (k-major per
+  // event, ones-row and zero-pad baked in) so a B fragment is a single
+  // 16-byte ds_read_b128; then lw [k][EST_BE+4] f32.
   extern __shared__ float lds[];
-  const int zrow = EST_BE + 8;   // bf16 elements per z row (pad)
-  const int lrow = EST_BE + 4;   // f32 elements per logw row
-  __hip_bfloat16* zs = (__hip_bfloat16*)lds;            // [d][zrow]
-  float* lw = lds + ((int64_t)d * zrow + 1) / 2;        // [k][lrow]
+  const int lrow = EST_BE + 4;
+  __hip_bfloat16* zs = (__hip_bfloat16*)lds;
+  float* lw = lds + (EST_BE * EST_ZROW) / 2;
 
   const int wave = threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
   const int64_t e0 = (int64_t)blockIdx.x * EST_BE;
   const int cnt = (int)min((int64_t)EST_BE, n - e0);
 
-  for (int idx = threadIdx.x; idx < d * EST_BE; idx += NT) {
-    const int di = idx / EST_BE, ei = idx % EST_BE;
-    zs[di * zrow + ei] = (ei < cnt)
-        ? z[(int64_t)di * n + e0 + ei]
-        : __hip_bfloat16(0.0f);
+  for (int idx = threadIdx.x; idx < EST_BE * 32; idx += NT) {
+    const int ei = idx >> 5, kk = idx & 31;
+    float v = 0.0f;
+    if (ei < cnt) {
+      if (kk < d) v = __bfloat162float(z[(int64_t)kk * n + e0 + ei]);
+      else if (kk == d) v = 1.0f;
+    }
+    zs[ei * EST_ZROW + kk] = __float2bfloat16(v);
   }
   __syncthreads();
 
-  // B fragments for this wave's two 16-event tiles (reused across clusters)
-  const int j_loc = lane & 15;
-  const int kbase = 8 * (lane >> 4);
-  bf16x8 bfrag[2];
-#pragma unroll
-  for (int t = 0; t < 2; ++t) {
-    const int e = wave * 32 + t * 16 + j_loc;
-#pragma unroll
-    for (int u = 0; u < 8; ++u) {
-      const int krow = mfma_b16_k(lane >> 4, u);
-      float v;
-      if (krow < d) v = __bfloat162float(zs[krow * zrow + e]);
-      else if (krow == d) v = 1.0f;
-      else v = 0.0f;
-      bfrag[t][u] = (__bf16)v;
-    }
-  }
-
-  const int dp = d + 1;
-  const int rt2 = dp > 16;
+  // 32x32x16 bf16 MFMA: one 32-row tile covers all of M (Dp <= 32) and 32
+  // events; WAVES SPLIT THE CLUSTER LOOP (c = wave, wave+4, ...) so each
+  // cluster's factor fragments are fetched once per block, not once per
+  // wave — the A-fragment L2 traffic was the previous bottleneck.
+  // A lane l -> A[i=l&31][kk=8*(l>>5)+u] per 16-deep chunk; B lane l ->
+  // B[kk][j=l&31]; C/D col=l&31, row=(reg&3)+8*(reg>>2)+4*(l>>5) (guide §3).
+  const int j32 = lane & 31;
+  const int g2 = lane >> 5;
+  typedef __attribute__((ext_vector_type(16))) float f32x16;
   const bf16x8* mf = (const bf16x8*)mfac;  // rows of 32 bf16 = 4 frags each
-  const int i_loc = lane & 15;
-  const int fq = kbase / 8;
+  const int fq0 = g2;      // chunk 0 covers k [0,16): slots {0,1}
+  const int fq1 = 2 + g2;  // chunk 1 covers k [16,32): slots {2,3}
+  const int nwaves = NT / WAVE;
 
-  // software-pipelined A-fragment loads: cluster c+1's factors stream from
-  // L2 while cluster c's MFMAs run (the per-cluster load latency otherwise
-  // serializes the K loop)
-  bf16x8 nx_hi0, nx_lo0, nx_hi1, nx_lo1;
+  bf16x8 nx_h0, nx_l0, nx_h1, nx_l1;
   float nx_add;
   auto load_a = [&](int c) {
     const int64_t base = ((int64_t)c * 2) * 32 * 4;  // in bf16x8 units
-    nx_hi0 = mf[base + i_loc * 4 + fq];
-    nx_lo0 = mf[base + 32 * 4 + i_loc * 4 + fq];
-    if (rt2) {
-      nx_hi1 = mf[base + (16 + i_loc) * 4 + fq];
-      nx_lo1 = mf[base + 32 * 4 + (16 + i_loc) * 4 + fq];
-    }
+    nx_h0 = mf[base + j32 * 4 + fq0];
+    nx_l0 = mf[base + 32 * 4 + j32 * 4 + fq0];
+    nx_h1 = mf[base + j32 * 4 + fq1];
+    nx_l1 = mf[base + 32 * 4 + j32 * 4 + fq1];
     nx_add = add[c];
   };
-  load_a(0);
+  if (wave < k) load_a(wave);
 
-  for (int c = 0; c < k; ++c) {
-    const bf16x8 a_hi0 = nx_hi0, a_lo0 = nx_lo0;
-    const bf16x8 a_hi1 = nx_hi1, a_lo1 = nx_lo1;
+  for (int c = wave; c < k; c += nwaves) {
+    const bf16x8 a_h0 = nx_h0, a_l0 = nx_l0, a_h1 = nx_h1, a_l1 = nx_l1;
     const float addc = nx_add;
-    if (c + 1 < k) load_a(c + 1);
-    // issue all four independent MFMA chains (2 event-tiles x 2 row-tiles)
-    // before any epilogue math: the dependent-accumulator latency and the
-    // square/shuffle epilogue overlap across chains instead of serializing
-    f32x4 y00 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-        a_hi0, bfrag[0], (f32x4){0, 0, 0, 0}, 0, 0, 0);
-    f32x4 y01 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-        a_hi0, bfrag[1], (f32x4){0, 0, 0, 0}, 0, 0, 0);
-    f32x4 y10, y11;
-    if (rt2) {
-      y10 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-          a_hi1, bfrag[0], (f32x4){0, 0, 0, 0}, 0, 0, 0);
-      y11 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-          a_hi1, bfrag[1], (f32x4){0, 0, 0, 0}, 0, 0, 0);
-    }
-    y00 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_lo0, bfrag[0], y00, 0, 0, 0);
-    y01 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_lo0, bfrag[1], y01, 0, 0, 0);
-    if (rt2) {
-      y10 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_lo1, bfrag[0], y10, 0, 0, 0);
-      y11 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_lo1, bfrag[1], y11, 0, 0, 0);
-    }
-    float s0 = y00[0] * y00[0] + y00[1] * y00[1] + y00[2] * y00[2] +
-               y00[3] * y00[3];
-    float s1 = y01[0] * y01[0] + y01[1] * y01[1] + y01[2] * y01[2] +
-               y01[3] * y01[3];
-    if (rt2) {
-      s0 += y10[0] * y10[0] + y10[1] * y10[1] + y10[2] * y10[2] +
-            y10[3] * y10[3];
-      s1 += y11[0] * y11[0] + y11[1] * y11[1] + y11[2] * y11[2] +
-            y11[3] * y11[3];
-    }
-    // rows of the padded Y are spread over lane groups: sum across them
-    s0 += __shfl_xor(s0, 16, WAVE);
-    s0 += __shfl_xor(s0, 32, WAVE);
-    s1 += __shfl_xor(s1, 16, WAVE);
-    s1 += __shfl_xor(s1, 32, WAVE);
-    if (lane < 16) {
-      const int e = wave * 32 + lane;
-      lw[c * lrow + e] = -0.5f * s0 + addc;
-      lw[c * lrow + e + 16] = -0.5f * s1 + addc;
+    if (c + nwaves < k) load_a(c + nwaves);
+#pragma unroll
+    for (int t = 0; t < EST_BE / 32; ++t) {
+      // B fragments: contiguous 16 B of the transposed z row
+      const bf16x8 b0 =
+          *(const bf16x8*)(zs + (t * 32 + j32) * EST_ZROW + 8 * g2);
+      const bf16x8 b1 =
+          *(const bf16x8*)(zs + (t * 32 + j32) * EST_ZROW + 16 + 8 * g2);
+      f32x16 y0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+          a_h0, b0, (f32x16)(0.0f), 0, 0, 0);
+      f32x16 y1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+          a_h1, b1, (f32x16)(0.0f), 0, 0, 0);
+      y0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a_l0, b0, y0, 0, 0, 0);
+      y1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a_l1, b1, y1, 0, 0, 0);
+      float s = 0.0f;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const float yv = y0[r] + y1[r];
+        s = fmaf(yv, yv, s);
+      }
+      // the 32 Y rows live across the two lane halves: one cross-half sum
+      s += __shfl_xor(s, 32, WAVE);
+      if (lane < 32) lw[c * lrow + t * 32 + j32] = -0.5f * s + addc;
     }
   }
   __syncthreads();
 
-  // pass 2: posteriors + likelihood over this block's events
+  // pass 2: posteriors + likelihood; one thread per event (EST_BE == NT)
   float acc = 0.0f;
   if (threadIdx.x < EST_BE && threadIdx.x < cnt) {
     const int t = threadIdx.x;

@@ -289,3 +289,19 @@ def test_constants_emits_valid_cholesky(device):
             u0, -(u @ means[c].cpu().numpy()), rtol=2e-2, atol=2e-2)
         assert np.abs(mm[c][:, d + 1:]).max() == 0.0
         assert np.abs(mm[c][d:, :]).max() == 0.0
+
+
+def test_mfma_bf16_probe32(device):
+    """32x32x16 bf16 fragment layout (used by the fused E-step)."""
+    from cuda_gmm_mpi_amd.ops.backend import hip_ext
+    rng = np.random.default_rng(78)
+    a = rng.standard_normal((32, 16)).astype(np.float32)
+    b = (rng.standard_normal((16, 32)) + np.arange(32)[None, :]).astype(
+        np.float32)
+    at = torch.from_numpy(a).to(device).to(torch.bfloat16)
+    bt = torch.from_numpy(b).to(device).to(torch.bfloat16)
+    c = torch.zeros(32, 32, dtype=torch.float32, device=device)
+    hip_ext().mfma_probe32(at.contiguous(), bt.contiguous(), c)
+    ref = at.float() @ bt.float()
+    np.testing.assert_allclose(c.cpu().numpy(), ref.cpu().numpy(),
+                               rtol=2e-2, atol=5e-2)
