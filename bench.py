@@ -56,8 +56,11 @@ def main() -> int:
     # (no dataset downloads; random-init mixture per BASELINE.json)
     data, _ = make_blobs(n_total, d, k, seed=1234)
 
-    cfg = GmmConfig(num_clusters=k, target_num_clusters=k,
-                    estep_dtype=("bf16" if args.dtype == "bf16" else "fp32"))
+    cfg = GmmConfig(
+        num_clusters=k, target_num_clusters=k,
+        estep_dtype=("bf16" if args.dtype == "bf16" else "fp32"),
+        mstep_precision=("bf16x3" if args.dtype == "bf16" else "fp32"),
+    )
     engine = build_engine(data, cfg, device=device)
 
     def sync():
@@ -106,7 +109,8 @@ def main() -> int:
                 "parallelism": f"dp{world}",
                 "device": device,
                 "note": ("bf16 E-step data reads, fp32 accumulate; "
-                         "fp32 M-step/constants (BASELINE config 2/3)"),
+                         "split-precision bf16x3 M-step moments; "
+                         "fp32 constants/finalize (BASELINE config 2/3)"),
             },
         }
         print(json.dumps(out))

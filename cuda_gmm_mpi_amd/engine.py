@@ -172,7 +172,8 @@ class EmEngine:
         """
         st = self.state.shrink(k)
         with self.profile.time("m_step"):
-            packed = F.mstep_moments(self.x, self.w[:k])
+            packed = F.mstep_moments(self.x, self.w[:k],
+                                     precision=self.cfg.mstep_precision)
         with self.profile.time("comm"):
             pdist.all_reduce_(packed)
         with self.profile.time("m_step"):

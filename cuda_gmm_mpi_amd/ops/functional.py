@@ -158,7 +158,8 @@ def estep_fused_available(device: torch.device, dtype: str, d: int,
 
 
 def mstep_moments(x: torch.Tensor, w: torch.Tensor,
-                  nchunk: int | None = None) -> torch.Tensor:
+                  nchunk: int | None = None,
+                  precision: str = "fp32") -> torch.Tensor:
     """Fused augmented sufficient statistics, packed lower triangle of
     T_c = sum_e w_ce [x;1][x;1]^T per cluster: [K, Dp*(Dp+1)/2] with layout
     [S_tri (D rows) | mean numerators (row D) | N (corner)] — the single
@@ -182,7 +183,10 @@ def mstep_moments(x: torch.Tensor, w: torch.Tensor,
         nchunk = int(min(nchunk, tiles))
         partials = torch.empty((nchunk, k, pp), dtype=torch.float32,
                                device=x.device)
-        hip_ext().mstep_moments(x, w, partials)
+        if precision == "bf16x3":
+            hip_ext().mstep_moments_b16(x, w, partials)
+        else:
+            hip_ext().mstep_moments(x, w, partials)
         return partials.sum(dim=0)
     packed = torch.empty((k, pp), dtype=torch.float32, device=x.device)
     if x.is_cuda:

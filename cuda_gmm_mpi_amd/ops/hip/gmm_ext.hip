@@ -233,6 +233,30 @@ void mstep_moments(torch::Tensor x, torch::Tensor w, torch::Tensor partials) {
   HIP_CHECK(hipGetLastError());
 }
 
+void mstep_moments_b16(torch::Tensor x, torch::Tensor w,
+                       torch::Tensor partials) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() &&
+                  x.scalar_type() == torch::kFloat32,
+              "x must be contiguous fp32");
+  check_f32(w, "w");
+  check_f32(partials, "partials");
+  const int d = (int)x.size(0);
+  const int64_t n = x.size(1);
+  const int k = (int)w.size(0);
+  const int nchunk = (int)partials.size(0);
+  const int dp = d + 1;
+  TORCH_CHECK(d <= 31, "mstep_moments_b16 needs D <= 31");
+  TORCH_CHECK(partials.size(1) == k &&
+                  partials.size(2) == dp * (dp + 1) / 2,
+              "partials must be [nchunk, K, Dp*(Dp+1)/2]");
+  const size_t lds = 32 * 132 * 4 + 2 * 32 * 136 * 2 + 4 * 128 * 4;
+  dim3 grid((k + 3) / 4, nchunk);
+  hipLaunchKernelGGL(gmm::mstep_moments_b16_kernel, grid, dim3(kNT), lds,
+                     stream(), x.data_ptr<float>(), w.data_ptr<float>(),
+                     partials.data_ptr<float>(), d, k, n, nchunk);
+  HIP_CHECK(hipGetLastError());
+}
+
 void estep_fused(torch::Tensor z, torch::Tensor mfac, torch::Tensor add,
                  torch::Tensor w_out, torch::Tensor partial) {
   TORCH_CHECK(z.is_cuda() && z.is_contiguous() &&
@@ -296,6 +320,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "packed weighted second-moment partials [nchunk,K,P]");
   m.def("constants", &constants,
         "batched no-pivot LU inverse + ln|det| + bf16 Cholesky factors");
+  m.def("mstep_moments_b16", &mstep_moments_b16,
+        "split-precision bf16x3 augmented moments");
   m.def("mstep_moments", &mstep_moments,
         "fused augmented moments [S|mean_num|N] via f32 MFMA");
   m.def("estep_fused", &estep_fused,

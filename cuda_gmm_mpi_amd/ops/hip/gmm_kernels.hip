@@ -45,6 +45,8 @@ __device__ inline void tri_row_col(int t, int* i, int* j) {
 // variant (4*(l>>4) + u%4 + 16*(u/4)) if the probe ever disagrees.
 __device__ inline int mfma_b16_k(int group, int u) { return 8 * group + u; }
 
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+
 // Stage cluster c's means + packed/pre-symmetrized Rinv into LDS.
 // lds layout: [0, d) means, [d, d + d(d+1)/2) packed rinv.
 __device__ inline void stage_cluster_params(
@@ -247,14 +249,23 @@ mstep_cov_kernel(const T* __restrict__ x, const float* __restrict__ w,
     const int64_t e0 = tile * te;
     const int cnt = (int)min((int64_t)te, n - e0);
     __syncthreads();
-    // stage x tile (coalesced per dimension row) and w tile
-    for (int i = threadIdx.x; i < d * te; i += NT) {
-      const int di = i / te, ei = i % te;
-      xs[di * row + ei] =
-          (ei < cnt) ? load_x(x, (int64_t)di * n + e0 + ei) : 0.0f;
+    // stage x tile (coalesced per dimension row) and w tile; branchless
+    // when the tile is full (guide §5 trap 4c)
+    if (cnt == te) {
+      for (int i = threadIdx.x; i < d * te; i += NT)
+        xs[(i / te) * row + i % te] =
+            load_x(x, (int64_t)(i / te) * n + e0 + i % te);
+      for (int ei = threadIdx.x; ei < te; ei += NT)
+        wt[ei] = w[(int64_t)c * n + e0 + ei];
+    } else {
+      for (int i = threadIdx.x; i < d * te; i += NT) {
+        const int di = i / te, ei = i % te;
+        xs[di * row + ei] =
+            (ei < cnt) ? load_x(x, (int64_t)di * n + e0 + ei) : 0.0f;
+      }
+      for (int ei = threadIdx.x; ei < te; ei += NT)
+        wt[ei] = (ei < cnt) ? w[(int64_t)c * n + e0 + ei] : 0.0f;
     }
-    for (int ei = threadIdx.x; ei < te; ei += NT)
-      wt[ei] = (ei < cnt) ? w[(int64_t)c * n + e0 + ei] : 0.0f;
     __syncthreads();
 
 #pragma unroll
@@ -504,16 +515,28 @@ mstep_moments_kernel(const T* __restrict__ x, const float* __restrict__ w,
     const int64_t e0 = tile * MOM_BK;
     const int cnt = (int)min((int64_t)MOM_BK, n - e0);
     __syncthreads();
-    for (int idx = threadIdx.x; idx < d * MOM_BK; idx += NT) {
-      const int di = idx / MOM_BK, ei = idx % MOM_BK;
-      xs[di * row + ei] =
-          (ei < cnt) ? load_x(x, (int64_t)di * n + e0 + ei) : 0.0f;
-    }
-    for (int idx = threadIdx.x; idx < 4 * MOM_BK; idx += NT) {
-      const int wv = idx / MOM_BK, ei = idx % MOM_BK;
-      const int cw = blockIdx.x * 4 + wv;
-      wt[wv * MOM_BK + ei] =
-          (cw < k && ei < cnt) ? w[(int64_t)cw * n + e0 + ei] : 0.0f;
+    // branchless full-tile staging: a per-element bounds ternary makes
+    // hipcc serialize every load behind a vmcnt(0) (guide §5 trap 4c)
+    const int nv = min(4, k - blockIdx.x * 4) * MOM_BK;
+    if (cnt == MOM_BK) {
+      for (int idx = threadIdx.x; idx < d * MOM_BK; idx += NT)
+        xs[(idx / MOM_BK) * row + idx % MOM_BK] =
+            load_x(x, (int64_t)(idx / MOM_BK) * n + e0 + idx % MOM_BK);
+      for (int idx = threadIdx.x; idx < nv; idx += NT)
+        wt[idx] = w[(int64_t)(blockIdx.x * 4 + idx / MOM_BK) * n + e0 +
+                    idx % MOM_BK];
+    } else {
+      for (int idx = threadIdx.x; idx < d * MOM_BK; idx += NT) {
+        const int di = idx / MOM_BK, ei = idx % MOM_BK;
+        xs[di * row + ei] =
+            (ei < cnt) ? load_x(x, (int64_t)di * n + e0 + ei) : 0.0f;
+      }
+      for (int idx = threadIdx.x; idx < 4 * MOM_BK; idx += NT) {
+        const int wv = idx / MOM_BK, ei = idx % MOM_BK;
+        const int cw = blockIdx.x * 4 + wv;
+        wt[wv * MOM_BK + ei] =
+            (cw < k && ei < cnt) ? w[(int64_t)cw * n + e0 + ei] : 0.0f;
+      }
     }
     __syncthreads();
 
@@ -564,6 +587,135 @@ mstep_moments_kernel(const T* __restrict__ x, const float* __restrict__ w,
 }
 
 // ---------------------------------------------------------------------------
+// Split-precision (bf16x3) augmented moments: same packed output as
+// mstep_moments_kernel but on v_mfma_f32_32x32x16_bf16 at ~15x the f32
+// MFMA rate. Each operand is split into bf16 hi+lo parts; the three
+// products hi*hi + hi*lo + lo*hi are accumulated in fp32 and the lo*lo
+// term (relative magnitude ~2^-16 per element) is dropped — ~1e-5-class
+// relative accuracy on S vs the exact-fp32 kernel.
+//   A[i=dim][kk=event] = w[e] * z_i[e]   (split per cluster)
+//   B[kk=event][j=dim] = z_j[e]          (split once, shared by 4 clusters)
+// ---------------------------------------------------------------------------
+#define MB_BK 128
+
+typedef __attribute__((ext_vector_type(16))) float f32x16_t;
+
+__global__ void __launch_bounds__(NT)
+mstep_moments_b16_kernel(const float* __restrict__ x,
+                         const float* __restrict__ w,
+                         float* __restrict__ partials, int d, int k,
+                         int64_t n, int nchunk) {
+  // LDS (all rows padded to 32 dims, ones-row at d, zeros above):
+  //   z32 [32][MB_BK] f32, zhi/zlo [32][MB_BK] bf16, wt [4][MB_BK] f32
+  extern __shared__ float lds[];
+  // rows padded so per-dim b128 reads spread across banks: an unpadded
+  // 128-element stride puts every lane of a 16-lane group on one bank slot
+  constexpr int Z32R = MB_BK + 4;  // f32 row stride
+  constexpr int ZBR = MB_BK + 8;   // bf16 row stride (keeps 16B row align)
+  float* z32 = lds;                               // 32*Z32R floats
+  __bf16* zhi = (__bf16*)(lds + 32 * Z32R);
+  __bf16* zlo = zhi + 32 * ZBR;
+  float* wt = (float*)(zlo + 32 * ZBR);           // 4*MB_BK floats
+
+  const int wave = threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int j32 = lane & 31;
+  const int g2 = lane >> 5;
+  const int c = blockIdx.x * 4 + wave;
+  const int chunk = blockIdx.y;
+  const int dp = d + 1;
+
+  f32x16_t accA = (f32x16_t)(0.0f);   // hi*hi
+  f32x16_t accB = (f32x16_t)(0.0f);   // hi*lo
+  f32x16_t accC = (f32x16_t)(0.0f);   // lo*hi
+
+  const int64_t tiles = (n + MB_BK - 1) / MB_BK;
+  for (int64_t tile = chunk; tile < tiles; tile += nchunk) {
+    const int64_t e0 = tile * MB_BK;
+    const int cnt = (int)min((int64_t)MB_BK, n - e0);
+    __syncthreads();
+    // branchless full-tile staging (guide §5 trap 4c); constant rows
+    // (ones at d, zeros above) written once per block, re-fixed only
+    // after a tail tile (the tail is always the globally last tile)
+    auto put = [&](int idx, float v) {
+      // native __bf16 casts lower to the hardware cvt (the
+      // __float2bfloat16 helper is a multi-instruction software RNE)
+      const int di = idx / MB_BK, ei = idx % MB_BK;
+      const __bf16 hi = (__bf16)v;
+      z32[di * Z32R + ei] = v;
+      zhi[di * ZBR + ei] = hi;
+      zlo[di * ZBR + ei] = (__bf16)(v - (float)hi);
+    };
+    const int nv = min(4, k - blockIdx.x * 4) * MB_BK;
+    if (cnt == MB_BK) {
+      for (int idx = threadIdx.x; idx < d * MB_BK; idx += NT)
+        put(idx, x[(int64_t)(idx / MB_BK) * n + e0 + idx % MB_BK]);
+      if (tile == chunk) {  // first tile: fill the constant rows
+        for (int idx = d * MB_BK + threadIdx.x; idx < 32 * MB_BK; idx += NT)
+          put(idx, (idx / MB_BK == d) ? 1.0f : 0.0f);
+      }
+      for (int idx = threadIdx.x; idx < nv; idx += NT)
+        wt[idx] = w[(int64_t)(blockIdx.x * 4 + idx / MB_BK) * n + e0 +
+                    idx % MB_BK];
+    } else {
+      for (int idx = threadIdx.x; idx < 32 * MB_BK; idx += NT) {
+        const int di = idx / MB_BK, ei = idx % MB_BK;
+        float v = 0.0f;
+        if (ei < cnt) {
+          if (di < d) v = x[(int64_t)di * n + e0 + ei];
+          else if (di == d) v = 1.0f;
+        }
+        put(idx, v);
+      }
+      for (int idx = threadIdx.x; idx < 4 * MB_BK; idx += NT) {
+        const int wv = idx / MB_BK, ei = idx % MB_BK;
+        const int cw = blockIdx.x * 4 + wv;
+        wt[idx] = (cw < k && ei < cnt) ? w[(int64_t)cw * n + e0 + ei] : 0.0f;
+      }
+    }
+    __syncthreads();
+
+#pragma unroll 2
+    for (int ch = 0; ch < MB_BK / 16; ++ch) {
+      const int eb = ch * 16 + 8 * g2;  // this lane's 8 events
+      // B fragments: contiguous events of this lane's dim row
+      const bf16x8 b_hi = *(const bf16x8*)(zhi + j32 * ZBR + eb);
+      const bf16x8 b_lo = *(const bf16x8*)(zlo + j32 * ZBR + eb);
+      // A fragments: w-weighted z, split on the fly
+      const float4 zv0 = *(const float4*)(z32 + j32 * Z32R + eb);
+      const float4 zv1 = *(const float4*)(z32 + j32 * Z32R + eb + 4);
+      const float4 wv0 = *(const float4*)(wt + wave * MB_BK + eb);
+      const float4 wv1 = *(const float4*)(wt + wave * MB_BK + eb + 4);
+      float av[8] = {zv0.x * wv0.x, zv0.y * wv0.y, zv0.z * wv0.z,
+                     zv0.w * wv0.w, zv1.x * wv1.x, zv1.y * wv1.y,
+                     zv1.z * wv1.z, zv1.w * wv1.w};
+      bf16x8 a_hi, a_lo;
+#pragma unroll
+      for (int u = 0; u < 8; ++u) {
+        const __bf16 hi = (__bf16)av[u];
+        a_hi[u] = hi;
+        a_lo[u] = (__bf16)(av[u] - (float)hi);
+      }
+      accA = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a_hi, b_hi, accA, 0, 0, 0);
+      accB = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a_hi, b_lo, accB, 0, 0, 0);
+      accC = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a_lo, b_hi, accC, 0, 0, 0);
+    }
+  }
+
+  if (c >= k) return;
+  const int p_aug = dp * (dp + 1) / 2;
+  float* out = partials + ((int64_t)chunk * k + c) * p_aug;
+  // C/D layout: col = l&31, row = (reg&3) + 8*(reg>>2) + 4*(l>>5)
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int gi = (r & 3) + 8 * (r >> 2) + 4 * g2;
+    const int gj = j32;
+    if (gi < dp && gj <= gi)
+      out[gi * (gi + 1) / 2 + gj] = accA[r] + accB[r] + accC[r];
+  }
+}
+
+// ---------------------------------------------------------------------------
 // MFMA fused E-step (replaces estep1 + estep2 in one pass for bf16, D<=31,
 // moderate K): logw via  q = ||M_c z||^2  with M_c = [U_c | -U_c mu_c],
 // U_c^T U_c = Rinv_c (Cholesky, emitted by the constants kernel), split into
@@ -577,7 +729,6 @@ mstep_moments_kernel(const T* __restrict__ x, const float* __restrict__ w,
 // ---------------------------------------------------------------------------
 #define EST_BE 128
 
-typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 
 
 // Layout probe: D = A(16x32) @ B(32x16) in bf16 via one MFMA.
@@ -640,14 +791,26 @@ estep_fused_kernel(const __hip_bfloat16* __restrict__ z,
   const int64_t e0 = (int64_t)blockIdx.x * EST_BE;
   const int cnt = (int)min((int64_t)EST_BE, n - e0);
 
-  for (int idx = threadIdx.x; idx < EST_BE * 32; idx += NT) {
-    const int ei = idx >> 5, kk = idx & 31;
-    float v = 0.0f;
-    if (ei < cnt) {
-      if (kk < d) v = __bfloat162float(z[(int64_t)kk * n + e0 + ei]);
-      else if (kk == d) v = 1.0f;
+  __bf16* zsb = (__bf16*)zs;
+  if (cnt == EST_BE) {
+    // branchless staging (guide §5 trap 4c): coalesced reads of the d data
+    // rows, transposed scatter into LDS; then the constant rows
+    for (int idx = threadIdx.x; idx < d * EST_BE; idx += NT) {
+      const int kk = idx / EST_BE, ei = idx % EST_BE;
+      zsb[ei * EST_ZROW + kk] =
+          (__bf16)__bfloat162float(z[(int64_t)kk * n + e0 + ei]);
     }
-    zs[ei * EST_ZROW + kk] = __float2bfloat16(v);
+  } else {
+    for (int idx = threadIdx.x; idx < d * EST_BE; idx += NT) {
+      const int kk = idx / EST_BE, ei = idx % EST_BE;
+      zsb[ei * EST_ZROW + kk] = (__bf16)(
+          (ei < cnt) ? __bfloat162float(z[(int64_t)kk * n + e0 + ei]) : 0.0f);
+    }
+  }
+  for (int idx = threadIdx.x; idx < (32 - d) * EST_BE; idx += NT) {
+    const int kk = d + idx / EST_BE, ei = idx % EST_BE;
+    zsb[ei * EST_ZROW + kk] =
+        (__bf16)((kk == d && ei < cnt) ? 1.0f : 0.0f);
   }
   __syncthreads();
 

@@ -56,6 +56,11 @@ class GmmConfig:
     # is always fp32. BASELINE.json config 2 names a bf16 E-step.
     estep_dtype: str = "fp32"
 
+    # M-step sufficient-statistics precision: "fp32" = exact fp32 fmaf
+    # chain (f32 MFMA, bitwise a VALU loop); "bf16x3" = split-precision
+    # bf16 MFMA (~1e-5 relative on S, ~15x the instruction rate).
+    mstep_precision: str = "fp32"
+
     # Internally center data by the global per-dimension mean before EM.
     # Translation-invariant math (covariance, quadratic forms) is unchanged;
     # output means get the center added back. This removes the catastrophic
@@ -81,6 +86,9 @@ class GmmConfig:
             )
         if self.estep_dtype not in ("fp32", "bf16"):
             raise ValueError(f"estep_dtype must be fp32|bf16: {self.estep_dtype}")
+        if self.mstep_precision not in ("fp32", "bf16x3"):
+            raise ValueError(
+                f"mstep_precision must be fp32|bf16x3: {self.mstep_precision}")
 
     @property
     def stop_number(self) -> int:

@@ -305,3 +305,25 @@ def test_mfma_bf16_probe32(device):
     ref = at.float() @ bt.float()
     np.testing.assert_allclose(c.cpu().numpy(), ref.cpu().numpy(),
                                rtol=2e-2, atol=5e-2)
+
+
+@pytest.mark.parametrize("d,n", [(2, 3001), (24, 100000), (31, 4097)])
+def test_mstep_moments_b16_matches_cpu(device, d, n):
+    """Split-precision bf16x3 moments: ~1e-5-class relative accuracy."""
+    from cuda_gmm_mpi_amd.ops import functional as F
+    rng = np.random.default_rng(d * 31 + 5)
+    k = 6
+    x = rng.standard_normal((d, n)).astype(np.float32) * 3
+    w = rng.uniform(0, 1, (k, n)).astype(np.float32)
+    packed = F.mstep_moments(torch.from_numpy(x).to(device),
+                             torch.from_numpy(w).to(device),
+                             precision="bf16x3")
+    n_c, mean_num, s = F.moments_views(packed, d)
+    rn, rm, rs = cpu.mstep_sufficient_stats(torch.from_numpy(x).double(),
+                                            torch.from_numpy(w).double())
+    np.testing.assert_allclose(n_c.cpu().numpy(), rn.numpy(), rtol=1e-4)
+    np.testing.assert_allclose(mean_num.cpu().numpy(), rm.numpy(),
+                               rtol=3e-4, atol=1e-2)
+    scale = float(rs.abs().max())
+    np.testing.assert_allclose(s.cpu().numpy(), rs.numpy(),
+                               rtol=3e-4, atol=3e-4 * scale)
