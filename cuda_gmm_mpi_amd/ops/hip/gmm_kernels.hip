@@ -851,18 +851,16 @@ estep_fused_kernel(const __hip_bfloat16* __restrict__ z,
           *(const bf16x8*)(zs + (t * 32 + j32) * EST_ZROW + 8 * g2);
       const bf16x8 b1 =
           *(const bf16x8*)(zs + (t * 32 + j32) * EST_ZROW + 16 + 8 * g2);
-      f32x16 y0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+      // one accumulator chain: Y = (Mhi+Mlo)(chunk0+chunk1) summed by the
+      // MFMAs themselves (the VALU epilogue was the measured bottleneck)
+      f32x16 y = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
           a_h0, b0, (f32x16)(0.0f), 0, 0, 0);
-      f32x16 y1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-          a_h1, b1, (f32x16)(0.0f), 0, 0, 0);
-      y0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a_l0, b0, y0, 0, 0, 0);
-      y1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a_l1, b1, y1, 0, 0, 0);
+      y = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a_h1, b1, y, 0, 0, 0);
+      y = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a_l0, b0, y, 0, 0, 0);
+      y = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a_l1, b1, y, 0, 0, 0);
       float s = 0.0f;
 #pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        const float yv = y0[r] + y1[r];
-        s = fmaf(yv, yv, s);
-      }
+      for (int r = 0; r < 16; ++r) s = fmaf(y[r], y[r], s);
       // the 32 Y rows live across the two lane halves: one cross-half sum
       s += __shfl_xor(s, 32, WAVE);
       if (lane < 32) lw[c * lrow + t * 32 + j32] = -0.5f * s + addc;
@@ -870,18 +868,24 @@ estep_fused_kernel(const __hip_bfloat16* __restrict__ z,
   }
   __syncthreads();
 
-  // pass 2: posteriors + likelihood; one thread per event (EST_BE == NT)
+  // pass 2: posteriors + likelihood; one exp per (cluster, event): the
+  // normalize pass rescales the already-computed exp(lw - m) by
+  // exp(m - denom) instead of re-exponentiating
   float acc = 0.0f;
   if (threadIdx.x < EST_BE && threadIdx.x < cnt) {
     const int t = threadIdx.x;
     float m = lw[t];
     for (int c = 1; c < k; ++c) m = fmaxf(m, lw[c * lrow + t]);
     float s = 0.0f;
-    for (int c = 0; c < k; ++c) s += __expf(lw[c * lrow + t] - m);
-    const float denom = m + __logf(s);
+    for (int c = 0; c < k; ++c) {
+      const float e = __expf(lw[c * lrow + t] - m);
+      lw[c * lrow + t] = e;
+      s += e;
+    }
+    const float inv = 1.0f / s;
     for (int c = 0; c < k; ++c)
-      w_out[(int64_t)c * n + e0 + t] = __expf(lw[c * lrow + t] - denom);
-    acc = denom;
+      w_out[(int64_t)c * n + e0 + t] = lw[c * lrow + t] * inv;
+    acc = m + __logf(s);
   }
   __shared__ float wsum[NT / WAVE];
   for (int off = WAVE / 2; off > 0; off >>= 1)
