@@ -47,11 +47,18 @@ def make_parser() -> argparse.ArgumentParser:
                    help="write only the .summary file")
     p.add_argument("--print", dest="enable_print", action="store_true")
     p.add_argument("--estep-dtype", choices=["fp32", "bf16"], default="fp32")
+    p.add_argument("--mstep-precision", choices=["fp32", "bf16x3"],
+                   default="fp32",
+                   help="M-step sufficient-statistics precision (bf16x3 = "
+                        "split-precision MFMA, ~1e-5 relative, ~1.5x faster)")
     p.add_argument("--no-center", dest="center_data", action="store_false")
     p.add_argument("--device", default=None,
                    help="cpu | cuda (default: cuda when available)")
     p.add_argument("--profile", action="store_true",
                    help="print the per-GPU timing report (gaussian.cu:967)")
+    p.add_argument("--gpus", type=int, default=None,
+                   help="spawn N single-GPU worker processes on this node "
+                        "(one rank per GPU over RCCL)")
     return p
 
 
@@ -64,6 +71,7 @@ def config_from_args(args) -> GmmConfig:
         diag_only=args.diag_only, bug_compat=args.bug_compat,
         enable_print=args.enable_print, enable_output=args.enable_output,
         estep_dtype=args.estep_dtype, center_data=args.center_data,
+        mstep_precision=args.mstep_precision,
     )
     cfg.validate()
     return cfg
@@ -107,7 +115,14 @@ def run_clustering(data: np.ndarray, cfg: GmmConfig, outfile: str,
 
 
 def main(argv=None) -> int:
-    args = make_parser().parse_args(argv)
+    if argv is None:
+        argv = sys.argv[1:]
+    # self-launch N local ranks when asked and not already inside a world
+    from .parallel.launcher import launch_workers, strip_gpus_arg
+    rest, ngpus = strip_gpus_arg(list(argv))
+    if ngpus and ngpus > 1 and "WORLD_SIZE" not in __import__("os").environ:
+        return launch_workers(rest, ngpus)
+    args = make_parser().parse_args(rest)
     if not (1 <= args.num_clusters <= MAX_CLUSTERS):
         print("Invalid number of starting clusters\n")
         return 1

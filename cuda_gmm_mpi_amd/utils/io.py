@@ -128,17 +128,25 @@ def write_summary(path: str, state, enable_output: bool = True) -> None:
 
 
 def write_results(path: str, data_by_event: np.ndarray,
-                  memberships: np.ndarray) -> None:
+                  memberships: np.ndarray,
+                  chunk: int = 65536) -> None:
     """Writes <out>.results (gaussian.cu:1042-1059).
 
     Per event: comma-joined %f data values, a tab, comma-joined %f
-    memberships (cluster-major array indexed [c, e]).
+    memberships (cluster-major array indexed [c, e]). Vectorized chunked
+    formatting — the reference's per-value fprintf is minutes at N=1M.
     """
     n_events, n_dims = data_by_event.shape
-    n_clusters = memberships.shape[0]
     with open(path, "w") as f:
-        for e in range(n_events):
-            f.write(",".join(f"{data_by_event[e, d]:f}" for d in range(n_dims)))
-            f.write("\t")
-            f.write(",".join(f"{memberships[c, e]:f}" for c in range(n_clusters)))
+        for s in range(0, n_events, chunk):
+            e = min(s + chunk, n_events)
+            left = _csv_block(data_by_event[s:e])
+            right = _csv_block(memberships[:, s:e].T)
+            f.write("\n".join(a + "\t" + b for a, b in zip(left, right)))
             f.write("\n")
+
+
+def _csv_block(arr: np.ndarray) -> list[str]:
+    """Rows of ``arr`` as %f comma-joined strings (C printf %f == .6f)."""
+    flat = np.char.mod("%f", arr.astype(np.float64))
+    return [",".join(row) for row in flat]

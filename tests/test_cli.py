@@ -80,3 +80,25 @@ def test_cli_no_output_creates_empty_summary(tmp_path, csv_file):
                "--device", "cpu", "--no-output"])
     assert rc == 0
     assert open(out + ".summary").read() == ""
+
+
+def test_cli_gpus_launcher_cpu_world2(tmp_path, csv_file, monkeypatch):
+    """--gpus 2 spawns two gloo ranks on CPU and produces one output."""
+    import os
+    monkeypatch.delenv("WORLD_SIZE", raising=False)
+    path, _ = csv_file
+    out = str(tmp_path / "mp")
+    rc = main(["3", path, out, "3", "--min-iters", "3", "--max-iters", "3",
+               "--device", "cpu", "--no-results", "--gpus", "2"])
+    assert rc == 0
+    assert open(out + ".summary").read().count("Cluster #") == 3
+
+
+def test_strip_gpus_arg():
+    from cuda_gmm_mpi_amd.parallel.launcher import strip_gpus_arg
+    rest, n = strip_gpus_arg(["3", "in.csv", "out", "--gpus", "4", "--print"])
+    assert rest == ["3", "in.csv", "out", "--print"] and n == 4
+    rest, n = strip_gpus_arg(["3", "x", "y", "--gpus=8"])
+    assert rest == ["3", "x", "y"] and n == 8
+    rest, n = strip_gpus_arg(["3", "x", "y"])
+    assert n is None
