@@ -56,6 +56,11 @@ def make_parser() -> argparse.ArgumentParser:
                    help="cpu | cuda (default: cuda when available)")
     p.add_argument("--profile", action="store_true",
                    help="print the per-GPU timing report (gaussian.cu:967)")
+    p.add_argument("--checkpoint-dir", default=None,
+                   help="directory for MDL-sweep checkpoints; an existing "
+                        "checkpoint there is resumed automatically")
+    p.add_argument("--metrics-out", default=None,
+                   help="write machine-readable run metrics (JSON) here")
     p.add_argument("--gpus", type=int, default=None,
                    help="spawn N single-GPU worker processes on this node "
                         "(one rank per GPU over RCCL)")
@@ -72,6 +77,7 @@ def config_from_args(args) -> GmmConfig:
         enable_print=args.enable_print, enable_output=args.enable_output,
         estep_dtype=args.estep_dtype, center_data=args.center_data,
         mstep_precision=args.mstep_precision,
+        checkpoint_dir=args.checkpoint_dir,
     )
     cfg.validate()
     return cfg
@@ -150,6 +156,20 @@ def main(argv=None) -> int:
             data, cfg, args.outfile, device,
             write_results=args.write_results, profile_report=args.profile,
         )
+        if rank == 0 and args.metrics_out:
+            import json
+            with open(args.metrics_out, "w") as f:
+                json.dump({
+                    "num_clusters": result["num_clusters"],
+                    "rissanen": result["rissanen"],
+                    "likelihood": result["likelihood"],
+                    "rissanen_by_k": {
+                        str(kk): v
+                        for kk, v in result["rissanen_by_k"].items()
+                    },
+                    "world_size": world,
+                    "device": device,
+                }, f, indent=2)
         if rank == 0 and args.enable_print:
             print(f"Ideal clusters: {result['num_clusters']} "
                   f"(rissanen {result['rissanen']:.4f})")

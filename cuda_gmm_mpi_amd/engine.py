@@ -266,7 +266,12 @@ class EmEngine:
         st.Rinv.copy_(torch.from_numpy(np.ascontiguousarray(hc.Rinv[:k])))
 
     def sweep(self) -> SweepResult:
-        """MDL model-order sweep K0 -> stop (gaussian.cu:479-960)."""
+        """MDL model-order sweep K0 -> stop (gaussian.cu:479-960).
+
+        With ``cfg.checkpoint_dir`` set, every completed K (post-merge)
+        writes a parameter checkpoint; an existing checkpoint is resumed
+        from automatically.
+        """
         cfg = self.cfg
         k = cfg.num_clusters
         stop = cfg.stop_number
@@ -275,6 +280,29 @@ class EmEngine:
         min_rissanen = float("inf")
         best_lik = 0.0
         riss_by_k: dict[int, float] = {}
+
+        if cfg.checkpoint_dir:
+            from .utils.checkpoint import load_sweep_checkpoint
+            ck = load_sweep_checkpoint(cfg.checkpoint_dir)
+            if ck and ck["state"] is not None and ck["k"] <= k:
+                k = ck["k"]
+                st = ck["state"].to(self.device)
+                self._load_host_clusters(
+                    HostClusters(
+                        N=st.N.cpu().numpy(), pi=st.pi.cpu().numpy(),
+                        constant=st.constant.cpu().numpy(),
+                        avgvar=st.avgvar.cpu().numpy(),
+                        means=st.means.cpu().numpy(), R=st.R.cpu().numpy(),
+                        Rinv=st.Rinv.cpu().numpy(),
+                    ), k)
+                if self.mfac is not None:
+                    # regenerate the fused-E-step factors for resumed params
+                    self._update_constants(self.state.shrink(k))
+                best_state = ck["best"]
+                best_k = ck["best_k"]
+                min_rissanen = ck["min_rissanen"]
+                best_lik = ck["best_lik"]
+                riss_by_k = dict(ck["rissanen_by_k"])
 
         while k >= stop:
             lik = self.run_em(k)
@@ -317,6 +345,11 @@ class EmEngine:
                         self.state.shrink(new_k).load_param_vector(vec)
             self.profile.count("reduce")
             k = new_k
+            if cfg.checkpoint_dir and self.rank == 0:
+                from .utils.checkpoint import save_sweep_checkpoint
+                save_sweep_checkpoint(
+                    cfg.checkpoint_dir, self.state.shrink(k), k,
+                    best_state, best_k, min_rissanen, best_lik, riss_by_k)
 
         assert best_state is not None
         if cfg.enable_print and self.rank == 0:

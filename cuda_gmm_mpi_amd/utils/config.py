@@ -72,6 +72,10 @@ class GmmConfig:
     # Device-side: memberships stay shard-resident except for this output.
     verbose: bool = False
 
+    # Directory for MDL-sweep checkpoints (.npz per completed K); None
+    # disables. Resume happens automatically when a checkpoint exists.
+    checkpoint_dir: str | None = None
+
     def validate(self) -> None:
         if not (1 <= self.num_clusters <= MAX_CLUSTERS):
             raise ValueError(

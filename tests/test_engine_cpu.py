@@ -141,3 +141,48 @@ def test_diag_only_engine_runs():
     for c in range(2):
         off = r[c] - np.diag(np.diag(r[c]))
         assert np.abs(off).max() == 0.0
+
+
+def test_sweep_checkpoint_resume(tmp_path):
+    """A sweep resumed from a mid-point checkpoint matches a direct run."""
+    data, _ = make_blobs(2500, 2, 3, seed=29)
+    ckdir = str(tmp_path / "ck")
+    # phase 1: sweep 6 -> 4, writing checkpoints after each merge
+    cfg1 = GmmConfig(num_clusters=6, target_num_clusters=4,
+                     min_iters=4, max_iters=4, checkpoint_dir=ckdir)
+    eng1 = build_engine(data, cfg1, device="cpu")
+    eng1.sweep()
+    # phase 2: resume down to 2
+    cfg2 = GmmConfig(num_clusters=6, target_num_clusters=2,
+                     min_iters=4, max_iters=4, checkpoint_dir=ckdir)
+    eng2 = build_engine(data, cfg2, device="cpu")
+    res_resumed = eng2.sweep()
+    # direct run without checkpoints
+    cfg3 = GmmConfig(num_clusters=6, target_num_clusters=2,
+                     min_iters=4, max_iters=4)
+    eng3 = build_engine(data, cfg3, device="cpu")
+    res_direct = eng3.sweep()
+    assert res_resumed.num_clusters == res_direct.num_clusters
+    assert res_resumed.min_rissanen == pytest.approx(
+        res_direct.min_rissanen, rel=1e-5)
+    np.testing.assert_allclose(res_resumed.state.means.numpy(),
+                               res_direct.state.means.numpy(), rtol=1e-4,
+                               atol=1e-4)
+    assert res_resumed.rissanen_by_k.keys() == res_direct.rissanen_by_k.keys()
+
+
+def test_metrics_out(tmp_path):
+    import json
+    from cuda_gmm_mpi_amd.cli import main
+    from cuda_gmm_mpi_amd.utils import io as gio
+    data, _ = make_blobs(600, 2, 3, seed=4)
+    binpath = str(tmp_path / "d.bin")
+    gio.write_bin(binpath, data)
+    mpath = str(tmp_path / "metrics.json")
+    rc = main(["3", binpath, str(tmp_path / "o"), "3", "--min-iters", "3",
+               "--max-iters", "3", "--device", "cpu", "--no-results",
+               "--metrics-out", mpath])
+    assert rc == 0
+    m = json.load(open(mpath))
+    assert m["num_clusters"] == 3
+    assert "3" in m["rissanen_by_k"]
