@@ -26,7 +26,7 @@ import math
 import numpy as np
 import torch
 
-from .models.merge import HostClusters, reduce_order
+from .models.merge import HostClusters, reduce_order_batched
 from .models.seed import seed_means_host, seed_state
 from .models.state import GmmState
 from .ops import functional as F
@@ -330,7 +330,9 @@ class EmEngine:
                 new_k = k
                 if self.rank == 0:
                     hc = self._host_clusters(k)
-                    new_k, _, _ = reduce_order(hc, bug_compat=cfg.bug_compat)
+                    new_k, _, _ = reduce_order_batched(
+                        hc, bug_compat=cfg.bug_compat,
+                        device=str(self.device))
                 if self.world > 1:
                     nk = torch.tensor([new_k], dtype=torch.int64)
                     pdist.broadcast_(nk, src=0)

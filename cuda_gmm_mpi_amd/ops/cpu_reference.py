@@ -180,6 +180,26 @@ def lu_invert_nopivot(a: torch.Tensor) -> tuple[torch.Tensor, torch.Tensor]:
     return out, logdet
 
 
+def lu_logdet_nopivot(a: torch.Tensor) -> torch.Tensor:
+    """Batched ln|det| via the same no-pivot elimination as
+    lu_invert_nopivot, without the inversion phases. a: [K, D, D]."""
+    k, d, _ = a.shape
+    data = a.clone()
+    if d == 1:
+        return torch.log(data[:, 0, 0].abs())
+    data[:, 0, 1:] /= data[:, 0, 0].unsqueeze(1)
+    for i in range(1, d):
+        data[:, i:, i] -= torch.einsum("kjp,kp->kj", data[:, i:, :i], data[:, :i, i])
+        if i == d - 1:
+            continue
+        data[:, i, i + 1:] = (
+            data[:, i, i + 1:]
+            - torch.einsum("kp,kpj->kj", data[:, i, :i], data[:, :i, i + 1:])
+        ) / data[:, i, i].unsqueeze(1)
+    diag = torch.diagonal(data, dim1=1, dim2=2).abs()
+    return torch.log(diag).sum(dim=1)
+
+
 def compute_constants(r: torch.Tensor,
                       diag_only: bool = False) -> tuple[torch.Tensor, torch.Tensor]:
     """Rinv + per-cluster constant (constants_kernel, gaussian_kernel.cu:196-243).

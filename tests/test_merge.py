@@ -114,3 +114,26 @@ def make_copy(hc):
         avgvar=hc.avgvar.copy(), means=hc.means.copy(), R=hc.R.copy(),
         Rinv=hc.Rinv.copy(),
     )
+
+
+def test_reduce_order_batched_matches_exhaustive(rng):
+    from cuda_gmm_mpi_amd.models.merge import reduce_order_batched
+    for trial in range(5):
+        hc_a = make_clusters(rng, 8, 5)
+        hc_b = make_copy(hc_a)
+        k_a, c1_a, c2_a = reduce_order(hc_a)
+        k_b, c1_b, c2_b = reduce_order_batched(hc_b)
+        assert (k_a, c1_a, c2_a) == (k_b, c1_b, c2_b)
+        np.testing.assert_allclose(hc_a.means[:k_a], hc_b.means[:k_b],
+                                   rtol=1e-6)
+        np.testing.assert_allclose(hc_a.R[:k_a], hc_b.R[:k_b], rtol=1e-6)
+        np.testing.assert_array_equal(hc_a.N[:k_a], hc_b.N[:k_b])
+
+
+def test_reduce_order_batched_no_bugcompat(rng):
+    from cuda_gmm_mpi_amd.models.merge import reduce_order_batched
+    hc_a = make_clusters(rng, 6, 4)
+    hc_b = make_copy(hc_a)
+    res_a = reduce_order(hc_a, bug_compat=False)
+    res_b = reduce_order_batched(hc_b, bug_compat=False)
+    assert res_a == res_b
