@@ -73,14 +73,18 @@ class EmEngine:
             self.x_estep = self.x
 
         k0 = config.num_clusters
-        # fused bf16-MFMA E-step path (D <= 31, LDS-bounded K)
+        # fused bf16-MFMA E-step path (D <= 31, LDS-bounded K); big-D MFMA
+        # logw path (31 < D <= 143); otherwise the VALU kernels
         self.use_fused_estep = F.estep_fused_available(
             self.device, config.estep_dtype, self.d, k0,
         ) and not config.diag_only
+        self.use_big_estep = F.estep_big_available(
+            self.device, config.estep_dtype, self.d,
+        ) and not config.diag_only
         self.mfac = (
-            torch.empty(k0, 2, 32, 32, dtype=torch.bfloat16,
+            torch.empty(k0, *F.mfac_shape(self.d), dtype=torch.bfloat16,
                         device=self.device)
-            if self.use_fused_estep else None
+            if (self.use_fused_estep or self.use_big_estep) else None
         )
 
         self.state = GmmState.empty(k0, self.d, self.device)
@@ -124,6 +128,11 @@ class EmEngine:
                 add = st.constant + torch.log(st.pi)
                 w, lik = F.estep_fused(self.x_estep, self.mfac[:k], add,
                                        self.w[:k])
+            elif self.use_big_estep:
+                add = st.constant + torch.log(st.pi)
+                logw = F.estep_logw_big(self.x_estep, self.mfac[:k], add,
+                                        self.w[:k])
+                w, lik = F.estep_posteriors(logw)
             else:
                 logw = F.estep_logw(
                     self.x_estep, st.means, st.Rinv, st.constant, st.pi,

@@ -157,6 +157,26 @@ def estep_fused_available(device: torch.device, dtype: str, d: int,
     return 128 * 40 * 2 + 4 * k * 132 <= 64 * 1024
 
 
+def estep_big_available(device: torch.device, dtype: str, d: int) -> bool:
+    """Big-D MFMA logw path gate (31 < D <= 143, bf16 data)."""
+    return device.type == "cuda" and dtype == "bf16" and 31 < d <= 143
+
+
+def mfac_shape(d: int) -> tuple[int, int, int]:
+    """Factor tensor shape [2, RT*32, KCT*16] for estep MFMA kernels."""
+    rows = ((d + 31) // 32) * 32
+    kc = (d + 1 + 15) // 16
+    kct = 2 if kc <= 2 else 3 if kc == 3 else 5 if kc <= 5 else 9
+    return 2, rows, kct * 16
+
+
+def estep_logw_big(z: torch.Tensor, mfac: torch.Tensor, add: torch.Tensor,
+                   out: torch.Tensor) -> torch.Tensor:
+    """Big-D MFMA log-weights into out [K, N] (CUDA, bf16, D > 31)."""
+    hip_ext().estep_logw_big(z, mfac, add, out)
+    return out
+
+
 def mstep_moments(x: torch.Tensor, w: torch.Tensor,
                   nchunk: int | None = None,
                   precision: str = "fp32") -> torch.Tensor:
@@ -187,6 +207,15 @@ def mstep_moments(x: torch.Tensor, w: torch.Tensor,
             hip_ext().mstep_moments_b16(x, w, partials)
         else:
             hip_ext().mstep_moments(x, w, partials)
+        return partials.sum(dim=0)
+    if x.is_cuda and precision == "bf16x3" and d <= 159:
+        tiles = (n + 63) // 64  # MBB_BK
+        if nchunk is None:
+            nchunk = max(1, min(256, (64 << 20) // (4 * k * pp)))
+        nchunk = int(min(nchunk, tiles))
+        partials = torch.empty((nchunk, k, pp), dtype=torch.float32,
+                               device=x.device)
+        hip_ext().mstep_moments_big(x, w, partials)
         return partials.sum(dim=0)
     packed = torch.empty((k, pp), dtype=torch.float32, device=x.device)
     if x.is_cuda:

@@ -175,10 +175,13 @@ void constants(torch::Tensor r, torch::Tensor means, torch::Tensor rinv,
   const bool make_mfac = mfac.numel() > 0;
   __hip_bfloat16* mp = nullptr;
   if (make_mfac) {
+    const int rows = ((d + 31) / 32) * 32;
+    const int kct = d + 1 <= 32 ? 2 : d + 1 <= 48 ? 3 : d + 1 <= 80 ? 5 : 9;
     TORCH_CHECK(mfac.is_cuda() && mfac.is_contiguous() &&
                     mfac.scalar_type() == torch::kBFloat16 &&
-                    mfac.numel() == (int64_t)k * 2 * 32 * 32 && d <= 31,
-                "mfac must be bf16 [K,2,32,32] with D <= 31");
+                    mfac.numel() == (int64_t)k * 2 * rows * kct * 16 &&
+                    d <= 143,
+                "mfac must be bf16 [K,2,RT*32,KCT*16] with D <= 143");
     mp = reinterpret_cast<__hip_bfloat16*>(mfac.data_ptr());
   }
   auto s = stream();
@@ -287,6 +290,56 @@ void estep_fused(torch::Tensor z, torch::Tensor mfac, torch::Tensor add,
   HIP_CHECK(hipGetLastError());
 }
 
+void estep_logw_big(torch::Tensor z, torch::Tensor mfac, torch::Tensor add,
+                    torch::Tensor logw) {
+  TORCH_CHECK(z.is_cuda() && z.is_contiguous() &&
+                  z.scalar_type() == torch::kBFloat16,
+              "z must be contiguous bf16 [D,N]");
+  TORCH_CHECK(mfac.is_contiguous() && mfac.scalar_type() == torch::kBFloat16);
+  check_f32(add, "add");
+  check_f32(logw, "logw");
+  const int d = (int)z.size(0);
+  const int64_t n = z.size(1);
+  const int k = (int)add.size(0);
+  TORCH_CHECK(d > 31 && d <= 143, "estep_logw_big is the 31 < D <= 143 path");
+  TORCH_CHECK(logw.size(0) == k && logw.size(1) == n, "logw shape");
+  const int kct = d + 1 <= 48 ? 3 : d + 1 <= 80 ? 5 : 9;
+  const size_t lds = (size_t)256 * (kct * 16 + 8) * 2;
+  dim3 grid((uint32_t)((n + 255) / 256), (k + 3) / 4);
+  auto s = stream();
+#define LAUNCH_ELB(KCT)                                                        do {                                                                           if (lds > 64 * 1024) {                                                         HIP_CHECK(hipFuncSetAttribute(                                                   reinterpret_cast<const void*>(&gmm::estep_logw_big_kernel<KCT>),             hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds));                }                                                                            hipLaunchKernelGGL((gmm::estep_logw_big_kernel<KCT>), grid, dim3(kNT),                          lds, s,                                                                      reinterpret_cast<const __hip_bfloat16*>(z.data_ptr()),                       reinterpret_cast<const __hip_bfloat16*>(                                         mfac.data_ptr()),                                                        add.data_ptr<float>(), logw.data_ptr<float>(), d, k,                         n);                                                     } while (0)
+  if (kct == 3) LAUNCH_ELB(3);
+  else if (kct == 5) LAUNCH_ELB(5);
+  else LAUNCH_ELB(9);
+#undef LAUNCH_ELB
+  HIP_CHECK(hipGetLastError());
+}
+
+void mstep_moments_big(torch::Tensor x, torch::Tensor w,
+                       torch::Tensor partials) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() &&
+                  x.scalar_type() == torch::kFloat32,
+              "x must be contiguous fp32");
+  check_f32(w, "w");
+  check_f32(partials, "partials");
+  const int d = (int)x.size(0);
+  const int64_t n = x.size(1);
+  const int k = (int)w.size(0);
+  const int nchunk = (int)partials.size(0);
+  const int dp = d + 1;
+  TORCH_CHECK(d > 31 && d <= 159, "mstep_moments_big is the D > 31 path");
+  TORCH_CHECK(partials.size(1) == k &&
+                  partials.size(2) == dp * (dp + 1) / 2,
+              "partials must be [nchunk, K, Dp*(Dp+1)/2]");
+  const int rows = ((dp + 31) / 32) * 32;
+  const size_t lds = (size_t)2 * rows * (64 + 8) * 2 + 2 * 64 * 4;
+  dim3 grid((k + 1) / 2, nchunk);
+  hipLaunchKernelGGL(gmm::mstep_moments_big_kernel, grid, dim3(kNT), lds,
+                     stream(), x.data_ptr<float>(), w.data_ptr<float>(),
+                     partials.data_ptr<float>(), d, k, n, nchunk);
+  HIP_CHECK(hipGetLastError());
+}
+
 void mfma_probe32(torch::Tensor a, torch::Tensor b, torch::Tensor c) {
   TORCH_CHECK(a.scalar_type() == torch::kBFloat16 && a.numel() == 32 * 16);
   TORCH_CHECK(b.scalar_type() == torch::kBFloat16 && b.numel() == 16 * 32);
@@ -320,6 +373,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "packed weighted second-moment partials [nchunk,K,P]");
   m.def("constants", &constants,
         "batched no-pivot LU inverse + ln|det| + bf16 Cholesky factors");
+  m.def("estep_logw_big", &estep_logw_big,
+        "big-D MFMA log-weights (31 < D <= 143)");
+  m.def("mstep_moments_big", &mstep_moments_big,
+        "big-D split-precision moments");
   m.def("mstep_moments_b16", &mstep_moments_b16,
         "split-precision bf16x3 augmented moments");
   m.def("mstep_moments", &mstep_moments,

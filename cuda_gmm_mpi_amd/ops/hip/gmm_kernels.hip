@@ -47,6 +47,13 @@ __device__ inline int mfma_b16_k(int group, int u) { return 8 * group + u; }
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 
+// k-chunk count rounded to the estep-big template tiers (D <= 143); the
+// factor storage is allocated at the tier width with zero-filled padding
+// so templated kernels can unroll a fixed chunk count.
+__host__ __device__ inline int kc_tier(int kc) {
+  return kc <= 2 ? 2 : kc == 3 ? 3 : kc <= 5 ? 5 : 9;
+}
+
 // Stage cluster c's means + packed/pre-symmetrized Rinv into LDS.
 // lds layout: [0, d) means, [d, d + d(d+1)/2) packed rinv.
 __device__ inline void stage_cluster_params(
@@ -332,11 +339,16 @@ __device__ inline void emit_mfac(float* u, const float* __restrict__ means,
     u0[i] = -s;
   }
   __syncthreads();
-  // write hi/lo bf16 fragments, rows padded to 32, k-order per mfma_b16_k
-  __hip_bfloat16* out = mfac + (int64_t)c * 2 * 32 * 32;
-  for (int t = tid; t < 32 * 32; t += NT) {
-    const int i = t / 32, slot = t % 32;
-    const int kx = mfma_b16_k(slot / 8, slot % 8);
+  // write hi/lo bf16 fragments: rows padded to RT*32, k-slots to KC*16
+  // (RT = ceil(d/32) row-tiles, KC = ceil((d+1)/16) MFMA k-chunks; for
+  // d <= 31 this is the original [32][32] layout). k-order is contiguous
+  // (identity fragment map, hardware-verified by the probes).
+  const int rows = ((d + 31) / 32) * 32;
+  const int cols = kc_tier((d + 1 + 15) / 16) * 16;
+  const int cells = rows * cols;
+  __hip_bfloat16* out = mfac + (int64_t)c * 2 * cells;
+  for (int t = tid; t < cells; t += NT) {
+    const int i = t / cols, kx = t % cols;
     float v = 0.0f;
     if (i < d) {
       if (kx < d) v = (kx >= i) ? u[i * d + kx] : 0.0f;
@@ -344,7 +356,7 @@ __device__ inline void emit_mfac(float* u, const float* __restrict__ means,
     }
     const __hip_bfloat16 hi = __float2bfloat16(v);
     out[t] = hi;
-    out[32 * 32 + t] = __float2bfloat16(v - __bfloat162float(hi));
+    out[cells + t] = __float2bfloat16(v - __bfloat162float(hi));
   }
 }
 
@@ -896,6 +908,246 @@ estep_fused_kernel(const __hip_bfloat16* __restrict__ z,
     float total = 0.0f;
     for (int wv = 0; wv < NT / WAVE; ++wv) total += wsum[wv];
     partial[blockIdx.x] = total;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Big-D split-precision moments (31 < D <= 159): same packed output as the
+// small-D kernels. Row-tiles of 32 cover the padded (D+1) dims; the
+// RT2*(RT2+1)/2 tile-pairs of the symmetric output are split across two
+// waves per cluster (two clusters per block share the staged z tile).
+// Grid (ceil(K/2), nchunk); BK = 64 events per tile.
+// ---------------------------------------------------------------------------
+#define MBB_BK 64
+#define MBB_PMAX 8  // max tile-pairs per wave (RT2 <= 5 -> TP <= 15)
+
+__global__ void __launch_bounds__(NT)
+mstep_moments_big_kernel(const float* __restrict__ x,
+                         const float* __restrict__ w,
+                         float* __restrict__ partials, int d, int k,
+                         int64_t n, int nchunk) {
+  typedef __attribute__((ext_vector_type(16))) float f32x16;
+  const int dp = d + 1;
+  const int rt2 = (dp + 31) / 32;
+  const int rows = rt2 * 32;
+  const int tp = rt2 * (rt2 + 1) / 2;
+  const int zbr = MBB_BK + 8;  // bf16 row stride
+  extern __shared__ float lds[];
+  __bf16* zhi = (__bf16*)lds;            // [rows][zbr]
+  __bf16* zlo = zhi + rows * zbr;
+  float* wt = (float*)(zlo + rows * zbr);  // [2][MBB_BK]
+
+  const int wave = threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int j32 = lane & 31;
+  const int g2 = lane >> 5;
+  const int cw = wave >> 1;              // which of the block's 2 clusters
+  const int half = wave & 1;             // which half of the tile-pairs
+  const int c = blockIdx.x * 2 + cw;
+  const int chunk = blockIdx.y;
+  const int p_lo = half ? (tp + 1) / 2 : 0;
+  const int p_hi = half ? tp : (tp + 1) / 2;
+
+  f32x16 acc[MBB_PMAX];
+#pragma unroll
+  for (int pp = 0; pp < MBB_PMAX; ++pp) acc[pp] = (f32x16)(0.0f);
+
+  const int64_t tiles = (n + MBB_BK - 1) / MBB_BK;
+  for (int64_t tile = chunk; tile < tiles; tile += nchunk) {
+    const int64_t e0 = tile * MBB_BK;
+    const int cnt = (int)min((int64_t)MBB_BK, n - e0);
+    __syncthreads();
+    auto put = [&](int di, int ei, float v) {
+      const __bf16 hi = (__bf16)v;
+      zhi[di * zbr + ei] = hi;
+      zlo[di * zbr + ei] = (__bf16)(v - (float)hi);
+    };
+    if (cnt == MBB_BK) {
+      for (int idx = threadIdx.x; idx < d * MBB_BK; idx += NT)
+        put(idx / MBB_BK, idx % MBB_BK,
+            x[(int64_t)(idx / MBB_BK) * n + e0 + idx % MBB_BK]);
+      if (tile == chunk) {
+        for (int idx = d * MBB_BK + threadIdx.x; idx < rows * MBB_BK;
+             idx += NT)
+          put(idx / MBB_BK, idx % MBB_BK,
+              (idx / MBB_BK == d) ? 1.0f : 0.0f);
+      }
+      for (int idx = threadIdx.x; idx < 2 * MBB_BK; idx += NT) {
+        const int ci = blockIdx.x * 2 + idx / MBB_BK;
+        if (ci < k)
+          wt[idx] = w[(int64_t)ci * n + e0 + idx % MBB_BK];
+      }
+    } else {
+      for (int idx = threadIdx.x; idx < rows * MBB_BK; idx += NT) {
+        const int di = idx / MBB_BK, ei = idx % MBB_BK;
+        float v = 0.0f;
+        if (ei < cnt) {
+          if (di < d) v = x[(int64_t)di * n + e0 + ei];
+          else if (di == d) v = 1.0f;
+        }
+        put(di, ei, v);
+      }
+      for (int idx = threadIdx.x; idx < 2 * MBB_BK; idx += NT) {
+        const int ci = blockIdx.x * 2 + idx / MBB_BK;
+        const int ei = idx % MBB_BK;
+        wt[idx] = (ci < k && ei < cnt) ? w[(int64_t)ci * n + e0 + ei] : 0.0f;
+      }
+    }
+    __syncthreads();
+    if (c >= k) continue;
+
+#pragma unroll 2
+    for (int ch = 0; ch < MBB_BK / 16; ++ch) {
+      const int eb = ch * 16 + 8 * g2;
+      const float4 wv0 = *(const float4*)(wt + cw * MBB_BK + eb);
+      const float4 wv1 = *(const float4*)(wt + cw * MBB_BK + eb + 4);
+      const float wv[8] = {wv0.x, wv0.y, wv0.z, wv0.w,
+                           wv1.x, wv1.y, wv1.z, wv1.w};
+#pragma unroll
+      for (int pp = 0; pp < MBB_PMAX; ++pp) {
+        const int p = p_lo + pp;
+        if (p >= p_hi) break;
+        int tr, tc;
+        tri_row_col(p, &tr, &tc);
+        const bf16x8 zah = *(const bf16x8*)(zhi + (tr * 32 + j32) * zbr + eb);
+        const bf16x8 zal = *(const bf16x8*)(zlo + (tr * 32 + j32) * zbr + eb);
+        bf16x8 a_hi, a_lo;
+#pragma unroll
+        for (int u = 0; u < 8; ++u) {
+          const float zf = (float)zah[u] + (float)zal[u];
+          const float av = wv[u] * zf;
+          const __bf16 hi = (__bf16)av;
+          a_hi[u] = hi;
+          a_lo[u] = (__bf16)(av - (float)hi);
+        }
+        const bf16x8 b_hi = *(const bf16x8*)(zhi + (tc * 32 + j32) * zbr + eb);
+        const bf16x8 b_lo = *(const bf16x8*)(zlo + (tc * 32 + j32) * zbr + eb);
+        acc[pp] =
+            __builtin_amdgcn_mfma_f32_32x32x16_bf16(a_hi, b_hi, acc[pp], 0, 0, 0);
+        acc[pp] =
+            __builtin_amdgcn_mfma_f32_32x32x16_bf16(a_hi, b_lo, acc[pp], 0, 0, 0);
+        acc[pp] =
+            __builtin_amdgcn_mfma_f32_32x32x16_bf16(a_lo, b_hi, acc[pp], 0, 0, 0);
+      }
+    }
+  }
+
+  if (c >= k) return;
+  const int p_aug = dp * (dp + 1) / 2;
+  float* out = partials + ((int64_t)chunk * k + c) * p_aug;
+#pragma unroll
+  for (int pp = 0; pp < MBB_PMAX; ++pp) {
+    const int p = p_lo + pp;
+    if (p >= p_hi) break;
+    int tr, tc;
+    tri_row_col(p, &tr, &tc);
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int gi = tr * 32 + (r & 3) + 8 * (r >> 2) + 4 * g2;
+      const int gj = tc * 32 + j32;
+      if (gi < dp && gj <= gi)
+        out[gi * (gi + 1) / 2 + gj] = acc[pp][r];
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Big-D MFMA E-step logw (31 < D <= 143): q = ||M_c z||^2 with the
+// generalized factor layout [K][2][RT*32][KCT*16]. Writes logw to global
+// (the separate posteriors kernel normalizes); replaces the VALU
+// estep_logw_gen path which is L1/L2-latency-bound at D^2/2 re-reads.
+// Grid (ceil(n/256), ceil(K/4)); 4 waves, one cluster per wave, 256 events
+// staged transposed in LDS; per (cluster, row-tile) the KCT A-fragment
+// pairs live in registers (compile-time KCT => no scratch).
+// ---------------------------------------------------------------------------
+#define ESB_BE 256
+
+template <int KCT>
+__global__ void __launch_bounds__(NT)
+estep_logw_big_kernel(const __hip_bfloat16* __restrict__ z,
+                      const __hip_bfloat16* __restrict__ mfac,
+                      const float* __restrict__ add,
+                      float* __restrict__ logw, int d, int k, int64_t n) {
+  typedef __attribute__((ext_vector_type(16))) float f32x16;
+  const int zrowk = KCT * 16 + 8;  // bf16 slots per transposed event row
+  extern __shared__ float lds[];
+  __bf16* zs = (__bf16*)lds;       // [ESB_BE][zrowk]
+
+  const int wave = threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int j32 = lane & 31;
+  const int g2 = lane >> 5;
+  const int64_t e0 = (int64_t)blockIdx.x * ESB_BE;
+  const int cnt = (int)min((int64_t)ESB_BE, n - e0);
+
+  // transposed staging with ones-row and zero-pad baked in (branchless
+  // for full tiles)
+  if (cnt == ESB_BE) {
+    for (int idx = threadIdx.x; idx < d * ESB_BE; idx += NT) {
+      const int kk = idx / ESB_BE, ei = idx % ESB_BE;
+      zs[ei * zrowk + kk] =
+          (__bf16)__bfloat162float(z[(int64_t)kk * n + e0 + ei]);
+    }
+  } else {
+    for (int idx = threadIdx.x; idx < d * ESB_BE; idx += NT) {
+      const int kk = idx / ESB_BE, ei = idx % ESB_BE;
+      zs[ei * zrowk + kk] = (__bf16)(
+          (ei < cnt) ? __bfloat162float(z[(int64_t)kk * n + e0 + ei]) : 0.0f);
+    }
+  }
+  for (int idx = threadIdx.x; idx < (KCT * 16 - d) * ESB_BE; idx += NT) {
+    const int kk = d + idx / ESB_BE, ei = idx % ESB_BE;
+    zs[ei * zrowk + kk] = (__bf16)((kk == d && ei < cnt) ? 1.0f : 0.0f);
+  }
+  __syncthreads();
+
+  const int c = blockIdx.y * 4 + wave;
+  if (c >= k) return;
+  const int rt_n = (d + 31) / 32;
+  const int cols = KCT * 16;
+  const int64_t cells = (int64_t)rt_n * 32 * cols;
+  const bf16x8* mf_hi =
+      (const bf16x8*)(mfac + (int64_t)c * 2 * cells);
+  const bf16x8* mf_lo = mf_hi + cells / 8;
+  const float addc = add[c];
+
+  float q[ESB_BE / 32];
+#pragma unroll
+  for (int t = 0; t < ESB_BE / 32; ++t) q[t] = 0.0f;
+
+  for (int rt = 0; rt < rt_n; ++rt) {
+    // A fragments for this row-tile: KCT chunks x hi/lo, in registers
+    bf16x8 ah[KCT], al[KCT];
+    const int row = rt * 32 + j32;
+    const int rowfr = row * (cols / 8);
+#pragma unroll
+    for (int kc = 0; kc < KCT; ++kc) {
+      ah[kc] = mf_hi[rowfr + kc * 2 + g2];
+      al[kc] = mf_lo[rowfr + kc * 2 + g2];
+    }
+#pragma unroll
+    for (int t = 0; t < ESB_BE / 32; ++t) {
+      const __bf16* zrow = zs + (t * 32 + j32) * zrowk;
+      f32x16 y = (f32x16)(0.0f);
+#pragma unroll
+      for (int kc = 0; kc < KCT; ++kc) {
+        const bf16x8 b = *(const bf16x8*)(zrow + kc * 16 + 8 * g2);
+        y = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ah[kc], b, y, 0, 0, 0);
+        y = __builtin_amdgcn_mfma_f32_32x32x16_bf16(al[kc], b, y, 0, 0, 0);
+      }
+      float s = 0.0f;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) s = fmaf(y[r], y[r], s);
+      q[t] += s;
+    }
+  }
+#pragma unroll
+  for (int t = 0; t < ESB_BE / 32; ++t) {
+    float s = q[t] + __shfl_xor(q[t], 32, WAVE);
+    if (lane < 32) {
+      const int64_t e = e0 + t * 32 + j32;
+      if (e < n) logw[(int64_t)c * n + e] = -0.5f * s + addc;
+    }
   }
 }
 

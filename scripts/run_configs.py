@@ -77,10 +77,10 @@ def main():
         n = int(4_000_000 * args.scale)
         data, _ = make_blobs(n, 128, 64, seed=7)
         cfg = GmmConfig(num_clusters=256, target_num_clusters=256,
-                        estep_dtype="bf16")
+                        estep_dtype="bf16", mstep_precision="bf16x3")
         ips, eng = time_em(data, cfg, dev, 256, 5, 2)
         emit(4, {"device": dev, "n": n, "em_iters_per_sec": ips,
-                 "fused_estep": eng.use_fused_estep})
+                 "big_estep": eng.use_big_estep})
 
     if 5 in configs:
         n = int(2_000_000 * args.scale)

@@ -327,3 +327,70 @@ def test_mstep_moments_b16_matches_cpu(device, d, n):
     scale = float(rs.abs().max())
     np.testing.assert_allclose(s.cpu().numpy(), rs.numpy(),
                                rtol=3e-4, atol=3e-4 * scale)
+
+
+@pytest.mark.parametrize("d", [40, 64, 100, 128])
+def test_estep_logw_big_matches_cpu(device, d):
+    """Big-D MFMA logw vs fp32 torch reference (bf16-class tolerance)."""
+    from cuda_gmm_mpi_amd.ops import functional as F
+    rng = np.random.default_rng(d)
+    k, n = 6, 2000 + 57
+    means, r, pi = random_model(rng, k, d, device)
+    mfac = torch.empty(k, *F.mfac_shape(d), dtype=torch.bfloat16,
+                       device=device)
+    rinv_g, const_g = F.constants(r, means, False, mfac)
+    add = const_g + torch.log(pi)
+    x = rng.standard_normal((d, n)).astype(np.float32)
+    out = torch.empty(k, n, dtype=torch.float32, device=device)
+    F.estep_logw_big(torch.from_numpy(x).to(device).to(torch.bfloat16),
+                     mfac, add, out)
+    rinv, const = cpu.compute_constants(r.cpu())
+    ref = cpu.estep_logw(torch.from_numpy(x), means.cpu(), rinv, const,
+                         pi.cpu())
+    # |logw| grows with D; tolerance scales accordingly
+    np.testing.assert_allclose(out.cpu().numpy(), ref.numpy(),
+                               rtol=5e-2, atol=2.0)
+
+
+@pytest.mark.parametrize("d,n", [(40, 5000), (64, 3001), (128, 2000)])
+def test_mstep_moments_big_matches_cpu(device, d, n):
+    from cuda_gmm_mpi_amd.ops import functional as F
+    rng = np.random.default_rng(d * 7 + 1)
+    k = 5
+    x = rng.standard_normal((d, n)).astype(np.float32) * 2
+    w = rng.uniform(0, 1, (k, n)).astype(np.float32)
+    packed = F.mstep_moments(torch.from_numpy(x).to(device),
+                             torch.from_numpy(w).to(device),
+                             precision="bf16x3")
+    n_c, mean_num, s = F.moments_views(packed, d)
+    rn, rm, rs = cpu.mstep_sufficient_stats(torch.from_numpy(x).double(),
+                                            torch.from_numpy(w).double())
+    np.testing.assert_allclose(n_c.cpu().numpy(), rn.numpy(), rtol=1e-4)
+    np.testing.assert_allclose(mean_num.cpu().numpy(), rm.numpy(),
+                               rtol=3e-4, atol=1e-2)
+    scale = float(rs.abs().max())
+    np.testing.assert_allclose(s.cpu().numpy(), rs.numpy(),
+                               rtol=5e-4, atol=5e-4 * scale)
+
+
+def test_engine_big_d_bf16_matches_fp32(device=None):
+    """Config-4-shaped engine run (scaled down): big-D MFMA paths vs the
+    fp32 VALU/compose paths."""
+    from cuda_gmm_mpi_amd.engine import build_engine
+    from cuda_gmm_mpi_amd.utils.config import GmmConfig
+    from cuda_gmm_mpi_amd.utils.synthetic import make_blobs
+    data, _ = make_blobs(30000, 64, 8, seed=61)
+    res = {}
+    for name, ed, mp in (("fp32", "fp32", "fp32"),
+                         ("bf16", "bf16", "bf16x3")):
+        cfg = GmmConfig(num_clusters=8, target_num_clusters=8,
+                        min_iters=6, max_iters=6, estep_dtype=ed,
+                        mstep_precision=mp)
+        eng = build_engine(data, cfg, device="cuda")
+        lik = eng.run_em(8)
+        res[name] = (lik, eng.state.means.cpu().numpy().copy(),
+                     eng.use_big_estep)
+    assert res["bf16"][2] is True and res["fp32"][2] is False
+    assert res["bf16"][0] == pytest.approx(res["fp32"][0], rel=2e-3)
+    np.testing.assert_allclose(res["bf16"][1], res["fp32"][1],
+                               rtol=5e-2, atol=5e-1)
