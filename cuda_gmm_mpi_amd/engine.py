@@ -71,6 +71,13 @@ class EmEngine:
             self.x_estep: torch.Tensor = self.x.to(torch.bfloat16)
         else:
             self.x_estep = self.x
+        # persistent hi/lo bf16 planes for the split-precision M-step
+        self.x_split = (
+            F.split_bf16_planes(self.x)
+            if (config.mstep_precision == "bf16x3"
+                and self.device.type == "cuda")
+            else None
+        )
 
         k0 = config.num_clusters
         # fused bf16-MFMA E-step path (D <= 31, LDS-bounded K); big-D MFMA
@@ -182,7 +189,8 @@ class EmEngine:
         st = self.state.shrink(k)
         with self.profile.time("m_step"):
             packed = F.mstep_moments(self.x, self.w[:k],
-                                     precision=self.cfg.mstep_precision)
+                                     precision=self.cfg.mstep_precision,
+                                     x_split=self.x_split)
         with self.profile.time("comm"):
             pdist.all_reduce_(packed)
         with self.profile.time("m_step"):

@@ -177,9 +177,18 @@ def estep_logw_big(z: torch.Tensor, mfac: torch.Tensor, add: torch.Tensor,
     return out
 
 
+def split_bf16_planes(x: torch.Tensor) -> tuple[torch.Tensor, torch.Tensor]:
+    """Persistent hi/lo bf16 split of fp32 data (x = hi + lo + O(2^-18 x))."""
+    hi = x.to(torch.bfloat16)
+    lo = (x - hi.to(torch.float32)).to(torch.bfloat16)
+    return hi.contiguous(), lo.contiguous()
+
+
 def mstep_moments(x: torch.Tensor, w: torch.Tensor,
                   nchunk: int | None = None,
-                  precision: str = "fp32") -> torch.Tensor:
+                  precision: str = "fp32",
+                  x_split: tuple[torch.Tensor, torch.Tensor] | None = None
+                  ) -> torch.Tensor:
     """Fused augmented sufficient statistics, packed lower triangle of
     T_c = sum_e w_ce [x;1][x;1]^T per cluster: [K, Dp*(Dp+1)/2] with layout
     [S_tri (D rows) | mean numerators (row D) | N (corner)] — the single
@@ -204,7 +213,9 @@ def mstep_moments(x: torch.Tensor, w: torch.Tensor,
         partials = torch.empty((nchunk, k, pp), dtype=torch.float32,
                                device=x.device)
         if precision == "bf16x3":
-            hip_ext().mstep_moments_b16(x, w, partials)
+            if x_split is None:
+                x_split = split_bf16_planes(x)
+            hip_ext().mstep_moments_b16(x_split[0], x_split[1], w, partials)
         else:
             hip_ext().mstep_moments(x, w, partials)
         return partials.sum(dim=0)

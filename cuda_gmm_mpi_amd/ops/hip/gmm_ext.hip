@@ -236,15 +236,16 @@ void mstep_moments(torch::Tensor x, torch::Tensor w, torch::Tensor partials) {
   HIP_CHECK(hipGetLastError());
 }
 
-void mstep_moments_b16(torch::Tensor x, torch::Tensor w,
-                       torch::Tensor partials) {
-  TORCH_CHECK(x.is_cuda() && x.is_contiguous() &&
-                  x.scalar_type() == torch::kFloat32,
-              "x must be contiguous fp32");
+void mstep_moments_b16(torch::Tensor xhi, torch::Tensor xlo,
+                       torch::Tensor w, torch::Tensor partials) {
+  TORCH_CHECK(xhi.is_cuda() && xhi.is_contiguous() && xlo.is_contiguous() &&
+                  xhi.scalar_type() == torch::kBFloat16 &&
+                  xlo.scalar_type() == torch::kBFloat16,
+              "xhi/xlo must be contiguous bf16 planes");
   check_f32(w, "w");
   check_f32(partials, "partials");
-  const int d = (int)x.size(0);
-  const int64_t n = x.size(1);
+  const int d = (int)xhi.size(0);
+  const int64_t n = xhi.size(1);
   const int k = (int)w.size(0);
   const int nchunk = (int)partials.size(0);
   const int dp = d + 1;
@@ -252,11 +253,14 @@ void mstep_moments_b16(torch::Tensor x, torch::Tensor w,
   TORCH_CHECK(partials.size(1) == k &&
                   partials.size(2) == dp * (dp + 1) / 2,
               "partials must be [nchunk, K, Dp*(Dp+1)/2]");
-  const size_t lds = 32 * 132 * 4 + 2 * 32 * 136 * 2 + 4 * 128 * 4;
+  const size_t lds = 2 * 32 * 136 * 2 + 4 * 128 * 4;
   dim3 grid((k + 3) / 4, nchunk);
   hipLaunchKernelGGL(gmm::mstep_moments_b16_kernel, grid, dim3(kNT), lds,
-                     stream(), x.data_ptr<float>(), w.data_ptr<float>(),
-                     partials.data_ptr<float>(), d, k, n, nchunk);
+                     stream(),
+                     reinterpret_cast<const __hip_bfloat16*>(xhi.data_ptr()),
+                     reinterpret_cast<const __hip_bfloat16*>(xlo.data_ptr()),
+                     w.data_ptr<float>(), partials.data_ptr<float>(), d, k,
+                     n, nchunk);
   HIP_CHECK(hipGetLastError());
 }
 
@@ -334,7 +338,7 @@ void mstep_moments_big(torch::Tensor x, torch::Tensor w,
   const int rows = ((dp + 31) / 32) * 32;
   const size_t lds = (size_t)2 * rows * (64 + 8) * 2 + 2 * 64 * 4;
   dim3 grid((k + 1) / 2, nchunk);
-  hipLaunchKernelGGL(gmm::mstep_moments_big_kernel, grid, dim3(kNT), lds,
+  hipLaunchKernelGGL(gmm::mstep_moments_big_kernel, grid, dim3(512), lds,
                      stream(), x.data_ptr<float>(), w.data_ptr<float>(),
                      partials.data_ptr<float>(), d, k, n, nchunk);
   HIP_CHECK(hipGetLastError());

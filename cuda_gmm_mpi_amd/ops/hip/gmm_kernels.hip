@@ -613,19 +613,18 @@ mstep_moments_kernel(const T* __restrict__ x, const float* __restrict__ w,
 typedef __attribute__((ext_vector_type(16))) float f32x16_t;
 
 __global__ void __launch_bounds__(NT)
-mstep_moments_b16_kernel(const float* __restrict__ x,
+mstep_moments_b16_kernel(const __hip_bfloat16* __restrict__ xhi,
+                         const __hip_bfloat16* __restrict__ xlo,
                          const float* __restrict__ w,
                          float* __restrict__ partials, int d, int k,
                          int64_t n, int nchunk) {
-  // LDS (all rows padded to 32 dims, ones-row at d, zeros above):
-  //   z32 [32][MB_BK] f32, zhi/zlo [32][MB_BK] bf16, wt [4][MB_BK] f32
+  // x arrives pre-split into persistent global hi/lo bf16 planes (x never
+  // changes across EM iterations; splitting at staging time was the
+  // measured VALU hog). LDS rows padded so per-dim b128 reads spread
+  // across banks: zhi/zlo [32][136] bf16, wt [4][MB_BK] f32
   extern __shared__ float lds[];
-  // rows padded so per-dim b128 reads spread across banks: an unpadded
-  // 128-element stride puts every lane of a 16-lane group on one bank slot
-  constexpr int Z32R = MB_BK + 4;  // f32 row stride
   constexpr int ZBR = MB_BK + 8;   // bf16 row stride (keeps 16B row align)
-  float* z32 = lds;                               // 32*Z32R floats
-  __bf16* zhi = (__bf16*)(lds + 32 * Z32R);
+  __bf16* zhi = (__bf16*)lds;
   __bf16* zlo = zhi + 32 * ZBR;
   float* wt = (float*)(zlo + 32 * ZBR);           // 4*MB_BK floats
 
@@ -649,22 +648,19 @@ mstep_moments_b16_kernel(const float* __restrict__ x,
     // branchless full-tile staging (guide §5 trap 4c); constant rows
     // (ones at d, zeros above) written once per block, re-fixed only
     // after a tail tile (the tail is always the globally last tile)
-    auto put = [&](int idx, float v) {
-      // native __bf16 casts lower to the hardware cvt (the
-      // __float2bfloat16 helper is a multi-instruction software RNE)
-      const int di = idx / MB_BK, ei = idx % MB_BK;
-      const __bf16 hi = (__bf16)v;
-      z32[di * Z32R + ei] = v;
-      zhi[di * ZBR + ei] = hi;
-      zlo[di * ZBR + ei] = (__bf16)(v - (float)hi);
-    };
     const int nv = min(4, k - blockIdx.x * 4) * MB_BK;
     if (cnt == MB_BK) {
-      for (int idx = threadIdx.x; idx < d * MB_BK; idx += NT)
-        put(idx, x[(int64_t)(idx / MB_BK) * n + e0 + idx % MB_BK]);
+      for (int idx = threadIdx.x; idx < d * MB_BK; idx += NT) {
+        const int di = idx / MB_BK, ei = idx % MB_BK;
+        zhi[di * ZBR + ei] = *(const __bf16*)&xhi[(int64_t)di * n + e0 + ei];
+        zlo[di * ZBR + ei] = *(const __bf16*)&xlo[(int64_t)di * n + e0 + ei];
+      }
       if (tile == chunk) {  // first tile: fill the constant rows
-        for (int idx = d * MB_BK + threadIdx.x; idx < 32 * MB_BK; idx += NT)
-          put(idx, (idx / MB_BK == d) ? 1.0f : 0.0f);
+        for (int idx = d * MB_BK + threadIdx.x; idx < 32 * MB_BK; idx += NT) {
+          const int di = idx / MB_BK, ei = idx % MB_BK;
+          zhi[di * ZBR + ei] = (__bf16)(di == d ? 1.0f : 0.0f);
+          zlo[di * ZBR + ei] = (__bf16)0.0f;
+        }
       }
       for (int idx = threadIdx.x; idx < nv; idx += NT)
         wt[idx] = w[(int64_t)(blockIdx.x * 4 + idx / MB_BK) * n + e0 +
@@ -672,12 +668,17 @@ mstep_moments_b16_kernel(const float* __restrict__ x,
     } else {
       for (int idx = threadIdx.x; idx < 32 * MB_BK; idx += NT) {
         const int di = idx / MB_BK, ei = idx % MB_BK;
-        float v = 0.0f;
+        float hi = 0.0f, lo = 0.0f;
         if (ei < cnt) {
-          if (di < d) v = x[(int64_t)di * n + e0 + ei];
-          else if (di == d) v = 1.0f;
+          if (di < d) {
+            hi = __bfloat162float(xhi[(int64_t)di * n + e0 + ei]);
+            lo = __bfloat162float(xlo[(int64_t)di * n + e0 + ei]);
+          } else if (di == d) {
+            hi = 1.0f;
+          }
         }
-        put(idx, v);
+        zhi[di * ZBR + ei] = (__bf16)hi;
+        zlo[di * ZBR + ei] = (__bf16)lo;
       }
       for (int idx = threadIdx.x; idx < 4 * MB_BK; idx += NT) {
         const int wv = idx / MB_BK, ei = idx % MB_BK;
@@ -693,20 +694,19 @@ mstep_moments_b16_kernel(const float* __restrict__ x,
       // B fragments: contiguous events of this lane's dim row
       const bf16x8 b_hi = *(const bf16x8*)(zhi + j32 * ZBR + eb);
       const bf16x8 b_lo = *(const bf16x8*)(zlo + j32 * ZBR + eb);
-      // A fragments: w-weighted z, split on the fly
-      const float4 zv0 = *(const float4*)(z32 + j32 * Z32R + eb);
-      const float4 zv1 = *(const float4*)(z32 + j32 * Z32R + eb + 4);
+      // A fragments: w-weighted z (reconstructed hi+lo), split on the fly
       const float4 wv0 = *(const float4*)(wt + wave * MB_BK + eb);
       const float4 wv1 = *(const float4*)(wt + wave * MB_BK + eb + 4);
-      float av[8] = {zv0.x * wv0.x, zv0.y * wv0.y, zv0.z * wv0.z,
-                     zv0.w * wv0.w, zv1.x * wv1.x, zv1.y * wv1.y,
-                     zv1.z * wv1.z, zv1.w * wv1.w};
+      const float wv[8] = {wv0.x, wv0.y, wv0.z, wv0.w,
+                           wv1.x, wv1.y, wv1.z, wv1.w};
       bf16x8 a_hi, a_lo;
 #pragma unroll
       for (int u = 0; u < 8; ++u) {
-        const __bf16 hi = (__bf16)av[u];
+        const float zf = (float)b_hi[u] + (float)b_lo[u];
+        const float av = wv[u] * zf;
+        const __bf16 hi = (__bf16)av;
         a_hi[u] = hi;
-        a_lo[u] = (__bf16)(av[u] - (float)hi);
+        a_lo[u] = (__bf16)(av - (float)hi);
       }
       accA = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a_hi, b_hi, accA, 0, 0, 0);
       accB = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a_hi, b_lo, accB, 0, 0, 0);
@@ -919,9 +919,10 @@ estep_fused_kernel(const __hip_bfloat16* __restrict__ z,
 // Grid (ceil(K/2), nchunk); BK = 64 events per tile.
 // ---------------------------------------------------------------------------
 #define MBB_BK 64
-#define MBB_PMAX 8  // max tile-pairs per wave (RT2 <= 5 -> TP <= 15)
+#define MBB_NT 512  // 8 waves: 2 clusters x 4 pair-quarters
+#define MBB_PMAX 4  // tile-pairs per wave (RT2 <= 5 -> TP <= 15 -> 4/quarter)
 
-__global__ void __launch_bounds__(NT)
+__global__ void __launch_bounds__(MBB_NT)
 mstep_moments_big_kernel(const float* __restrict__ x,
                          const float* __restrict__ w,
                          float* __restrict__ partials, int d, int k,
@@ -941,12 +942,20 @@ mstep_moments_big_kernel(const float* __restrict__ x,
   const int lane = threadIdx.x & (WAVE - 1);
   const int j32 = lane & 31;
   const int g2 = lane >> 5;
-  const int cw = wave >> 1;              // which of the block's 2 clusters
-  const int half = wave & 1;             // which half of the tile-pairs
+  const int cw = wave >> 2;              // which of the block's 2 clusters
+  const int quarter = wave & 3;          // which quarter of the tile-pairs
   const int c = blockIdx.x * 2 + cw;
   const int chunk = blockIdx.y;
-  const int p_lo = half ? (tp + 1) / 2 : 0;
-  const int p_hi = half ? tp : (tp + 1) / 2;
+  const int p_lo = quarter * MBB_PMAX;
+  const int p_hi = min(tp, p_lo + MBB_PMAX);
+  // hoisted pair->tile mapping (tri_row_col has a sqrtf: keep it out of
+  // the chunk loop)
+  int ptr[MBB_PMAX], ptc[MBB_PMAX];
+#pragma unroll
+  for (int pp = 0; pp < MBB_PMAX; ++pp) {
+    if (p_lo + pp < tp) tri_row_col(p_lo + pp, &ptr[pp], &ptc[pp]);
+    else { ptr[pp] = 0; ptc[pp] = 0; }
+  }
 
   f32x16 acc[MBB_PMAX];
 #pragma unroll
@@ -963,22 +972,22 @@ mstep_moments_big_kernel(const float* __restrict__ x,
       zlo[di * zbr + ei] = (__bf16)(v - (float)hi);
     };
     if (cnt == MBB_BK) {
-      for (int idx = threadIdx.x; idx < d * MBB_BK; idx += NT)
+      for (int idx = threadIdx.x; idx < d * MBB_BK; idx += MBB_NT)
         put(idx / MBB_BK, idx % MBB_BK,
             x[(int64_t)(idx / MBB_BK) * n + e0 + idx % MBB_BK]);
       if (tile == chunk) {
         for (int idx = d * MBB_BK + threadIdx.x; idx < rows * MBB_BK;
-             idx += NT)
+             idx += MBB_NT)
           put(idx / MBB_BK, idx % MBB_BK,
               (idx / MBB_BK == d) ? 1.0f : 0.0f);
       }
-      for (int idx = threadIdx.x; idx < 2 * MBB_BK; idx += NT) {
+      for (int idx = threadIdx.x; idx < 2 * MBB_BK; idx += MBB_NT) {
         const int ci = blockIdx.x * 2 + idx / MBB_BK;
         if (ci < k)
           wt[idx] = w[(int64_t)ci * n + e0 + idx % MBB_BK];
       }
     } else {
-      for (int idx = threadIdx.x; idx < rows * MBB_BK; idx += NT) {
+      for (int idx = threadIdx.x; idx < rows * MBB_BK; idx += MBB_NT) {
         const int di = idx / MBB_BK, ei = idx % MBB_BK;
         float v = 0.0f;
         if (ei < cnt) {
@@ -987,7 +996,7 @@ mstep_moments_big_kernel(const float* __restrict__ x,
         }
         put(di, ei, v);
       }
-      for (int idx = threadIdx.x; idx < 2 * MBB_BK; idx += NT) {
+      for (int idx = threadIdx.x; idx < 2 * MBB_BK; idx += MBB_NT) {
         const int ci = blockIdx.x * 2 + idx / MBB_BK;
         const int ei = idx % MBB_BK;
         wt[idx] = (ci < k && ei < cnt) ? w[(int64_t)ci * n + e0 + ei] : 0.0f;
@@ -1005,10 +1014,8 @@ mstep_moments_big_kernel(const float* __restrict__ x,
                            wv1.x, wv1.y, wv1.z, wv1.w};
 #pragma unroll
       for (int pp = 0; pp < MBB_PMAX; ++pp) {
-        const int p = p_lo + pp;
-        if (p >= p_hi) break;
-        int tr, tc;
-        tri_row_col(p, &tr, &tc);
+        if (p_lo + pp >= p_hi) break;
+        const int tr = ptr[pp], tc = ptc[pp];
         const bf16x8 zah = *(const bf16x8*)(zhi + (tr * 32 + j32) * zbr + eb);
         const bf16x8 zal = *(const bf16x8*)(zlo + (tr * 32 + j32) * zbr + eb);
         bf16x8 a_hi, a_lo;
@@ -1037,10 +1044,8 @@ mstep_moments_big_kernel(const float* __restrict__ x,
   float* out = partials + ((int64_t)chunk * k + c) * p_aug;
 #pragma unroll
   for (int pp = 0; pp < MBB_PMAX; ++pp) {
-    const int p = p_lo + pp;
-    if (p >= p_hi) break;
-    int tr, tc;
-    tri_row_col(p, &tr, &tc);
+    if (p_lo + pp >= p_hi) break;
+    const int tr = ptr[pp], tc = ptc[pp];
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const int gi = tr * 32 + (r & 3) + 8 * (r >> 2) + 4 * g2;
@@ -1125,7 +1130,7 @@ estep_logw_big_kernel(const __hip_bfloat16* __restrict__ z,
       ah[kc] = mf_hi[rowfr + kc * 2 + g2];
       al[kc] = mf_lo[rowfr + kc * 2 + g2];
     }
-#pragma unroll
+#pragma unroll 2
     for (int t = 0; t < ESB_BE / 32; ++t) {
       const __bf16* zrow = zs + (t * 32 + j32) * zrowk;
       f32x16 y = (f32x16)(0.0f);

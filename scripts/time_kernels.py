@@ -12,7 +12,8 @@ def t(f, iters=30):
     for _ in range(iters): f()
     torch.cuda.synchronize(); return (time.perf_counter()-t0)/iters*1e3
 def m32(): return F.mstep_moments(x, w)
-def mb16(): return F.mstep_moments(x, w, precision="bf16x3")
+xs = F.split_bf16_planes(x)
+def mb16(): return F.mstep_moments(x, w, precision="bf16x3", x_split=xs)
 print("moments fp32 : %.3f ms" % t(m32))
 print("moments b16x3: %.3f ms" % t(mb16))
 means = torch.randn(k, d, device="cuda")
