@@ -253,7 +253,8 @@ void mstep_moments_b16(torch::Tensor xhi, torch::Tensor xlo,
   TORCH_CHECK(partials.size(1) == k &&
                   partials.size(2) == dp * (dp + 1) / 2,
               "partials must be [nchunk, K, Dp*(Dp+1)/2]");
-  const size_t lds = 2 * 32 * 136 * 2 + 4 * 128 * 4;
+  // two buffers x (zhi+zlo planes + w tile)
+  const size_t lds = 2 * (2 * 32 * 136 * 2 + 4 * 128 * 4);
   dim3 grid((k + 3) / 4, nchunk);
   hipLaunchKernelGGL(gmm::mstep_moments_b16_kernel, grid, dim3(kNT), lds,
                      stream(),

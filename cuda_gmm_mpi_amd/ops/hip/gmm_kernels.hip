@@ -46,6 +46,7 @@ __device__ inline void tri_row_col(int t, int* i, int* j) {
 __device__ inline int mfma_b16_k(int group, int u) { return 8 * group + u; }
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(16))) float f32x16_t;
 
 // k-chunk count rounded to the estep-big template tiers (D <= 143); the
 // factor storage is allocated at the tier width with zero-filled padding
@@ -609,8 +610,7 @@ mstep_moments_kernel(const T* __restrict__ x, const float* __restrict__ w,
 //   B[kk=event][j=dim] = z_j[e]          (split once, shared by 4 clusters)
 // ---------------------------------------------------------------------------
 #define MB_BK 128
-
-typedef __attribute__((ext_vector_type(16))) float f32x16_t;
+#define MB_CPB 4  // clusters per block (one per wave)
 
 __global__ void __launch_bounds__(NT)
 mstep_moments_b16_kernel(const __hip_bfloat16* __restrict__ xhi,
@@ -618,83 +618,132 @@ mstep_moments_b16_kernel(const __hip_bfloat16* __restrict__ xhi,
                          const float* __restrict__ w,
                          float* __restrict__ partials, int d, int k,
                          int64_t n, int nchunk) {
-  // x arrives pre-split into persistent global hi/lo bf16 planes (x never
-  // changes across EM iterations; splitting at staging time was the
-  // measured VALU hog). LDS rows padded so per-dim b128 reads spread
-  // across banks: zhi/zlo [32][136] bf16, wt [4][MB_BK] f32
+  // Split-precision moments with double-buffered, register-staged tiles
+  // (guide T14): tile t+1's global loads are issued before tile t's MFMA
+  // work and written to the other LDS buffer after it — the measured 52%
+  // SQ_WAIT share was staging latency exposed at the tile barrier.
+  // x arrives pre-split into persistent hi/lo bf16 planes.
+  // LDS per buffer: zhi/zlo [32][136] bf16, wt [MB_CPB][MB_BK] f32.
   extern __shared__ float lds[];
-  constexpr int ZBR = MB_BK + 8;   // bf16 row stride (keeps 16B row align)
-  __bf16* zhi = (__bf16*)lds;
-  __bf16* zlo = zhi + 32 * ZBR;
-  float* wt = (float*)(zlo + 32 * ZBR);           // 4*MB_BK floats
+  constexpr int ZBR = MB_BK + 8;
+  constexpr int PLANE = 32 * ZBR;          // bf16 elements per z plane
+  constexpr int BUFB = 2 * PLANE;          // zhi+zlo per buffer (bf16)
+  __bf16* zbuf = (__bf16*)lds;             // [2][2*PLANE]
+  float* wbuf = (float*)(zbuf + 2 * BUFB); // [2][MB_CPB*MB_BK]
 
   const int wave = threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
   const int j32 = lane & 31;
   const int g2 = lane >> 5;
-  const int c = blockIdx.x * 4 + wave;
+  const int c = blockIdx.x * MB_CPB + wave;
   const int chunk = blockIdx.y;
   const int dp = d + 1;
+
+  // this thread's staging assignment: quads of 4 consecutive events
+  // (x: d*MB_BK/4 quads; w: MB_CPB*MB_BK/4 float4, first quarter of threads)
+  const int xq_total = d * (MB_BK / 4);
+  const int nxq = (xq_total + NT - 1) / NT;  // quads per thread (<= 4 at D<=31)
 
   f32x16_t accA = (f32x16_t)(0.0f);   // hi*hi
   f32x16_t accB = (f32x16_t)(0.0f);   // hi*lo
   f32x16_t accC = (f32x16_t)(0.0f);   // lo*hi
 
   const int64_t tiles = (n + MB_BK - 1) / MB_BK;
-  for (int64_t tile = chunk; tile < tiles; tile += nchunk) {
+  const int64_t my_tiles =
+      chunk < tiles ? (tiles - chunk + nchunk - 1) / nchunk : 0;
+
+  // ---- staging helpers -------------------------------------------------
+  uint2 rxh[4], rxl[4];
+  float4 rw;
+  auto issue_loads = [&](int64_t tile) {
     const int64_t e0 = tile * MB_BK;
-    const int cnt = (int)min((int64_t)MB_BK, n - e0);
-    __syncthreads();
-    // branchless full-tile staging (guide §5 trap 4c); constant rows
-    // (ones at d, zeros above) written once per block, re-fixed only
-    // after a tail tile (the tail is always the globally last tile)
-    const int nv = min(4, k - blockIdx.x * 4) * MB_BK;
-    if (cnt == MB_BK) {
-      for (int idx = threadIdx.x; idx < d * MB_BK; idx += NT) {
-        const int di = idx / MB_BK, ei = idx % MB_BK;
-        zhi[di * ZBR + ei] = *(const __bf16*)&xhi[(int64_t)di * n + e0 + ei];
-        zlo[di * ZBR + ei] = *(const __bf16*)&xlo[(int64_t)di * n + e0 + ei];
-      }
-      if (tile == chunk) {  // first tile: fill the constant rows
-        for (int idx = d * MB_BK + threadIdx.x; idx < 32 * MB_BK; idx += NT) {
-          const int di = idx / MB_BK, ei = idx % MB_BK;
-          zhi[di * ZBR + ei] = (__bf16)(di == d ? 1.0f : 0.0f);
-          zlo[di * ZBR + ei] = (__bf16)0.0f;
-        }
-      }
-      for (int idx = threadIdx.x; idx < nv; idx += NT)
-        wt[idx] = w[(int64_t)(blockIdx.x * 4 + idx / MB_BK) * n + e0 +
-                    idx % MB_BK];
-    } else {
-      for (int idx = threadIdx.x; idx < 32 * MB_BK; idx += NT) {
-        const int di = idx / MB_BK, ei = idx % MB_BK;
-        float hi = 0.0f, lo = 0.0f;
-        if (ei < cnt) {
-          if (di < d) {
-            hi = __bfloat162float(xhi[(int64_t)di * n + e0 + ei]);
-            lo = __bfloat162float(xlo[(int64_t)di * n + e0 + ei]);
-          } else if (di == d) {
-            hi = 1.0f;
+    const bool full = (n - e0) >= MB_BK;
+#pragma unroll
+    for (int s = 0; s < 4; ++s) {
+      const int q = threadIdx.x + s * NT;
+      if (s < nxq && q < xq_total) {
+        const int di = q / (MB_BK / 4), eq = q % (MB_BK / 4);
+        const int64_t g = (int64_t)di * n + e0 + eq * 4;
+        if (full || eq * 4 + 3 < (int)(n - e0)) {
+          rxh[s] = *(const uint2*)&xhi[g];
+          rxl[s] = *(const uint2*)&xlo[g];
+        } else {  // tail tile: element-wise guarded
+          __hip_bfloat16 h[4], l[4];
+          for (int u = 0; u < 4; ++u) {
+            const bool ok = e0 + eq * 4 + u < n;
+            h[u] = ok ? xhi[g + u] : __hip_bfloat16(0.0f);
+            l[u] = ok ? xlo[g + u] : __hip_bfloat16(0.0f);
           }
+          rxh[s] = *(uint2*)h;
+          rxl[s] = *(uint2*)l;
         }
-        zhi[di * ZBR + ei] = (__bf16)hi;
-        zlo[di * ZBR + ei] = (__bf16)lo;
-      }
-      for (int idx = threadIdx.x; idx < 4 * MB_BK; idx += NT) {
-        const int wv = idx / MB_BK, ei = idx % MB_BK;
-        const int cw = blockIdx.x * 4 + wv;
-        wt[idx] = (cw < k && ei < cnt) ? w[(int64_t)cw * n + e0 + ei] : 0.0f;
       }
     }
-    __syncthreads();
+    const int wq_total = MB_CPB * (MB_BK / 4);
+    if (threadIdx.x < wq_total) {
+      const int wv = threadIdx.x / (MB_BK / 4);
+      const int eq = threadIdx.x % (MB_BK / 4);
+      const int cw = blockIdx.x * MB_CPB + wv;
+      if (cw < k) {
+        const int64_t g = (int64_t)cw * n + e0 + eq * 4;
+        if (full) {
+          rw = *(const float4*)&w[g];
+        } else {
+          float v[4];
+          for (int u = 0; u < 4; ++u)
+            v[u] = (e0 + eq * 4 + u < n) ? w[g + u] : 0.0f;
+          rw = *(float4*)v;
+        }
+      } else {
+        rw = (float4){0, 0, 0, 0};
+      }
+    }
+  };
+  auto write_buf = [&](int buf) {
+    __bf16* zh = zbuf + buf * BUFB;
+    __bf16* zl = zh + PLANE;
+#pragma unroll
+    for (int s = 0; s < 4; ++s) {
+      const int q = threadIdx.x + s * NT;
+      if (s < nxq && q < xq_total) {
+        const int di = q / (MB_BK / 4), eq = q % (MB_BK / 4);
+        *(uint2*)(zh + di * ZBR + eq * 4) = rxh[s];
+        *(uint2*)(zl + di * ZBR + eq * 4) = rxl[s];
+      }
+    }
+    if (threadIdx.x < MB_CPB * (MB_BK / 4))
+      *(float4*)(wbuf + buf * MB_CPB * MB_BK + threadIdx.x * 4) = rw;
+  };
+  // constant rows (ones at d, zeros above) in BOTH buffers, written once;
+  // a tail tile is always the globally last so no re-fix is needed
+  for (int b = 0; b < 2; ++b) {
+    __bf16* zh = zbuf + b * BUFB;
+    __bf16* zl = zh + PLANE;
+    for (int idx = d * MB_BK + threadIdx.x; idx < 32 * MB_BK; idx += NT) {
+      const int di = idx / MB_BK, ei = idx % MB_BK;
+      zh[di * ZBR + ei] = (__bf16)(di == d ? 1.0f : 0.0f);
+      zl[di * ZBR + ei] = (__bf16)0.0f;
+    }
+  }
+
+  if (my_tiles > 0) {
+    issue_loads(chunk);
+    write_buf(0);
+  }
+  __syncthreads();
+
+  int cur = 0;
+  for (int64_t ti = 0; ti < my_tiles; ++ti) {
+    if (ti + 1 < my_tiles) issue_loads(chunk + (ti + 1) * nchunk);
+    const __bf16* zh = zbuf + cur * BUFB;
+    const __bf16* zl = zh + PLANE;
+    const float* wt = wbuf + cur * MB_CPB * MB_BK;
 
 #pragma unroll 2
     for (int ch = 0; ch < MB_BK / 16; ++ch) {
-      const int eb = ch * 16 + 8 * g2;  // this lane's 8 events
-      // B fragments: contiguous events of this lane's dim row
-      const bf16x8 b_hi = *(const bf16x8*)(zhi + j32 * ZBR + eb);
-      const bf16x8 b_lo = *(const bf16x8*)(zlo + j32 * ZBR + eb);
-      // A fragments: w-weighted z (reconstructed hi+lo), split on the fly
+      const int eb = ch * 16 + 8 * g2;
+      const bf16x8 b_hi = *(const bf16x8*)(zh + j32 * ZBR + eb);
+      const bf16x8 b_lo = *(const bf16x8*)(zl + j32 * ZBR + eb);
       const float4 wv0 = *(const float4*)(wt + wave * MB_BK + eb);
       const float4 wv1 = *(const float4*)(wt + wave * MB_BK + eb + 4);
       const float wv[8] = {wv0.x, wv0.y, wv0.z, wv0.w,
@@ -712,6 +761,10 @@ mstep_moments_b16_kernel(const __hip_bfloat16* __restrict__ xhi,
       accB = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a_hi, b_lo, accB, 0, 0, 0);
       accC = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a_lo, b_hi, accC, 0, 0, 0);
     }
+    __syncthreads();  // everyone done reading buf[cur]
+    if (ti + 1 < my_tiles) write_buf(cur ^ 1);
+    __syncthreads();  // buf[cur^1] ready
+    cur ^= 1;
   }
 
   if (c >= k) return;
