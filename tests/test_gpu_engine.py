@@ -91,3 +91,31 @@ def test_gpu_cli_end_to_end(tmp_path):
     assert open(out + ".summary").read().count("Cluster #") == 3
     lines = open(out + ".results").read().splitlines()
     assert len(lines) == 5000
+
+
+def test_gpu_cli_sweep_checkpoint_metrics(tmp_path):
+    """Full user path on GPU: bin input, MDL sweep, checkpoint, metrics."""
+    import json
+    from cuda_gmm_mpi_amd.cli import main
+    from cuda_gmm_mpi_amd.utils import io as gio
+    data, _ = make_blobs(40000, 10, 6, seed=71)
+    binpath = str(tmp_path / "d.bin")
+    gio.write_bin(binpath, data)
+    out = str(tmp_path / "o")
+    rc = main(["8", binpath, out, "4", "--min-iters", "10", "--max-iters",
+               "10", "--device", "cuda", "--estep-dtype", "bf16",
+               "--mstep-precision", "bf16x3",
+               "--checkpoint-dir", str(tmp_path / "ck"),
+               "--metrics-out", str(tmp_path / "m.json")])
+    assert rc == 0
+    m = json.load(open(tmp_path / "m.json"))
+    # empty-cluster elimination may jump past the target K, in which case
+    # the first saved model is kept (reference semantics, gaussian.cu:839)
+    kk = m["num_clusters"]
+    assert kk == 4 or kk == 8
+    assert open(out + ".summary").read().count("Cluster #") == kk
+    lines = open(out + ".results").read().splitlines()
+    assert len(lines) == 40000
+    assert len(lines[0].split("\t")[1].split(",")) == kk
+    import os
+    assert os.path.exists(tmp_path / "ck" / "gmm_sweep.npz")
