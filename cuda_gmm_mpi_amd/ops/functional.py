@@ -144,7 +144,8 @@ def estep_fused(z: torch.Tensor, mfac: torch.Tensor, add: torch.Tensor,
     touches HBM."""
     n = z.shape[1]
     nblk = (n + 127) // 128
-    partial = torch.zeros(nblk, dtype=torch.float32, device=z.device)
+    # every launched block writes its partial slot: no zero-fill needed
+    partial = torch.empty(nblk, dtype=torch.float32, device=z.device)
     hip_ext().estep_fused(z, mfac, add, w_out, partial)
     return w_out, partial.sum()
 

@@ -228,7 +228,7 @@ void mstep_moments(torch::Tensor x, torch::Tensor w, torch::Tensor partials) {
   TORCH_CHECK(partials.size(1) == k &&
                   partials.size(2) == dp * (dp + 1) / 2,
               "partials must be [nchunk, K, Dp*(Dp+1)/2]");
-  const size_t lds = sizeof(float) * ((size_t)d * (128 + 4) + 4 * 128);
+  const size_t lds = 2 * sizeof(float) * ((size_t)d * (128 + 4) + 4 * 128);
   dim3 grid((k + 3) / 4, nchunk);
   hipLaunchKernelGGL((gmm::mstep_moments_kernel<float>), grid, dim3(kNT), lds,
                      stream(), x.data_ptr<float>(), w.data_ptr<float>(),
