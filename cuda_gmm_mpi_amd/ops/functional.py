@@ -206,10 +206,11 @@ def mstep_moments(x: torch.Tensor, w: torch.Tensor,
     if x.is_cuda and d <= 31:
         tiles = (n + 127) // 128
         if nchunk is None:
-            # enough chunk-parallelism to fill 8 XCDs x 32 CUs (measured:
-            # 512 chunks beat 64 by ~25% at K=64), capped so the partial
-            # buffer stays <= 64 MB
-            nchunk = max(1, min(512, (64 << 20) // (4 * k * pp)))
+            # enough chunk-parallelism to fill 8 XCDs x 32 CUs, capped so
+            # the partial buffer stays <= 64 MB (measured sweeps: bf16x3
+            # fastest at ~256 chunks, fp32 at ~512, K=64 N=1M)
+            target = 256 if precision == "bf16x3" else 512
+            nchunk = max(1, min(target, (64 << 20) // (4 * k * pp)))
         nchunk = int(min(nchunk, tiles))
         partials = torch.empty((nchunk, k, pp), dtype=torch.float32,
                                device=x.device)
