@@ -80,18 +80,27 @@ class EmEngine:
         )
 
         k0 = config.num_clusters
-        # fused bf16-MFMA E-step path (D <= 31, LDS-bounded K); big-D MFMA
-        # logw path (31 < D <= 143); otherwise the VALU kernels
+        # fused MFMA E-step paths (D <= 31: bf16 split-factor or exact-f32;
+        # 31 < D <= 143: big-D bf16 logw); otherwise the VALU kernels
         self.use_fused_estep = F.estep_fused_available(
             self.device, config.estep_dtype, self.d, k0,
         ) and not config.diag_only
         self.use_big_estep = F.estep_big_available(
             self.device, config.estep_dtype, self.d,
         ) and not config.diag_only
+        need_bf16_fac = (
+            (self.use_fused_estep and config.estep_dtype == "bf16")
+            or self.use_big_estep
+        )
+        need_f32_fac = self.use_fused_estep and config.estep_dtype == "fp32"
         self.mfac = (
             torch.empty(k0, *F.mfac_shape(self.d), dtype=torch.bfloat16,
                         device=self.device)
-            if (self.use_fused_estep or self.use_big_estep) else None
+            if (need_bf16_fac or need_f32_fac) else None
+        )
+        self.mfac32 = (
+            torch.empty(k0, 32, 32, dtype=torch.float32, device=self.device)
+            if need_f32_fac else None
         )
 
         self.state = GmmState.empty(k0, self.d, self.device)
@@ -119,8 +128,9 @@ class EmEngine:
         k = st.num_clusters
         with self.profile.time("constants"):
             mfac = self.mfac[:k] if self.mfac is not None else None
+            mfac32 = self.mfac32[:k] if self.mfac32 is not None else None
             rinv, const = F.constants(st.R, st.means, self.cfg.diag_only,
-                                      mfac)
+                                      mfac, mfac32)
             st.Rinv.copy_(rinv)
             st.constant.copy_(const)
         self.profile.count("constants")
@@ -133,8 +143,12 @@ class EmEngine:
         with self.profile.time("e_step"):
             if self.use_fused_estep:
                 add = st.constant + torch.log(st.pi)
-                w, lik = F.estep_fused(self.x_estep, self.mfac[:k], add,
-                                       self.w[:k])
+                if self.mfac32 is not None:
+                    w, lik = F.estep_fused_f32(self.x_estep, self.mfac32[:k],
+                                               add, self.w[:k])
+                else:
+                    w, lik = F.estep_fused(self.x_estep, self.mfac[:k], add,
+                                           self.w[:k])
             elif self.use_big_estep:
                 add = st.constant + torch.log(st.pi)
                 logw = F.estep_logw_big(self.x_estep, self.mfac[:k], add,

@@ -115,7 +115,8 @@ def mstep_covariance_s(x: torch.Tensor, w: torch.Tensor,
 
 
 def constants(r: torch.Tensor, means: torch.Tensor | None = None,
-              diag_only: bool = False, mfac: torch.Tensor | None = None
+              diag_only: bool = False, mfac: torch.Tensor | None = None,
+              mfac32: torch.Tensor | None = None
               ) -> tuple[torch.Tensor, torch.Tensor]:
     """(Rinv [K,D,D], constant [K]) via no-pivot LU + ln|det|.
 
@@ -131,7 +132,10 @@ def constants(r: torch.Tensor, means: torch.Tensor | None = None,
             means = torch.zeros(k, d, dtype=torch.float32, device=r.device)
         if mfac is None:
             mfac = torch.empty(0, dtype=torch.bfloat16, device=r.device)
-        hip_ext().constants(r, means, rinv, logdet, mfac, bool(diag_only))
+        if mfac32 is None:
+            mfac32 = torch.empty(0, dtype=torch.float32, device=r.device)
+        hip_ext().constants(r, means, rinv, logdet, mfac, mfac32,
+                            bool(diag_only))
         const = -d * 0.5 * cpu.LOG_2PI - 0.5 * logdet
         return rinv, const
     return cpu.compute_constants(r, diag_only)
@@ -152,10 +156,22 @@ def estep_fused(z: torch.Tensor, mfac: torch.Tensor, add: torch.Tensor,
 
 def estep_fused_available(device: torch.device, dtype: str, d: int,
                           k: int) -> bool:
-    """Fused path gate (mirrors the LDS check in gmm_ext.hip)."""
-    if device.type != "cuda" or dtype != "bf16" or d > 31:
+    """Fused path gate (mirrors the LDS checks in gmm_ext.hip)."""
+    if device.type != "cuda" or d > 31:
         return False
-    return 128 * 40 * 2 + 4 * k * 132 <= 64 * 1024
+    if dtype == "bf16":
+        return 128 * 40 * 2 + 4 * k * 132 <= 64 * 1024
+    return 4 * (128 * 33 + k * 132) <= 64 * 1024  # exact-f32 variant
+
+
+def estep_fused_f32(z: torch.Tensor, mfac32: torch.Tensor, add: torch.Tensor,
+                    w_out: torch.Tensor) -> tuple[torch.Tensor, torch.Tensor]:
+    """Exact-f32 MFMA fused E-step (CUDA, D <= 31)."""
+    n = z.shape[1]
+    nblk = (n + 127) // 128
+    partial = torch.empty(nblk, dtype=torch.float32, device=z.device)
+    hip_ext().estep_fused_f32(z, mfac32, add, w_out, partial)
+    return w_out, partial.sum()
 
 
 def estep_big_available(device: torch.device, dtype: str, d: int) -> bool:

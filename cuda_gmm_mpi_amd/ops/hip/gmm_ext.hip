@@ -165,7 +165,8 @@ void mstep_covariance_partials(torch::Tensor x, torch::Tensor w,
 }
 
 void constants(torch::Tensor r, torch::Tensor means, torch::Tensor rinv,
-               torch::Tensor logdet, torch::Tensor mfac, bool diag_only) {
+               torch::Tensor logdet, torch::Tensor mfac, torch::Tensor mfac32,
+               bool diag_only) {
   check_f32(r, "r");
   check_f32(means, "means");
   check_f32(rinv, "rinv");
@@ -174,6 +175,13 @@ void constants(torch::Tensor r, torch::Tensor means, torch::Tensor rinv,
   const int d = (int)r.size(1);
   const bool make_mfac = mfac.numel() > 0;
   __hip_bfloat16* mp = nullptr;
+  float* mp32 = nullptr;
+  if (mfac32.numel() > 0) {
+    TORCH_CHECK(mfac32.is_cuda() && mfac32.is_contiguous() &&
+                    mfac32.scalar_type() == torch::kFloat32,
+                "mfac32 must be contiguous fp32");
+    mp32 = mfac32.data_ptr<float>();
+  }
   if (make_mfac) {
     const int rows = ((d + 31) / 32) * 32;
     const int kct = d + 1 <= 32 ? 2 : d + 1 <= 48 ? 3 : d + 1 <= 80 ? 5 : 9;
@@ -195,7 +203,7 @@ void constants(torch::Tensor r, torch::Tensor means, torch::Tensor rinv,
       hipLaunchKernelGGL(gmm::emit_mfac_from_rinv_kernel, dim3(k), dim3(kNT),
                          sizeof(float) * (2 * (size_t)d * d + d), s,
                          rinv.data_ptr<float>(), means.data_ptr<float>(), mp,
-                         d);
+                         mp32, d);
     }
   } else {
     // working buffer + read-only LU snapshot (+ u0 scratch for the factor)
@@ -208,7 +216,7 @@ void constants(torch::Tensor r, torch::Tensor means, torch::Tensor rinv,
     hipLaunchKernelGGL(gmm::constants_lu_kernel, dim3(k), dim3(kNT), lds, s,
                        r.data_ptr<float>(), means.data_ptr<float>(),
                        rinv.data_ptr<float>(), logdet.data_ptr<float>(), mp,
-                       d);
+                       mp32, d);
   }
   HIP_CHECK(hipGetLastError());
 }
@@ -356,6 +364,33 @@ void mfma_probe32(torch::Tensor a, torch::Tensor b, torch::Tensor c) {
   HIP_CHECK(hipGetLastError());
 }
 
+void estep_fused_f32(torch::Tensor z, torch::Tensor mfac32,
+                     torch::Tensor add, torch::Tensor w_out,
+                     torch::Tensor partial) {
+  check_f32(z, "z");
+  check_f32(mfac32, "mfac32");
+  check_f32(add, "add");
+  check_f32(w_out, "w_out");
+  check_f32(partial, "partial");
+  const int d = (int)z.size(0);
+  const int64_t n = z.size(1);
+  const int k = (int)add.size(0);
+  TORCH_CHECK(d <= 31, "estep_fused_f32 needs D <= 31");
+  TORCH_CHECK(mfac32.numel() >= (int64_t)k * 32 * 32, "mfac32 too small");
+  TORCH_CHECK(w_out.size(0) == k && w_out.size(1) == n, "w_out shape");
+  const int64_t nblk = (n + 128 - 1) / 128;
+  TORCH_CHECK(partial.size(0) >= nblk, "partial buffer too small");
+  const size_t lds =
+      sizeof(float) * ((size_t)128 * 33 + (size_t)k * (128 + 4));
+  TORCH_CHECK(lds <= 64 * 1024, "estep_fused_f32 LDS budget exceeded");
+  hipLaunchKernelGGL(gmm::estep_fused_f32_kernel, dim3((uint32_t)nblk),
+                     dim3(kNT), lds, stream(), z.data_ptr<float>(),
+                     mfac32.data_ptr<float>(), add.data_ptr<float>(),
+                     w_out.data_ptr<float>(), partial.data_ptr<float>(), d,
+                     k, n);
+  HIP_CHECK(hipGetLastError());
+}
+
 void mfma_probe(torch::Tensor a, torch::Tensor b, torch::Tensor c) {
   TORCH_CHECK(a.scalar_type() == torch::kBFloat16 && a.numel() == 16 * 32);
   TORCH_CHECK(b.scalar_type() == torch::kBFloat16 && b.numel() == 32 * 16);
@@ -386,6 +421,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "split-precision bf16x3 augmented moments");
   m.def("mstep_moments", &mstep_moments,
         "fused augmented moments [S|mean_num|N] via f32 MFMA");
+  m.def("estep_fused_f32", &estep_fused_f32,
+        "exact-f32 MFMA fused E-step (D <= 31)");
   m.def("estep_fused", &estep_fused,
         "fused bf16-MFMA E-step: posteriors + likelihood partials");
   m.def("mfma_probe", &mfma_probe, "bf16 MFMA fragment-layout probe");

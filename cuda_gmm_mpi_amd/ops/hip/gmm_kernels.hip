@@ -313,8 +313,8 @@ mstep_cov_kernel(const T* __restrict__ x, const float* __restrict__ w,
 // A-fragment k-order (mfma_b16_k). `u` is scratch LDS holding Rinv [d*d]
 // (upper triangle becomes U in place) + u0 [d].
 __device__ inline void emit_mfac(float* u, const float* __restrict__ means,
-                                 __hip_bfloat16* __restrict__ mfac, int c,
-                                 int d) {
+                                 __hip_bfloat16* __restrict__ mfac,
+                                 float* __restrict__ mfac32, int c, int d) {
   const int tid = threadIdx.x;
   float* u0 = u + d * d;
   // in-place upper Cholesky of Rinv (SPD up to rounding; clamped pivots)
@@ -348,6 +348,7 @@ __device__ inline void emit_mfac(float* u, const float* __restrict__ means,
   const int cols = kc_tier((d + 1 + 15) / 16) * 16;
   const int cells = rows * cols;
   __hip_bfloat16* out = mfac + (int64_t)c * 2 * cells;
+  float* out32 = mfac32 ? mfac32 + (int64_t)c * cells : nullptr;
   for (int t = tid; t < cells; t += NT) {
     const int i = t / cols, kx = t % cols;
     float v = 0.0f;
@@ -358,6 +359,7 @@ __device__ inline void emit_mfac(float* u, const float* __restrict__ means,
     const __hip_bfloat16 hi = __float2bfloat16(v);
     out[t] = hi;
     out[cells + t] = __float2bfloat16(v - __bfloat162float(hi));
+    if (out32) out32[t] = v;
   }
 }
 
@@ -365,7 +367,8 @@ __global__ void __launch_bounds__(NT)
 constants_lu_kernel(const float* __restrict__ r,
                     const float* __restrict__ means, float* __restrict__ rinv,
                     float* __restrict__ logdet,
-                    __hip_bfloat16* __restrict__ mfac, int d) {
+                    __hip_bfloat16* __restrict__ mfac,
+                    float* __restrict__ mfac32, int d) {
   // lds: a[d*d] working buffer, o[d*d] read-only snapshot of the LU factor.
   // The reference's in-place triangular inversion reads a mix of original
   // and already-inverted entries in a serial order; parallelized across
@@ -388,7 +391,7 @@ constants_lu_kernel(const float* __restrict__ r,
       o[0] = oc[0];
     }
     __syncthreads();
-    if (mfac != nullptr) emit_mfac(o, means, mfac, c, 1);
+    if (mfac != nullptr) emit_mfac(o, means, mfac, mfac32, c, 1);
     return;
   }
 
@@ -477,7 +480,7 @@ constants_lu_kernel(const float* __restrict__ r,
   }
   if (mfac != nullptr) {
     __syncthreads();
-    emit_mfac(o, means, mfac, c, d);
+    emit_mfac(o, means, mfac, mfac32, c, d);
   }
 }
 
@@ -965,6 +968,118 @@ estep_fused_kernel(const __hip_bfloat16* __restrict__ z,
 }
 
 // ---------------------------------------------------------------------------
+// Exact-fp32 fused E-step (D <= 31): same structure as estep_fused_kernel
+// but on v_mfma_f32_32x32x2_f32 — f32 in / f32 accumulate, bitwise an fmaf
+// chain (guide §3), consuming the fp32 factor plane. 16 dependent MFMAs
+// per (cluster, 32-event tile); issue interval == dependent latency (64),
+// so the chain runs at the f32 matrix rate. This makes the CLI's default
+// exact mode ~4x faster than the VALU quadratic-form path.
+// ---------------------------------------------------------------------------
+__global__ void __launch_bounds__(NT)
+estep_fused_f32_kernel(const float* __restrict__ z,
+                       const float* __restrict__ mfac32,  // [K][32][32]
+                       const float* __restrict__ add,
+                       float* __restrict__ w_out, float* __restrict__ partial,
+                       int d, int k, int64_t n) {
+  typedef __attribute__((ext_vector_type(16))) float f32x16;
+  constexpr int ZR = 33;  // f32 slots per transposed event row (32 + pad)
+  extern __shared__ float lds[];
+  float* zs = lds;                         // [EST_BE][ZR]
+  const int lrow = EST_BE + 4;
+  float* lw = lds + EST_BE * ZR;           // [k][lrow]
+
+  const int wave = threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int j32 = lane & 31;
+  const int g2 = lane >> 5;
+  const int64_t e0 = (int64_t)blockIdx.x * EST_BE;
+  const int cnt = (int)min((int64_t)EST_BE, n - e0);
+
+  if (cnt == EST_BE) {
+    for (int idx = threadIdx.x; idx < d * EST_BE; idx += NT) {
+      const int kk = idx / EST_BE, ei = idx % EST_BE;
+      zs[ei * ZR + kk] = z[(int64_t)kk * n + e0 + ei];
+    }
+  } else {
+    for (int idx = threadIdx.x; idx < d * EST_BE; idx += NT) {
+      const int kk = idx / EST_BE, ei = idx % EST_BE;
+      zs[ei * ZR + kk] =
+          (ei < cnt) ? z[(int64_t)kk * n + e0 + ei] : 0.0f;
+    }
+  }
+  for (int idx = threadIdx.x; idx < (32 - d) * EST_BE; idx += NT) {
+    const int kk = d + idx / EST_BE, ei = idx % EST_BE;
+    zs[ei * ZR + kk] = (kk == d && ei < cnt) ? 1.0f : 0.0f;
+  }
+  __syncthreads();
+
+  const int nwaves = NT / WAVE;
+
+  // prefetched A rows: 16 f32 per lane (row j32, k-slots 2*ch + g2)
+  float nx_a[16];
+  float nx_add;
+  auto load_a = [&](int c) {
+    const float* row = mfac32 + ((int64_t)c * 32 + j32) * 32;
+#pragma unroll
+    for (int ch = 0; ch < 16; ++ch) nx_a[ch] = row[2 * ch + g2];
+    nx_add = add[c];
+  };
+  if (wave < k) load_a(wave);
+
+  for (int c = wave; c < k; c += nwaves) {
+    float a[16];
+#pragma unroll
+    for (int ch = 0; ch < 16; ++ch) a[ch] = nx_a[ch];
+    const float addc = nx_add;
+    if (c + nwaves < k) load_a(c + nwaves);
+#pragma unroll 2
+    for (int t = 0; t < EST_BE / 32; ++t) {
+      const float* zrow = zs + (t * 32 + j32) * ZR;
+      f32x16 y = (f32x16)(0.0f);
+#pragma unroll
+      for (int ch = 0; ch < 16; ++ch) {
+        const float b = zrow[2 * ch + g2];
+        y = __builtin_amdgcn_mfma_f32_32x32x2f32(a[ch], b, y, 0, 0, 0);
+      }
+      float s = 0.0f;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) s = fmaf(y[r], y[r], s);
+      s += __shfl_xor(s, 32, WAVE);
+      if (lane < 32) lw[c * lrow + t * 32 + j32] = -0.5f * s + addc;
+    }
+  }
+  __syncthreads();
+
+  // pass 2: identical to the bf16 kernel
+  float acc = 0.0f;
+  if (threadIdx.x < EST_BE && threadIdx.x < cnt) {
+    const int t = threadIdx.x;
+    float m = lw[t];
+    for (int c = 1; c < k; ++c) m = fmaxf(m, lw[c * lrow + t]);
+    float s = 0.0f;
+    for (int c = 0; c < k; ++c) {
+      const float e = __expf(lw[c * lrow + t] - m);
+      lw[c * lrow + t] = e;
+      s += e;
+    }
+    const float inv = 1.0f / s;
+    for (int c = 0; c < k; ++c)
+      w_out[(int64_t)c * n + e0 + t] = lw[c * lrow + t] * inv;
+    acc = m + __logf(s);
+  }
+  __shared__ float wsum2[NT / WAVE];
+  for (int off = WAVE / 2; off > 0; off >>= 1)
+    acc += __shfl_down(acc, off, WAVE);
+  if (lane == 0) wsum2[wave] = acc;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float total = 0.0f;
+    for (int wv = 0; wv < NT / WAVE; ++wv) total += wsum2[wv];
+    partial[blockIdx.x] = total;
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Big-D split-precision moments (31 < D <= 159): same packed output as the
 // small-D kernels. Row-tiles of 32 cover the padded (D+1) dims; the
 // RT2*(RT2+1)/2 tile-pairs of the symmetric output are split across two
@@ -1214,13 +1329,14 @@ estep_logw_big_kernel(const __hip_bfloat16* __restrict__ z,
 __global__ void __launch_bounds__(NT)
 emit_mfac_from_rinv_kernel(const float* __restrict__ rinv,
                            const float* __restrict__ means,
-                           __hip_bfloat16* __restrict__ mfac, int d) {
+                           __hip_bfloat16* __restrict__ mfac,
+                           float* __restrict__ mfac32, int d) {
   extern __shared__ float buf[];
   const int c = blockIdx.x;
   for (int t = threadIdx.x; t < d * d; t += NT)
     buf[t] = rinv[(int64_t)c * d * d + t];
   __syncthreads();
-  emit_mfac(buf, means, mfac, c, d);
+  emit_mfac(buf, means, mfac, mfac32, c, d);
 }
 
 // DIAG_ONLY constants (gaussian_kernel.cu:215-223)

