@@ -213,8 +213,8 @@ void constants(torch::Tensor r, torch::Tensor means, torch::Tensor rinv,
           reinterpret_cast<const void*>(&gmm::constants_lu_kernel),
           hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds));
     }
-    hipLaunchKernelGGL(gmm::constants_lu_kernel, dim3(k), dim3(kNT), lds, s,
-                       r.data_ptr<float>(), means.data_ptr<float>(),
+    hipLaunchKernelGGL(gmm::constants_lu_kernel, dim3(k), dim3(kNT),
+                       lds, s, r.data_ptr<float>(), means.data_ptr<float>(),
                        rinv.data_ptr<float>(), logdet.data_ptr<float>(), mp,
                        mp32, d);
   }

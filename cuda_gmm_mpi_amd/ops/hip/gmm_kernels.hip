@@ -326,7 +326,7 @@ __device__ inline void emit_mfac(float* u, const float* __restrict__ means,
     }
     __syncthreads();
     const float piv = u[j * d + j];
-    for (int i = j + 1 + tid; i < d; i += NT) {
+    for (int i = j + 1 + tid; i < d; i += blockDim.x) {
       float s = u[j * d + i];
       for (int kk = 0; kk < j; ++kk) s -= u[kk * d + j] * u[kk * d + i];
       u[j * d + i] = s / piv;
@@ -334,7 +334,7 @@ __device__ inline void emit_mfac(float* u, const float* __restrict__ means,
     __syncthreads();
   }
   // u0 = -U mu (mu = centered cluster means)
-  for (int i = tid; i < d; i += NT) {
+  for (int i = tid; i < d; i += blockDim.x) {
     float s = 0.0f;
     for (int j = i; j < d; ++j) s += u[i * d + j] * means[c * d + j];
     u0[i] = -s;
@@ -349,7 +349,7 @@ __device__ inline void emit_mfac(float* u, const float* __restrict__ means,
   const int cells = rows * cols;
   __hip_bfloat16* out = mfac + (int64_t)c * 2 * cells;
   float* out32 = mfac32 ? mfac32 + (int64_t)c * cells : nullptr;
-  for (int t = tid; t < cells; t += NT) {
+  for (int t = tid; t < cells; t += blockDim.x) {
     const int i = t / cols, kx = t % cols;
     float v = 0.0f;
     if (i < d) {
@@ -381,7 +381,7 @@ constants_lu_kernel(const float* __restrict__ r,
   const float* rc = r + (int64_t)c * d * d;
   float* oc = rinv + (int64_t)c * d * d;
 
-  for (int t = tid; t < d * d; t += NT) a[t] = rc[t];
+  for (int t = tid; t < d * d; t += blockDim.x) a[t] = rc[t];
   __syncthreads();
 
   if (d == 1) {
@@ -396,12 +396,12 @@ constants_lu_kernel(const float* __restrict__ r,
   }
 
   // normalize row 0 (invert_matrix.cpp:42 / gaussian_kernel.cu:120)
-  for (int j = 1 + tid; j < d; j += NT) a[j] /= a[0];
+  for (int j = 1 + tid; j < d; j += blockDim.x) a[j] /= a[0];
   __syncthreads();
 
   for (int i = 1; i < d; ++i) {
     // column i of L: rows j >= i in parallel
-    for (int j = i + tid; j < d; j += NT) {
+    for (int j = i + tid; j < d; j += blockDim.x) {
       float s = 0.0f;
       for (int kk = 0; kk < i; ++kk) s = fmaf(a[j * d + kk], a[kk * d + i], s);
       a[j * d + i] -= s;
@@ -410,7 +410,7 @@ constants_lu_kernel(const float* __restrict__ r,
     if (i == d - 1) break;
     // row i of U: cols j > i in parallel
     const float pivot = a[i * d + i];
-    for (int j = i + 1 + tid; j < d; j += NT) {
+    for (int j = i + 1 + tid; j < d; j += blockDim.x) {
       float s = 0.0f;
       for (int kk = 0; kk < i; ++kk) s = fmaf(a[i * d + kk], a[kk * d + j], s);
       a[i * d + j] = (a[i * d + j] - s) / pivot;
@@ -421,28 +421,29 @@ constants_lu_kernel(const float* __restrict__ r,
   // ln|det| = sum ln|diag| (natural log, gaussian_kernel.cu:139)
   {
     __shared__ float wsum[NT / WAVE];
+    const int nw = (blockDim.x + WAVE - 1) / WAVE;
     float acc = 0.0f;
-    for (int i = tid; i < d; i += NT) acc += __logf(fabsf(a[i * d + i]));
+    for (int i = tid; i < d; i += blockDim.x) acc += __logf(fabsf(a[i * d + i]));
     for (int off = WAVE / 2; off > 0; off >>= 1)
       acc += __shfl_down(acc, off, WAVE);
     if ((tid & (WAVE - 1)) == 0) wsum[tid / WAVE] = acc;
     __syncthreads();
     if (tid == 0) {
       float total = 0.0f;
-      for (int wv = 0; wv < NT / WAVE; ++wv) total += wsum[wv];
+      for (int wv = 0; wv < nw; ++wv) total += wsum[wv];
       logdet[c] = total;
     }
   }
   __syncthreads();
   // snapshot the LU factor: the inversion below reads original L/U values
   // that the in-place writes would otherwise clobber across threads
-  for (int t = tid; t < d * d; t += NT) o[t] = a[t];
+  for (int t = tid; t < d * d; t += blockDim.x) o[t] = a[t];
   __syncthreads();
 
   // invert L: column i per thread, serial down rows; cross-column reads and
   // the diagonal divisor come from the snapshot (gaussian_kernel.cu:142-151:
   // data[j,k] for k>i and data[j,j] are pre-inversion values there)
-  for (int i = tid; i < d; i += NT) {
+  for (int i = tid; i < d; i += blockDim.x) {
     for (int j = i; j < d; ++j) {
       float xv = 1.0f;
       if (i != j) {
@@ -457,7 +458,7 @@ constants_lu_kernel(const float* __restrict__ r,
   // snapshot (gaussian_kernel.cu:152-159: data[k,j] for k>i pre-inversion),
   // row reads from this thread's own inverted values. Disjoint from the L
   // writes (strict upper vs lower+diag) so no barrier is needed between.
-  for (int i = tid; i < d; i += NT) {
+  for (int i = tid; i < d; i += blockDim.x) {
     for (int j = i + 1; j < d; ++j) {
       float s = 0.0f;
       for (int kk = i; kk < j; ++kk)
@@ -470,7 +471,7 @@ constants_lu_kernel(const float* __restrict__ r,
   // (gaussian_kernel.cu:160-166) — each output element independent; write
   // straight to global (reads see the pre-write LDS values, same as the
   // reference's read-before-write order)
-  for (int t = tid; t < d * d; t += NT) {
+  for (int t = tid; t < d * d; t += blockDim.x) {
     const int j = t / d, i = t % d;
     float s = 0.0f;
     for (int kk = (i > j ? i : j); kk < d; ++kk)
@@ -535,19 +536,19 @@ mstep_moments_kernel(const T* __restrict__ x, const float* __restrict__ w,
     // hipcc serialize every load behind a vmcnt(0) (guide §5 trap 4c)
     const int nv = min(4, k - blockIdx.x * 4) * MOM_BK;
     if (cnt == MOM_BK) {
-      for (int idx = threadIdx.x; idx < d * MOM_BK; idx += NT)
+      for (int idx = threadIdx.x; idx < d * MOM_BK; idx += blockDim.x)
         xs[(idx / MOM_BK) * row + idx % MOM_BK] =
             load_x(x, (int64_t)(idx / MOM_BK) * n + e0 + idx % MOM_BK);
-      for (int idx = threadIdx.x; idx < nv; idx += NT)
+      for (int idx = threadIdx.x; idx < nv; idx += blockDim.x)
         wt[idx] = w[(int64_t)(blockIdx.x * 4 + idx / MOM_BK) * n + e0 +
                     idx % MOM_BK];
     } else {
-      for (int idx = threadIdx.x; idx < d * MOM_BK; idx += NT) {
+      for (int idx = threadIdx.x; idx < d * MOM_BK; idx += blockDim.x) {
         const int di = idx / MOM_BK, ei = idx % MOM_BK;
         xs[di * row + ei] =
             (ei < cnt) ? load_x(x, (int64_t)di * n + e0 + ei) : 0.0f;
       }
-      for (int idx = threadIdx.x; idx < 4 * MOM_BK; idx += NT) {
+      for (int idx = threadIdx.x; idx < 4 * MOM_BK; idx += blockDim.x) {
         const int wv = idx / MOM_BK, ei = idx % MOM_BK;
         const int cw = blockIdx.x * 4 + wv;
         wt[wv * MOM_BK + ei] =
@@ -722,7 +723,7 @@ mstep_moments_b16_kernel(const __hip_bfloat16* __restrict__ xhi,
   for (int b = 0; b < 2; ++b) {
     __bf16* zh = zbuf + b * BUFB;
     __bf16* zl = zh + PLANE;
-    for (int idx = d * MB_BK + threadIdx.x; idx < 32 * MB_BK; idx += NT) {
+    for (int idx = d * MB_BK + threadIdx.x; idx < 32 * MB_BK; idx += blockDim.x) {
       const int di = idx / MB_BK, ei = idx % MB_BK;
       zh[di * ZBR + ei] = (__bf16)(di == d ? 1.0f : 0.0f);
       zl[di * ZBR + ei] = (__bf16)0.0f;
@@ -863,19 +864,19 @@ estep_fused_kernel(const __hip_bfloat16* __restrict__ z,
   if (cnt == EST_BE) {
     // branchless staging (guide §5 trap 4c): coalesced reads of the d data
     // rows, transposed scatter into LDS; then the constant rows
-    for (int idx = threadIdx.x; idx < d * EST_BE; idx += NT) {
+    for (int idx = threadIdx.x; idx < d * EST_BE; idx += blockDim.x) {
       const int kk = idx / EST_BE, ei = idx % EST_BE;
       zsb[ei * EST_ZROW + kk] =
           (__bf16)__bfloat162float(z[(int64_t)kk * n + e0 + ei]);
     }
   } else {
-    for (int idx = threadIdx.x; idx < d * EST_BE; idx += NT) {
+    for (int idx = threadIdx.x; idx < d * EST_BE; idx += blockDim.x) {
       const int kk = idx / EST_BE, ei = idx % EST_BE;
       zsb[ei * EST_ZROW + kk] = (__bf16)(
           (ei < cnt) ? __bfloat162float(z[(int64_t)kk * n + e0 + ei]) : 0.0f);
     }
   }
-  for (int idx = threadIdx.x; idx < (32 - d) * EST_BE; idx += NT) {
+  for (int idx = threadIdx.x; idx < (32 - d) * EST_BE; idx += blockDim.x) {
     const int kk = d + idx / EST_BE, ei = idx % EST_BE;
     zsb[ei * EST_ZROW + kk] =
         (__bf16)((kk == d && ei < cnt) ? 1.0f : 0.0f);
@@ -996,18 +997,18 @@ estep_fused_f32_kernel(const float* __restrict__ z,
   const int cnt = (int)min((int64_t)EST_BE, n - e0);
 
   if (cnt == EST_BE) {
-    for (int idx = threadIdx.x; idx < d * EST_BE; idx += NT) {
+    for (int idx = threadIdx.x; idx < d * EST_BE; idx += blockDim.x) {
       const int kk = idx / EST_BE, ei = idx % EST_BE;
       zs[ei * ZR + kk] = z[(int64_t)kk * n + e0 + ei];
     }
   } else {
-    for (int idx = threadIdx.x; idx < d * EST_BE; idx += NT) {
+    for (int idx = threadIdx.x; idx < d * EST_BE; idx += blockDim.x) {
       const int kk = idx / EST_BE, ei = idx % EST_BE;
       zs[ei * ZR + kk] =
           (ei < cnt) ? z[(int64_t)kk * n + e0 + ei] : 0.0f;
     }
   }
-  for (int idx = threadIdx.x; idx < (32 - d) * EST_BE; idx += NT) {
+  for (int idx = threadIdx.x; idx < (32 - d) * EST_BE; idx += blockDim.x) {
     const int kk = d + idx / EST_BE, ei = idx % EST_BE;
     zs[ei * ZR + kk] = (kk == d && ei < cnt) ? 1.0f : 0.0f;
   }
@@ -1256,19 +1257,19 @@ estep_logw_big_kernel(const __hip_bfloat16* __restrict__ z,
   // transposed staging with ones-row and zero-pad baked in (branchless
   // for full tiles)
   if (cnt == ESB_BE) {
-    for (int idx = threadIdx.x; idx < d * ESB_BE; idx += NT) {
+    for (int idx = threadIdx.x; idx < d * ESB_BE; idx += blockDim.x) {
       const int kk = idx / ESB_BE, ei = idx % ESB_BE;
       zs[ei * zrowk + kk] =
           (__bf16)__bfloat162float(z[(int64_t)kk * n + e0 + ei]);
     }
   } else {
-    for (int idx = threadIdx.x; idx < d * ESB_BE; idx += NT) {
+    for (int idx = threadIdx.x; idx < d * ESB_BE; idx += blockDim.x) {
       const int kk = idx / ESB_BE, ei = idx % ESB_BE;
       zs[ei * zrowk + kk] = (__bf16)(
           (ei < cnt) ? __bfloat162float(z[(int64_t)kk * n + e0 + ei]) : 0.0f);
     }
   }
-  for (int idx = threadIdx.x; idx < (KCT * 16 - d) * ESB_BE; idx += NT) {
+  for (int idx = threadIdx.x; idx < (KCT * 16 - d) * ESB_BE; idx += blockDim.x) {
     const int kk = d + idx / ESB_BE, ei = idx % ESB_BE;
     zs[ei * zrowk + kk] = (__bf16)((kk == d && ei < cnt) ? 1.0f : 0.0f);
   }
