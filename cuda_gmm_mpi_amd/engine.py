@@ -124,6 +124,24 @@ class EmEngine:
         self._graphs: dict[int, object] = {}
         self._lik_dev = torch.zeros(1, dtype=torch.float32, device=self.device)
 
+    def _refresh_mfac(self, k: int) -> None:
+        """Re-emit the fused-E-step factors from the CURRENT Rinv without
+        touching the constants — after an MDL merge or a checkpoint resume
+        the merged cluster's constant/Rinv come from the host path and must
+        feed the next E-step unchanged, but the factor
+        tables must match the compacted Rinv."""
+        if self.mfac is None and self.mfac32 is None:
+            return
+        from .ops.backend import hip_ext
+        st = self.state.shrink(k)
+        empty_b = torch.empty(0, dtype=torch.bfloat16, device=self.device)
+        empty_f = torch.empty(0, dtype=torch.float32, device=self.device)
+        hip_ext().emit_factors(
+            st.Rinv.contiguous(), st.means.contiguous(),
+            self.mfac[:k] if self.mfac is not None else empty_b,
+            self.mfac32[:k] if self.mfac32 is not None else empty_f,
+        )
+
     def _update_constants(self, st: GmmState) -> None:
         k = st.num_clusters
         with self.profile.time("constants"):
@@ -326,9 +344,9 @@ class EmEngine:
                         means=st.means.cpu().numpy(), R=st.R.cpu().numpy(),
                         Rinv=st.Rinv.cpu().numpy(),
                     ), k)
-                if self.mfac is not None:
-                    # regenerate the fused-E-step factors for resumed params
-                    self._update_constants(self.state.shrink(k))
+                # regenerate the fused-E-step factors for the resumed
+                # params without recomputing the saved constants
+                self._refresh_mfac(k)
                 best_state = ck["best"]
                 best_k = ck["best_k"]
                 min_rissanen = ck["min_rissanen"]
@@ -376,6 +394,9 @@ class EmEngine:
                         pdist.broadcast_(vec, src=0)
                     if self.rank != 0:
                         self.state.shrink(new_k).load_param_vector(vec)
+            # the merged/compacted Rinv needs fresh E-step factors; the
+            # constants stay as the host merge path produced them
+            self._refresh_mfac(new_k)
             self.profile.count("reduce")
             k = new_k
             if cfg.checkpoint_dir and self.rank == 0:

@@ -364,6 +364,34 @@ void mfma_probe32(torch::Tensor a, torch::Tensor b, torch::Tensor c) {
   HIP_CHECK(hipGetLastError());
 }
 
+void emit_factors(torch::Tensor rinv, torch::Tensor means,
+                  torch::Tensor mfac, torch::Tensor mfac32) {
+  check_f32(rinv, "rinv");
+  check_f32(means, "means");
+  const int k = (int)rinv.size(0);
+  const int d = (int)rinv.size(1);
+  __hip_bfloat16* mp = nullptr;
+  float* mp32 = nullptr;
+  if (mfac.numel() > 0) {
+    TORCH_CHECK(mfac.scalar_type() == torch::kBFloat16 && mfac.is_contiguous());
+    mp = reinterpret_cast<__hip_bfloat16*>(mfac.data_ptr());
+  }
+  if (mfac32.numel() > 0) {
+    check_f32(mfac32, "mfac32");
+    mp32 = mfac32.data_ptr<float>();
+  }
+  const size_t lds = sizeof(float) * (2 * (size_t)d * d + d);
+  if (lds > 64 * 1024) {
+    HIP_CHECK(hipFuncSetAttribute(
+        reinterpret_cast<const void*>(&gmm::emit_mfac_from_rinv_kernel),
+        hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds));
+  }
+  hipLaunchKernelGGL(gmm::emit_mfac_from_rinv_kernel, dim3(k), dim3(kNT),
+                     lds, stream(), rinv.data_ptr<float>(),
+                     means.data_ptr<float>(), mp, mp32, d);
+  HIP_CHECK(hipGetLastError());
+}
+
 void estep_fused_f32(torch::Tensor z, torch::Tensor mfac32,
                      torch::Tensor add, torch::Tensor w_out,
                      torch::Tensor partial) {
@@ -421,6 +449,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "split-precision bf16x3 augmented moments");
   m.def("mstep_moments", &mstep_moments,
         "fused augmented moments [S|mean_num|N] via f32 MFMA");
+  m.def("emit_factors", &emit_factors,
+        "re-emit E-step factors from an existing Rinv (post-merge/resume)");
   m.def("estep_fused_f32", &estep_fused_f32,
         "exact-f32 MFMA fused E-step (D <= 31)");
   m.def("estep_fused", &estep_fused,
