@@ -79,11 +79,13 @@ def main() -> int:
     sync()
     elapsed = time.perf_counter() - t0
 
-    # MAX over ranks == min iterations/sec (use the slowest rank's time)
-    t = torch.tensor([elapsed], dtype=torch.float64)
+    # MAX over ranks == min iterations/sec (use the slowest rank's time).
+    # NCCL/RCCL reduces device tensors only.
     if world > 1:
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device="cuda" if has_gpu else "cpu")
         torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
-    elapsed = float(t.item())
+        elapsed = float(t.item())
 
     iters_per_sec = args.steps / elapsed
     if rank == 0:

@@ -56,6 +56,10 @@ def make_parser() -> argparse.ArgumentParser:
                    help="cpu | cuda (default: cuda when available)")
     p.add_argument("--profile", action="store_true",
                    help="print the per-GPU timing report (gaussian.cu:967)")
+    p.add_argument("--nearest-target", action="store_true",
+                   help="if empty-cluster elimination jumps past the target "
+                        "K, keep the closest completed K instead of the "
+                        "reference's first-saved model")
     p.add_argument("--checkpoint-dir", default=None,
                    help="directory for MDL-sweep checkpoints; an existing "
                         "checkpoint there is resumed automatically")
@@ -78,6 +82,7 @@ def config_from_args(args) -> GmmConfig:
         estep_dtype=args.estep_dtype, center_data=args.center_data,
         mstep_precision=args.mstep_precision,
         checkpoint_dir=args.checkpoint_dir,
+        nearest_target=args.nearest_target,
     )
     cfg.validate()
     return cfg

@@ -205,9 +205,10 @@ class EmEngine:
         if self.device.type != "cuda":
             return False
         if self.world > 1:
-            import torch.distributed as dist
-            if dist.get_backend() != "nccl":
-                return False
+            # RCCL collectives inside hipGraph capture are unexercised in
+            # this environment (single-GPU boxes); the eager path costs
+            # ~3% and cannot hang. Opt in with GMM_DIST_GRAPHS=1.
+            return bool(os.environ.get("GMM_DIST_GRAPHS"))
         return True
 
     def _mstep(self, k: int) -> None:
@@ -364,6 +365,10 @@ class EmEngine:
                 k == cfg.num_clusters
                 or (riss < min_rissanen and cfg.target_num_clusters == 0)
                 or k == cfg.target_num_clusters
+                # nearest-target mode: keep updating while above the target
+                # so a jumped-over target yields the closest completed K
+                or (cfg.nearest_target and cfg.target_num_clusters > 0
+                    and k >= cfg.target_num_clusters)
             )
             if save:
                 min_rissanen = riss
@@ -383,7 +388,9 @@ class EmEngine:
                         hc, bug_compat=cfg.bug_compat,
                         device=str(self.device))
                 if self.world > 1:
-                    nk = torch.tensor([new_k], dtype=torch.int64)
+                    # NCCL/RCCL broadcasts device tensors only
+                    nk = torch.tensor([new_k], dtype=torch.int64,
+                                      device=self.device)
                     pdist.broadcast_(nk, src=0)
                     new_k = int(nk.item())
                 if self.rank == 0:

@@ -72,6 +72,11 @@ class GmmConfig:
     # Device-side: memberships stay shard-resident except for this output.
     verbose: bool = False
 
+    # When the empty-cluster elimination jumps past target_num_clusters,
+    # the reference keeps the FIRST saved model (K0). With this flag the
+    # sweep instead saves the last completed K nearest the target.
+    nearest_target: bool = False
+
     # Directory for MDL-sweep checkpoints (.npz per completed K); None
     # disables. Resume happens automatically when a checkpoint exists.
     checkpoint_dir: str | None = None

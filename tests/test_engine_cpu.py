@@ -186,3 +186,22 @@ def test_metrics_out(tmp_path):
     m = json.load(open(mpath))
     assert m["num_clusters"] == 3
     assert "3" in m["rissanen_by_k"]
+
+
+def test_nearest_target_mode():
+    """When elimination jumps past the target, nearest_target keeps the
+    closest completed K instead of the first-saved model."""
+    rng = np.random.default_rng(3)
+    # data with genuinely collapsing clusters: few blobs, many clusters
+    data, _ = make_blobs(1200, 2, 2, seed=3)
+    base = dict(num_clusters=8, target_num_clusters=3,
+                min_iters=4, max_iters=4)
+    eng_a = build_engine(data, GmmConfig(**base), device="cpu")
+    res_a = eng_a.sweep()
+    eng_b = build_engine(data, GmmConfig(**base, nearest_target=True),
+                         device="cpu")
+    res_b = eng_b.sweep()
+    # nearest-target never does worse than the reference-faithful choice
+    assert res_b.num_clusters <= res_a.num_clusters or \
+        res_a.num_clusters == 3
+    assert res_b.num_clusters >= 3
