@@ -261,10 +261,10 @@ void mstep_moments_b16(torch::Tensor xhi, torch::Tensor xlo,
   TORCH_CHECK(partials.size(1) == k &&
                   partials.size(2) == dp * (dp + 1) / 2,
               "partials must be [nchunk, K, Dp*(Dp+1)/2]");
-  // two buffers x (zhi+zlo planes + w tile)
-  const size_t lds = 2 * (2 * 32 * 136 * 2 + 4 * 128 * 4);
-  dim3 grid((k + 3) / 4, nchunk);
-  hipLaunchKernelGGL(gmm::mstep_moments_b16_kernel, grid, dim3(kNT), lds,
+  // two buffers x (zhi+zlo planes + w tiles for 8 clusters)
+  const size_t lds = 2 * (2 * 32 * 136 * 2 + 8 * 128 * 4);
+  dim3 grid((k + 7) / 8, nchunk);
+  hipLaunchKernelGGL(gmm::mstep_moments_b16_kernel, grid, dim3(512), lds,
                      stream(),
                      reinterpret_cast<const __hip_bfloat16*>(xhi.data_ptr()),
                      reinterpret_cast<const __hip_bfloat16*>(xlo.data_ptr()),

@@ -614,9 +614,10 @@ mstep_moments_kernel(const T* __restrict__ x, const float* __restrict__ w,
 //   B[kk=event][j=dim] = z_j[e]          (split once, shared by 4 clusters)
 // ---------------------------------------------------------------------------
 #define MB_BK 128
-#define MB_CPB 4  // clusters per block (one per wave)
+#define MB_NT 512
+#define MB_CPB 8  // clusters per block (one per wave; 8 halves x re-reads)
 
-__global__ void __launch_bounds__(NT)
+__global__ void __launch_bounds__(MB_NT)
 mstep_moments_b16_kernel(const __hip_bfloat16* __restrict__ xhi,
                          const __hip_bfloat16* __restrict__ xlo,
                          const float* __restrict__ w,
@@ -646,7 +647,7 @@ mstep_moments_b16_kernel(const __hip_bfloat16* __restrict__ xhi,
   // this thread's staging assignment: quads of 4 consecutive events
   // (x: d*MB_BK/4 quads; w: MB_CPB*MB_BK/4 float4, first quarter of threads)
   const int xq_total = d * (MB_BK / 4);
-  const int nxq = (xq_total + NT - 1) / NT;  // quads per thread (<= 4 at D<=31)
+  const int nxq = (xq_total + MB_NT - 1) / MB_NT;  // <= 2 at D <= 31
 
   f32x16_t accA = (f32x16_t)(0.0f);   // hi*hi
   f32x16_t accB = (f32x16_t)(0.0f);   // hi*lo
@@ -657,14 +658,14 @@ mstep_moments_b16_kernel(const __hip_bfloat16* __restrict__ xhi,
       chunk < tiles ? (tiles - chunk + nchunk - 1) / nchunk : 0;
 
   // ---- staging helpers -------------------------------------------------
-  uint2 rxh[4], rxl[4];
+  uint2 rxh[2], rxl[2];
   float4 rw;
   auto issue_loads = [&](int64_t tile) {
     const int64_t e0 = tile * MB_BK;
     const bool full = (n - e0) >= MB_BK;
 #pragma unroll
-    for (int s = 0; s < 4; ++s) {
-      const int q = threadIdx.x + s * NT;
+    for (int s = 0; s < 2; ++s) {
+      const int q = threadIdx.x + s * MB_NT;
       if (s < nxq && q < xq_total) {
         const int di = q / (MB_BK / 4), eq = q % (MB_BK / 4);
         const int64_t g = (int64_t)di * n + e0 + eq * 4;
@@ -707,8 +708,8 @@ mstep_moments_b16_kernel(const __hip_bfloat16* __restrict__ xhi,
     __bf16* zh = zbuf + buf * BUFB;
     __bf16* zl = zh + PLANE;
 #pragma unroll
-    for (int s = 0; s < 4; ++s) {
-      const int q = threadIdx.x + s * NT;
+    for (int s = 0; s < 2; ++s) {
+      const int q = threadIdx.x + s * MB_NT;
       if (s < nxq && q < xq_total) {
         const int di = q / (MB_BK / 4), eq = q % (MB_BK / 4);
         *(uint2*)(zh + di * ZBR + eq * 4) = rxh[s];
