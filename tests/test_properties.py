@@ -1,0 +1,70 @@
+"""Property-based tests (hypothesis) over the IO and numerics layers."""
+import numpy as np
+import torch
+from hypothesis import given, settings, strategies as st
+
+from cuda_gmm_mpi_amd.ops import cpu_reference as cpu
+from cuda_gmm_mpi_amd.utils import io as gio
+
+
+@settings(max_examples=25, deadline=None)
+@given(
+    n=st.integers(1, 40), d=st.integers(1, 8),
+    seed=st.integers(0, 2**31 - 1),
+)
+def test_bin_roundtrip_property(tmp_path_factory, n, d, seed):
+    rng = np.random.default_rng(seed)
+    data = (rng.standard_normal((n, d)) * 100).astype(np.float32)
+    path = str(tmp_path_factory.mktemp("bin") / "x.bin")
+    gio.write_bin(path, data)
+    np.testing.assert_array_equal(gio.read_bin(path), data)
+
+
+@settings(max_examples=25, deadline=None)
+@given(k=st.integers(1, 12), n=st.integers(1, 200),
+       scale=st.floats(0.1, 50), seed=st.integers(0, 2**31 - 1))
+def test_posteriors_always_normalized(k, n, scale, seed):
+    rng = np.random.default_rng(seed)
+    logw = torch.from_numpy(
+        (rng.standard_normal((k, n)) * scale).astype(np.float32))
+    w, lik = cpu.estep_posteriors(logw.clone())
+    assert torch.isfinite(w).all()
+    np.testing.assert_allclose(w.sum(dim=0).numpy(), np.ones(n), rtol=1e-4)
+    assert float(lik) == float(lik)  # finite, not NaN
+
+
+@settings(max_examples=20, deadline=None)
+@given(d=st.integers(1, 16), seed=st.integers(0, 2**31 - 1),
+       cond=st.floats(1.0, 100.0))
+def test_lu_invert_property(d, seed, cond):
+    rng = np.random.default_rng(seed)
+    a = rng.standard_normal((d, d))
+    m = (a @ a.T + cond * d * np.eye(d)).astype(np.float32)
+    inv, logdet = cpu.lu_invert_nopivot(torch.from_numpy(m)[None])
+    ref = np.linalg.slogdet(m.astype(np.float64))[1]
+    assert abs(float(logdet[0]) - ref) < max(1e-2, 2e-3 * abs(ref))
+    np.testing.assert_allclose(
+        (inv[0].numpy() @ m), np.eye(d), atol=5e-2)
+
+
+@settings(max_examples=15, deadline=None)
+@given(st.lists(st.floats(-1e6, 1e6, allow_nan=False, width=32),
+                min_size=2, max_size=20),
+       st.integers(2, 6))
+def test_csv_roundtrip_values(values, d):
+    import io as _io
+    import tempfile, os
+    rows = [values[i:i + d] for i in range(0, len(values) - d + 1, d)]
+    if not rows:
+        return
+    with tempfile.TemporaryDirectory() as td:
+        path = os.path.join(td, "x.csv")
+        with open(path, "w") as f:
+            f.write(",".join("h" * d) or "h")
+            f.write("\n")
+            for r in rows:
+                f.write(",".join(f"{v!r}" for v in r) + "\n")
+        data = gio.read_csv(path)
+    assert data.shape == (len(rows), d)
+    np.testing.assert_allclose(
+        data, np.array(rows, dtype=np.float32), rtol=1e-6)
