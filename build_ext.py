@@ -25,7 +25,9 @@ def torch_paths():
     return ce.include_paths(), ce.library_paths(), torch._C._GLIBCXX_USE_CXX11_ABI
 
 
-def build(verbose: bool = True) -> str:
+def build(verbose: bool = True, asan: bool = False) -> str:
+    """asan=True adds host-side AddressSanitizer to the extension build
+    (SURVEY §5 race/sanitizer row; device code is unaffected)."""
     includes, libdirs, cxx11_abi = torch_paths()
     cmd = [
         HIPCC, "-O3", "-std=c++17", "-fPIC", "-shared",
@@ -40,6 +42,8 @@ def build(verbose: bool = True) -> str:
     ]
     for inc in includes + [sysconfig.get_paths()["include"]]:
         cmd.append(f"-I{inc}")
+    if asan:
+        cmd += ["-fsanitize=address", "-shared-libasan"]
     for lib in libdirs:
         cmd += [f"-L{lib}", f"-Wl,-rpath,{lib}"]
     cmd += ["-ltorch", "-ltorch_cpu", "-ltorch_hip", "-lc10", "-lc10_hip",
@@ -51,6 +55,7 @@ def build(verbose: bool = True) -> str:
 
 
 if __name__ == "__main__":
-    build()
+    import sys as _sys
+    build(asan="--asan" in _sys.argv)
     print(f"built {OUT}")
     sys.exit(0)

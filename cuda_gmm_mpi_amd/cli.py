@@ -46,6 +46,8 @@ def make_parser() -> argparse.ArgumentParser:
     p.add_argument("--no-results", dest="write_results", action="store_false",
                    help="write only the .summary file")
     p.add_argument("--print", dest="enable_print", action="store_true")
+    p.add_argument("--verbose", action="store_true",
+                   help="per-iteration likelihood lines (DEBUG-style)")
     p.add_argument("--estep-dtype", choices=["fp32", "bf16"], default="fp32")
     p.add_argument("--mstep-precision", choices=["fp32", "bf16x3"],
                    default="fp32",
@@ -80,6 +82,7 @@ def config_from_args(args) -> GmmConfig:
         diag_only=args.diag_only, bug_compat=args.bug_compat,
         enable_print=args.enable_print, enable_output=args.enable_output,
         estep_dtype=args.estep_dtype, center_data=args.center_data,
+        verbose=args.verbose,
         mstep_precision=args.mstep_precision,
         checkpoint_dir=args.checkpoint_dir,
         nearest_target=args.nearest_target,
@@ -151,6 +154,12 @@ def main(argv=None) -> int:
         device = "cuda" if torch.cuda.is_available() else "cpu"
 
     rank, local_rank, world = pdist.init_process_group()
+    if cfg.enable_print:
+        import socket
+        dev_name = (torch.cuda.get_device_name(0)
+                    if device == "cuda" else "cpu")
+        print(f"Rank {rank} of {world} on {socket.gethostname()} "
+              f"using {dev_name}")
     try:
         try:
             data = gio.read_data(args.infile)

@@ -254,6 +254,9 @@ class EmEngine:
             lik = float(self._lik_dev.item())
             change = lik - old_lik
             iters += 1
+            if cfg.verbose and self.rank == 0:
+                print(f"[K={k}] iter {iters}: likelihood {lik:e} "
+                      f"(change {change:e})")
         self.likelihood = lik
         return lik
 
