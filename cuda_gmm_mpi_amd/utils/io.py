@@ -51,6 +51,20 @@ def write_bin(path: str, data: np.ndarray) -> None:
 
 
 def read_csv(path: str) -> np.ndarray:
+    """CSV reader; fast pandas path for well-formed files, quirk-faithful
+    fallback (atof semantics) otherwise."""
+    try:
+        import pandas as pd
+        df = pd.read_csv(path, header=0, sep=",", skip_blank_lines=True,
+                         dtype=np.float32)
+        if df.shape[0] > 0 and df.shape[1] > 0 and not df.isna().any().any():
+            return np.ascontiguousarray(df.to_numpy(dtype=np.float32))
+    except Exception:  # noqa: BLE001 — fall back to the faithful parser
+        pass
+    return _read_csv_faithful(path)
+
+
+def _read_csv_faithful(path: str) -> np.ndarray:
     with open(path, "r") as f:
         lines = [ln for ln in (line.strip("\n") for line in f) if ln != ""]
     if not lines:
