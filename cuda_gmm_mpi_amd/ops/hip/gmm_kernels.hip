@@ -650,8 +650,7 @@ mstep_moments_b16_kernel(const __hip_bfloat16* __restrict__ xhi,
   const int nxq = (xq_total + MB_NT - 1) / MB_NT;  // <= 2 at D <= 31
 
   f32x16_t accA = (f32x16_t)(0.0f);   // hi*hi
-  f32x16_t accB = (f32x16_t)(0.0f);   // hi*lo
-  f32x16_t accC = (f32x16_t)(0.0f);   // lo*hi
+  f32x16_t accB = (f32x16_t)(0.0f);   // hi*lo + lo*hi (shared accumulator)
 
   const int64_t tiles = (n + MB_BK - 1) / MB_BK;
   const int64_t my_tiles =
@@ -764,7 +763,7 @@ mstep_moments_b16_kernel(const __hip_bfloat16* __restrict__ xhi,
       }
       accA = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a_hi, b_hi, accA, 0, 0, 0);
       accB = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a_hi, b_lo, accB, 0, 0, 0);
-      accC = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a_lo, b_hi, accC, 0, 0, 0);
+      accB = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a_lo, b_hi, accB, 0, 0, 0);
     }
     __syncthreads();  // everyone done reading buf[cur]
     if (ti + 1 < my_tiles) write_buf(cur ^ 1);
@@ -781,7 +780,7 @@ mstep_moments_b16_kernel(const __hip_bfloat16* __restrict__ xhi,
     const int gi = (r & 3) + 8 * (r >> 2) + 4 * g2;
     const int gj = j32;
     if (gi < dp && gj <= gi)
-      out[gi * (gi + 1) / 2 + gj] = accA[r] + accB[r] + accC[r];
+      out[gi * (gi + 1) / 2 + gj] = accA[r] + accB[r];
   }
 }
 
