@@ -169,3 +169,36 @@ def test_sweep_bf16_fused_post_merge_factors():
     assert res_g.num_clusters == res_c.num_clusters
     # bf16-class tolerance on the score trajectory
     assert res_g.min_rissanen == pytest.approx(res_c.min_rissanen, rel=2e-2)
+
+
+def test_gpu_soak_100_iterations_monotone():
+    """Reference-length EM (100 iterations) at scale: likelihood monotone
+    (exact fp32 path) and posteriors normalized throughout."""
+    data, _ = make_blobs(500_000, 24, 64, seed=97)
+    cfg = GmmConfig(num_clusters=64, target_num_clusters=64,
+                    min_iters=1, max_iters=1)
+    eng = build_engine(data, cfg, device="cuda")
+    liks = [eng._reduce_likelihood(eng._estep(64))]
+    for _ in range(100):
+        eng.em_iteration(64)
+        liks.append(float(eng._lik_dev.item()))
+    liks = np.array(liks)
+    tol = abs(liks[-1]) * 1e-6
+    assert (np.diff(liks) > -tol).all(), \
+        f"non-monotone at {np.argmin(np.diff(liks))}"
+    s = eng.w[:64].sum(dim=0)
+    assert float((s - 1).abs().max()) < 1e-3
+
+
+def test_gpu_soak_bf16_close_to_fp32():
+    """bf16 fast path tracks the fp32 path over a 50-iteration run."""
+    data, _ = make_blobs(300_000, 24, 32, seed=101)
+    liks = {}
+    for name, ed, mp in (("fp32", "fp32", "fp32"),
+                         ("bf16", "bf16", "bf16x3")):
+        cfg = GmmConfig(num_clusters=32, target_num_clusters=32,
+                        min_iters=50, max_iters=50, estep_dtype=ed,
+                        mstep_precision=mp)
+        eng = build_engine(data, cfg, device="cuda")
+        liks[name] = eng.run_em(32)
+    assert liks["bf16"] == pytest.approx(liks["fp32"], rel=5e-3)

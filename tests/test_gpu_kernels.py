@@ -418,3 +418,33 @@ def test_estep_fused_f32_matches_cpu(device):
     np.testing.assert_allclose(w.cpu().numpy(), ref_w.numpy(),
                                rtol=2e-3, atol=2e-4)
     assert float(lik) == pytest.approx(float(ref_lik), rel=1e-4)
+
+
+def test_fuzz_kernels_vs_cpu(device):
+    """Randomized shape fuzz: fused/big E-step + moments vs CPU reference."""
+    from cuda_gmm_mpi_amd.ops import functional as F
+    rng = np.random.default_rng(2024)
+    for trial in range(12):
+        d = int(rng.integers(1, 32)) if trial % 2 == 0 else int(
+            rng.integers(32, 129))
+        k = int(rng.integers(1, 20))
+        n = int(rng.integers(50, 4000))
+        x = (rng.standard_normal((d, n)) * rng.uniform(0.5, 5)).astype(
+            np.float32)
+        w = rng.uniform(0, 1, (k, n)).astype(np.float32)
+        xt = torch.from_numpy(x).to(device)
+        wt = torch.from_numpy(w).to(device)
+        packed = F.mstep_moments(xt, wt, precision="bf16x3")
+        n_c, mean_num, s = F.moments_views(packed, d)
+        rn, rm, rs = cpu.mstep_sufficient_stats(
+            torch.from_numpy(x).double(), torch.from_numpy(w).double())
+        np.testing.assert_allclose(n_c.cpu().numpy(), rn.numpy(),
+                                   rtol=1e-3, atol=1e-3)
+        scale = float(rs.abs().max()) + 1e-6
+        np.testing.assert_allclose(s.cpu().numpy(), rs.numpy(),
+                                   rtol=1e-3, atol=1e-3 * scale)
+        # exact fp32 moments too (D <= 31 fast path or compose path)
+        packed32 = F.mstep_moments(xt, wt, precision="fp32")
+        n32, m32, s32 = F.moments_views(packed32, d)
+        np.testing.assert_allclose(s32.cpu().numpy(), rs.numpy(),
+                                   rtol=1e-4, atol=1e-4 * scale)
