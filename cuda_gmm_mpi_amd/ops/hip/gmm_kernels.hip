@@ -1088,7 +1088,8 @@ estep_fused_f32_kernel(const float* __restrict__ z,
 // Grid (ceil(K/2), nchunk); BK = 64 events per tile.
 // ---------------------------------------------------------------------------
 #define MBB_BK 64
-#define MBB_NT 512  // 8 waves: 2 clusters x 4 pair-quarters
+#define MBB_NT 1024  // 16 waves: 4 clusters x 4 pair-quarters
+#define MBB_CPB 4
 #define MBB_PMAX 4  // tile-pairs per wave (RT2 <= 5 -> TP <= 15 -> 4/quarter)
 
 __global__ void __launch_bounds__(MBB_NT)
@@ -1105,15 +1106,15 @@ mstep_moments_big_kernel(const float* __restrict__ x,
   extern __shared__ float lds[];
   __bf16* zhi = (__bf16*)lds;            // [rows][zbr]
   __bf16* zlo = zhi + rows * zbr;
-  float* wt = (float*)(zlo + rows * zbr);  // [2][MBB_BK]
+  float* wt = (float*)(zlo + rows * zbr);  // [MBB_CPB][MBB_BK]
 
   const int wave = threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
   const int j32 = lane & 31;
   const int g2 = lane >> 5;
-  const int cw = wave >> 2;              // which of the block's 2 clusters
+  const int cw = wave >> 2;              // which of the block's 4 clusters
   const int quarter = wave & 3;          // which quarter of the tile-pairs
-  const int c = blockIdx.x * 2 + cw;
+  const int c = blockIdx.x * MBB_CPB + cw;
   const int chunk = blockIdx.y;
   const int p_lo = quarter * MBB_PMAX;
   const int p_hi = min(tp, p_lo + MBB_PMAX);
@@ -1150,8 +1151,8 @@ mstep_moments_big_kernel(const float* __restrict__ x,
           put(idx / MBB_BK, idx % MBB_BK,
               (idx / MBB_BK == d) ? 1.0f : 0.0f);
       }
-      for (int idx = threadIdx.x; idx < 2 * MBB_BK; idx += MBB_NT) {
-        const int ci = blockIdx.x * 2 + idx / MBB_BK;
+      for (int idx = threadIdx.x; idx < MBB_CPB * MBB_BK; idx += MBB_NT) {
+        const int ci = blockIdx.x * MBB_CPB + idx / MBB_BK;
         if (ci < k)
           wt[idx] = w[(int64_t)ci * n + e0 + idx % MBB_BK];
       }
@@ -1165,8 +1166,8 @@ mstep_moments_big_kernel(const float* __restrict__ x,
         }
         put(di, ei, v);
       }
-      for (int idx = threadIdx.x; idx < 2 * MBB_BK; idx += MBB_NT) {
-        const int ci = blockIdx.x * 2 + idx / MBB_BK;
+      for (int idx = threadIdx.x; idx < MBB_CPB * MBB_BK; idx += MBB_NT) {
+        const int ci = blockIdx.x * MBB_CPB + idx / MBB_BK;
         const int ei = idx % MBB_BK;
         wt[idx] = (ci < k && ei < cnt) ? w[(int64_t)ci * n + e0 + ei] : 0.0f;
       }
