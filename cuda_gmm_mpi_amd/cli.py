@@ -101,6 +101,9 @@ def run_clustering(data: np.ndarray, cfg: GmmConfig, outfile: str,
         local_rank = torch.cuda.current_device()
     prof = Profile(device)
     engine = build_engine(data, cfg, device=device, profile=prof)
+    if profile_report:
+        # graph replay bypasses the per-bucket hipEvent timers
+        engine.use_graphs = False
     result = engine.sweep()
 
     # de-center the saved model for output
