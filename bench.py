@@ -62,6 +62,9 @@ def main() -> int:
         mstep_precision=("bf16x3" if args.dtype == "bf16" else "fp32"),
     )
     engine = build_engine(data, cfg, device=device)
+    if args.profile:
+        # per-bucket hipEvent timers need the eager path
+        engine.use_graphs = False
 
     def sync():
         pdist.barrier()

@@ -139,7 +139,12 @@ def main(argv=None) -> int:
     rest, ngpus = strip_gpus_arg(list(argv))
     if ngpus and ngpus > 1 and "WORLD_SIZE" not in __import__("os").environ:
         return launch_workers(rest, ngpus)
-    args = make_parser().parse_args(rest)
+    try:
+        args = make_parser().parse_args(rest)
+    except SystemExit:
+        # reference prints usage and returns 1 on bad arguments
+        # (gaussian.cu:1111-1166)
+        return 1
     if not (1 <= args.num_clusters <= MAX_CLUSTERS):
         print("Invalid number of starting clusters\n")
         return 1

@@ -102,3 +102,9 @@ def test_strip_gpus_arg():
     assert rest == ["3", "x", "y"] and n == 8
     rest, n = strip_gpus_arg(["3", "x", "y"])
     assert n is None
+
+
+def test_cli_bad_args_exit_code():
+    # reference prints usage and returns 1 (gaussian.cu:1162-1165)
+    assert main([]) == 1
+    assert main(["notanumber", "x", "y"]) == 1

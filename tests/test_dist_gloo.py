@@ -132,3 +132,26 @@ def _single_worker(fn_name, q):
     for var in ("RANK", "WORLD_SIZE", "LOCAL_RANK"):
         os.environ.pop(var, None)
     q.put(globals()[fn_name]())
+
+
+def _fit_world4():
+    from cuda_gmm_mpi_amd.engine import build_engine
+    from cuda_gmm_mpi_amd.utils.config import GmmConfig
+    data, _ = make_blobs(1003, 3, 3, seed=37)  # odd N: remainder shards
+    cfg = GmmConfig(num_clusters=4, target_num_clusters=2,
+                    min_iters=3, max_iters=3,
+                    covariance_dynamic_range=1e15)
+    eng = build_engine(data, cfg, device="cpu")
+    res = eng.sweep()
+    return {"k": res.num_clusters, "rissanen": res.min_rissanen,
+            "means": res.state.means.numpy().copy()}
+
+
+@pytest.mark.timeout(300)
+def test_world4_sweep_matches_single():
+    single = _run_single("_fit_world4")
+    multi = run_world(4, "_fit_world4", port=29813)
+    assert multi["k"] == single["k"]
+    assert multi["rissanen"] == pytest.approx(single["rissanen"], rel=1e-3)
+    np.testing.assert_allclose(multi["means"], single["means"],
+                               rtol=2e-3, atol=2e-3)

@@ -205,3 +205,32 @@ def test_nearest_target_mode():
     assert res_b.num_clusters <= res_a.num_clusters or \
         res_a.num_clusters == 3
     assert res_b.num_clusters >= 3
+
+
+def test_single_cluster_k1():
+    """K=1 is a legal starting count in the reference (gaussian.cu:1121)."""
+    data, _ = make_blobs(800, 3, 2, seed=31)
+    cfg = GmmConfig(num_clusters=1, target_num_clusters=1,
+                    min_iters=4, max_iters=4)
+    eng = build_engine(data, cfg, device="cpu")
+    res = eng.sweep()
+    assert res.num_clusters == 1
+    assert eng.state.pi[0] == pytest.approx(1.0)
+    # single-cluster means ~ global mean, R ~ global covariance
+    mu = (eng.state.means[0] + eng.center).numpy()
+    np.testing.assert_allclose(mu, data.mean(axis=0), rtol=1e-3, atol=1e-2)
+
+
+def test_one_dimensional_data():
+    """D=1 exercises the scalar branches of every op."""
+    rng = np.random.default_rng(7)
+    data = np.concatenate([rng.normal(0, 1, 500),
+                           rng.normal(10, 2, 500)]).astype(np.float32)
+    data = data.reshape(-1, 1)
+    cfg = GmmConfig(num_clusters=2, target_num_clusters=2,
+                    min_iters=15, max_iters=15)
+    eng = build_engine(data, cfg, device="cpu")
+    lik = eng.run_em(2)
+    assert np.isfinite(lik)
+    mu = np.sort((eng.state.means + eng.center).numpy().ravel())
+    np.testing.assert_allclose(mu, [0.0, 10.0], atol=0.5)

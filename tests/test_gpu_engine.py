@@ -202,3 +202,25 @@ def test_gpu_soak_bf16_close_to_fp32():
         eng = build_engine(data, cfg, device="cuda")
         liks[name] = eng.run_em(32)
     assert liks["bf16"] == pytest.approx(liks["fp32"], rel=5e-3)
+
+
+def test_gpu_k1_and_d1():
+    """Degenerate shapes on every GPU kernel path."""
+    rng = np.random.default_rng(9)
+    d1 = np.concatenate([rng.normal(0, 1, 2000),
+                         rng.normal(8, 1, 2000)]).astype(np.float32)
+    for ed in ("fp32", "bf16"):
+        cfg = GmmConfig(num_clusters=2, target_num_clusters=2,
+                        min_iters=10, max_iters=10, estep_dtype=ed)
+        eng = build_engine(d1.reshape(-1, 1), cfg, device="cuda")
+        lik = eng.run_em(2)
+        assert np.isfinite(lik)
+        mu = np.sort((eng.state.means.cpu() + eng.center.cpu()).numpy().ravel())
+        np.testing.assert_allclose(mu, [0.0, 8.0], atol=0.5)
+    data, _ = make_blobs(5000, 4, 2, seed=3)
+    cfg = GmmConfig(num_clusters=1, target_num_clusters=1,
+                    min_iters=4, max_iters=4, estep_dtype="bf16",
+                    mstep_precision="bf16x3")
+    eng = build_engine(data, cfg, device="cuda")
+    res = eng.sweep()
+    assert res.num_clusters == 1
