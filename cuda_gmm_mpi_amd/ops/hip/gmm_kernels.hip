@@ -46,6 +46,7 @@ __device__ inline void tri_row_col(int t, int* i, int* j) {
 __device__ inline int mfma_b16_k(int group, int u) { return 8 * group + u; }
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
 typedef __attribute__((ext_vector_type(16))) float f32x16_t;
 
 // k-chunk count rounded to the estep-big template tiers (D <= 143); the
@@ -503,78 +504,125 @@ constants_lu_kernel(const float* __restrict__ r,
 // ---------------------------------------------------------------------------
 #define MOM_BK 128
 
-typedef __attribute__((ext_vector_type(4))) float f32x4;
-
 template <typename T>
 __global__ void __launch_bounds__(NT)
 mstep_moments_kernel(const T* __restrict__ x, const float* __restrict__ w,
                      float* __restrict__ partials, int d, int k, int64_t n,
                      int nchunk) {
+  // Exact-fp32 augmented moments on v_mfma_f32_16x16x4_f32, with
+  // double-buffered register-staged tiles (T14) like the bf16x3 kernel.
+  // LDS per buffer: xs [d][MOM_BK+4] f32, wt [4][MOM_BK] f32.
   extern __shared__ float lds[];
   const int row = MOM_BK + 4;
-  float* xs = lds;                       // [d][MOM_BK+4]
-  float* wt = lds + (int64_t)d * row;    // [4][MOM_BK]
+  const int plane = d * row;
+  float* xs0 = lds;
+  float* wt0 = lds + plane;
+  float* xs1 = wt0 + 4 * MOM_BK;
+  float* wt1 = xs1 + plane;
 
   const int wave = threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
   const int c = blockIdx.x * 4 + wave;
   const int chunk = blockIdx.y;
   const int dp = d + 1;
-  const int rt2 = dp > 16;               // second row-tile present?
+  const int rt2 = dp > 16;
 
   f32x4 acc00 = {0, 0, 0, 0}, acc10 = {0, 0, 0, 0}, acc11 = {0, 0, 0, 0};
-
   const int i_loc = lane & 15;
   const int kk = lane >> 4;              // event sub-index 0..3
 
+  const int xq_total = d * (MOM_BK / 4);
+  const int nxq = (xq_total + NT - 1) / NT;  // <= 4 at D <= 31
   const int64_t tiles = (n + MOM_BK - 1) / MOM_BK;
-  for (int64_t tile = chunk; tile < tiles; tile += nchunk) {
+  const int64_t my_tiles =
+      chunk < tiles ? (tiles - chunk + nchunk - 1) / nchunk : 0;
+
+  float4 rx[4];
+  float4 rw;
+  auto issue_loads = [&](int64_t tile) {
     const int64_t e0 = tile * MOM_BK;
-    const int cnt = (int)min((int64_t)MOM_BK, n - e0);
-    __syncthreads();
-    // branchless full-tile staging: a per-element bounds ternary makes
-    // hipcc serialize every load behind a vmcnt(0) (guide §5 trap 4c)
-    const int nv = min(4, k - blockIdx.x * 4) * MOM_BK;
-    if (cnt == MOM_BK) {
-      for (int idx = threadIdx.x; idx < d * MOM_BK; idx += blockDim.x)
-        xs[(idx / MOM_BK) * row + idx % MOM_BK] =
-            load_x(x, (int64_t)(idx / MOM_BK) * n + e0 + idx % MOM_BK);
-      for (int idx = threadIdx.x; idx < nv; idx += blockDim.x)
-        wt[idx] = w[(int64_t)(blockIdx.x * 4 + idx / MOM_BK) * n + e0 +
-                    idx % MOM_BK];
-    } else {
-      for (int idx = threadIdx.x; idx < d * MOM_BK; idx += blockDim.x) {
-        const int di = idx / MOM_BK, ei = idx % MOM_BK;
-        xs[di * row + ei] =
-            (ei < cnt) ? load_x(x, (int64_t)di * n + e0 + ei) : 0.0f;
-      }
-      for (int idx = threadIdx.x; idx < 4 * MOM_BK; idx += blockDim.x) {
-        const int wv = idx / MOM_BK, ei = idx % MOM_BK;
-        const int cw = blockIdx.x * 4 + wv;
-        wt[wv * MOM_BK + ei] =
-            (cw < k && ei < cnt) ? w[(int64_t)cw * n + e0 + ei] : 0.0f;
+    const bool full = (n - e0) >= MOM_BK;
+#pragma unroll
+    for (int s = 0; s < 4; ++s) {
+      const int q = threadIdx.x + s * NT;
+      if (s < nxq && q < xq_total) {
+        const int di = q / (MOM_BK / 4), eq = q % (MOM_BK / 4);
+        const int64_t g = (int64_t)di * n + e0 + eq * 4;
+        if (full || eq * 4 + 3 < (int)(n - e0)) {
+          rx[s] = make_float4(load_x(x, g), load_x(x, g + 1),
+                              load_x(x, g + 2), load_x(x, g + 3));
+        } else {
+          float v[4];
+          for (int u = 0; u < 4; ++u)
+            v[u] = (e0 + eq * 4 + u < n) ? load_x(x, g + u) : 0.0f;
+          rx[s] = *(float4*)v;
+        }
       }
     }
-    __syncthreads();
+    if (threadIdx.x < 4 * (MOM_BK / 4)) {
+      const int wv = threadIdx.x / (MOM_BK / 4);
+      const int eq = threadIdx.x % (MOM_BK / 4);
+      const int cw = blockIdx.x * 4 + wv;
+      if (cw < k) {
+        const int64_t g = (int64_t)cw * n + e0 + eq * 4;
+        if (full) {
+          rw = *(const float4*)&w[g];
+        } else {
+          float v[4];
+          for (int u = 0; u < 4; ++u)
+            v[u] = (e0 + eq * 4 + u < n) ? w[g + u] : 0.0f;
+          rw = *(float4*)v;
+        }
+      } else {
+        rw = (float4){0, 0, 0, 0};
+      }
+    }
+  };
+  auto write_buf = [&](float* xs, float* wt) {
+#pragma unroll
+    for (int s = 0; s < 4; ++s) {
+      const int q = threadIdx.x + s * NT;
+      if (s < nxq && q < xq_total) {
+        const int di = q / (MOM_BK / 4), eq = q % (MOM_BK / 4);
+        *(float4*)(xs + di * row + eq * 4) = rx[s];
+      }
+    }
+    if (threadIdx.x < 4 * (MOM_BK / 4))
+      *(float4*)(wt + threadIdx.x * 4) = rw;
+  };
 
-#pragma unroll 4
+  if (my_tiles > 0) {
+    issue_loads(chunk);
+    write_buf(xs0, wt0);
+  }
+  __syncthreads();
+
+  int cur = 0;
+  for (int64_t ti = 0; ti < my_tiles; ++ti) {
+    if (ti + 1 < my_tiles) issue_loads(chunk + (ti + 1) * nchunk);
+    const float* xs = cur ? xs1 : xs0;
+    const float* wt = cur ? wt1 : wt0;
+
     for (int ks = 0; ks < MOM_BK / 4; ++ks) {
       const int e = ks * 4 + kk;
       const float we = wt[wave * MOM_BK + e];
-      // z values for this lane's row position in each row-tile
       const int g0 = i_loc;             // row-tile 0 rows 0..15
       const int g1 = 16 + i_loc;        // row-tile 1 rows 16..31
       const float z0 = (g0 < d) ? xs[g0 * row + e] : (g0 == d ? 1.0f : 0.0f);
       const float z1 = (g1 < d) ? xs[g1 * row + e] : (g1 == d ? 1.0f : 0.0f);
       const float a0 = we * z0;
-      const float b0 = z0;
-      acc00 = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b0, acc00, 0, 0, 0);
+      acc00 = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, z0, acc00, 0, 0, 0);
       if (rt2) {
         const float a1 = we * z1;
-        acc10 = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b0, acc10, 0, 0, 0);
+        acc10 = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, z0, acc10, 0, 0, 0);
         acc11 = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, z1, acc11, 0, 0, 0);
       }
     }
+    // single barrier per tile: the write targets the buffer nobody reads
+    // this iteration, and the previous barrier already ordered its readers
+    if (ti + 1 < my_tiles) write_buf(cur ? xs0 : xs1, cur ? wt0 : wt1);
+    __syncthreads();
+    cur ^= 1;
   }
 
   if (c >= k) return;
@@ -765,9 +813,10 @@ mstep_moments_b16_kernel(const __hip_bfloat16* __restrict__ xhi,
       accB = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a_hi, b_lo, accB, 0, 0, 0);
       accB = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a_lo, b_hi, accB, 0, 0, 0);
     }
-    __syncthreads();  // everyone done reading buf[cur]
+    // single barrier per tile: the write targets the buffer nobody reads
+    // this iteration, and the previous barrier already ordered its readers
     if (ti + 1 < my_tiles) write_buf(cur ^ 1);
-    __syncthreads();  // buf[cur^1] ready
+    __syncthreads();  // buf[cur^1] ready for the next iteration
     cur ^= 1;
   }
 
