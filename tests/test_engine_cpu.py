@@ -234,3 +234,44 @@ def test_one_dimensional_data():
     assert np.isfinite(lik)
     mu = np.sort((eng.state.means + eng.center).numpy().ravel())
     np.testing.assert_allclose(mu, [0.0, 10.0], atol=0.5)
+
+
+def test_corrupted_checkpoint_is_ignored(tmp_path):
+    """A damaged checkpoint must not break the sweep start."""
+    import os
+    from cuda_gmm_mpi_amd.utils.checkpoint import load_sweep_checkpoint
+    ckdir = str(tmp_path / "ck")
+    os.makedirs(ckdir)
+    with open(os.path.join(ckdir, "gmm_sweep.npz"), "wb") as f:
+        f.write(b"not an npz at all")
+    try:
+        ck = load_sweep_checkpoint(ckdir)
+    except Exception:
+        ck = None
+    assert ck is None or isinstance(ck, dict)
+
+
+def test_fused_gate_boundaries():
+    """LDS gate: bf16 fused path admits K up to 104, not beyond."""
+    import torch as t
+    from cuda_gmm_mpi_amd.ops import functional as F
+    dev = t.device("cuda")  # gate logic only; no GPU work
+    assert F.estep_fused_available(dev, "bf16", 24, 104)
+    assert not F.estep_fused_available(dev, "bf16", 24, 105)
+    assert not F.estep_fused_available(dev, "bf16", 32, 8)   # D > 31
+    assert F.estep_big_available(dev, "bf16", 32)
+    assert F.estep_big_available(dev, "bf16", 143)
+    assert not F.estep_big_available(dev, "bf16", 144)
+    assert not F.estep_big_available(dev, "fp32", 64)
+    # fp32 fused gate is tighter (f32 z tile + lw in LDS)
+    assert F.estep_fused_available(dev, "fp32", 24, 85)
+    assert not F.estep_fused_available(dev, "fp32", 24, 95)
+
+
+def test_mfac_shape_tiers():
+    from cuda_gmm_mpi_amd.ops.functional import mfac_shape
+    assert mfac_shape(24) == (2, 32, 32)
+    assert mfac_shape(31) == (2, 32, 32)
+    assert mfac_shape(32) == (2, 32, 48)  # 32 rows, 3 k-chunks
+    assert mfac_shape(64) == (2, 64, 80)
+    assert mfac_shape(128) == (2, 128, 144)
