@@ -175,8 +175,10 @@ def estep_fused_f32(z: torch.Tensor, mfac32: torch.Tensor, add: torch.Tensor,
 
 
 def estep_big_available(device: torch.device, dtype: str, d: int) -> bool:
-    """Big-D MFMA logw path gate (31 < D <= 143, bf16 data)."""
-    return device.type == "cuda" and dtype == "bf16" and 31 < d <= 143
+    """Big-D MFMA logw path gate (bf16 data, D <= 143). The engine prefers
+    the fused kernel when its LDS gate admits K; this path also covers
+    D <= 31 with large K (posteriors via the separate kernel)."""
+    return device.type == "cuda" and dtype == "bf16" and 1 <= d <= 143
 
 
 def mfac_shape(d: int) -> tuple[int, int, int]:

@@ -314,14 +314,16 @@ void estep_logw_big(torch::Tensor z, torch::Tensor mfac, torch::Tensor add,
   const int d = (int)z.size(0);
   const int64_t n = z.size(1);
   const int k = (int)add.size(0);
-  TORCH_CHECK(d > 31 && d <= 143, "estep_logw_big is the 31 < D <= 143 path");
+  TORCH_CHECK(d >= 1 && d <= 143, "estep_logw_big supports D <= 143");
   TORCH_CHECK(logw.size(0) == k && logw.size(1) == n, "logw shape");
-  const int kct = d + 1 <= 48 ? 3 : d + 1 <= 80 ? 5 : 9;
+  const int kct =
+      d + 1 <= 32 ? 2 : d + 1 <= 48 ? 3 : d + 1 <= 80 ? 5 : 9;
   const size_t lds = (size_t)256 * (kct * 16 + 8) * 2;
   dim3 grid((uint32_t)((n + 255) / 256), (k + 3) / 4);
   auto s = stream();
 #define LAUNCH_ELB(KCT)                                                        do {                                                                           if (lds > 64 * 1024) {                                                         HIP_CHECK(hipFuncSetAttribute(                                                   reinterpret_cast<const void*>(&gmm::estep_logw_big_kernel<KCT>),             hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds));                }                                                                            hipLaunchKernelGGL((gmm::estep_logw_big_kernel<KCT>), grid, dim3(kNT),                          lds, s,                                                                      reinterpret_cast<const __hip_bfloat16*>(z.data_ptr()),                       reinterpret_cast<const __hip_bfloat16*>(                                         mfac.data_ptr()),                                                        add.data_ptr<float>(), logw.data_ptr<float>(), d, k,                         n);                                                     } while (0)
-  if (kct == 3) LAUNCH_ELB(3);
+  if (kct == 2) LAUNCH_ELB(2);
+  else if (kct == 3) LAUNCH_ELB(3);
   else if (kct == 5) LAUNCH_ELB(5);
   else LAUNCH_ELB(9);
 #undef LAUNCH_ELB

@@ -260,6 +260,7 @@ def test_fused_gate_boundaries():
     assert not F.estep_fused_available(dev, "bf16", 24, 105)
     assert not F.estep_fused_available(dev, "bf16", 32, 8)   # D > 31
     assert F.estep_big_available(dev, "bf16", 32)
+    assert F.estep_big_available(dev, "bf16", 24)  # large-K fallback tier
     assert F.estep_big_available(dev, "bf16", 143)
     assert not F.estep_big_available(dev, "bf16", 144)
     assert not F.estep_big_available(dev, "fp32", 64)

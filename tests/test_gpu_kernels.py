@@ -448,3 +448,25 @@ def test_fuzz_kernels_vs_cpu(device):
         n32, m32, s32 = F.moments_views(packed32, d)
         np.testing.assert_allclose(s32.cpu().numpy(), rs.numpy(),
                                    rtol=1e-4, atol=1e-4 * scale)
+
+
+def test_estep_big_small_d_large_k(device):
+    """D <= 31 with K beyond the fused LDS gate routes through the big-D
+    MFMA kernel (KCT=2 tier)."""
+    from cuda_gmm_mpi_amd.ops import functional as F
+    rng = np.random.default_rng(29)
+    k, d, n = 7, 12, 3000
+    means, r, pi = random_model(rng, k, d, device)
+    mfac = torch.empty(k, *F.mfac_shape(d), dtype=torch.bfloat16,
+                       device=device)
+    rinv_g, const_g = F.constants(r, means, False, mfac)
+    add = const_g + torch.log(pi)
+    x = rng.standard_normal((d, n)).astype(np.float32)
+    out = torch.empty(k, n, dtype=torch.float32, device=device)
+    F.estep_logw_big(torch.from_numpy(x).to(device).to(torch.bfloat16),
+                     mfac, add, out)
+    rinv, const = cpu.compute_constants(r.cpu())
+    ref = cpu.estep_logw(torch.from_numpy(x), means.cpu(), rinv, const,
+                         pi.cpu())
+    np.testing.assert_allclose(out.cpu().numpy(), ref.numpy(),
+                               rtol=5e-2, atol=1.0)
