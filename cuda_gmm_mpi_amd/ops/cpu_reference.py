@@ -148,7 +148,7 @@ def lu_invert_nopivot(a: torch.Tensor) -> tuple[torch.Tensor, torch.Tensor]:
         ) / data[:, i, i].unsqueeze(1)
     diag = torch.diagonal(data, dim1=1, dim2=2).abs()
     logdet = torch.log(diag).sum(dim=1)
-    # invert L (stored below diag; unit diagonal NOT stored — diag holds U)
+    # invert L (Crout storage: L holds the diagonal, U is unit-diagonal)
     for i in range(d):
         for j in range(i, d):
             if i == j:
