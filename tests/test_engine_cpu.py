@@ -276,3 +276,22 @@ def test_mfac_shape_tiers():
     assert mfac_shape(32) == (2, 32, 48)  # 32 rows, 3 k-chunks
     assert mfac_shape(64) == (2, 64, 80)
     assert mfac_shape(128) == (2, 128, 144)
+
+
+def test_profile_report_shape():
+    """Timing report mirrors the reference's per-GPU block
+    (gaussian.cu:967)."""
+    from cuda_gmm_mpi_amd.utils.timers import Profile
+    data, _ = make_blobs(500, 2, 2, seed=1)
+    cfg = GmmConfig(num_clusters=2, target_num_clusters=2,
+                    min_iters=2, max_iters=2)
+    prof = Profile("cpu")
+    eng = build_engine(data, cfg, device="cpu", profile=prof)
+    eng.run_em(2)
+    rep = prof.report(rank=0, gpu=0)
+    for token in ("Node 00 GPU 0:", "E-step Kernel:", "M-step Kernel:",
+                  "Consts Kernel:", "Order Reduce:", "GPU Memcpy:",
+                  "CPU:", "Comm:"):
+        assert token in rep
+    # per-iteration averages present (count column == iterations run)
+    assert "\t3\t" in rep or "\t2\t" in rep
