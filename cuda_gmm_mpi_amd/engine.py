@@ -103,13 +103,19 @@ class EmEngine:
             if need_f32_fac else None
         )
 
+        # persistent constant+ln(pi) buffer for the fused E-step kernels
+        self._add = (
+            torch.empty(k0, dtype=torch.float32, device=self.device)
+            if self.device.type == "cuda" else None
+        )
+
         self.state = GmmState.empty(k0, self.d, self.device)
         seed_state(
             self.state, seed_means - center.unsqueeze(0), var_per_dim,
             num_events_total, config.covariance_dynamic_range,
         )
         # constants for the seeded R=I state (constants_kernel after seeding,
-        # gaussian.cu:404)
+        # gaussian.cu:404); also fills the add buffer
         self._update_constants(self.state)
 
         # membership / logw buffer, cluster-major [K, n_shard]
@@ -123,6 +129,7 @@ class EmEngine:
         self.use_graphs = True
         self._graphs: dict[int, object] = {}
         self._lik_dev = torch.zeros(1, dtype=torch.float32, device=self.device)
+
 
     def _refresh_mfac(self, k: int) -> None:
         """Re-emit the fused-E-step factors from the CURRENT Rinv without
@@ -147,11 +154,21 @@ class EmEngine:
         with self.profile.time("constants"):
             mfac = self.mfac[:k] if self.mfac is not None else None
             mfac32 = self.mfac32[:k] if self.mfac32 is not None else None
+            pi_add = (
+                (st.pi, self._add[:k]) if self._add is not None else None
+            )
             rinv, const = F.constants(st.R, st.means, self.cfg.diag_only,
-                                      mfac, mfac32)
+                                      mfac, mfac32, pi_add=pi_add)
             st.Rinv.copy_(rinv)
             st.constant.copy_(const)
         self.profile.count("constants")
+
+    def _sync_add(self, k: int) -> None:
+        """Recompute constant+ln(pi) after host-side param loads (merge /
+        resume) where constants must NOT be recomputed (quirk #8)."""
+        if self._add is not None:
+            st = self.state.shrink(k)
+            self._add[:k] = st.constant + torch.log(st.pi)
 
     # ------------------------------------------------------------------ EM
 
@@ -160,7 +177,7 @@ class EmEngine:
         st = self.state.shrink(k)
         with self.profile.time("e_step"):
             if self.use_fused_estep:
-                add = st.constant + torch.log(st.pi)
+                add = self._add[:k]
                 if self.mfac32 is not None:
                     w, lik = F.estep_fused_f32(self.x_estep, self.mfac32[:k],
                                                add, self.w[:k])
@@ -168,7 +185,7 @@ class EmEngine:
                     w, lik = F.estep_fused(self.x_estep, self.mfac[:k], add,
                                            self.w[:k])
             elif self.use_big_estep:
-                add = st.constant + torch.log(st.pi)
+                add = self._add[:k]
                 logw = F.estep_logw_big(self.x_estep, self.mfac[:k], add,
                                         self.w[:k])
                 w, lik = F.estep_posteriors(logw)
@@ -227,15 +244,26 @@ class EmEngine:
         with self.profile.time("comm"):
             pdist.all_reduce_(packed)
         with self.profile.time("m_step"):
-            n_c, mean_num, s = F.moments_views(packed, self.d)
-            st.N.copy_(n_c)
-            st.means.copy_(F.finalize_means(n_c, mean_num))
-            st.R.copy_(F.finalize_covariance(
-                n_c, st.means, s, st.avgvar, self.world, self.cfg.diag_only,
-            ))
+            if self.device.type == "cuda":
+                # one kernel: N, means, R, pi from the packed moments with
+                # the reference's exact finalize rules
+                from .ops.backend import hip_ext
+                hip_ext().mstep_finalize(
+                    packed, st.avgvar, self.world, st.N, st.means, st.R,
+                    st.pi, bool(self.cfg.diag_only),
+                )
+            else:
+                n_c, mean_num, s = F.moments_views(packed, self.d)
+                st.N.copy_(n_c)
+                st.means.copy_(F.finalize_means(n_c, mean_num))
+                st.R.copy_(F.finalize_covariance(
+                    n_c, st.means, s, st.avgvar, self.world,
+                    self.cfg.diag_only,
+                ))
         self.profile.count("params")
         self._update_constants(st)
-        st.pi.copy_(F.compute_pi(st.N))
+        if self.device.type != "cuda":
+            st.pi.copy_(F.compute_pi(st.N))
 
     def run_em(self, k: int) -> float:
         """Full EM at fixed K (the inner loop of gaussian.cu:479-755).
@@ -351,6 +379,7 @@ class EmEngine:
                 # regenerate the fused-E-step factors for the resumed
                 # params without recomputing the saved constants
                 self._refresh_mfac(k)
+                self._sync_add(k)
                 best_state = ck["best"]
                 best_k = ck["best_k"]
                 min_rissanen = ck["min_rissanen"]
@@ -407,6 +436,7 @@ class EmEngine:
             # the merged/compacted Rinv needs fresh E-step factors; the
             # constants stay as the host merge path produced them
             self._refresh_mfac(new_k)
+            self._sync_add(new_k)
             self.profile.count("reduce")
             k = new_k
             if cfg.checkpoint_dir and self.rank == 0:

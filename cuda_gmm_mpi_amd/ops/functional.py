@@ -116,7 +116,8 @@ def mstep_covariance_s(x: torch.Tensor, w: torch.Tensor,
 
 def constants(r: torch.Tensor, means: torch.Tensor | None = None,
               diag_only: bool = False, mfac: torch.Tensor | None = None,
-              mfac32: torch.Tensor | None = None
+              mfac32: torch.Tensor | None = None,
+              pi_add: tuple[torch.Tensor, torch.Tensor] | None = None
               ) -> tuple[torch.Tensor, torch.Tensor]:
     """(Rinv [K,D,D], constant [K]) via no-pivot LU + ln|det|.
 
@@ -134,8 +135,18 @@ def constants(r: torch.Tensor, means: torch.Tensor | None = None,
             mfac = torch.empty(0, dtype=torch.bfloat16, device=r.device)
         if mfac32 is None:
             mfac32 = torch.empty(0, dtype=torch.float32, device=r.device)
-        hip_ext().constants(r, means, rinv, logdet, mfac, mfac32,
-                            bool(diag_only))
+        empty = torch.empty(0, dtype=torch.float32, device=r.device)
+        if pi_add is not None:
+            pi_t, add_t = pi_add
+            const_t = torch.empty(k, dtype=torch.float32, device=r.device)
+            hip_ext().constants(r, means, pi_t, rinv, logdet, const_t,
+                                add_t, mfac, mfac32, bool(diag_only))
+            if diag_only:  # diag kernel does not emit constant/add
+                const_t = -d * 0.5 * cpu.LOG_2PI - 0.5 * logdet
+                add_t.copy_(const_t + torch.log(pi_t))
+            return rinv, const_t
+        hip_ext().constants(r, means, empty, rinv, logdet, empty, empty,
+                            mfac, mfac32, bool(diag_only))
         const = -d * 0.5 * cpu.LOG_2PI - 0.5 * logdet
         return rinv, const
     return cpu.compute_constants(r, diag_only)

@@ -164,9 +164,10 @@ void mstep_covariance_partials(torch::Tensor x, torch::Tensor w,
   }
 }
 
-void constants(torch::Tensor r, torch::Tensor means, torch::Tensor rinv,
-               torch::Tensor logdet, torch::Tensor mfac, torch::Tensor mfac32,
-               bool diag_only) {
+void constants(torch::Tensor r, torch::Tensor means, torch::Tensor pi,
+               torch::Tensor rinv, torch::Tensor logdet,
+               torch::Tensor constant, torch::Tensor add, torch::Tensor mfac,
+               torch::Tensor mfac32, bool diag_only) {
   check_f32(r, "r");
   check_f32(means, "means");
   check_f32(rinv, "rinv");
@@ -193,6 +194,10 @@ void constants(torch::Tensor r, torch::Tensor means, torch::Tensor rinv,
     mp = reinterpret_cast<__hip_bfloat16*>(mfac.data_ptr());
   }
   auto s = stream();
+  float* pip = pi.numel() > 0 ? pi.data_ptr<float>() : nullptr;
+  float* cst = constant.numel() > 0 ? constant.data_ptr<float>() : nullptr;
+  float* addp = add.numel() > 0 ? add.data_ptr<float>() : nullptr;
+  TORCH_CHECK(addp == nullptr || pip != nullptr, "add output requires pi");
   if (diag_only) {
     hipLaunchKernelGGL(gmm::constants_diag_kernel, dim3(k), dim3(kNT), 0, s,
                        r.data_ptr<float>(), rinv.data_ptr<float>(),
@@ -215,8 +220,8 @@ void constants(torch::Tensor r, torch::Tensor means, torch::Tensor rinv,
     }
     hipLaunchKernelGGL(gmm::constants_lu_kernel, dim3(k), dim3(kNT),
                        lds, s, r.data_ptr<float>(), means.data_ptr<float>(),
-                       rinv.data_ptr<float>(), logdet.data_ptr<float>(), mp,
-                       mp32, d);
+                       pip, rinv.data_ptr<float>(), logdet.data_ptr<float>(),
+                       cst, addp, mp, mp32, d);
   }
   HIP_CHECK(hipGetLastError());
 }
@@ -366,6 +371,29 @@ void mfma_probe32(torch::Tensor a, torch::Tensor b, torch::Tensor c) {
   HIP_CHECK(hipGetLastError());
 }
 
+void mstep_finalize(torch::Tensor packed, torch::Tensor avgvar,
+                    int64_t world, torch::Tensor n_out, torch::Tensor means,
+                    torch::Tensor r_out, torch::Tensor pi, bool diag_only) {
+  check_f32(packed, "packed");
+  check_f32(avgvar, "avgvar");
+  check_f32(n_out, "n_out");
+  check_f32(means, "means");
+  check_f32(r_out, "r_out");
+  check_f32(pi, "pi");
+  const int k = (int)means.size(0);
+  const int d = (int)means.size(1);
+  TORCH_CHECK(packed.size(0) == k &&
+                  packed.size(1) == (d + 1) * (d + 2) / 2,
+              "packed shape mismatch");
+  hipLaunchKernelGGL(gmm::mstep_finalize_kernel, dim3(k), dim3(kNT),
+                     sizeof(float) * d, stream(), packed.data_ptr<float>(),
+                     avgvar.data_ptr<float>(), (int)world,
+                     n_out.data_ptr<float>(), means.data_ptr<float>(),
+                     r_out.data_ptr<float>(), pi.data_ptr<float>(), d, k,
+                     diag_only ? 1 : 0);
+  HIP_CHECK(hipGetLastError());
+}
+
 void emit_factors(torch::Tensor rinv, torch::Tensor means,
                   torch::Tensor mfac, torch::Tensor mfac32) {
   check_f32(rinv, "rinv");
@@ -451,6 +479,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "split-precision bf16x3 augmented moments");
   m.def("mstep_moments", &mstep_moments,
         "fused augmented moments [S|mean_num|N] via f32 MFMA");
+  m.def("mstep_finalize", &mstep_finalize,
+        "finalize N/means/R/pi from all-reduced packed moments");
   m.def("emit_factors", &emit_factors,
         "re-emit E-step factors from an existing Rinv (post-merge/resume)");
   m.def("estep_fused_f32", &estep_fused_f32,

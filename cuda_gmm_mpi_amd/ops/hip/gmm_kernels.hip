@@ -309,6 +309,57 @@ mstep_cov_kernel(const T* __restrict__ x, const float* __restrict__ w,
 // sized for D up to 128 — no NUM_DIMENSIONS cap).
 // One workgroup per cluster. Crout-style: L keeps the diagonal, U unit.
 // ---------------------------------------------------------------------------
+// ---------------------------------------------------------------------------
+// M-step finalize (one workgroup per cluster): from the all-reduced packed
+// moments [K, Dp(Dp+1)/2] produce N, means, R and pi with the reference's
+// exact rules (cov zero at N<1, G*avgvar on the diagonal,
+// divide at N>0.5 else identity; means zero at N<=0.5; pi floor 1e-10).
+// Replaces ~15 eager torch ops per iteration.
+// ---------------------------------------------------------------------------
+__global__ void __launch_bounds__(NT)
+mstep_finalize_kernel(const float* __restrict__ packed,
+                      const float* __restrict__ avgvar, int world,
+                      float* __restrict__ n_out, float* __restrict__ means,
+                      float* __restrict__ r_out, float* __restrict__ pi,
+                      int d, int k, int diag_only) {
+  extern __shared__ float mu[];  // [d] this cluster's means
+  const int c = blockIdx.x;
+  const int dp = d + 1;
+  const int pp = dp * (dp + 1) / 2;
+  const int p = d * (d + 1) / 2;
+  const float* row = packed + (int64_t)c * pp;
+  const float n_c = row[pp - 1];
+  const bool ge1 = n_c >= 1.0f;
+  const bool gt05 = n_c > 0.5f;
+
+  if (threadIdx.x == 0) {
+    n_out[c] = n_c;
+    // pi = N/sum(N) with the 1e-10 floor (compute_pi semantics)
+    float total = 0.0f;
+    for (int cc = 0; cc < k; ++cc) total += packed[(int64_t)cc * pp + pp - 1];
+    pi[c] = (n_c < 0.5f) ? 1e-10f : n_c / total;
+  }
+  for (int i = threadIdx.x; i < d; i += NT) {
+    const float m = gt05 ? row[p + i] / n_c : 0.0f;
+    mu[i] = m;
+    means[(int64_t)c * d + i] = m;
+  }
+  __syncthreads();
+  const float reg = world * avgvar[c];
+  for (int t = threadIdx.x; t < p; t += NT) {
+    int i, j;
+    tri_row_col(t, &i, &j);
+    float cov = ge1 ? row[t] - n_c * mu[i] * mu[j] : 0.0f;
+    if (diag_only && i != j) cov = 0.0f;
+    if (i == j) cov += reg;
+    float rv;
+    if (gt05) rv = cov / n_c;
+    else rv = (i == j) ? 1.0f : 0.0f;
+    r_out[((int64_t)c * d + i) * d + j] = rv;
+    if (i != j) r_out[((int64_t)c * d + j) * d + i] = rv;
+  }
+}
+
 // Emit the fused-E-step factor M = [U | -U mu] (U^T U = Rinv, upper
 // Cholesky) as bf16 hi/lo pairs into mfac[c][2][32][32], stored in MFMA
 // A-fragment k-order (mfma_b16_k). `u` is scratch LDS holding Rinv [d*d]
@@ -366,8 +417,10 @@ __device__ inline void emit_mfac(float* u, const float* __restrict__ means,
 
 __global__ void __launch_bounds__(NT)
 constants_lu_kernel(const float* __restrict__ r,
-                    const float* __restrict__ means, float* __restrict__ rinv,
+                    const float* __restrict__ means,
+                    const float* __restrict__ pi, float* __restrict__ rinv,
                     float* __restrict__ logdet,
+                    float* __restrict__ constant, float* __restrict__ add,
                     __hip_bfloat16* __restrict__ mfac,
                     float* __restrict__ mfac32, int d) {
   // lds: a[d*d] working buffer, o[d*d] read-only snapshot of the LU factor.
@@ -387,7 +440,11 @@ constants_lu_kernel(const float* __restrict__ r,
 
   if (d == 1) {
     if (tid == 0) {
-      logdet[c] = __logf(a[0]);
+      const float ld = __logf(a[0]);
+      logdet[c] = ld;
+      const float cst = -0.5f * 1.8378770664093453f - 0.5f * ld;  // ln(2pi)
+      if (constant) constant[c] = cst;
+      if (add) add[c] = cst + __logf(pi[c]);
       oc[0] = 1.0f / a[0];
       o[0] = oc[0];
     }
@@ -433,6 +490,10 @@ constants_lu_kernel(const float* __restrict__ r,
       float total = 0.0f;
       for (int wv = 0; wv < nw; ++wv) total += wsum[wv];
       logdet[c] = total;
+      // constant = -D/2 ln(2pi) - 0.5 ln|R| (gaussian_kernel.cu:241)
+      const float cst = -d * 0.5f * 1.8378770664093453f - 0.5f * total;
+      if (constant) constant[c] = cst;
+      if (add) add[c] = cst + __logf(pi[c]);
     }
   }
   __syncthreads();

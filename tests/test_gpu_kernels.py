@@ -470,3 +470,38 @@ def test_estep_big_small_d_large_k(device):
                          pi.cpu())
     np.testing.assert_allclose(out.cpu().numpy(), ref.numpy(),
                                rtol=5e-2, atol=1.0)
+
+
+def test_mstep_finalize_kernel_matches_cpu(device):
+    """The one-kernel finalize vs the torch reference chain (exact rules:
+    empty-cluster resets, G*avgvar, pi floor)."""
+    from cuda_gmm_mpi_amd.ops import functional as F
+    from cuda_gmm_mpi_amd.ops.backend import hip_ext
+    rng = np.random.default_rng(33)
+    k, d, n = 6, 24, 5000
+    x = rng.standard_normal((d, n)).astype(np.float32)
+    w = rng.uniform(0, 1, (k, n)).astype(np.float32)
+    w[3] *= 0.0       # empty cluster (N=0 -> identity reset, pi floor)
+    w[4] *= 1.5e-4    # 0.5 < N < 1 edge
+    packed = F.mstep_moments(torch.from_numpy(x).to(device),
+                             torch.from_numpy(w).to(device))
+    avgvar = torch.full((k,), 0.37, device=device)
+    n_o = torch.empty(k, device=device)
+    mu_o = torch.empty(k, d, device=device)
+    r_o = torch.empty(k, d, d, device=device)
+    pi_o = torch.empty(k, device=device)
+    hip_ext().mstep_finalize(packed, avgvar, 3, n_o, mu_o, r_o, pi_o, False)
+    # reference chain
+    n_c, mean_num, s = F.moments_views(packed, d)
+    ref_mu = cpu.finalize_means(n_c, mean_num)
+    ref_r = cpu.finalize_covariance(n_c, ref_mu, s, avgvar, world_size=3)
+    ref_pi = cpu.compute_pi(n_c)
+    np.testing.assert_allclose(n_o.cpu().numpy(), n_c.cpu().numpy())
+    np.testing.assert_allclose(mu_o.cpu().numpy(), ref_mu.cpu().numpy(),
+                               rtol=1e-5, atol=1e-6)
+    np.testing.assert_allclose(r_o.cpu().numpy(), ref_r.cpu().numpy(),
+                               rtol=1e-4, atol=1e-5)
+    np.testing.assert_allclose(pi_o.cpu().numpy(), ref_pi.cpu().numpy(),
+                               rtol=1e-6)
+    assert pi_o.cpu().numpy()[3] == 1e-10
+    np.testing.assert_allclose(r_o[3].cpu().numpy(), np.eye(d))
