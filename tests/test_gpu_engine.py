@@ -224,3 +224,19 @@ def test_gpu_k1_and_d1():
     eng = build_engine(data, cfg, device="cuda")
     res = eng.sweep()
     assert res.num_clusters == 1
+
+
+def test_gpu_max_clusters_512():
+    """MAX_CLUSTERS=512 (reference bound) runs through the large-K paths
+    (big-D estep tier + separate posteriors)."""
+    data, _ = make_blobs(60000, 16, 32, seed=111)
+    cfg = GmmConfig(num_clusters=512, target_num_clusters=512,
+                    min_iters=2, max_iters=2, estep_dtype="bf16",
+                    mstep_precision="bf16x3")
+    eng = build_engine(data, cfg, device="cuda")
+    assert not eng.use_fused_estep and eng.use_big_estep
+    lik = eng.run_em(512)
+    assert np.isfinite(lik)
+    s = eng.w[:512].sum(dim=0)
+    assert float((s - 1).abs().max()) < 1e-2
+    assert abs(float(eng.state.pi.sum()) - 1.0) < 1e-2
