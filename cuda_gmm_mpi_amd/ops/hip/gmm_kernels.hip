@@ -1047,24 +1047,46 @@ estep_fused_kernel(const __hip_bfloat16* __restrict__ z,
   }
   __syncthreads();
 
-  // pass 2: posteriors + likelihood; one exp per (cluster, event): the
-  // normalize pass rescales the already-computed exp(lw - m) by
-  // exp(m - denom) instead of re-exponentiating
+  // pass 2: posteriors + likelihood, cluster loop split across BOTH
+  // thread halves (the serial 64-deep store loop was 24% of the kernel —
+  // store-issue-bound): threads t and t+EST_BE each handle half the
+  // clusters of event t, combining max/sum through LDS.
+  __shared__ float pmax[2][EST_BE];
+  __shared__ float psum[2][EST_BE];
   float acc = 0.0f;
-  if (threadIdx.x < EST_BE && threadIdx.x < cnt) {
-    const int t = threadIdx.x;
-    float m = lw[t];
-    for (int c = 1; c < k; ++c) m = fmaxf(m, lw[c * lrow + t]);
-    float s = 0.0f;
-    for (int c = 0; c < k; ++c) {
-      const float e = __expf(lw[c * lrow + t] - m);
-      lw[c * lrow + t] = e;
-      s += e;
+  {
+    const int t = threadIdx.x & (EST_BE - 1);
+    const int half = threadIdx.x >> 7;       // EST_BE == 128
+    const int mid = (k + 1) / 2;   // half 0 never empty (K=1, odd K)
+    const int c_lo = half * mid;
+    const int c_hi = half ? k : mid;
+    if (t < cnt && c_lo < c_hi) {
+      float m = lw[c_lo * lrow + t];
+      for (int c = c_lo + 1; c < c_hi; ++c)
+        m = fmaxf(m, lw[c * lrow + t]);
+      pmax[half][t] = m;
+    } else if (t < EST_BE) {
+      pmax[half][t] = -3.0e38f;
     }
-    const float inv = 1.0f / s;
-    for (int c = 0; c < k; ++c)
-      w_out[(int64_t)c * n + e0 + t] = lw[c * lrow + t] * inv;
-    acc = m + __logf(s);
+    __syncthreads();
+    const float m = fmaxf(pmax[0][t], pmax[1][t]);
+    float s = 0.0f;
+    if (t < cnt) {
+      for (int c = c_lo; c < c_hi; ++c) {
+        const float e = __expf(lw[c * lrow + t] - m);
+        lw[c * lrow + t] = e;
+        s += e;
+      }
+    }
+    psum[half][t] = s;
+    __syncthreads();
+    const float total = psum[0][t] + psum[1][t];
+    if (t < cnt && c_lo < c_hi) {
+      const float inv = 1.0f / total;
+      for (int c = c_lo; c < c_hi; ++c)
+        w_out[(int64_t)c * n + e0 + t] = lw[c * lrow + t] * inv;
+      if (half == 0) acc = m + __logf(total);
+    }
   }
   __shared__ float wsum[NT / WAVE];
   for (int off = WAVE / 2; off > 0; off >>= 1)

@@ -505,3 +505,27 @@ def test_mstep_finalize_kernel_matches_cpu(device):
                                rtol=1e-6)
     assert pi_o.cpu().numpy()[3] == 1e-10
     np.testing.assert_allclose(r_o[3].cpu().numpy(), np.eye(d))
+
+
+@pytest.mark.parametrize("k", [1, 3, 64, 104])
+def test_estep_fused_odd_k(device, k):
+    """Pass-2 half-split must cover odd K and K=1 (likelihood non-zero)."""
+    from cuda_gmm_mpi_amd.ops import functional as F
+    rng = np.random.default_rng(k)
+    d, n = 8, 1000
+    means, r, pi = random_model(rng, k, d, device)
+    mfac = torch.empty(k, *F.mfac_shape(d), dtype=torch.bfloat16,
+                       device=device)
+    rinv_g, const_g = F.constants(r, means, False, mfac)
+    add = const_g + torch.log(pi)
+    x = rng.standard_normal((d, n)).astype(np.float32)
+    w_out = torch.empty(k, n, dtype=torch.float32, device=device)
+    w, lik = F.estep_fused(
+        torch.from_numpy(x).to(device).to(torch.bfloat16), mfac, add, w_out)
+    rinv, const = cpu.compute_constants(r.cpu())
+    ref_logw = cpu.estep_logw(torch.from_numpy(x), means.cpu(), rinv, const,
+                              pi.cpu())
+    ref_w, ref_lik = cpu.estep_posteriors(ref_logw)
+    assert float(lik) == pytest.approx(float(ref_lik), rel=1e-2)
+    np.testing.assert_allclose(w.sum(dim=0).cpu().numpy(), np.ones(n),
+                               rtol=1e-3)
