@@ -253,7 +253,11 @@ def mstep_moments(x: torch.Tensor, w: torch.Tensor,
     if x.is_cuda and precision == "bf16x3" and d <= 159:
         tiles = (n + 63) // 64  # MBB_BK
         if nchunk is None:
-            nchunk = max(1, min(256, (64 << 20) // (4 * k * pp)))
+            # big-D packed rows are large (Pp ~ 8k at D=128): a small byte
+            # cap starves chunk-parallelism (7 chunks = 448 blocks at
+            # K=256). The partial buffer is cheap in 288 GB and its sum is
+            # ~90 us/GB; keep >= 8 XCDs x 32 CUs of blocks instead.
+            nchunk = max(1, min(256, (512 << 20) // (4 * k * pp)))
         nchunk = int(min(nchunk, tiles))
         partials = torch.empty((nchunk, k, pp), dtype=torch.float32,
                                device=x.device)
