@@ -108,3 +108,16 @@ def test_cli_bad_args_exit_code():
     # reference prints usage and returns 1 (gaussian.cu:1162-1165)
     assert main([]) == 1
     assert main(["notanumber", "x", "y"]) == 1
+
+
+def test_cli_no_bug_compat_and_verbose(tmp_path, csv_file, capsys):
+    """--no-bug-compat (corrected merge constant) + --verbose progress."""
+    path, _ = csv_file
+    out = str(tmp_path / "fix")
+    rc = main(["5", path, out, "2", "--min-iters", "3", "--max-iters", "3",
+               "--device", "cpu", "--no-results", "--no-bug-compat",
+               "--verbose"])
+    assert rc == 0
+    assert open(out + ".summary").read().count("Cluster #") == 2
+    logged = capsys.readouterr().out
+    assert "clusters" in logged or "likelihood" in logged.lower()
