@@ -155,3 +155,63 @@ def test_world4_sweep_matches_single():
     assert multi["rissanen"] == pytest.approx(single["rissanen"], rel=1e-3)
     np.testing.assert_allclose(multi["means"], single["means"],
                                rtol=2e-3, atol=2e-3)
+
+
+def _ck_cfg(**kw):
+    from cuda_gmm_mpi_amd.utils.config import GmmConfig
+    return GmmConfig(num_clusters=6, min_iters=3, max_iters=3,
+                     covariance_dynamic_range=1e15, **kw)
+
+
+def _resume_data():
+    data, _ = make_blobs(1801, 2, 3, seed=41)
+    return data
+
+
+def _fit_ck_phase1():
+    """World-1 sweep 6 -> 4, leaving checkpoints behind."""
+    from cuda_gmm_mpi_amd.engine import build_engine
+    cfg = _ck_cfg(target_num_clusters=4,
+                  checkpoint_dir=os.environ["GMM_TEST_CKDIR"])
+    eng = build_engine(_resume_data(), cfg, device="cpu")
+    eng.sweep()
+    return True
+
+
+def _fit_ck_resume():
+    """Resume the phase-1 checkpoint down to 2 (any world size)."""
+    from cuda_gmm_mpi_amd.engine import build_engine
+    cfg = _ck_cfg(target_num_clusters=2,
+                  checkpoint_dir=os.environ["GMM_TEST_CKDIR"])
+    eng = build_engine(_resume_data(), cfg, device="cpu")
+    res = eng.sweep()
+    return {"k": res.num_clusters, "rissanen": res.min_rissanen,
+            "means": res.state.means.numpy().copy()}
+
+
+def _fit_ck_direct():
+    from cuda_gmm_mpi_amd.engine import build_engine
+    eng = build_engine(_resume_data(), _ck_cfg(target_num_clusters=2),
+                       device="cpu")
+    res = eng.sweep()
+    return {"k": res.num_clusters, "rissanen": res.min_rissanen,
+            "means": res.state.means.numpy().copy()}
+
+
+@pytest.mark.timeout(300)
+def test_world2_resume_from_world1_checkpoint(tmp_path):
+    """Checkpoints hold model parameters, not shards, so a sweep
+    checkpointed at world 1 resumes under world 2 (re-sharded data) and
+    matches a direct single-process run (avgvar ridge neutralized — it is
+    world-size dependent by reference design, SURVEY 2.6 #5)."""
+    os.environ["GMM_TEST_CKDIR"] = str(tmp_path / "ck")
+    try:
+        _run_single("_fit_ck_phase1")
+        multi = run_world(2, "_fit_ck_resume", port=29814)
+        single = _run_single("_fit_ck_direct")
+    finally:
+        os.environ.pop("GMM_TEST_CKDIR", None)
+    assert multi["k"] == single["k"]
+    assert multi["rissanen"] == pytest.approx(single["rissanen"], rel=1e-3)
+    np.testing.assert_allclose(multi["means"], single["means"],
+                               rtol=1e-3, atol=1e-3)
