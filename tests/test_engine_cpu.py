@@ -295,3 +295,20 @@ def test_profile_report_shape():
         assert token in rep
     # per-iteration averages present (count column == iterations run)
     assert "\t3\t" in rep or "\t2\t" in rep
+
+
+def test_em_likelihood_monotone():
+    """EM ascent property: total log-likelihood is
+    non-decreasing across iterations on well-conditioned data, up to
+    fp32 reduction noise."""
+    data, _ = make_blobs(5000, 4, 3, seed=7)
+    cfg = GmmConfig(num_clusters=3, target_num_clusters=3,
+                    min_iters=1, max_iters=1)
+    eng = build_engine(data, cfg, device="cpu")
+    eng.run_em(3)
+    liks = [eng.likelihood]
+    for _ in range(25):
+        eng.em_iteration(3)
+        liks.append(float(eng._lik_dev.item()))
+    diffs = np.diff(np.array(liks))
+    assert np.all(diffs >= -1e-6 * abs(liks[-1])), diffs
