@@ -257,3 +257,20 @@ def test_gpu_diag_only_engine():
     for c in range(3):
         off = r[c] - np.diag(np.diag(r[c]))
         assert np.abs(off).max() == 0.0
+
+
+def test_gpu_em_likelihood_monotone_fp32():
+    """EM ascent on the exact-fp32 GPU path (fused f32 MFMA E-step +
+    fp32 moments): likelihood non-decreasing up to reduction noise."""
+    data, _ = make_blobs(20000, 6, 4, seed=9)
+    cfg = GmmConfig(num_clusters=4, target_num_clusters=4,
+                    min_iters=1, max_iters=1,
+                    estep_dtype="fp32", mstep_precision="fp32")
+    eng = build_engine(data, cfg, device="cuda")
+    eng.run_em(4)
+    liks = [eng.likelihood]
+    for _ in range(25):
+        eng.em_iteration(4)
+        liks.append(float(eng._lik_dev.item()))
+    diffs = np.diff(np.array(liks))
+    assert np.all(diffs >= -1e-6 * abs(liks[-1])), diffs
