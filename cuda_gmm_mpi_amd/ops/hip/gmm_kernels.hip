@@ -380,6 +380,7 @@ __device__ inline void emit_mfac(float* u, const float* __restrict__ means,
     const float piv = u[j * d + j];
     for (int i = j + 1 + tid; i < d; i += blockDim.x) {
       float s = u[j * d + i];
+#pragma unroll 8
       for (int kk = 0; kk < j; ++kk) s -= u[kk * d + j] * u[kk * d + i];
       u[j * d + i] = s / piv;
     }
@@ -388,6 +389,7 @@ __device__ inline void emit_mfac(float* u, const float* __restrict__ means,
   // u0 = -U mu (mu = centered cluster means)
   for (int i = tid; i < d; i += blockDim.x) {
     float s = 0.0f;
+#pragma unroll 8
     for (int j = i; j < d; ++j) s += u[i * d + j] * means[c * d + j];
     u0[i] = -s;
   }
@@ -461,6 +463,7 @@ constants_lu_kernel(const float* __restrict__ r,
     // column i of L: rows j >= i in parallel
     for (int j = i + tid; j < d; j += blockDim.x) {
       float s = 0.0f;
+#pragma unroll 8
       for (int kk = 0; kk < i; ++kk) s = fmaf(a[j * d + kk], a[kk * d + i], s);
       a[j * d + i] -= s;
     }
@@ -470,6 +473,7 @@ constants_lu_kernel(const float* __restrict__ r,
     const float pivot = a[i * d + i];
     for (int j = i + 1 + tid; j < d; j += blockDim.x) {
       float s = 0.0f;
+#pragma unroll 8
       for (int kk = 0; kk < i; ++kk) s = fmaf(a[i * d + kk], a[kk * d + j], s);
       a[i * d + j] = (a[i * d + j] - s) / pivot;
     }
@@ -510,6 +514,7 @@ constants_lu_kernel(const float* __restrict__ r,
       float xv = 1.0f;
       if (i != j) {
         xv = 0.0f;
+#pragma unroll 8
         for (int kk = i; kk < j; ++kk)
           xv -= o[j * d + kk] * a[kk * d + i];
       }
@@ -523,6 +528,7 @@ constants_lu_kernel(const float* __restrict__ r,
   for (int i = tid; i < d; i += blockDim.x) {
     for (int j = i + 1; j < d; ++j) {
       float s = 0.0f;
+#pragma unroll 8
       for (int kk = i; kk < j; ++kk)
         s += o[kk * d + j] * ((i == kk) ? 1.0f : a[i * d + kk]);
       a[i * d + j] = -s;
@@ -536,6 +542,7 @@ constants_lu_kernel(const float* __restrict__ r,
   for (int t = tid; t < d * d; t += blockDim.x) {
     const int j = t / d, i = t % d;
     float s = 0.0f;
+#pragma unroll 8
     for (int kk = (i > j ? i : j); kk < d; ++kk)
       s = fmaf((j == kk) ? 1.0f : a[j * d + kk], a[kk * d + i], s);
     oc[j * d + i] = s;
