@@ -1069,6 +1069,7 @@ estep_fused_kernel(const __hip_bfloat16* __restrict__ z,
     const int c_hi = half ? k : mid;
     if (t < cnt && c_lo < c_hi) {
       float m = lw[c_lo * lrow + t];
+#pragma unroll 4
       for (int c = c_lo + 1; c < c_hi; ++c)
         m = fmaxf(m, lw[c * lrow + t]);
       pmax[half][t] = m;
@@ -1079,6 +1080,7 @@ estep_fused_kernel(const __hip_bfloat16* __restrict__ z,
     const float m = fmaxf(pmax[0][t], pmax[1][t]);
     float s = 0.0f;
     if (t < cnt) {
+#pragma unroll 4
       for (int c = c_lo; c < c_hi; ++c) {
         const float e = __expf(lw[c * lrow + t] - m);
         lw[c * lrow + t] = e;
@@ -1090,6 +1092,7 @@ estep_fused_kernel(const __hip_bfloat16* __restrict__ z,
     const float total = psum[0][t] + psum[1][t];
     if (t < cnt && c_lo < c_hi) {
       const float inv = 1.0f / total;
+#pragma unroll 4
       for (int c = c_lo; c < c_hi; ++c)
         w_out[(int64_t)c * n + e0 + t] = lw[c * lrow + t] * inv;
       if (half == 0) acc = m + __logf(total);
