@@ -192,10 +192,13 @@ estep_posteriors_kernel(float* __restrict__ logw, float* __restrict__ partial,
   for (int64_t e = (int64_t)blockIdx.x * NT + threadIdx.x; e < n;
        e += (int64_t)gridDim.x * NT) {
     float m = logw[e];
+#pragma unroll 4
     for (int c = 1; c < k; ++c) m = fmaxf(m, logw[(int64_t)c * n + e]);
     float s = 0.0f;
+#pragma unroll 4
     for (int c = 0; c < k; ++c) s += __expf(logw[(int64_t)c * n + e] - m);
     const float denom = m + __logf(s);
+#pragma unroll 4
     for (int c = 0; c < k; ++c) {
       const int64_t idx = (int64_t)c * n + e;
       logw[idx] = __expf(logw[idx] - denom);
@@ -1198,14 +1201,17 @@ estep_fused_f32_kernel(const float* __restrict__ z,
   if (threadIdx.x < EST_BE && threadIdx.x < cnt) {
     const int t = threadIdx.x;
     float m = lw[t];
+#pragma unroll 4
     for (int c = 1; c < k; ++c) m = fmaxf(m, lw[c * lrow + t]);
     float s = 0.0f;
+#pragma unroll 4
     for (int c = 0; c < k; ++c) {
       const float e = __expf(lw[c * lrow + t] - m);
       lw[c * lrow + t] = e;
       s += e;
     }
     const float inv = 1.0f / s;
+#pragma unroll 4
     for (int c = 0; c < k; ++c)
       w_out[(int64_t)c * n + e0 + t] = lw[c * lrow + t] * inv;
     acc = m + __logf(s);
