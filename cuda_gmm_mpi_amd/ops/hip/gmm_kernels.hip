@@ -138,6 +138,7 @@ estep_logw_gen_kernel(const T* __restrict__ x, const float* __restrict__ means,
     for (int i = 0; i < d; ++i) {
       const float dxi = load_x(x, (int64_t)i * n + e) - lds_means[i];
       float qi = 0.0f;
+#pragma unroll 4
       for (int j = 0; j < i; ++j) {
         const float dxj = load_x(x, (int64_t)j * n + e) - lds_means[j];
         qi = fmaf(lds_rp[t + j], dxj, qi);  // pre-summed (R_ij + R_ji)
@@ -287,6 +288,7 @@ mstep_cov_kernel(const T* __restrict__ x, const float* __restrict__ w,
       const float4* xj = (const float4*)(xs + pc[u] * row);
       const float4* wv = (const float4*)wt;
       float a = acc[u];
+#pragma unroll 2
       for (int e4 = 0; e4 < te / 4; ++e4) {
         const float4 vi = xi[e4], vj = xj[e4], vw = wv[e4];
         a = fmaf(vw.x * vi.x, vj.x, a);
