@@ -70,6 +70,12 @@ def make_parser() -> argparse.ArgumentParser:
     p.add_argument("--gpus", type=int, default=None,
                    help="spawn N single-GPU worker processes on this node "
                         "(one rank per GPU over RCCL)")
+    p.add_argument("--scatter-input", action="store_true",
+                   help="only rank 0 reads the input file; event shards "
+                        "are scattered point-to-point and seeding stats "
+                        "broadcast (no shared filesystem needed; the "
+                        "reference MPI_Bcasts the whole dataset, "
+                        "gaussian.cu:193-200)")
     return p
 
 
@@ -91,16 +97,31 @@ def config_from_args(args) -> GmmConfig:
     return cfg
 
 
-def run_clustering(data: np.ndarray, cfg: GmmConfig, outfile: str,
+def run_clustering(data: np.ndarray | None, cfg: GmmConfig, outfile: str,
                    device: str, write_results: bool = True,
-                   profile_report: bool = False) -> dict:
+                   profile_report: bool = False,
+                   scatter_input: bool = False) -> dict:
     """Full pipeline on already-initialized process group. Returns a result
-    dict (rank 0) with num_clusters / rissanen / likelihood."""
+    dict (rank 0) with num_clusters / rissanen / likelihood.
+
+    With scatter_input, only rank 0 passes data (others None): shards are
+    scattered and seeding stats broadcast (parallel.dist.distribute_input).
+    """
     rank, local_rank, world = pdist.rank(), 0, pdist.world_size()
     if device == "cuda":
         local_rank = torch.cuda.current_device()
     prof = Profile(device)
-    engine = build_engine(data, cfg, device=device, profile=prof)
+    if scatter_input:
+        from .engine import build_engine_sharded
+        full = (torch.from_numpy(np.ascontiguousarray(data, np.float32))
+                if rank == 0 else None)
+        shard, n_tot, mean, var, seed_means = pdist.distribute_input(
+            full, cfg.num_clusters)
+        engine = build_engine_sharded(shard, cfg, n_tot, mean, var,
+                                      seed_means, device=device,
+                                      profile=prof)
+    else:
+        engine = build_engine(data, cfg, device=device, profile=prof)
     if profile_report:
         # graph replay bypasses the per-bucket hipEvent timers
         engine.use_graphs = False
@@ -169,14 +190,27 @@ def main(argv=None) -> int:
         print(f"Rank {rank} of {world} on {socket.gethostname()} "
               f"using {dev_name}")
     try:
-        try:
-            data = gio.read_data(args.infile)
-        except (OSError, ValueError) as e:
-            print(f"Invalid infile. ({e})\n")
-            return 2
+        scatter = args.scatter_input and world > 1
+        data = None
+        read_err = 0
+        if not scatter or rank == 0:
+            try:
+                data = gio.read_data(args.infile)
+            except (OSError, ValueError) as e:
+                print(f"Invalid infile. ({e})\n")
+                if not scatter:
+                    return 2
+                read_err = 1
+        if scatter:
+            # rank 0's read outcome gates everyone (no hang on bad input)
+            flag = torch.tensor([read_err], dtype=torch.long)
+            pdist.broadcast_(flag)
+            if int(flag.item()):
+                return 2
         result = run_clustering(
             data, cfg, args.outfile, device,
             write_results=args.write_results, profile_report=args.profile,
+            scatter_input=scatter,
         )
         if rank == 0 and args.metrics_out:
             import json

@@ -523,3 +523,21 @@ def build_engine(data_by_event: np.ndarray | torch.Tensor, config: GmmConfig,
         data[s:e], config, n, center, seed_means, var.float(),
         device=device, profile=profile,
     )
+
+
+def build_engine_sharded(shard: torch.Tensor, config: GmmConfig,
+                         n_total: int, mean: torch.Tensor,
+                         var: torch.Tensor, seed_means: torch.Tensor,
+                         device: torch.device | str = "cpu",
+                         profile: Profile | None = None) -> EmEngine:
+    """Build an engine from a pre-distributed shard (rank-0-read input:
+    parallel.dist.distribute_input). `mean`/`var`/`seed_means` are the
+    full-data statistics broadcast from rank 0, so results are identical
+    to the shared-filesystem `build_engine` path."""
+    config.validate()
+    d = shard.shape[1]
+    center = mean.float() if config.center_data else torch.zeros(d)
+    return EmEngine(
+        shard.to(torch.float32), config, n_total, center, seed_means,
+        var.float(), device=device, profile=profile,
+    )

@@ -94,3 +94,60 @@ def shard_bounds(num_events: int, world: int, r: int) -> tuple[int, int]:
 def destroy() -> None:
     if is_dist():
         dist.destroy_process_group()
+
+
+def distribute_input(data: torch.Tensor | None, num_clusters: int):
+    """Rank-0-read input distribution (reference: MPI_Bcast of the whole
+    dataset, gaussian.cu:193-200 — here each rank receives only its SHARD
+    via point-to-point sends, sized for N >> full-replication).
+
+    Rank 0 passes the full [N, D] float32 tensor; other ranks pass None.
+    Returns (shard, n_total, mean, var, seed_means): the per-rank event
+    shard plus the global seeding statistics, computed on rank 0 from the
+    full data with exactly the same code as the shared-filesystem path
+    (engine.build_engine) so both paths produce identical results.
+    """
+    from ..models.seed import seed_means_host
+
+    w, r = world_size(), rank()
+    if w == 1:
+        n, d = data.shape
+        mean = data.double().mean(dim=0).float()
+        var = (data.double().pow(2).mean(dim=0)
+               - data.double().mean(dim=0).pow(2)).float()
+        return data, n, mean, var, seed_means_host(data, num_clusters)
+
+    if r == 0:
+        if data is None or data.dim() != 2:
+            raise ValueError("rank 0 must provide the full [N, D] data")
+        shape = torch.tensor(list(data.shape), dtype=torch.long)
+    else:
+        shape = torch.zeros(2, dtype=torch.long)
+    broadcast_(shape)
+    n, d = int(shape[0]), int(shape[1])
+
+    k = num_clusters
+    stats = torch.zeros(2 * d + k * d, dtype=torch.float32)
+    if r == 0:
+        dmean = data.double().mean(dim=0)
+        dvar = data.double().pow(2).mean(dim=0) - dmean * dmean
+        stats[:d] = dmean.float()
+        stats[d:2 * d] = dvar.float()
+        stats[2 * d:] = seed_means_host(data, k).reshape(-1)
+    broadcast_(stats)
+    mean, var = stats[:d], stats[d:2 * d]
+    seed_means = stats[2 * d:].reshape(k, d)
+
+    s, e = shard_bounds(n, w, r)
+    if r == 0:
+        shard = data[s:e].contiguous()
+        reqs = []
+        for dst in range(1, w):
+            ds_, de_ = shard_bounds(n, w, dst)
+            reqs.append(dist.isend(data[ds_:de_].contiguous(), dst=dst))
+        for rq in reqs:
+            rq.wait()
+    else:
+        shard = torch.empty(e - s, d, dtype=torch.float32)
+        dist.recv(shard, src=0)
+    return shard, n, mean, var, seed_means

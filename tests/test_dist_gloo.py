@@ -28,7 +28,8 @@ def _worker(rank, world, port, fn_name, out_q):
         if rank == 0:
             out_q.put(result)
     finally:
-        dist.destroy_process_group()
+        if dist.is_initialized():  # CLI fns destroy the group themselves
+            dist.destroy_process_group()
 
 
 def run_world(world, fn_name, port):
@@ -215,3 +216,68 @@ def test_world2_resume_from_world1_checkpoint(tmp_path):
     assert multi["rissanen"] == pytest.approx(single["rissanen"], rel=1e-3)
     np.testing.assert_allclose(multi["means"], single["means"],
                                rtol=1e-3, atol=1e-3)
+
+
+def _cli_scatter(flag=True):
+    """Run the CLI with/without --scatter-input; return the .summary text."""
+    import tempfile
+    from cuda_gmm_mpi_amd.cli import main
+    from cuda_gmm_mpi_amd.utils import io as gio
+    work = os.environ["GMM_TEST_SCATTER_DIR"]
+    binpath = os.path.join(work, "data.bin")
+    out = os.path.join(work, f"out_{'sc' if flag else 'fs'}_{os.environ.get('RANK', '0')}")
+    argv = ["4", binpath, out, "3", "--min-iters", "4", "--max-iters", "4",
+            "--device", "cpu", "--no-results"]
+    if flag:
+        argv.append("--scatter-input")
+    rc = main(argv)
+    assert rc == 0
+    if os.environ.get("RANK", "0") == "0":
+        return open(out + ".summary").read()
+    return None
+
+
+def _cli_scatter_on():
+    return _cli_scatter(True)
+
+
+def _cli_scatter_off():
+    return _cli_scatter(False)
+
+
+def _cli_scatter_missing():
+    """Bad input path under --scatter-input must fail on ALL ranks, not
+    hang the non-root ranks in distribute_input."""
+    from cuda_gmm_mpi_amd.cli import main
+    work = os.environ["GMM_TEST_SCATTER_DIR"]
+    rc = main(["3", os.path.join(work, "nope.bin"),
+               os.path.join(work, "x"), "--device", "cpu",
+               "--scatter-input"])
+    return rc
+
+
+@pytest.mark.timeout(300)
+def test_scatter_input_matches_shared_fs(tmp_path):
+    """--scatter-input (rank-0 read + shard scatter + stats broadcast)
+    produces byte-identical output to the every-rank-reads path."""
+    from cuda_gmm_mpi_amd.utils import io as gio
+    data, _ = make_blobs(1507, 3, 4, seed=53)
+    os.environ["GMM_TEST_SCATTER_DIR"] = str(tmp_path)
+    try:
+        gio.write_bin(str(tmp_path / "data.bin"), data)
+        sc = run_world(2, "_cli_scatter_on", port=29815)
+        fs = run_world(2, "_cli_scatter_off", port=29816)
+    finally:
+        os.environ.pop("GMM_TEST_SCATTER_DIR", None)
+    assert sc is not None and sc == fs
+    assert sc.count("Cluster #") == 3
+
+
+@pytest.mark.timeout(300)
+def test_scatter_input_missing_file_no_hang(tmp_path):
+    os.environ["GMM_TEST_SCATTER_DIR"] = str(tmp_path)
+    try:
+        rc = run_world(2, "_cli_scatter_missing", port=29817)
+    finally:
+        os.environ.pop("GMM_TEST_SCATTER_DIR", None)
+    assert rc == 2
