@@ -70,6 +70,11 @@ def make_parser() -> argparse.ArgumentParser:
     p.add_argument("--gpus", type=int, default=None,
                    help="spawn N single-GPU worker processes on this node "
                         "(one rank per GPU over RCCL)")
+    p.add_argument("--striped-results", action="store_true",
+                   help="each rank writes its own contiguous "
+                        "<out>.results.<rank> stripe (no gather; "
+                        "concatenating stripes in rank order equals the "
+                        "single gathered file)")
     p.add_argument("--scatter-input", action="store_true",
                    help="only rank 0 reads the input file; event shards "
                         "are scattered point-to-point and seeding stats "
@@ -100,7 +105,8 @@ def config_from_args(args) -> GmmConfig:
 def run_clustering(data: np.ndarray | None, cfg: GmmConfig, outfile: str,
                    device: str, write_results: bool = True,
                    profile_report: bool = False,
-                   scatter_input: bool = False) -> dict:
+                   scatter_input: bool = False,
+                   striped_results: bool = False) -> dict:
     """Full pipeline on already-initialized process group. Returns a result
     dict (rank 0) with num_clusters / rissanen / likelihood.
 
@@ -111,6 +117,7 @@ def run_clustering(data: np.ndarray | None, cfg: GmmConfig, outfile: str,
     if device == "cuda":
         local_rank = torch.cuda.current_device()
     prof = Profile(device)
+    shard = None
     if scatter_input:
         from .engine import build_engine_sharded
         full = (torch.from_numpy(np.ascontiguousarray(data, np.float32))
@@ -132,12 +139,28 @@ def run_clustering(data: np.ndarray | None, cfg: GmmConfig, outfile: str,
     out_state.means += engine.center.cpu().unsqueeze(0)
 
     w_shard = engine.recompute_memberships(result.state)
-    memberships = engine.gather_memberships(w_shard)
-
-    if rank == 0:
-        gio.write_summary(outfile + ".summary", out_state, cfg.enable_output)
-        if cfg.enable_output and write_results and memberships is not None:
-            gio.write_results(outfile + ".results", data, memberships)
+    memberships = None
+    if striped_results:
+        # SURVEY §5 long-context plan: per-rank file stripes — shard-local
+        # posteriors and data rows, no gather. Stripes are contiguous in
+        # rank order, each ends with a newline, so
+        # `cat out.results.0 out.results.1 ...` == the gathered file.
+        if cfg.enable_output and write_results:
+            s0, e0 = pdist.shard_bounds(engine.n_total, world, rank)
+            local = (shard.numpy() if shard is not None
+                     else np.asarray(data[s0:e0], dtype=np.float32))
+            gio.write_results(f"{outfile}.results.{rank}", local,
+                              w_shard.cpu().numpy())
+        if rank == 0:
+            gio.write_summary(outfile + ".summary", out_state,
+                              cfg.enable_output)
+    else:
+        memberships = engine.gather_memberships(w_shard)
+        if rank == 0:
+            gio.write_summary(outfile + ".summary", out_state,
+                              cfg.enable_output)
+            if cfg.enable_output and write_results and memberships is not None:
+                gio.write_results(outfile + ".results", data, memberships)
 
     if profile_report:
         print(engine.profile.report(rank, local_rank))
@@ -210,7 +233,7 @@ def main(argv=None) -> int:
         result = run_clustering(
             data, cfg, args.outfile, device,
             write_results=args.write_results, profile_report=args.profile,
-            scatter_input=scatter,
+            scatter_input=scatter, striped_results=args.striped_results,
         )
         if rank == 0 and args.metrics_out:
             import json

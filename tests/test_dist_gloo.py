@@ -281,3 +281,45 @@ def test_scatter_input_missing_file_no_hang(tmp_path):
     finally:
         os.environ.pop("GMM_TEST_SCATTER_DIR", None)
     assert rc == 2
+
+
+def _cli_results(striped):
+    from cuda_gmm_mpi_amd.cli import main
+    work = os.environ["GMM_TEST_SCATTER_DIR"]
+    out = os.path.join(work, "st" if striped else "ga")
+    argv = ["3", os.path.join(work, "data.bin"), out, "3",
+            "--min-iters", "3", "--max-iters", "3", "--device", "cpu"]
+    if striped:
+        argv += ["--striped-results", "--scatter-input"]
+    assert main(argv) == 0
+    return True
+
+
+def _cli_results_striped():
+    return _cli_results(True)
+
+
+def _cli_results_gathered():
+    return _cli_results(False)
+
+
+@pytest.mark.timeout(300)
+def test_striped_results_concat_equals_gathered(tmp_path):
+    """--striped-results: concatenating per-rank stripes in rank order is
+    byte-identical to the single gathered .results file (here combined
+    with --scatter-input: no gather AND no shared-fs read)."""
+    from cuda_gmm_mpi_amd.utils import io as gio
+    data, _ = make_blobs(1207, 3, 3, seed=61)
+    os.environ["GMM_TEST_SCATTER_DIR"] = str(tmp_path)
+    try:
+        gio.write_bin(str(tmp_path / "data.bin"), data)
+        run_world(2, "_cli_results_striped", port=29818)
+        run_world(2, "_cli_results_gathered", port=29819)
+    finally:
+        os.environ.pop("GMM_TEST_SCATTER_DIR", None)
+    stripes = b"".join(
+        open(tmp_path / f"st.results.{r}", "rb").read() for r in range(2))
+    gathered = open(tmp_path / "ga.results", "rb").read()
+    assert stripes == gathered
+    assert (open(tmp_path / "st.summary").read()
+            == open(tmp_path / "ga.summary").read())
