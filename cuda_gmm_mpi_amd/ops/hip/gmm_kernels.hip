@@ -172,6 +172,7 @@ estep_logw_diag_kernel(const T* __restrict__ x, const float* __restrict__ means,
   for (int64_t e = (int64_t)blockIdx.x * NT + threadIdx.x; e < n;
        e += (int64_t)gridDim.x * NT) {
     float q = 0.0f;
+#pragma unroll 4
     for (int i = 0; i < d; ++i) {
       const float dx = load_x(x, (int64_t)i * n + e) - lds_means[i];
       q = fmaf(dx * dx, lds_rd[i], q);
