@@ -68,3 +68,22 @@ def test_csv_roundtrip_values(values, d):
     assert data.shape == (len(rows), d)
     np.testing.assert_allclose(
         data, np.array(rows, dtype=np.float32), rtol=1e-6)
+
+
+@given(n=st.integers(0, 10_000_000), world=st.integers(1, 64))
+@settings(max_examples=200, deadline=None)
+def test_shard_bounds_partition(n, world):
+    """shard_bounds partitions [0, n) exactly: contiguous, ordered, no
+    dropped or double-counted tail events (the reference's remainder bugs,
+    SURVEY 2.6 #4). The split follows the reference scheme — every rank
+    gets floor(n/world) events and the LAST rank absorbs the remainder
+    (gaussian.cu:348-352) — which gather_memberships also assumes."""
+    from cuda_gmm_mpi_amd.parallel.dist import shard_bounds
+    prev_end = 0
+    per = n // world
+    for r in range(world):
+        s, e = shard_bounds(n, world, r)
+        assert s == prev_end and e >= s
+        assert e - s == (per if r < world - 1 else n - per * (world - 1))
+        prev_end = e
+    assert prev_end == n
