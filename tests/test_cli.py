@@ -121,3 +121,15 @@ def test_cli_no_bug_compat_and_verbose(tmp_path, csv_file, capsys):
     assert open(out + ".summary").read().count("Cluster #") == 2
     logged = capsys.readouterr().out
     assert "clusters" in logged or "likelihood" in logged.lower()
+
+
+def test_cli_single_cluster(tmp_path, csv_file):
+    """K=1 end to end: no merging possible, one cluster summary."""
+    path, _ = csv_file
+    out = str(tmp_path / "k1")
+    rc = main(["1", path, out, "1", "--min-iters", "3", "--max-iters", "3",
+               "--device", "cpu", "--no-results"])
+    assert rc == 0
+    summary = open(out + ".summary").read()
+    assert summary.count("Cluster #") == 1
+    assert "Probability: 1.0" in summary
