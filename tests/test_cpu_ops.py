@@ -182,3 +182,26 @@ def test_mstep_moments_packed_layout(rng):
     np.testing.assert_allclose(s.numpy(), rs.numpy(), rtol=1e-3, atol=1e-3)
     # unpacked S symmetric by construction
     assert float((s - s.transpose(1, 2)).abs().max()) == 0.0
+
+
+def test_finalize_covariance_world_dependence_quirk():
+    """the reference's diagonal regularization scales with
+    the number of GPUs G (each GPU adds avgvar to its partial before the
+    global sum), so R(world=G) - R(world=1) == (G-1)*avgvar*I / N. This
+    test pins the quirk so it isn't 'fixed' accidentally."""
+    torch.manual_seed(3)
+    k, d = 3, 4
+    n_c = torch.tensor([50.0, 20.0, 10.0])
+    means = torch.randn(k, d)
+    a = torch.randn(k, d, 2 * d)
+    s = a @ a.transpose(1, 2) + n_c.view(k, 1, 1) * (
+        means.unsqueeze(2) * means.unsqueeze(1))
+    avgvar = torch.tensor([0.5, 1.5, 2.0])
+    r1 = cpu.finalize_covariance(n_c, means, s, avgvar, world_size=1)
+    r4 = cpu.finalize_covariance(n_c, means, s, avgvar, world_size=4)
+    diff = r4 - r1
+    expect = 3.0 * avgvar.view(k, 1) / n_c.view(k, 1)
+    eye = torch.eye(d).bool()
+    np.testing.assert_allclose(
+        diff[:, eye].numpy(), expect.expand(k, d).numpy(), rtol=1e-5)
+    assert torch.all(diff[:, ~eye] == 0)
