@@ -323,3 +323,19 @@ def test_striped_results_concat_equals_gathered(tmp_path):
     assert stripes == gathered
     assert (open(tmp_path / "st.summary").read()
             == open(tmp_path / "ga.summary").read())
+
+
+@pytest.mark.timeout(300)
+def test_scatter_input_world4(tmp_path):
+    """Scatter with 3 concurrent isends + uneven tail shard (N=1205 over
+    4 ranks: 301+301+301+302 per the reference split) matches shared-fs."""
+    from cuda_gmm_mpi_amd.utils import io as gio
+    data, _ = make_blobs(1205, 2, 3, seed=67)
+    os.environ["GMM_TEST_SCATTER_DIR"] = str(tmp_path)
+    try:
+        gio.write_bin(str(tmp_path / "data.bin"), data)
+        sc = run_world(4, "_cli_scatter_on", port=29820)
+        fs = run_world(4, "_cli_scatter_off", port=29821)
+    finally:
+        os.environ.pop("GMM_TEST_SCATTER_DIR", None)
+    assert sc is not None and sc == fs
