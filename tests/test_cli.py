@@ -133,3 +133,16 @@ def test_cli_single_cluster(tmp_path, csv_file):
     summary = open(out + ".summary").read()
     assert summary.count("Cluster #") == 1
     assert "Probability: 1.0" in summary
+
+
+def test_cli_profile_report(tmp_path, csv_file, capsys):
+    """--profile prints the reference-shaped per-GPU timing report
+    (gaussian.cu:967) after the run."""
+    path, _ = csv_file
+    out = str(tmp_path / "prof")
+    rc = main(["2", path, out, "2", "--min-iters", "2", "--max-iters", "2",
+               "--device", "cpu", "--no-results", "--profile"])
+    assert rc == 0
+    text = capsys.readouterr().out
+    assert "E-step Kernel:" in text and "M-step Kernel:" in text
+    assert "Consts Kernel:" in text
