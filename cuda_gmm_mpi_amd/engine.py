@@ -324,6 +324,74 @@ class EmEngine:
 
     # ------------------------------------------------- MDL sweep / merging
 
+    def _bcast_checkpoint(self, ck: dict | None) -> dict | None:
+        """Rank-0-authoritative checkpoint resume. Only rank 0 reads the
+        checkpoint directory — it is the only rank that writes checkpoints,
+        and on a rank-0-only filesystem (--scatter-input deployments) the
+        only rank that can have one. The payload is broadcast as tensors
+        so every rank resumes identically (a per-rank read could diverge:
+        rank 0 resuming at k' while others start at k0 deadlocks the
+        collectives)."""
+        dev = self.device
+        if self.rank == 0:
+            found = int(bool(ck and ck.get("state") is not None))
+            hdr_vals = [
+                found,
+                int(ck["k"]) if found else 0,
+                int(ck["best_k"]) if found else 0,
+                0 if (found and ck.get("best") is not None) else 1,
+                len(ck["rissanen_by_k"]) if found else 0,
+            ]
+            hdr = torch.tensor(hdr_vals, dtype=torch.int64, device=dev)
+        else:
+            hdr = torch.zeros(5, dtype=torch.int64, device=dev)
+        pdist.broadcast_(hdr)
+        found, k, best_k, best_none, nk = (int(x) for x in hdr.cpu())
+        if not found:
+            return None
+        if self.rank == 0:
+            scal = torch.tensor([ck["min_rissanen"], ck["best_lik"]],
+                                dtype=torch.float64, device=dev)
+        else:
+            scal = torch.zeros(2, dtype=torch.float64, device=dev)
+        pdist.broadcast_(scal)
+        keys = torch.zeros(max(nk, 1), dtype=torch.int64, device=dev)
+        vals = torch.zeros(max(nk, 1), dtype=torch.float64, device=dev)
+        if self.rank == 0 and nk:
+            kk = sorted(ck["rissanen_by_k"])
+            keys[:nk] = torch.tensor(kk, dtype=torch.int64)
+            vals[:nk] = torch.tensor([ck["rissanen_by_k"][x] for x in kk],
+                                     dtype=torch.float64)
+        pdist.broadcast_(keys)
+        pdist.broadcast_(vals)
+
+        def bcast_state(st_src, kc):
+            size = 4 * kc + kc * self.d + 2 * kc * self.d * self.d
+            if self.rank == 0:
+                vec = st_src.param_vector().to(dev)
+            else:
+                vec = torch.zeros(size, dtype=torch.float32, device=dev)
+            pdist.broadcast_(vec)
+            if self.rank == 0:
+                return st_src
+            st = GmmState.empty(kc, self.d)
+            st.load_param_vector(vec.cpu())
+            return st
+
+        state = bcast_state(ck["state"] if self.rank == 0 else None, k)
+        best = (None if best_none
+                else bcast_state(ck["best"] if self.rank == 0 else None,
+                                 best_k))
+        return {
+            "k": k, "state": state, "best": best, "best_k": best_k,
+            "min_rissanen": float(scal[0].item()),
+            "best_lik": float(scal[1].item()),
+            "rissanen_by_k": {
+                int(keys[i].item()): float(vals[i].item())
+                for i in range(nk)
+            },
+        }
+
     def _host_clusters(self, k: int) -> HostClusters:
         st = self.state.shrink(k)
         return HostClusters(
@@ -364,7 +432,10 @@ class EmEngine:
 
         if cfg.checkpoint_dir:
             from .utils.checkpoint import load_sweep_checkpoint
-            ck = load_sweep_checkpoint(cfg.checkpoint_dir)
+            ck = (load_sweep_checkpoint(cfg.checkpoint_dir)
+                  if self.rank == 0 else None)
+            if self.world > 1:
+                ck = self._bcast_checkpoint(ck)
             if ck and ck["state"] is not None and ck["k"] <= k:
                 k = ck["k"]
                 st = ck["state"].to(self.device)

@@ -339,3 +339,31 @@ def test_scatter_input_world4(tmp_path):
     finally:
         os.environ.pop("GMM_TEST_SCATTER_DIR", None)
     assert sc is not None and sc == fs
+
+
+def _fit_ck_resume_rank0_only():
+    """Resume where ONLY rank 0 can access the checkpoint store — any
+    checkpoint I/O from a non-root rank trips an AssertionError."""
+    from cuda_gmm_mpi_amd.utils import checkpoint as ckmod
+    if os.environ.get("RANK") != "0":
+        def boom(*a, **k):
+            raise AssertionError("non-root rank touched the checkpoint dir")
+        ckmod.load_sweep_checkpoint = boom
+        ckmod.save_sweep_checkpoint = boom
+    return _fit_ck_resume()
+
+
+@pytest.mark.timeout(300)
+def test_world2_resume_rank0_only_checkpoint(tmp_path):
+    """Checkpoint resume is rank-0-authoritative (broadcast): works when
+    non-root ranks cannot read the checkpoint directory at all (the
+    --scatter-input / rank-0-only-filesystem deployment)."""
+    os.environ["GMM_TEST_CKDIR"] = str(tmp_path / "ck")
+    try:
+        _run_single("_fit_ck_phase1")
+        multi = run_world(2, "_fit_ck_resume_rank0_only", port=29822)
+        single = _run_single("_fit_ck_direct")
+    finally:
+        os.environ.pop("GMM_TEST_CKDIR", None)
+    assert multi["k"] == single["k"]
+    assert multi["rissanen"] == pytest.approx(single["rissanen"], rel=1e-3)
