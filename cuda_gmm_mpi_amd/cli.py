@@ -180,7 +180,11 @@ def main(argv=None) -> int:
         argv = sys.argv[1:]
     # self-launch N local ranks when asked and not already inside a world
     from .parallel.launcher import launch_workers, strip_gpus_arg
-    rest, ngpus = strip_gpus_arg(list(argv))
+    try:
+        rest, ngpus = strip_gpus_arg(list(argv))
+    except ValueError as e:
+        print(f"{e}\n")
+        return 1
     if ngpus and ngpus > 1 and "WORLD_SIZE" not in __import__("os").environ:
         return launch_workers(rest, ngpus)
     try:

@@ -54,6 +54,8 @@ def strip_gpus_arg(argv: list[str]) -> tuple[list[str], int | None]:
     while i < len(argv):
         a = argv[i]
         if a == "--gpus":
+            if i + 1 >= len(argv):
+                raise ValueError("--gpus requires a value")
             n = int(argv[i + 1])
             i += 2
             continue

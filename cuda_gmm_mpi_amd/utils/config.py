@@ -68,8 +68,8 @@ class GmmConfig:
     # E[x^2]-mean^2 variance (gaussian_kernel.cu:84-87).
     center_data: bool = True
 
-    # Write .results membership file (per-event posteriors).
-    # Device-side: memberships stay shard-resident except for this output.
+    # Per-iteration likelihood logging on rank 0 (runtime replacement for
+    # the reference's compile-time DEBUG printf macros, gaussian.h:44-54).
     verbose: bool = False
 
     # When the empty-cluster elimination jumps past target_num_clusters,
