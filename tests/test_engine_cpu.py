@@ -312,3 +312,25 @@ def test_em_likelihood_monotone():
         liks.append(float(eng._lik_dev.item()))
     diffs = np.diff(np.array(liks))
     assert np.all(diffs >= -1e-6 * abs(liks[-1])), diffs
+
+
+def test_state_param_vector_roundtrip():
+    """param_vector/load_param_vector (the fused merge broadcast payload)
+    is an exact roundtrip for all 7 parameter arrays."""
+    from cuda_gmm_mpi_amd.models.state import GmmState
+    rng = np.random.default_rng(11)
+    k, d = 5, 3
+    src = GmmState.empty(k, d)
+    for t in (src.N, src.pi, src.constant, src.avgvar, src.means, src.R,
+              src.Rinv):
+        t.copy_(torch.from_numpy(
+            rng.standard_normal(tuple(t.shape)).astype(np.float32)))
+    vec = src.param_vector()
+    assert vec.numel() == 4 * k + k * d + 2 * k * d * d
+    dst = GmmState.empty(k, d)
+    dst.load_param_vector(vec)
+    for a, b in ((src.N, dst.N), (src.pi, dst.pi),
+                 (src.constant, dst.constant), (src.avgvar, dst.avgvar),
+                 (src.means, dst.means), (src.R, dst.R),
+                 (src.Rinv, dst.Rinv)):
+        assert torch.equal(a, b)

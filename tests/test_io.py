@@ -82,3 +82,26 @@ def test_results_format(tmp_path):
     # (gaussian.cu:1046-1056)
     assert lines[0] == "1.000000,2.000000\t0.900000,0.100000"
     assert lines[1] == "3.500000,4.000000\t0.250000,0.750000"
+
+
+def test_csv_atof_junk_fallback(tmp_path):
+    """C atof semantics on malformed cells (readData.cpp:108): parse the
+    longest valid leading prefix, 0.0 on pure junk — via the faithful
+    parser fallback (pandas rejects the junk column)."""
+    path = tmp_path / "junk.csv"
+    path.write_text("a,b\n1.5xyz,2.0\nfoo,3e2\n-0.25,nan4\n")
+    data = gio.read_csv(str(path))
+    np.testing.assert_allclose(
+        data, [[1.5, 2.0], [0.0, 300.0], [-0.25, float("nan")]],
+        equal_nan=True)
+
+
+def test_csv_pandas_and_faithful_agree(tmp_path):
+    """Well-formed files parse identically through the fast pandas path
+    and the quirk-faithful fallback."""
+    path = tmp_path / "ok.csv"
+    rows = "\n".join(f"{i * 0.5},{-i},{i ** 2}" for i in range(1, 30))
+    path.write_text("h1,h2,h3\n" + rows + "\n")
+    fast = gio.read_csv(str(path))
+    faithful = gio._read_csv_faithful(str(path))
+    np.testing.assert_array_equal(fast, faithful)
