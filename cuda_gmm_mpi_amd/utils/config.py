@@ -93,6 +93,10 @@ class GmmConfig:
             raise ValueError(
                 "target_num_clusters must be less than equal to num_clusters"
             )
+        if self.min_iters < 0 or self.max_iters < self.min_iters:
+            raise ValueError(
+                "need 0 <= min_iters <= max_iters: "
+                f"{self.min_iters}..{self.max_iters}")
         if self.estep_dtype not in ("fp32", "bf16"):
             raise ValueError(f"estep_dtype must be fp32|bf16: {self.estep_dtype}")
         if self.mstep_precision not in ("fp32", "bf16x3"):

@@ -334,3 +334,11 @@ def test_state_param_vector_roundtrip():
                  (src.means, dst.means), (src.R, dst.R),
                  (src.Rinv, dst.Rinv)):
         assert torch.equal(a, b)
+
+
+def test_config_rejects_bad_iter_bounds():
+    with pytest.raises(ValueError):
+        GmmConfig(num_clusters=2, min_iters=-1).validate()
+    with pytest.raises(ValueError):
+        GmmConfig(num_clusters=2, min_iters=5, max_iters=3).validate()
+    GmmConfig(num_clusters=2, min_iters=0, max_iters=0).validate()
