@@ -137,3 +137,20 @@ def test_reduce_order_batched_no_bugcompat(rng):
     res_a = reduce_order(hc_a, bug_compat=False)
     res_b = reduce_order_batched(hc_b, bug_compat=False)
     assert res_a == res_b
+
+
+def test_reduce_order_after_mass_elimination(rng):
+    """5 clusters where 3 are empty: elimination leaves 2, which must then
+    merge into 1 in the same reduce_order step (the path that can jump
+    past target_num_clusters, gaussian.cu:866-907)."""
+    hc = make_clusters(rng, 5, 3)
+    hc.N[0] = 0.1
+    hc.N[2] = 0.4
+    hc.N[4] = 0.0
+    n_live = hc.N[1] + hc.N[3]
+    from cuda_gmm_mpi_amd.models.merge import reduce_order_batched
+    for fn in (reduce_order, reduce_order_batched):
+        k, c1, c2 = fn(make_copy(hc))
+        assert k == 1 and (c1, c2) == (0, 1)
+    k, _, _ = reduce_order(hc)
+    assert hc.N[0] == pytest.approx(n_live, rel=1e-6)
