@@ -342,3 +342,24 @@ def test_config_rejects_bad_iter_bounds():
     with pytest.raises(ValueError):
         GmmConfig(num_clusters=2, min_iters=5, max_iters=3).validate()
     GmmConfig(num_clusters=2, min_iters=0, max_iters=0).validate()
+
+
+def test_em_early_stop_on_convergence():
+    """With min_iters < max_iters the (reference-dead) epsilon comes
+    alive: EM stops once |likelihood change| <= epsilon, well before
+    max_iters (gaussian.cu:532 loop condition)."""
+    data, _ = make_blobs(4000, 3, 2, seed=13)
+    cfg = GmmConfig(num_clusters=2, target_num_clusters=2,
+                    min_iters=2, max_iters=500)
+    eng = build_engine(data, cfg, device="cpu")
+    calls = 0
+    orig = eng.em_iteration
+
+    def counting(k):
+        nonlocal calls
+        calls += 1
+        orig(k)
+
+    eng.em_iteration = counting
+    eng.run_em(2)
+    assert 2 <= calls < 500, calls
