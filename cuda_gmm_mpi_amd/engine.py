@@ -486,19 +486,28 @@ class EmEngine:
             with self.profile.time("reduce"):
                 new_k = k
                 if self.rank == 0:
+                    from .models.merge import eliminate_empty_clusters
                     hc = self._host_clusters(k)
-                    new_k, _, _ = reduce_order_batched(
-                        hc, bug_compat=cfg.bug_compat,
-                        device=str(self.device))
+                    k_live = eliminate_empty_clusters(hc)
+                    if k_live >= 2:
+                        new_k, _, _ = reduce_order_batched(
+                            hc, bug_compat=cfg.bug_compat,
+                            device=str(self.device))
+                    else:
+                        # elimination left < 2 clusters: nothing to merge
+                        # (the reference's pair scan would read
+                        # uninitialized merge state here); stop reducing
+                        # at the survivor count
+                        new_k = k_live
                 if self.world > 1:
                     # NCCL/RCCL broadcasts device tensors only
                     nk = torch.tensor([new_k], dtype=torch.int64,
                                       device=self.device)
                     pdist.broadcast_(nk, src=0)
                     new_k = int(nk.item())
-                if self.rank == 0:
+                if self.rank == 0 and new_k >= 1:
                     self._load_host_clusters(hc.truncated(new_k), new_k)
-                if self.world > 1:
+                if self.world > 1 and new_k >= 1:
                     with self.profile.time("comm"):
                         vec = self.state.shrink(new_k).param_vector().contiguous()
                         pdist.broadcast_(vec, src=0)
@@ -506,8 +515,9 @@ class EmEngine:
                         self.state.shrink(new_k).load_param_vector(vec)
             # the merged/compacted Rinv needs fresh E-step factors; the
             # constants stay as the host merge path produced them
-            self._refresh_mfac(new_k)
-            self._sync_add(new_k)
+            if new_k >= 1:
+                self._refresh_mfac(new_k)
+                self._sync_add(new_k)
             self.profile.count("reduce")
             k = new_k
             if cfg.checkpoint_dir and self.rank == 0:

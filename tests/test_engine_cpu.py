@@ -363,3 +363,16 @@ def test_em_early_stop_on_convergence():
     eng.em_iteration = counting
     eng.run_em(2)
     assert 2 <= calls < 500, calls
+
+
+def test_sweep_survives_mass_cluster_die_off():
+    """Few events + many clusters: empty-cluster elimination can leave
+    fewer than 2 clusters mid-sweep. The sweep must finish cleanly with
+    the best completed model instead of crashing in the pair scan (the
+    reference reads uninitialized merge state here)."""
+    data, _ = make_blobs(40, 1, 2, seed=1)
+    cfg = GmmConfig(num_clusters=8, target_num_clusters=0,
+                    min_iters=2, max_iters=2)
+    res = build_engine(data, cfg, device="cpu").sweep()
+    assert 1 <= res.num_clusters <= 8
+    assert np.isfinite(res.min_rissanen)
