@@ -87,3 +87,32 @@ def test_shard_bounds_partition(n, world):
         assert e - s == (per if r < world - 1 else n - per * (world - 1))
         prev_end = e
     assert prev_end == n
+
+
+@given(
+    n=st.integers(12, 300), d=st.integers(1, 8), k=st.integers(1, 12),
+    tgt_mode=st.integers(0, 2), diag=st.booleans(), bug=st.booleans(),
+    center=st.booleans(), seed=st.integers(0, 1000),
+)
+@settings(max_examples=25, deadline=None)
+def test_sweep_config_fuzz(n, d, k, tgt_mode, diag, bug, center, seed):
+    """Whole-sweep fuzz over the config space on tiny CPU problems: any
+    valid (N, D, K, target, diag, bug_compat, center) combination must
+    complete with a finite best model — no crashes in seeding, EM,
+    elimination, merging, or bookkeeping."""
+    from cuda_gmm_mpi_amd.engine import build_engine
+    from cuda_gmm_mpi_amd.utils.config import GmmConfig
+    from cuda_gmm_mpi_amd.utils.synthetic import make_blobs
+
+    target = 0 if tgt_mode == 0 else (1 if tgt_mode == 1 else min(k, 3))
+    data, _ = make_blobs(n, d, max(2, min(k, 4)), seed=seed)
+    cfg = GmmConfig(num_clusters=k, target_num_clusters=target,
+                    min_iters=1, max_iters=2, diag_only=diag,
+                    bug_compat=bug, center_data=center)
+    res = build_engine(data, cfg, device="cpu").sweep()
+    assert 1 <= res.num_clusters <= k
+    assert np.isfinite(res.min_rissanen)
+    assert all(np.isfinite(v) for v in res.rissanen_by_k.values())
+    st_best = res.state
+    assert int(st_best.num_clusters) == res.num_clusters
+    assert np.isfinite(st_best.means.numpy()).all()
