@@ -367,3 +367,24 @@ def test_world2_resume_rank0_only_checkpoint(tmp_path):
         os.environ.pop("GMM_TEST_CKDIR", None)
     assert multi["k"] == single["k"]
     assert multi["rissanen"] == pytest.approx(single["rissanen"], rel=1e-3)
+
+
+def _fit_die_off():
+    """Mass cluster die-off mid-sweep under distribution: the survivor
+    stop (new_k < 2) must keep both ranks in lockstep."""
+    from cuda_gmm_mpi_amd.engine import build_engine
+    from cuda_gmm_mpi_amd.utils.config import GmmConfig
+    data, _ = make_blobs(40, 1, 2, seed=1)
+    cfg = GmmConfig(num_clusters=8, target_num_clusters=0,
+                    min_iters=2, max_iters=2,
+                    covariance_dynamic_range=1e15)
+    res = build_engine(data, cfg, device="cpu").sweep()
+    return {"k": res.num_clusters, "riss": res.min_rissanen}
+
+
+@pytest.mark.timeout(300)
+def test_world2_mass_die_off_lockstep():
+    single = _run_single("_fit_die_off")
+    multi = run_world(2, "_fit_die_off", port=29823)
+    assert multi["k"] == single["k"]
+    assert np.isfinite(multi["riss"])
