@@ -376,3 +376,33 @@ def test_sweep_survives_mass_cluster_die_off():
     res = build_engine(data, cfg, device="cpu").sweep()
     assert 1 <= res.num_clusters <= 8
     assert np.isfinite(res.min_rissanen)
+
+
+def test_checkpoint_save_load_roundtrip(tmp_path):
+    """save_sweep_checkpoint / load_sweep_checkpoint preserve every field
+    (params, best model, bookkeeping) exactly."""
+    from cuda_gmm_mpi_amd.models.state import GmmState
+    from cuda_gmm_mpi_amd.utils.checkpoint import (
+        load_sweep_checkpoint, save_sweep_checkpoint)
+    rng = np.random.default_rng(21)
+    k, bk, d = 4, 6, 3
+
+    def rand_state(kc):
+        st = GmmState.empty(kc, d)
+        for t in (st.N, st.pi, st.constant, st.avgvar, st.means, st.R,
+                  st.Rinv):
+            t.copy_(torch.from_numpy(
+                rng.standard_normal(tuple(t.shape)).astype(np.float32)))
+        return st
+
+    cur, best = rand_state(k), rand_state(bk)
+    riss = {6: 123.5, 5: 117.25, 4: 120.0}
+    save_sweep_checkpoint(str(tmp_path), cur, k, best, bk, 117.25, -55.5,
+                          riss)
+    ck = load_sweep_checkpoint(str(tmp_path))
+    assert ck["k"] == k and ck["best_k"] == bk
+    assert ck["min_rissanen"] == 117.25 and ck["best_lik"] == -55.5
+    assert ck["rissanen_by_k"] == riss
+    for src, dst in ((cur, ck["state"]), (best, ck["best"])):
+        for name in ("N", "pi", "constant", "avgvar", "means", "R", "Rinv"):
+            assert torch.equal(getattr(src, name), getattr(dst, name)), name
