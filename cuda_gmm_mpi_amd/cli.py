@@ -116,7 +116,9 @@ def run_clustering(data: np.ndarray | None, cfg: GmmConfig, outfile: str,
     rank, local_rank, world = pdist.rank(), 0, pdist.world_size()
     if device == "cuda":
         local_rank = torch.cuda.current_device()
+    import time as _time
     prof = Profile(device)
+    t_start = _time.perf_counter()
     shard = None
     if scatter_input:
         from .engine import build_engine_sharded
@@ -165,6 +167,7 @@ def run_clustering(data: np.ndarray | None, cfg: GmmConfig, outfile: str,
     if profile_report:
         print(engine.profile.report(rank, local_rank))
 
+    elapsed = _time.perf_counter() - t_start
     return {
         "num_clusters": result.num_clusters,
         "rissanen": result.min_rissanen,
@@ -172,6 +175,11 @@ def run_clustering(data: np.ndarray | None, cfg: GmmConfig, outfile: str,
         "rissanen_by_k": result.rissanen_by_k,
         "state": out_state,
         "memberships": memberships,
+        "seconds": elapsed,
+        "total_em_iterations": engine.total_em_iterations,
+        "em_iterations_per_sec": (engine.total_em_iterations / elapsed
+                                  if elapsed > 0 else 0.0),
+        "n_events": engine.n_total,
     }
 
 
@@ -252,6 +260,11 @@ def main(argv=None) -> int:
                     },
                     "world_size": world,
                     "device": device,
+                    "seconds": result["seconds"],
+                    "total_em_iterations": result["total_em_iterations"],
+                    "em_iterations_per_sec":
+                        result["em_iterations_per_sec"],
+                    "n_events": result["n_events"],
                 }, f, indent=2)
         if rank == 0 and args.enable_print:
             print(f"Ideal clusters: {result['num_clusters']} "

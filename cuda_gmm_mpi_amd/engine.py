@@ -129,6 +129,7 @@ class EmEngine:
         self.use_graphs = True
         self._graphs: dict[int, object] = {}
         self._lik_dev = torch.zeros(1, dtype=torch.float32, device=self.device)
+        self.total_em_iterations = 0  # across the whole sweep (all Ks)
 
 
     def _refresh_mfac(self, k: int) -> None:
@@ -286,6 +287,7 @@ class EmEngine:
                 print(f"[K={k}] iter {iters}: likelihood {lik:e} "
                       f"(change {change:e})")
         self.likelihood = lik
+        self.total_em_iterations += iters
         return lik
 
     def em_iteration(self, k: int) -> None:

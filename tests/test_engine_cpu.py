@@ -184,6 +184,8 @@ def test_metrics_out(tmp_path):
                "--metrics-out", mpath])
     assert rc == 0
     m = json.load(open(mpath))
+    assert m["total_em_iterations"] == 3 and m["seconds"] > 0
+    assert m["em_iterations_per_sec"] > 0 and m["n_events"] == 600
     assert m["num_clusters"] == 3
     assert "3" in m["rissanen_by_k"]
 
