@@ -38,6 +38,11 @@ def init_process_group(backend: str | None = None,
     if not dist.is_initialized():
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", "29541")
+        # fail fast on collective errors/hangs (the reference
+        # aborts on any MPI/CUDA failure; RCCL equivalent is async error
+        # handling — a stuck collective aborts after `timeout_s` instead
+        # of hanging the job)
+        os.environ.setdefault("TORCH_NCCL_ASYNC_ERROR_HANDLING", "1")
         dist.init_process_group(
             backend=backend, rank=rank, world_size=world,
             timeout=datetime.timedelta(seconds=timeout_s),
