@@ -21,7 +21,6 @@ MI355X-native redesign of the reference driver (gaussian.cu:128-1106):
 from __future__ import annotations
 
 import dataclasses
-import math
 
 import numpy as np
 import torch

@@ -8,7 +8,6 @@ the torch fp32 reference implementations (`cpu_reference`).
 from __future__ import annotations
 
 import importlib
-import os
 
 _ext = None
 _ext_error: Exception | None = None
