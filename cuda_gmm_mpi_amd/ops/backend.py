@@ -33,7 +33,7 @@ def hip_ext():
             "cuda_gmm_mpi_amd HIP extension (_gmm_hip) is not built. "
             "GPU execution requires the hand-written gfx950 kernels — there "
             "is no eager fallback by design. Build with: "
-            "PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace "
+            "PYTORCH_ROCM_ARCH=gfx950 python build_ext.py "
             f"(import error: {_ext_error})"
         )
     return ext
