@@ -103,7 +103,8 @@ def main() -> int:
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": args.dtype,
+            # CPU smoke runs compute in fp32 regardless of the request
+            "dtype": args.dtype if has_gpu else "fp32",
             "data": "synthetic",
             "config": {
                 "model": "gmm_em",
