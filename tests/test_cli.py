@@ -84,7 +84,6 @@ def test_cli_no_output_creates_empty_summary(tmp_path, csv_file):
 
 def test_cli_gpus_launcher_cpu_world2(tmp_path, csv_file, monkeypatch):
     """--gpus 2 spawns two gloo ranks on CPU and produces one output."""
-    import os
     monkeypatch.delenv("WORLD_SIZE", raising=False)
     path, _ = csv_file
     out = str(tmp_path / "mp")

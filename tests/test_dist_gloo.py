@@ -220,7 +220,6 @@ def test_world2_resume_from_world1_checkpoint(tmp_path):
 
 def _cli_scatter(flag=True):
     """Run the CLI with/without --scatter-input; return the .summary text."""
-    import tempfile
     from cuda_gmm_mpi_amd.cli import main
     from cuda_gmm_mpi_amd.utils import io as gio
     work = os.environ["GMM_TEST_SCATTER_DIR"]

@@ -52,7 +52,6 @@ def test_lu_invert_property(d, seed, cond):
                 min_size=2, max_size=20),
        st.integers(2, 6))
 def test_csv_roundtrip_values(values, d):
-    import io as _io
     import tempfile, os
     rows = [values[i:i + d] for i in range(0, len(values) - d + 1, d)]
     if not rows:
