@@ -205,3 +205,38 @@ def test_finalize_covariance_world_dependence_quirk():
     np.testing.assert_allclose(
         diff[:, eye].numpy(), expect.expand(k, d).numpy(), rtol=1e-5)
     assert torch.all(diff[:, ~eye] == 0)
+
+
+def test_aux_utility_symbols():
+    """Small public utilities: availability probe, GEMM-shaped N+means
+    helper, logdet wrapper, timer totals, bench-data generator."""
+    from cuda_gmm_mpi_amd.ops.backend import has_hip_ext
+    from cuda_gmm_mpi_amd.ops.functional import mstep_n_means
+    from cuda_gmm_mpi_amd.ops.invert import log_det_lu_nopivot
+    from cuda_gmm_mpi_amd.utils.synthetic import make_bench_data
+    from cuda_gmm_mpi_amd.utils.timers import Profile
+
+    assert isinstance(has_hip_ext(), bool)
+
+    rng = np.random.default_rng(2)
+    x = rng.standard_normal((3, 50)).astype(np.float32)
+    w = rng.uniform(0, 1, (4, 50)).astype(np.float32)
+    x_aug = np.concatenate([x, np.ones((1, 50), np.float32)]).T
+    n_c, mean_num = mstep_n_means(torch.from_numpy(x_aug),
+                                  torch.from_numpy(w))
+    np.testing.assert_allclose(n_c.numpy(), w.sum(axis=1), rtol=1e-5)
+    np.testing.assert_allclose(mean_num.numpy(), w @ x.T, rtol=1e-5)
+
+    a = rng.standard_normal((3, 3)).astype(np.float32)
+    spd = a @ a.T + 3 * np.eye(3, dtype=np.float32)
+    ld = log_det_lu_nopivot(spd)
+    assert ld == pytest.approx(np.linalg.slogdet(spd)[1], rel=1e-4)
+
+    data = make_bench_data(100, 4, 3)
+    assert data.shape == (100, 4) and data.dtype == np.float32
+
+    p = Profile("cpu")
+    with p.time("cpu"):
+        pass
+    assert set(p.totals_ms()) == {"e_step", "m_step", "constants",
+                                  "reduce", "memcpy", "cpu", "comm"}
