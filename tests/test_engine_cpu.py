@@ -430,3 +430,24 @@ def test_single_event_degenerate_is_reference_faithful():
                     min_iters=1, max_iters=1)
     res = build_engine(data, cfg, device="cpu").sweep()
     assert res.num_clusters == 1
+
+
+def test_deep_sweep_and_resume_after_completion(tmp_path):
+    """40-cluster MDL sweep down to 1 with checkpointing (39 merges), then
+    a fresh engine resuming from the FINAL checkpoint — it must return the
+    same best model without re-running the whole sweep."""
+    data, _ = make_blobs(20000, 6, 8, seed=77)
+    cfg = GmmConfig(num_clusters=40, target_num_clusters=0,
+                    min_iters=3, max_iters=3,
+                    checkpoint_dir=str(tmp_path))
+    eng = build_engine(data, cfg, device="cpu")
+    res = eng.sweep()
+    assert len(res.rissanen_by_k) == 40
+    assert eng.total_em_iterations == 3 * 40
+    assert np.isfinite(res.min_rissanen)
+
+    eng2 = build_engine(data, cfg, device="cpu")
+    res2 = eng2.sweep()
+    assert res2.num_clusters == res.num_clusters
+    assert res2.min_rissanen == pytest.approx(res.min_rissanen, rel=1e-6)
+    assert eng2.total_em_iterations <= 3  # only the final K re-runs
