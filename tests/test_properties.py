@@ -33,18 +33,22 @@ def test_posteriors_always_normalized(k, n, scale, seed):
     assert float(lik) == float(lik)  # finite, not NaN
 
 
-@settings(max_examples=20, deadline=None)
+@settings(max_examples=30, deadline=None)
 @given(d=st.integers(1, 16), seed=st.integers(0, 2**31 - 1),
-       cond=st.floats(1.0, 100.0))
-def test_lu_invert_property(d, seed, cond):
+       ridge=st.floats(1e-2, 100.0))
+def test_lu_invert_property(d, seed, ridge):
+    """No-pivot LU inverse + logdet across conditioning: ridge down to
+    1e-2 gives condition numbers up to ~1e5 (fp32 headroom shrinks with
+    cond, so tolerances scale with the inverse's magnitude)."""
     rng = np.random.default_rng(seed)
     a = rng.standard_normal((d, d))
-    m = (a @ a.T + cond * d * np.eye(d)).astype(np.float32)
+    m = (a @ a.T + ridge * d * np.eye(d)).astype(np.float32)
     inv, logdet = cpu.lu_invert_nopivot(torch.from_numpy(m)[None])
     ref = np.linalg.slogdet(m.astype(np.float64))[1]
     assert abs(float(logdet[0]) - ref) < max(1e-2, 2e-3 * abs(ref))
+    scale = max(1.0, float(np.abs(inv[0].numpy()).max()))
     np.testing.assert_allclose(
-        (inv[0].numpy() @ m), np.eye(d), atol=5e-2)
+        (inv[0].numpy() @ m), np.eye(d), atol=5e-4 * scale * d)
 
 
 @settings(max_examples=15, deadline=None)
