@@ -62,17 +62,28 @@ def _load_state(z, prefix: str) -> GmmState | None:
 
 
 def load_sweep_checkpoint(directory: str) -> dict | None:
+    """Load the sweep checkpoint; a missing file returns None and an
+    unreadable/corrupt one WARNS and returns None (the sweep starts
+    fresh) — checkpoint writes are atomic (os.replace), so corruption
+    here means external damage, not an interrupted writer."""
     path = checkpoint_path(directory)
     if not os.path.exists(path):
         return None
-    z = np.load(path)
-    riss = {int(k): float(v) for k, v in zip(z["riss_keys"], z["riss_vals"])}
-    return {
-        "k": int(z["k"]),
-        "best_k": int(z["best_k"]),
-        "min_rissanen": float(z["min_rissanen"]),
-        "best_lik": float(z["best_lik"]),
-        "rissanen_by_k": riss,
-        "state": _load_state(z, "cur"),
-        "best": _load_state(z, "best"),
-    }
+    try:
+        z = np.load(path)
+        riss = {int(k): float(v)
+                for k, v in zip(z["riss_keys"], z["riss_vals"])}
+        return {
+            "k": int(z["k"]),
+            "best_k": int(z["best_k"]),
+            "min_rissanen": float(z["min_rissanen"]),
+            "best_lik": float(z["best_lik"]),
+            "rissanen_by_k": riss,
+            "state": _load_state(z, "cur"),
+            "best": _load_state(z, "best"),
+        }
+    except Exception as e:  # noqa: BLE001 — corrupt/alien file
+        import sys
+        print(f"WARNING: ignoring unreadable sweep checkpoint {path}: {e}",
+              file=sys.stderr)
+        return None

@@ -501,3 +501,19 @@ def test_checkpoint_crash_recovery_bit_exact(tmp_path):
     res = run(ck)  # resume (or re-confirm) to completion
     assert res["k"] == ref["k"]
     assert res["riss"] == ref["riss"]  # bit-exact
+
+
+def test_corrupt_checkpoint_starts_fresh(tmp_path, capsys):
+    """A corrupt checkpoint file warns and is ignored — the sweep runs
+    from scratch instead of crashing at resume."""
+    from cuda_gmm_mpi_amd.utils.checkpoint import checkpoint_path
+    ck = tmp_path / "ck"
+    ck.mkdir()
+    with open(checkpoint_path(str(ck)), "wb") as f:
+        f.write(b"this is not an npz file")
+    data, _ = make_blobs(800, 2, 3, seed=9)
+    cfg = GmmConfig(num_clusters=4, target_num_clusters=2,
+                    min_iters=2, max_iters=2, checkpoint_dir=str(ck))
+    res = build_engine(data, cfg, device="cpu").sweep()
+    assert res.num_clusters == 2
+    assert "ignoring unreadable sweep checkpoint" in capsys.readouterr().err
