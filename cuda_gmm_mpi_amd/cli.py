@@ -218,6 +218,25 @@ def main(argv=None) -> int:
         device = "cuda" if torch.cuda.is_available() else "cpu"
 
     rank, local_rank, world = pdist.init_process_group()
+    # outfile writability check BEFORE any compute — the reference's own
+    # check exists but is commented out (gaussian.cu:1135-1141, dead
+    # "return 3" path); revived here with its exit code. Probed on rank 0
+    # only (the writing rank) and broadcast so no rank hangs.
+    out_err = 0
+    if rank == 0:
+        try:
+            with open(args.outfile + ".summary", "a"):
+                pass
+        except OSError as e:
+            print(f"Unable to create output file. ({e})\n")
+            out_err = 1
+    if world > 1:
+        flag = torch.tensor([out_err], dtype=torch.long)
+        pdist.broadcast_(flag)
+        out_err = int(flag.item())
+    if out_err:
+        pdist.destroy()
+        return 3
     if cfg.enable_print:
         import socket
         dev_name = (torch.cuda.get_device_name(0)

@@ -145,3 +145,12 @@ def test_cli_profile_report(tmp_path, csv_file, capsys):
     text = capsys.readouterr().out
     assert "E-step Kernel:" in text and "M-step Kernel:" in text
     assert "Consts Kernel:" in text
+
+
+def test_cli_unwritable_outfile_exits_early(tmp_path, csv_file):
+    """Unwritable output path returns 3 BEFORE any compute (the
+    reference's outfile check is commented out, gaussian.cu:1135-1141;
+    revived with its dead exit code)."""
+    path, _ = csv_file
+    rc = main(["3", path, "/nonexistent_dir/out", "3", "--device", "cpu"])
+    assert rc == 3
