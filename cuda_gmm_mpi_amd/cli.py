@@ -261,6 +261,12 @@ def main(argv=None) -> int:
             pdist.broadcast_(flag)
             if int(flag.item()):
                 return 2
+        if data is not None and not np.isfinite(data).all():
+            # faithful behavior propagates them (the reference's atof
+            # parses "nan"/"inf"); warn so garbage results are explicable
+            bad = int((~np.isfinite(data)).sum())
+            print(f"WARNING: input contains {bad} non-finite values; "
+                  "results will be degenerate", file=sys.stderr)
         result = run_clustering(
             data, cfg, args.outfile, device,
             write_results=args.write_results, profile_report=args.profile,

@@ -154,3 +154,14 @@ def test_cli_unwritable_outfile_exits_early(tmp_path, csv_file):
     path, _ = csv_file
     rc = main(["3", path, "/nonexistent_dir/out", "3", "--device", "cpu"])
     assert rc == 3
+
+
+def test_cli_warns_on_nonfinite_input(tmp_path, capsys):
+    """NaN/Inf in the input propagate (reference atof semantics) but
+    produce a rank-0 warning at load time."""
+    path = tmp_path / "bad.csv"
+    path.write_text("a,b\n1.0,2.0\nnan,3.0\n4.0,inf\n")
+    rc = main(["2", str(path), str(tmp_path / "o"), "2", "--min-iters", "1",
+               "--max-iters", "1", "--device", "cpu", "--no-results"])
+    assert rc == 0
+    assert "non-finite" in capsys.readouterr().err
