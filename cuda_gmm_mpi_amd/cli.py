@@ -54,7 +54,7 @@ def make_parser() -> argparse.ArgumentParser:
                    help="M-step sufficient-statistics precision (bf16x3 = "
                         "split-precision MFMA, ~1e-5 relative, ~1.5x faster)")
     p.add_argument("--no-center", dest="center_data", action="store_false")
-    p.add_argument("--device", default=None,
+    p.add_argument("--device", default=None, choices=["cpu", "cuda"],
                    help="cpu | cuda (default: cuda when available)")
     p.add_argument("--profile", action="store_true",
                    help="print the per-GPU timing report (gaussian.cu:967)")
