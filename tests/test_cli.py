@@ -165,3 +165,18 @@ def test_cli_warns_on_nonfinite_input(tmp_path, capsys):
                "--max-iters", "1", "--device", "cpu", "--no-results"])
     assert rc == 0
     assert "non-finite" in capsys.readouterr().err
+
+
+def test_python_dash_m_package_entry(tmp_path, csv_file):
+    """`python -m cuda_gmm_mpi_amd` runs the same CLI."""
+    import subprocess
+    import sys
+    path, _ = csv_file
+    out = str(tmp_path / "m")
+    r = subprocess.run(
+        [sys.executable, "-m", "cuda_gmm_mpi_amd", "2", path, out, "2",
+         "--min-iters", "1", "--max-iters", "1", "--device", "cpu",
+         "--no-results"],
+        capture_output=True, timeout=240)
+    assert r.returncode == 0, r.stderr.decode()[-400:]
+    assert open(out + ".summary").read().count("Cluster #") == 2
