@@ -261,7 +261,7 @@ def main(argv=None) -> int:
             pdist.broadcast_(flag)
             if int(flag.item()):
                 return 2
-        if data is not None and not np.isfinite(data).all():
+        if rank == 0 and data is not None and not np.isfinite(data).all():
             # faithful behavior propagates them (the reference's atof
             # parses "nan"/"inf"); warn so garbage results are explicable
             bad = int((~np.isfinite(data)).sum())
