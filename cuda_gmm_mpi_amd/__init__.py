@@ -7,6 +7,6 @@ for the hot ops, and RCCL over xGMI for multi-GPU data parallelism.
 """
 __version__ = "0.1.0"
 
-from .engine import EmEngine, build_engine  # noqa: F401
+from .engine import EmEngine, build_engine, build_engine_sharded  # noqa: F401
 from .models.state import GmmState  # noqa: F401
 from .utils.config import GmmConfig  # noqa: F401
