@@ -136,11 +136,15 @@ def run_clustering(data: np.ndarray | None, cfg: GmmConfig, outfile: str,
         engine.use_graphs = False
     result = engine.sweep()
 
-    # de-center the saved model for output
-    out_state = result.state.to("cpu")
-    out_state.means += engine.center.cpu().unsqueeze(0)
-
+    # memberships first: recompute_memberships needs the saved means in the
+    # engine's internally-centered frame, so it must run before de-centering
     w_shard = engine.recompute_memberships(result.state)
+
+    # de-center the saved model for output. clone() is load-bearing: on CPU
+    # .to("cpu") returns aliased tensors and the += would corrupt the saved
+    # state in place.
+    out_state = result.state.to("cpu").clone(with_memberships=False)
+    out_state.means += engine.center.cpu().unsqueeze(0)
     memberships = None
     if striped_results:
         # SURVEY §5 long-context plan: per-rank file stripes — shard-local

@@ -29,9 +29,10 @@ def invert_cpu(a: np.ndarray, base10: bool = True) -> tuple[np.ndarray, float]:
     n = data.shape[0]
     assert data.shape == (n, n)
     if n == 1:
+        # the reference's 1x1 special case uses logf (natural log) even
+        # though its n>=2 path accumulates log10 (invert_matrix.cpp:39
+        # vs :61) — mirror that regardless of base10
         log_det = float(np.log(data[0, 0]))
-        if base10:
-            log_det = float(np.log10(data[0, 0]))
         return np.array([[1.0 / data[0, 0]]], dtype=np.float32), log_det
 
     # normalize row 0 (invert_matrix.cpp:42)

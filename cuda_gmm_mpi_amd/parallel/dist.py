@@ -34,7 +34,12 @@ def init_process_group(backend: str | None = None,
     if world == 1:
         return rank, local_rank, world
     if backend is None:
-        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        # device-specific backends: RCCL for device buffers, gloo for the
+        # CPU-tensor control-plane collectives (outfile/read flags,
+        # distribute_input shards + stats). An nccl-only group would raise
+        # "no backend for device type cpu" on every multi-rank GPU run.
+        backend = ("cpu:gloo,cuda:nccl" if torch.cuda.is_available()
+                   else "gloo")
     if not dist.is_initialized():
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", "29541")

@@ -37,6 +37,35 @@ def test_cli_end_to_end_csv(tmp_path, csv_file):
     assert memb.sum() == pytest.approx(1.0, abs=1e-4)
 
 
+@pytest.mark.parametrize("striped", [False, True])
+def test_cli_results_membership_content_on_offset_data(tmp_path, striped):
+    """Regression: on CPU runs .to("cpu") aliased the saved
+    state and the de-centering += shifted means before memberships were
+    recomputed — a correct 200/200 split became 400/0. Assert actual
+    assignment content on well-separated data with a nonzero mean."""
+    rng = np.random.default_rng(7)
+    a = rng.normal(loc=50.0, scale=0.5, size=(200, 2))
+    b = rng.normal(loc=80.0, scale=0.5, size=(200, 2))
+    data = np.vstack([a, b]).astype(np.float32)
+    binpath = str(tmp_path / "sep.bin")
+    gio.write_bin(binpath, data)
+    out = str(tmp_path / "sep")
+    argv = ["2", binpath, out, "2", "--min-iters", "8", "--max-iters", "8",
+            "--device", "cpu"]
+    if striped:
+        argv.append("--striped-results")
+    rc = main(argv)
+    assert rc == 0
+    results_path = out + (".results.0" if striped else ".results")
+    lines = open(results_path).read().splitlines()
+    assert len(lines) == 400
+    assign = [int(np.argmax([float(v) for v in ln.split("\t")[1].split(",")]))
+              for ln in lines]
+    first, second = assign[:200], assign[200:]
+    assert len(set(first)) == 1 and len(set(second)) == 1
+    assert first[0] != second[0]
+
+
 def test_cli_bin_input(tmp_path):
     data, _ = make_blobs(500, 3, 2, seed=6)
     binpath = str(tmp_path / "d.bin")
