@@ -271,6 +271,19 @@ class EmEngine:
         Returns the final global log-likelihood.
         """
         cfg = self.cfg
+        if cfg.min_iters >= cfg.max_iters and not cfg.verbose:
+            # epsilon is dead (reference default MIN_ITERS == MAX_ITERS,
+            # gaussian.h:26-27): the loop runs exactly min_iters times no
+            # matter what the likelihood does, so keep it on device and
+            # read the scalar ONCE per K instead of syncing the host every
+            # iteration (config 5's 22-K sweep paid ~2,200 syncs).
+            self._finish_likelihood(self._estep(k))
+            for _ in range(cfg.min_iters):
+                self.em_iteration(k)
+            lik = float(self._lik_dev.item())
+            self.likelihood = lik
+            self.total_em_iterations += cfg.min_iters
+            return lik
         lik = self._reduce_likelihood(self._estep(k))
         iters = 0
         change = self.epsilon * 2
@@ -557,23 +570,31 @@ class EmEngine:
         return w
 
     def gather_memberships(self, w_shard: torch.Tensor) -> np.ndarray | None:
-        """Gather per-event posteriors to rank 0 (event-major contiguous
-        reassembly). Returns [K, N_total] on rank 0, None elsewhere."""
+        """Gather per-event posteriors to rank 0 only, via exact-size
+        point-to-point sends (reference analog: the hand-rolled
+        MPI_Send/Recv gather, gaussian.cu:783-823).
+
+        Non-root ranks send their [K, n_shard] shard and allocate nothing;
+        rank 0 streams one shard-sized device buffer at a time into the
+        host [K, N_total] output — no padded all_gather, no full [K, N]
+        materialization anywhere but the writing rank's host buffer.
+        Returns [K, N_total] on rank 0, None elsewhere.
+        """
         k = int(w_shard.shape[0])
         if self.world == 1:
             return w_shard.cpu().numpy()
-        per = self.n_total // self.world
-        maxn = self.n_total - per * (self.world - 1)
-        pad = torch.zeros(k, maxn, dtype=torch.float32, device=w_shard.device)
-        pad[:, : self.n_shard] = w_shard
-        parts = [torch.empty_like(pad) for _ in range(self.world)]
-        torch.distributed.all_gather(parts, pad)
         if self.rank != 0:
+            torch.distributed.send(w_shard.contiguous(), dst=0)
             return None
         out = np.empty((k, self.n_total), dtype=np.float32)
-        for r in range(self.world):
+        s0, e0 = pdist.shard_bounds(self.n_total, self.world, 0)
+        out[:, s0:e0] = w_shard.cpu().numpy()
+        for r in range(1, self.world):
             s, e = pdist.shard_bounds(self.n_total, self.world, r)
-            out[:, s:e] = parts[r][:, : e - s].cpu().numpy()
+            buf = torch.empty(k, e - s, dtype=torch.float32,
+                              device=w_shard.device)
+            torch.distributed.recv(buf, src=r)
+            out[:, s:e] = buf.cpu().numpy()
         return out
 
 

@@ -25,13 +25,21 @@ def env_world() -> tuple[int, int, int]:
 
 
 def init_process_group(backend: str | None = None,
-                       timeout_s: float = 600.0) -> tuple[int, int, int]:
-    """Initialize torch.distributed from the environment (no-op at world 1).
+                       timeout_s: float = 600.0,
+                       force: bool = False) -> tuple[int, int, int]:
+    """Initialize torch.distributed from the environment.
+
+    A bare world-1 run (no launcher) skips initialization; under a
+    launcher (WORLD_SIZE set, e.g. torchrun --nproc-per-node 1) or with
+    ``force=True`` a real 1-rank communicator IS created, so RCCL init +
+    device-buffer collectives + graph capture execute even on a 1-GPU box
+    (world=1 communicator replaces the reference's untested
+    MPI path).
 
     Returns (rank, local_rank, world_size).
     """
     rank, local_rank, world = env_world()
-    if world == 1:
+    if world == 1 and "WORLD_SIZE" not in os.environ and not force:
         return rank, local_rank, world
     if backend is None:
         # device-specific backends: RCCL for device buffers, gloo for the

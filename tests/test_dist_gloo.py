@@ -387,3 +387,35 @@ def test_world2_mass_die_off_lockstep():
     multi = run_world(2, "_fit_die_off", port=29823)
     assert multi["k"] == single["k"]
     assert np.isfinite(multi["riss"])
+
+
+def _gather_bytes():
+    """Byte-exact rank-0 gather: shard w filled with the global event index
+    so the reassembled [K, N] must equal the analytic pattern EXACTLY."""
+    from cuda_gmm_mpi_amd.engine import build_engine
+    from cuda_gmm_mpi_amd.parallel import dist as pdist
+    from cuda_gmm_mpi_amd.utils.config import GmmConfig
+    n, k = 1003, 3
+    data, _ = make_blobs(n, 2, 2, seed=5)
+    cfg = GmmConfig(num_clusters=k, target_num_clusters=k,
+                    min_iters=1, max_iters=1)
+    eng = build_engine(data, cfg, device="cpu")
+    s, e = pdist.shard_bounds(n, eng.world, eng.rank)
+    w = (torch.arange(s, e, dtype=torch.float32).unsqueeze(0)
+         + 1000.0 * torch.arange(k, dtype=torch.float32).unsqueeze(1))
+    g = eng.gather_memberships(w)
+    if eng.rank != 0:
+        assert g is None  # non-root allocates/returns nothing
+        return None
+    expect = (np.arange(n, dtype=np.float32)[None, :]
+              + 1000.0 * np.arange(k, dtype=np.float32)[:, None])
+    return {"exact": bool((g == expect).all()), "shape": list(g.shape)}
+
+
+@pytest.mark.timeout(300)
+def test_world3_gather_memberships_byte_exact():
+    """The point-to-point rank-0 gather (no padded all_gather) reassembles
+    uneven shards byte-exactly in event order."""
+    res = run_world(3, "_gather_bytes", port=29824)
+    assert res["shape"] == [3, 1003]
+    assert res["exact"]
