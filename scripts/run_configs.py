@@ -82,9 +82,15 @@ def main():
         emit(4, {"device": dev, "n": n, "em_iters_per_sec": ips,
                  "big_estep": eng.use_big_estep})
 
+    rc = 0
     if 5 in configs:
+        # 100 equal-weight, separation-checked clusters: elimination
+        # cannot mass-kill, the merge path steps K down one at a time and
+        # the sweep LANDS on the target K=20 (the reference's save-target
+        # path, gaussian.cu:839) — asserted below, nonzero exit otherwise
+        from cuda_gmm_mpi_amd.utils.synthetic import make_supported_blobs
         n = int(2_000_000 * args.scale)
-        data, _ = make_blobs(n, 21, 100, seed=11)
+        data, _ = make_supported_blobs(n, 21, 100, seed=11)
         iters = args.iters if args.iters is not None else 100
         cfg = GmmConfig(num_clusters=100, target_num_clusters=20,
                         min_iters=iters, max_iters=iters,
@@ -93,14 +99,21 @@ def main():
         eng = build_engine(data, cfg, device=dev)
         res = eng.sweep()
         dt = time.perf_counter() - t0
-        total_iters = sum(iters + 1 for _ in res.rissanen_by_k)
+        total_iters = eng.total_em_iterations
+        reached = res.num_clusters == 20
         emit(5, {"device": dev, "n": n, "seconds": dt,
                  "final_k": res.num_clusters,
+                 "target_reached": reached,
                  "min_rissanen": res.min_rissanen,
                  "ks_swept": len(res.rissanen_by_k),
                  "em_iterations_total": total_iters,
                  "em_iters_per_sec_incl_merges": total_iters / dt})
+        if not reached:
+            print(f"config 5 FAILED to land on K=20 (final_k="
+                  f"{res.num_clusters})", file=sys.stderr)
+            rc = 1
+    return rc
 
 
 if __name__ == "__main__":
-    main()
+    sys.exit(main())

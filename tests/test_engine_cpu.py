@@ -517,3 +517,16 @@ def test_corrupt_checkpoint_starts_fresh(tmp_path, capsys):
     res = build_engine(data, cfg, device="cpu").sweep()
     assert res.num_clusters == 2
     assert "ignoring unreadable sweep checkpoint" in capsys.readouterr().err
+
+
+def test_sweep_lands_on_target_with_supported_blobs():
+    """BASELINE config-5 semantics in miniature: equal-weight separated
+    blobs keep every cluster supported, so the MDL sweep steps down and
+    LANDS on the target K (the reference's save-target path,
+    gaussian.cu:839) instead of jumping past it."""
+    from cuda_gmm_mpi_amd.utils.synthetic import make_supported_blobs
+    data, _ = make_supported_blobs(6000, 5, 30, seed=11)
+    cfg = GmmConfig(num_clusters=30, target_num_clusters=10,
+                    min_iters=4, max_iters=4)
+    res = build_engine(data, cfg, device="cpu").sweep()
+    assert res.num_clusters == 10
