@@ -155,10 +155,10 @@ def constants(r: torch.Tensor, means: torch.Tensor | None = None,
 def estep_fused(z: torch.Tensor, mfac: torch.Tensor, add: torch.Tensor,
                 w_out: torch.Tensor) -> tuple[torch.Tensor, torch.Tensor]:
     """Fused bf16-MFMA E-step (CUDA only): posteriors into w_out [K,N] and
-    the shard-partial likelihood. One pass over the data; logw never
-    touches HBM."""
+    the shard-partial likelihood. Online-softmax over 256-event blocks:
+    logw is staged through w_out (L2-hot) and normalized in-block."""
     n = z.shape[1]
-    nblk = (n + 127) // 128
+    nblk = (n + 255) // 256
     # every launched block writes its partial slot: no zero-fill needed
     partial = torch.empty(nblk, dtype=torch.float32, device=z.device)
     hip_ext().estep_fused(z, mfac, add, w_out, partial)
@@ -167,19 +167,16 @@ def estep_fused(z: torch.Tensor, mfac: torch.Tensor, add: torch.Tensor,
 
 def estep_fused_available(device: torch.device, dtype: str, d: int,
                           k: int) -> bool:
-    """Fused path gate (mirrors the LDS checks in gmm_ext.hip)."""
-    if device.type != "cuda" or d > 31:
-        return False
-    if dtype == "bf16":
-        return 128 * 40 * 2 + 4 * k * 132 <= 64 * 1024
-    return 4 * (128 * 33 + k * 132) <= 64 * 1024  # exact-f32 variant
+    """Fused path gate: D <= 31 (any K — the online-softmax redesign
+    removed the K-sized logw LDS buffer)."""
+    return device.type == "cuda" and d <= 31
 
 
 def estep_fused_f32(z: torch.Tensor, mfac32: torch.Tensor, add: torch.Tensor,
                     w_out: torch.Tensor) -> tuple[torch.Tensor, torch.Tensor]:
-    """Exact-f32 MFMA fused E-step (CUDA, D <= 31)."""
+    """Exact-f32 MFMA fused E-step (CUDA, D <= 31, any K)."""
     n = z.shape[1]
-    nblk = (n + 127) // 128
+    nblk = (n + 255) // 256
     partial = torch.empty(nblk, dtype=torch.float32, device=z.device)
     hip_ext().estep_fused_f32(z, mfac32, add, w_out, partial)
     return w_out, partial.sum()

@@ -294,11 +294,11 @@ void estep_fused(torch::Tensor z, torch::Tensor mfac, torch::Tensor add,
   TORCH_CHECK(d <= 31, "estep_fused needs D <= 31");
   TORCH_CHECK(mfac.numel() >= (int64_t)k * 2 * 32 * 32, "mfac too small");
   TORCH_CHECK(w_out.size(0) == k && w_out.size(1) == n, "w_out shape");
-  const int64_t nblk = (n + 128 - 1) / 128;
+  const int64_t nblk = (n + 256 - 1) / 256;
   TORCH_CHECK(partial.size(0) >= nblk, "partial buffer too small");
-  const size_t zbytes = (size_t)128 * 40 * 2;  // transposed z tile
-  const size_t lds = zbytes + sizeof(float) * (size_t)k * (128 + 4);
-  TORCH_CHECK(lds <= 64 * 1024, "estep_fused LDS budget exceeded (K too big)");
+  // transposed z tile + per-wave online-softmax state + lse (K-independent)
+  const size_t lds = (size_t)256 * 40 * 2 +
+                     sizeof(float) * (2 * (256 / 64) * 256 + 256);
   hipLaunchKernelGGL(gmm::estep_fused_kernel, dim3((uint32_t)nblk), dim3(kNT),
                      lds, stream(),
                      reinterpret_cast<const __hip_bfloat16*>(z.data_ptr()),
@@ -436,11 +436,10 @@ void estep_fused_f32(torch::Tensor z, torch::Tensor mfac32,
   TORCH_CHECK(d <= 31, "estep_fused_f32 needs D <= 31");
   TORCH_CHECK(mfac32.numel() >= (int64_t)k * 32 * 32, "mfac32 too small");
   TORCH_CHECK(w_out.size(0) == k && w_out.size(1) == n, "w_out shape");
-  const int64_t nblk = (n + 128 - 1) / 128;
+  const int64_t nblk = (n + 256 - 1) / 256;
   TORCH_CHECK(partial.size(0) >= nblk, "partial buffer too small");
   const size_t lds =
-      sizeof(float) * ((size_t)128 * 33 + (size_t)k * (128 + 4));
-  TORCH_CHECK(lds <= 64 * 1024, "estep_fused_f32 LDS budget exceeded");
+      sizeof(float) * ((size_t)256 * 33 + 2 * (256 / 64) * 256 + 256);
   hipLaunchKernelGGL(gmm::estep_fused_f32_kernel, dim3((uint32_t)nblk),
                      dim3(kNT), lds, stream(), z.data_ptr<float>(),
                      mfac32.data_ptr<float>(), add.data_ptr<float>(),

@@ -909,17 +909,23 @@ mstep_moments_b16_kernel(const __hip_bfloat16* __restrict__ xhi,
 
 // ---------------------------------------------------------------------------
 // MFMA fused E-step (replaces estep1 + estep2 in one pass for bf16, D<=31,
-// moderate K): logw via  q = ||M_c z||^2  with M_c = [U_c | -U_c mu_c],
+// any K): logw via  q = ||M_c z||^2  with M_c = [U_c | -U_c mu_c],
 // U_c^T U_c = Rinv_c (Cholesky, emitted by the constants kernel), split into
 // bf16 hi+lo parts so accuracy matches a bf16-data fp32-VALU quadratic form.
-// Per block: stage z tile once, loop clusters (B fragments built once and
-// reused across all K!), logw kept in LDS, then in-block posteriors +
-// likelihood partial — memberships are written ONCE and logw never touches
-// HBM (vs 3 reads + 2 writes of [K,N] in the two-kernel path).
+//
+// ONLINE-SOFTMAX redesign (round 2): 256-event blocks with NO logw-in-LDS
+// buffer. Each wave walks its cluster slice keeping a running per-event
+// (max, sum) pair in 8 KB of LDS; logw goes to w_out as scratch; after a
+// cross-wave combine producing the per-event log-sum-exp, a third in-block
+// pass reads the logw back (L2-hot: this block just wrote it) and writes
+// normalized posteriors. vs the 128-event version: HALF the factor-table
+// L2 traffic (the measured E-step floor) and 29.7 KB blocks -> 5
+// blocks/CU instead of 3 (the K*(BE+4) f32 logw buffer is gone, so K is
+// no longer LDS-bounded).
 // v_mfma_f32_16x16x32_bf16: A lane l -> A[i=l&15][k=8*(l>>4)+u];
-// B lane l -> B[k=8*(l>>4)+u][j=l&15]; C/D col=l&15, row=(l>>4)*4+reg.
+// B lane l -> B[k=8*(l>>4)+u][j=l&15]; C/D col=l&31 etc (guide §3).
 // ---------------------------------------------------------------------------
-#define EST_BE 128
+#define EST_BE 256
 
 
 
@@ -972,19 +978,22 @@ estep_fused_kernel(const __hip_bfloat16* __restrict__ z,
                    int d, int k, int64_t n) {
   // LDS: zs_t [EST_BE][EST_ZROW] bf16 — z staged TRANSPOSED (k-major per
   // event, ones-row and zero-pad baked in) so a B fragment is a single
-  // 16-byte ds_read_b128; then lw [k][EST_BE+4] f32.
+  // 16-byte ds_read_b128; then per-wave online-softmax state
+  // m/s [nwaves][EST_BE] f32 and the combined lse [EST_BE].
   extern __shared__ float lds[];
-  const int lrow = EST_BE + 4;
   __hip_bfloat16* zs = (__hip_bfloat16*)lds;
-  float* lw = lds + (EST_BE * EST_ZROW) / 2;
+  float* mstate = lds + (EST_BE * EST_ZROW) / 2;   // [nwaves][EST_BE]
+  float* sstate = mstate + (NT / WAVE) * EST_BE;   // [nwaves][EST_BE]
+  float* lse = sstate + (NT / WAVE) * EST_BE;      // [EST_BE]
 
   const int wave = threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
   const int64_t e0 = (int64_t)blockIdx.x * EST_BE;
   const int cnt = (int)min((int64_t)EST_BE, n - e0);
+  const bool full = cnt == EST_BE;
 
   __bf16* zsb = (__bf16*)zs;
-  if (cnt == EST_BE) {
+  if (full) {
     // branchless staging (guide §5 trap 4c): coalesced reads of the d data
     // rows, transposed scatter into LDS; then the constant rows
     for (int idx = threadIdx.x; idx < d * EST_BE; idx += blockDim.x) {
@@ -1004,12 +1013,16 @@ estep_fused_kernel(const __hip_bfloat16* __restrict__ z,
     zsb[ei * EST_ZROW + kk] =
         (__bf16)((kk == d && ei < cnt) ? 1.0f : 0.0f);
   }
+  for (int i = threadIdx.x; i < (NT / WAVE) * EST_BE; i += blockDim.x) {
+    mstate[i] = -3.0e38f;
+    sstate[i] = 0.0f;
+  }
   __syncthreads();
 
   // 32x32x16 bf16 MFMA: one 32-row tile covers all of M (Dp <= 32) and 32
   // events; WAVES SPLIT THE CLUSTER LOOP (c = wave, wave+4, ...) so each
   // cluster's factor fragments are fetched once per block, not once per
-  // wave — the A-fragment L2 traffic was the previous bottleneck.
+  // wave — the A-fragment L2 traffic is the measured E-step floor.
   // A lane l -> A[i=l&31][kk=8*(l>>5)+u] per 16-deep chunk; B lane l ->
   // B[kk][j=l&31]; C/D col=l&31, row=(reg&3)+8*(reg>>2)+4*(l>>5) (guide §3).
   const int j32 = lane & 31;
@@ -1019,6 +1032,8 @@ estep_fused_kernel(const __hip_bfloat16* __restrict__ z,
   const int fq0 = g2;      // chunk 0 covers k [0,16): slots {0,1}
   const int fq1 = 2 + g2;  // chunk 1 covers k [16,32): slots {2,3}
   const int nwaves = NT / WAVE;
+  float* mrow = mstate + wave * EST_BE;
+  float* srow = sstate + wave * EST_BE;
 
   bf16x8 nx_h0, nx_l0, nx_h1, nx_l1;
   float nx_add;
@@ -1055,74 +1070,76 @@ estep_fused_kernel(const __hip_bfloat16* __restrict__ z,
       for (int r = 0; r < 16; ++r) s = fmaf(y[r], y[r], s);
       // the 32 Y rows live across the two lane halves: one cross-half sum
       s += __shfl_xor(s, 32, WAVE);
-      if (lane < 32) lw[c * lrow + t * 32 + j32] = -0.5f * s + addc;
+      if (lane < 32) {
+        const int e = t * 32 + j32;
+        const float lwv = -0.5f * s + addc;
+        // logw to w_out as scratch (this block's slab stays L2-hot for
+        // the normalize pass); lane j32 owns event e exclusively within
+        // this wave, so the state update is race-free
+        if (full || e < cnt) w_out[(int64_t)c * n + e0 + e] = lwv;
+        const float mo = mrow[e];
+        if (lwv > mo) {
+          srow[e] = srow[e] * __expf(mo - lwv) + 1.0f;
+          mrow[e] = lwv;
+        } else {
+          srow[e] += __expf(lwv - mo);
+        }
+      }
     }
   }
   __syncthreads();
 
-  // pass 2: posteriors + likelihood, cluster loop split across BOTH
-  // thread halves (the serial 64-deep store loop was 24% of the kernel —
-  // store-issue-bound): threads t and t+EST_BE each handle half the
-  // clusters of event t, combining max/sum through LDS.
-  __shared__ float pmax[2][EST_BE];
-  __shared__ float psum[2][EST_BE];
-  float acc = 0.0f;
+  // combine the per-wave (m, s) into the per-event log-sum-exp
+  // (EST_BE == NT: each thread owns one event)
   {
-    const int t = threadIdx.x & (EST_BE - 1);
-    const int half = threadIdx.x >> 7;       // EST_BE == 128
-    const int mid = (k + 1) / 2;   // half 0 never empty (K=1, odd K)
-    const int c_lo = half * mid;
-    const int c_hi = half ? k : mid;
-    if (t < cnt && c_lo < c_hi) {
-      float m = lw[c_lo * lrow + t];
-#pragma unroll 4
-      for (int c = c_lo + 1; c < c_hi; ++c)
-        m = fmaxf(m, lw[c * lrow + t]);
-      pmax[half][t] = m;
-    } else if (t < EST_BE) {
-      pmax[half][t] = -3.0e38f;
-    }
+    const int e = threadIdx.x;
+    float m = mstate[e];
+#pragma unroll
+    for (int wv = 1; wv < nwaves; ++wv)
+      m = fmaxf(m, mstate[wv * EST_BE + e]);
+    float ssum = 0.0f;
+#pragma unroll
+    for (int wv = 0; wv < nwaves; ++wv)
+      ssum += sstate[wv * EST_BE + e] * __expf(mstate[wv * EST_BE + e] - m);
+    lse[e] = m + __logf(ssum);
+  }
+  __syncthreads();
+
+  // likelihood partial: sum of lse over live events
+  {
+    float acc = (threadIdx.x < cnt) ? lse[threadIdx.x] : 0.0f;
+    __shared__ float wsum[NT / WAVE];
+    for (int off = WAVE / 2; off > 0; off >>= 1)
+      acc += __shfl_down(acc, off, WAVE);
+    if (lane == 0) wsum[wave] = acc;
     __syncthreads();
-    const float m = fmaxf(pmax[0][t], pmax[1][t]);
-    float s = 0.0f;
-    if (t < cnt) {
-#pragma unroll 4
-      for (int c = c_lo; c < c_hi; ++c) {
-        const float e = __expf(lw[c * lrow + t] - m);
-        lw[c * lrow + t] = e;
-        s += e;
-      }
-    }
-    psum[half][t] = s;
-    __syncthreads();
-    const float total = psum[0][t] + psum[1][t];
-    if (t < cnt && c_lo < c_hi) {
-      const float inv = 1.0f / total;
-#pragma unroll 4
-      for (int c = c_lo; c < c_hi; ++c)
-        w_out[(int64_t)c * n + e0 + t] = lw[c * lrow + t] * inv;
-      if (half == 0) acc = m + __logf(total);
+    if (threadIdx.x == 0) {
+      float total = 0.0f;
+      for (int wv = 0; wv < NT / WAVE; ++wv) total += wsum[wv];
+      partial[blockIdx.x] = total;
     }
   }
-  __shared__ float wsum[NT / WAVE];
-  for (int off = WAVE / 2; off > 0; off >>= 1)
-    acc += __shfl_down(acc, off, WAVE);
-  if (lane == 0) wsum[wave] = acc;
-  __syncthreads();
-  if (threadIdx.x == 0) {
-    float total = 0.0f;
-    for (int wv = 0; wv < NT / WAVE; ++wv) total += wsum[wv];
-    partial[blockIdx.x] = total;
+
+  // normalize pass: read the scratch logw back (L2-hot) and overwrite with
+  // posteriors. Flat block-wide mapping (EST_BE == NT): thread t handles
+  // event t of every cluster row — fully coalesced 1 KB accesses.
+  const int e = threadIdx.x;
+  if (e < cnt) {
+    const float l = lse[e];
+    for (int c = 0; c < k; ++c) {
+      const int64_t idx = (int64_t)c * n + e0 + e;
+      w_out[idx] = __expf(w_out[idx] - l);
+    }
   }
 }
 
 // ---------------------------------------------------------------------------
-// Exact-fp32 fused E-step (D <= 31): same structure as estep_fused_kernel
-// but on v_mfma_f32_32x32x2_f32 — f32 in / f32 accumulate, bitwise an fmaf
-// chain (guide §3), consuming the fp32 factor plane. 16 dependent MFMAs
-// per (cluster, 32-event tile); issue interval == dependent latency (64),
-// so the chain runs at the f32 matrix rate. This makes the CLI's default
-// exact mode ~4x faster than the VALU quadratic-form path.
+// Exact-fp32 fused E-step (D <= 31): same online-softmax structure as
+// estep_fused_kernel but on v_mfma_f32_32x32x2_f32 — f32 in / f32
+// accumulate, bitwise an fmaf chain (guide §3), consuming the fp32 factor
+// plane. 16 dependent MFMAs per (cluster, 32-event tile); issue interval
+// == dependent latency (64), so the chain runs at the f32 matrix rate.
+// 256-event blocks, no logw LDS buffer: K is unbounded (42.8 KB LDS).
 // ---------------------------------------------------------------------------
 __global__ void __launch_bounds__(NT)
 estep_fused_f32_kernel(const float* __restrict__ z,
@@ -1133,9 +1150,10 @@ estep_fused_f32_kernel(const float* __restrict__ z,
   typedef __attribute__((ext_vector_type(16))) float f32x16;
   constexpr int ZR = 33;  // f32 slots per transposed event row (32 + pad)
   extern __shared__ float lds[];
-  float* zs = lds;                         // [EST_BE][ZR]
-  const int lrow = EST_BE + 4;
-  float* lw = lds + EST_BE * ZR;           // [k][lrow]
+  float* zs = lds;                                 // [EST_BE][ZR]
+  float* mstate = lds + EST_BE * ZR;               // [nwaves][EST_BE]
+  float* sstate = mstate + (NT / WAVE) * EST_BE;   // [nwaves][EST_BE]
+  float* lse = sstate + (NT / WAVE) * EST_BE;      // [EST_BE]
 
   const int wave = threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
@@ -1143,8 +1161,9 @@ estep_fused_f32_kernel(const float* __restrict__ z,
   const int g2 = lane >> 5;
   const int64_t e0 = (int64_t)blockIdx.x * EST_BE;
   const int cnt = (int)min((int64_t)EST_BE, n - e0);
+  const bool full = cnt == EST_BE;
 
-  if (cnt == EST_BE) {
+  if (full) {
     for (int idx = threadIdx.x; idx < d * EST_BE; idx += blockDim.x) {
       const int kk = idx / EST_BE, ei = idx % EST_BE;
       zs[ei * ZR + kk] = z[(int64_t)kk * n + e0 + ei];
@@ -1160,9 +1179,15 @@ estep_fused_f32_kernel(const float* __restrict__ z,
     const int kk = d + idx / EST_BE, ei = idx % EST_BE;
     zs[ei * ZR + kk] = (kk == d && ei < cnt) ? 1.0f : 0.0f;
   }
+  for (int i = threadIdx.x; i < (NT / WAVE) * EST_BE; i += blockDim.x) {
+    mstate[i] = -3.0e38f;
+    sstate[i] = 0.0f;
+  }
   __syncthreads();
 
   const int nwaves = NT / WAVE;
+  float* mrow = mstate + wave * EST_BE;
+  float* srow = sstate + wave * EST_BE;
 
   // prefetched A rows: 16 f32 per lane (row j32, k-slots 2*ch + g2)
   float nx_a[16];
@@ -1194,40 +1219,59 @@ estep_fused_f32_kernel(const float* __restrict__ z,
 #pragma unroll
       for (int r = 0; r < 16; ++r) s = fmaf(y[r], y[r], s);
       s += __shfl_xor(s, 32, WAVE);
-      if (lane < 32) lw[c * lrow + t * 32 + j32] = -0.5f * s + addc;
+      if (lane < 32) {
+        const int e = t * 32 + j32;
+        const float lwv = -0.5f * s + addc;
+        if (full || e < cnt) w_out[(int64_t)c * n + e0 + e] = lwv;
+        const float mo = mrow[e];
+        if (lwv > mo) {
+          srow[e] = srow[e] * __expf(mo - lwv) + 1.0f;
+          mrow[e] = lwv;
+        } else {
+          srow[e] += __expf(lwv - mo);
+        }
+      }
     }
   }
   __syncthreads();
 
-  // pass 2: identical to the bf16 kernel
-  float acc = 0.0f;
-  if (threadIdx.x < EST_BE && threadIdx.x < cnt) {
-    const int t = threadIdx.x;
-    float m = lw[t];
-#pragma unroll 4
-    for (int c = 1; c < k; ++c) m = fmaxf(m, lw[c * lrow + t]);
-    float s = 0.0f;
-#pragma unroll 4
-    for (int c = 0; c < k; ++c) {
-      const float e = __expf(lw[c * lrow + t] - m);
-      lw[c * lrow + t] = e;
-      s += e;
-    }
-    const float inv = 1.0f / s;
-#pragma unroll 4
-    for (int c = 0; c < k; ++c)
-      w_out[(int64_t)c * n + e0 + t] = lw[c * lrow + t] * inv;
-    acc = m + __logf(s);
+  // cross-wave combine -> per-event log-sum-exp (EST_BE == NT)
+  {
+    const int e = threadIdx.x;
+    float m = mstate[e];
+#pragma unroll
+    for (int wv = 1; wv < nwaves; ++wv)
+      m = fmaxf(m, mstate[wv * EST_BE + e]);
+    float ssum = 0.0f;
+#pragma unroll
+    for (int wv = 0; wv < nwaves; ++wv)
+      ssum += sstate[wv * EST_BE + e] * __expf(mstate[wv * EST_BE + e] - m);
+    lse[e] = m + __logf(ssum);
   }
-  __shared__ float wsum2[NT / WAVE];
-  for (int off = WAVE / 2; off > 0; off >>= 1)
-    acc += __shfl_down(acc, off, WAVE);
-  if (lane == 0) wsum2[wave] = acc;
   __syncthreads();
-  if (threadIdx.x == 0) {
-    float total = 0.0f;
-    for (int wv = 0; wv < NT / WAVE; ++wv) total += wsum2[wv];
-    partial[blockIdx.x] = total;
+
+  {
+    float acc = (threadIdx.x < cnt) ? lse[threadIdx.x] : 0.0f;
+    __shared__ float wsum2[NT / WAVE];
+    for (int off = WAVE / 2; off > 0; off >>= 1)
+      acc += __shfl_down(acc, off, WAVE);
+    if (lane == 0) wsum2[wave] = acc;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      float total = 0.0f;
+      for (int wv = 0; wv < NT / WAVE; ++wv) total += wsum2[wv];
+      partial[blockIdx.x] = total;
+    }
+  }
+
+  // normalize pass over the L2-hot scratch logw
+  const int e = threadIdx.x;
+  if (e < cnt) {
+    const float l = lse[e];
+    for (int c = 0; c < k; ++c) {
+      const int64_t idx = (int64_t)c * n + e0 + e;
+      w_out[idx] = __expf(w_out[idx] - l);
+    }
   }
 }
 
