@@ -79,26 +79,30 @@ class EmEngine:
         )
 
         k0 = config.num_clusters
-        # fused MFMA E-step paths (D <= 31: bf16 split-factor or exact-f32;
-        # 31 < D <= 143: big-D bf16 logw); otherwise the VALU kernels
+        # MFMA E-step paths (D <= 31: fused online-softmax, bf16 or
+        # exact-f32, any K; 31 < D <= 143: big-D logw, bf16 or exact-f32);
+        # otherwise the VALU kernels. DIAG_ONLY routes through the same
+        # factor path: a diagonal Rinv yields a diagonal Cholesky factor,
+        # so q = ||Uz+u0||^2 equals the diagonal quadratic form exactly.
         self.use_fused_estep = F.estep_fused_available(
             self.device, config.estep_dtype, self.d, k0,
-        ) and not config.diag_only
+        )
         self.use_big_estep = F.estep_big_available(
             self.device, config.estep_dtype, self.d,
-        ) and not config.diag_only
-        need_bf16_fac = (
-            (self.use_fused_estep and config.estep_dtype == "bf16")
-            or self.use_big_estep
-        )
-        need_f32_fac = self.use_fused_estep and config.estep_dtype == "fp32"
+        ) and not self.use_fused_estep
+        need_fac = self.use_fused_estep or self.use_big_estep
+        need_f32_fac = need_fac and config.estep_dtype == "fp32"
+        # the bf16 factor plane always exists when factors are emitted (it
+        # is what triggers emission in the constants kernel); the f32 plane
+        # is added for the exact-f32 MFMA kernels
         self.mfac = (
             torch.empty(k0, *F.mfac_shape(self.d), dtype=torch.bfloat16,
                         device=self.device)
-            if (need_bf16_fac or need_f32_fac) else None
+            if need_fac else None
         )
         self.mfac32 = (
-            torch.empty(k0, 32, 32, dtype=torch.float32, device=self.device)
+            torch.empty(k0, *F.mfac_shape(self.d)[1:], dtype=torch.float32,
+                        device=self.device)
             if need_f32_fac else None
         )
 
@@ -117,9 +121,17 @@ class EmEngine:
         # gaussian.cu:404); also fills the add buffer
         self._update_constants(self.state)
 
-        # membership / logw buffer, cluster-major [K, n_shard]
+        # membership / logw buffer, cluster-major [K, n_shard]. On CUDA
+        # this holds LOG weights after the E-step, with the per-event
+        # log-sum-exp in self._lse — the lse-aware M-step kernels apply
+        # exp(logw - lse) on the fly, so posteriors are never materialized
+        # during the sweep (use .posteriors() for normalized values).
         self.w = torch.empty(k0, self.n_shard, dtype=torch.float32,
                              device=self.device)
+        self._lse = (
+            torch.empty(self.n_shard, dtype=torch.float32, device=self.device)
+            if self.device.type == "cuda" else None
+        )
         self.epsilon = em_epsilon(self.d, num_events_total)
         self.likelihood = 0.0
         # hipGraph capture of the EM iteration (launch-overhead elimination);
@@ -180,15 +192,26 @@ class EmEngine:
                 add = self._add[:k]
                 if self.mfac32 is not None:
                     w, lik = F.estep_fused_f32(self.x_estep, self.mfac32[:k],
-                                               add, self.w[:k])
+                                               add, self.w[:k], self._lse)
                 else:
                     w, lik = F.estep_fused(self.x_estep, self.mfac[:k], add,
-                                           self.w[:k])
+                                           self.w[:k], self._lse)
             elif self.use_big_estep:
                 add = self._add[:k]
-                logw = F.estep_logw_big(self.x_estep, self.mfac[:k], add,
-                                        self.w[:k])
-                w, lik = F.estep_posteriors(logw)
+                if self.mfac32 is not None:
+                    logw = F.estep_logw_big_f32(self.x_estep,
+                                                self.mfac32[:k], add,
+                                                self.w[:k])
+                else:
+                    logw = F.estep_logw_big(self.x_estep, self.mfac[:k],
+                                            add, self.w[:k])
+                lik = F.estep_lse(logw, self._lse)
+            elif self.device.type == "cuda":
+                logw = F.estep_logw(
+                    self.x_estep, st.means, st.Rinv, st.constant, st.pi,
+                    self.cfg.diag_only, out=self.w[:k],
+                )
+                lik = F.estep_lse(logw, self._lse)
             else:
                 logw = F.estep_logw(
                     self.x_estep, st.means, st.Rinv, st.constant, st.pi,
@@ -240,7 +263,7 @@ class EmEngine:
         with self.profile.time("m_step"):
             packed = F.mstep_moments(self.x, self.w[:k],
                                      precision=self.cfg.mstep_precision,
-                                     x_split=self.x_split)
+                                     x_split=self.x_split, lse=self._lse)
         with self.profile.time("comm"):
             pdist.all_reduce_(packed)
         with self.profile.time("m_step"):
@@ -549,6 +572,13 @@ class EmEngine:
             min_rissanen=min_rissanen, likelihood=best_lik,
             rissanen_by_k=riss_by_k,
         )
+
+    def posteriors(self, k: int) -> torch.Tensor:
+        """Normalized posteriors [k, n_shard] for the current E-step state
+        (on CUDA the w buffer holds log weights; normalize with the lse)."""
+        if self._lse is None:
+            return self.w[:k]
+        return torch.exp(self.w[:k] - self._lse.unsqueeze(0))
 
     # ------------------------------------------------------------- output
 

@@ -48,6 +48,18 @@ def estep_posteriors(logw: torch.Tensor) -> tuple[torch.Tensor, torch.Tensor]:
     return logw, lik
 
 
+def estep_lse(logw: torch.Tensor, lse: torch.Tensor) -> torch.Tensor:
+    """Per-event log-sum-exp into lse [N] + likelihood scalar (CUDA).
+
+    logw is NOT normalized — the lse-aware M-step kernels apply
+    exp(logw - lse) while staging w, deleting the posterior write+read
+    round trip of the two-kernel path."""
+    nblocks = 1024
+    partial = torch.zeros(nblocks, dtype=torch.float32, device=logw.device)
+    hip_ext().estep_lse(logw, lse, partial)
+    return partial.sum()
+
+
 def mstep_n_means(x_aug_t: torch.Tensor, w: torch.Tensor
                   ) -> tuple[torch.Tensor, torch.Tensor]:
     """(N [K], mean numerators [K, D]) via one rocBLAS GEMM.
@@ -80,7 +92,8 @@ def _tri_unpack_index(d: int, device) -> torch.Tensor:
 
 def mstep_covariance_s(x: torch.Tensor, w: torch.Tensor,
                        out: torch.Tensor | None = None,
-                       nchunk: int = 64) -> torch.Tensor:
+                       nchunk: int = 64,
+                       lse: torch.Tensor | None = None) -> torch.Tensor:
     """Uncentered weighted second moments S [K, D, D].
 
     CUDA: custom LDS-tiled kernel (the covariance showpiece) producing
@@ -97,7 +110,9 @@ def mstep_covariance_s(x: torch.Tensor, w: torch.Tensor,
         nchunk = int(min(nchunk, tiles))
         partials = torch.empty((nchunk, k, p), dtype=torch.float32,
                                device=x.device)
-        hip_ext().mstep_covariance_partials(x, w, partials)
+        if lse is None:
+            lse = torch.empty(0, dtype=torch.float32, device=x.device)
+        hip_ext().mstep_covariance_partials(x, w, lse, partials)
         packed = partials.sum(dim=0)                      # [K, P]
         idx = _tri_unpack_index(d, x.device)
         s = packed[:, idx].view(k, d, d)
@@ -153,15 +168,17 @@ def constants(r: torch.Tensor, means: torch.Tensor | None = None,
 
 
 def estep_fused(z: torch.Tensor, mfac: torch.Tensor, add: torch.Tensor,
-                w_out: torch.Tensor) -> tuple[torch.Tensor, torch.Tensor]:
-    """Fused bf16-MFMA E-step (CUDA only): posteriors into w_out [K,N] and
-    the shard-partial likelihood. Online-softmax over 256-event blocks:
-    logw is staged through w_out (L2-hot) and normalized in-block."""
+                w_out: torch.Tensor, lse: torch.Tensor
+                ) -> tuple[torch.Tensor, torch.Tensor]:
+    """Fused bf16-MFMA E-step (CUDA only): LOG weights into w_out [K,N],
+    per-event log-sum-exp into lse [N], returns (w_out, likelihood).
+    Online-softmax over 256-event blocks; the lse-aware M-step applies
+    exp(w_out - lse) on the fly — posteriors are never materialized."""
     n = z.shape[1]
     nblk = (n + 255) // 256
     # every launched block writes its partial slot: no zero-fill needed
     partial = torch.empty(nblk, dtype=torch.float32, device=z.device)
-    hip_ext().estep_fused(z, mfac, add, w_out, partial)
+    hip_ext().estep_fused(z, mfac, add, w_out, lse, partial)
     return w_out, partial.sum()
 
 
@@ -173,20 +190,21 @@ def estep_fused_available(device: torch.device, dtype: str, d: int,
 
 
 def estep_fused_f32(z: torch.Tensor, mfac32: torch.Tensor, add: torch.Tensor,
-                    w_out: torch.Tensor) -> tuple[torch.Tensor, torch.Tensor]:
-    """Exact-f32 MFMA fused E-step (CUDA, D <= 31, any K)."""
+                    w_out: torch.Tensor, lse: torch.Tensor
+                    ) -> tuple[torch.Tensor, torch.Tensor]:
+    """Exact-f32 MFMA fused E-step (CUDA, D <= 31, any K): log weights
+    into w_out, per-event log-sum-exp into lse."""
     n = z.shape[1]
     nblk = (n + 255) // 256
     partial = torch.empty(nblk, dtype=torch.float32, device=z.device)
-    hip_ext().estep_fused_f32(z, mfac32, add, w_out, partial)
+    hip_ext().estep_fused_f32(z, mfac32, add, w_out, lse, partial)
     return w_out, partial.sum()
 
 
 def estep_big_available(device: torch.device, dtype: str, d: int) -> bool:
-    """Big-D MFMA logw path gate (bf16 data, D <= 143). The engine prefers
-    the fused kernel when its LDS gate admits K; this path also covers
-    D <= 31 with large K (posteriors via the separate kernel)."""
-    return device.type == "cuda" and dtype == "bf16" and 1 <= d <= 143
+    """Big-D MFMA logw path gate (D <= 143, bf16 or exact-f32 MFMA). The
+    engine prefers the fused kernel for D <= 31."""
+    return device.type == "cuda" and dtype in ("bf16", "fp32") and 1 <= d <= 143
 
 
 def mfac_shape(d: int) -> tuple[int, int, int]:
@@ -204,6 +222,13 @@ def estep_logw_big(z: torch.Tensor, mfac: torch.Tensor, add: torch.Tensor,
     return out
 
 
+def estep_logw_big_f32(z: torch.Tensor, mfac32: torch.Tensor,
+                       add: torch.Tensor, out: torch.Tensor) -> torch.Tensor:
+    """Exact-f32 big-D MFMA log-weights into out [K, N] (CUDA, D > 31)."""
+    hip_ext().estep_logw_big_f32(z, mfac32, add, out)
+    return out
+
+
 def split_bf16_planes(x: torch.Tensor) -> tuple[torch.Tensor, torch.Tensor]:
     """Persistent hi/lo bf16 split of fp32 data (x = hi + lo + O(2^-18 x))."""
     hi = x.to(torch.bfloat16)
@@ -214,8 +239,8 @@ def split_bf16_planes(x: torch.Tensor) -> tuple[torch.Tensor, torch.Tensor]:
 def mstep_moments(x: torch.Tensor, w: torch.Tensor,
                   nchunk: int | None = None,
                   precision: str = "fp32",
-                  x_split: tuple[torch.Tensor, torch.Tensor] | None = None
-                  ) -> torch.Tensor:
+                  x_split: tuple[torch.Tensor, torch.Tensor] | None = None,
+                  lse: torch.Tensor | None = None) -> torch.Tensor:
     """Fused augmented sufficient statistics, packed lower triangle of
     T_c = sum_e w_ce [x;1][x;1]^T per cluster: [K, Dp*(Dp+1)/2] with layout
     [S_tri (D rows) | mean numerators (row D) | N (corner)] — the single
@@ -240,12 +265,15 @@ def mstep_moments(x: torch.Tensor, w: torch.Tensor,
         nchunk = int(min(nchunk, tiles))
         partials = torch.empty((nchunk, k, pp), dtype=torch.float32,
                                device=x.device)
+        lse_t = (lse if lse is not None
+                 else torch.empty(0, dtype=torch.float32, device=x.device))
         if precision == "bf16x3":
             if x_split is None:
                 x_split = split_bf16_planes(x)
-            hip_ext().mstep_moments_b16(x_split[0], x_split[1], w, partials)
+            hip_ext().mstep_moments_b16(x_split[0], x_split[1], w, lse_t,
+                                        partials)
         else:
-            hip_ext().mstep_moments(x, w, partials)
+            hip_ext().mstep_moments(x, w, lse_t, partials)
         return partials.sum(dim=0)
     if x.is_cuda and precision == "bf16x3" and d <= 159:
         tiles = (n + 63) // 64  # MBB_BK
@@ -258,10 +286,17 @@ def mstep_moments(x: torch.Tensor, w: torch.Tensor,
         nchunk = int(min(nchunk, tiles))
         partials = torch.empty((nchunk, k, pp), dtype=torch.float32,
                                device=x.device)
-        hip_ext().mstep_moments_big(x, w, partials)
+        lse_t = (lse if lse is not None
+                 else torch.empty(0, dtype=torch.float32, device=x.device))
+        hip_ext().mstep_moments_big(x, w, lse_t, partials)
         return partials.sum(dim=0)
     packed = torch.empty((k, pp), dtype=torch.float32, device=x.device)
     if x.is_cuda:
+        if lse is not None:
+            # fallback path mixes rocBLAS GEMMs that need materialized
+            # posteriors: normalize w in place (it is rewritten by the
+            # next E-step)
+            w.sub_(lse.unsqueeze(0)).exp_()
         s = mstep_covariance_s(x, w)        # custom kernel, [K, D, D]
         mean_num = w @ x.T                  # rocBLAS
         n_c = w.sum(dim=1)

@@ -32,6 +32,15 @@ int grid_x_for(int64_t n) {
   return (int)std::min<int64_t>(tiles, 16384);
 }
 
+
+const float* lse_ptr(const torch::Tensor& lse, int64_t n) {
+  if (lse.numel() == 0) return nullptr;
+  TORCH_CHECK(lse.is_cuda() && lse.is_contiguous() &&
+                  lse.scalar_type() == torch::kFloat32 && lse.numel() == n,
+              "lse must be contiguous fp32 [N]");
+  return lse.data_ptr<float>();
+}
+
 template <typename T>
 const T* data_as(const torch::Tensor& t) {
   return reinterpret_cast<const T*>(t.data_ptr());
@@ -114,9 +123,25 @@ void estep_posteriors(torch::Tensor logw, torch::Tensor partial) {
   HIP_CHECK(hipGetLastError());
 }
 
+void estep_lse(torch::Tensor logw, torch::Tensor lse,
+               torch::Tensor partial) {
+  check_f32(logw, "logw");
+  check_f32(lse, "lse");
+  check_f32(partial, "partial");
+  const int k = (int)logw.size(0);
+  const int64_t n = logw.size(1);
+  TORCH_CHECK(lse.numel() == n, "lse must be [N]");
+  int grid = (int)std::min<int64_t>((n + kNT - 1) / kNT, partial.size(0));
+  TORCH_CHECK(grid >= 1, "empty logw");
+  hipLaunchKernelGGL(gmm::estep_lse_kernel, dim3(grid), dim3(kNT), 0,
+                     stream(), logw.data_ptr<float>(), lse.data_ptr<float>(),
+                     partial.data_ptr<float>(), k, n);
+  HIP_CHECK(hipGetLastError());
+}
+
 template <typename T>
 void mstep_cov_impl(const torch::Tensor& x, const torch::Tensor& w,
-                    torch::Tensor& partials) {
+                    const torch::Tensor& lse, torch::Tensor& partials) {
   const int d = (int)x.size(0);
   const int64_t n = x.size(1);
   const int k = (int)w.size(0);
@@ -135,10 +160,11 @@ void mstep_cov_impl(const torch::Tensor& x, const torch::Tensor& w,
   const int ppt = (p + kNT - 1) / kNT;
   const T* xp = data_as<T>(x);
   const float* wp = w.data_ptr<float>();
+  const float* lp = lse_ptr(lse, n);
   float* pp = partials.data_ptr<float>();
 #define LAUNCH_COV(PPT)                                                     \
   hipLaunchKernelGGL((gmm::mstep_cov_kernel<PPT, T>), grid, dim3(kNT), lds, \
-                     s, xp, wp, pp, d, k, n, te, nchunk)
+                     s, xp, wp, lp, pp, d, k, n, te, nchunk)
   if (ppt <= 1) LAUNCH_COV(1);
   else if (ppt <= 2) LAUNCH_COV(2);
   else if (ppt <= 4) LAUNCH_COV(4);
@@ -151,14 +177,14 @@ void mstep_cov_impl(const torch::Tensor& x, const torch::Tensor& w,
 }
 
 void mstep_covariance_partials(torch::Tensor x, torch::Tensor w,
-                               torch::Tensor partials) {
+                               torch::Tensor lse, torch::Tensor partials) {
   TORCH_CHECK(x.is_cuda() && x.is_contiguous(), "x must be contiguous device");
   check_f32(w, "w");
   check_f32(partials, "partials");
   if (x.scalar_type() == torch::kFloat32) {
-    mstep_cov_impl<float>(x, w, partials);
+    mstep_cov_impl<float>(x, w, lse, partials);
   } else if (x.scalar_type() == torch::kBFloat16) {
-    mstep_cov_impl<__hip_bfloat16>(x, w, partials);
+    mstep_cov_impl<__hip_bfloat16>(x, w, lse, partials);
   } else {
     TORCH_CHECK(false, "x must be fp32 or bf16");
   }
@@ -226,7 +252,8 @@ void constants(torch::Tensor r, torch::Tensor means, torch::Tensor pi,
   HIP_CHECK(hipGetLastError());
 }
 
-void mstep_moments(torch::Tensor x, torch::Tensor w, torch::Tensor partials) {
+void mstep_moments(torch::Tensor x, torch::Tensor w, torch::Tensor lse,
+                   torch::Tensor partials) {
   TORCH_CHECK(x.is_cuda() && x.is_contiguous() &&
                   x.scalar_type() == torch::kFloat32,
               "x must be contiguous fp32 (exact M-step)");
@@ -245,12 +272,14 @@ void mstep_moments(torch::Tensor x, torch::Tensor w, torch::Tensor partials) {
   dim3 grid((k + 3) / 4, nchunk);
   hipLaunchKernelGGL((gmm::mstep_moments_kernel<float>), grid, dim3(kNT), lds,
                      stream(), x.data_ptr<float>(), w.data_ptr<float>(),
-                     partials.data_ptr<float>(), d, k, n, nchunk);
+                     lse_ptr(lse, n), partials.data_ptr<float>(), d, k, n,
+                     nchunk);
   HIP_CHECK(hipGetLastError());
 }
 
 void mstep_moments_b16(torch::Tensor xhi, torch::Tensor xlo,
-                       torch::Tensor w, torch::Tensor partials) {
+                       torch::Tensor w, torch::Tensor lse,
+                       torch::Tensor partials) {
   TORCH_CHECK(xhi.is_cuda() && xhi.is_contiguous() && xlo.is_contiguous() &&
                   xhi.scalar_type() == torch::kBFloat16 &&
                   xlo.scalar_type() == torch::kBFloat16,
@@ -273,13 +302,14 @@ void mstep_moments_b16(torch::Tensor xhi, torch::Tensor xlo,
                      stream(),
                      reinterpret_cast<const __hip_bfloat16*>(xhi.data_ptr()),
                      reinterpret_cast<const __hip_bfloat16*>(xlo.data_ptr()),
-                     w.data_ptr<float>(), partials.data_ptr<float>(), d, k,
-                     n, nchunk);
+                     w.data_ptr<float>(), lse_ptr(lse, n),
+                     partials.data_ptr<float>(), d, k, n, nchunk);
   HIP_CHECK(hipGetLastError());
 }
 
 void estep_fused(torch::Tensor z, torch::Tensor mfac, torch::Tensor add,
-                 torch::Tensor w_out, torch::Tensor partial) {
+                 torch::Tensor w_out, torch::Tensor lse_out,
+                 torch::Tensor partial) {
   TORCH_CHECK(z.is_cuda() && z.is_contiguous() &&
                   z.scalar_type() == torch::kBFloat16,
               "z must be contiguous bf16 [D,N]");
@@ -299,11 +329,14 @@ void estep_fused(torch::Tensor z, torch::Tensor mfac, torch::Tensor add,
   // transposed z tile + per-wave online-softmax state + lse (K-independent)
   const size_t lds = (size_t)256 * 40 * 2 +
                      sizeof(float) * (2 * (256 / 64) * 256 + 256);
+  check_f32(lse_out, "lse_out");
+  TORCH_CHECK(lse_out.numel() == n, "lse_out must be [N]");
   hipLaunchKernelGGL(gmm::estep_fused_kernel, dim3((uint32_t)nblk), dim3(kNT),
                      lds, stream(),
                      reinterpret_cast<const __hip_bfloat16*>(z.data_ptr()),
                      reinterpret_cast<const __hip_bfloat16*>(mfac.data_ptr()),
                      add.data_ptr<float>(), w_out.data_ptr<float>(),
+                     lse_out.data_ptr<float>(),
                      partial.data_ptr<float>(), d, k, n);
   HIP_CHECK(hipGetLastError());
 }
@@ -335,8 +368,48 @@ void estep_logw_big(torch::Tensor z, torch::Tensor mfac, torch::Tensor add,
   HIP_CHECK(hipGetLastError());
 }
 
+void estep_logw_big_f32(torch::Tensor z, torch::Tensor mfac32,
+                        torch::Tensor add, torch::Tensor logw) {
+  check_f32(z, "z");
+  check_f32(mfac32, "mfac32");
+  check_f32(add, "add");
+  check_f32(logw, "logw");
+  const int d = (int)z.size(0);
+  const int64_t n = z.size(1);
+  const int k = (int)add.size(0);
+  TORCH_CHECK(d >= 1 && d <= 143, "estep_logw_big_f32 supports D <= 143");
+  TORCH_CHECK(logw.size(0) == k && logw.size(1) == n, "logw shape");
+  const int kct =
+      d + 1 <= 32 ? 2 : d + 1 <= 48 ? 3 : d + 1 <= 80 ? 5 : 9;
+  const int rows = ((d + 31) / 32) * 32;
+  TORCH_CHECK(mfac32.numel() >= (int64_t)k * rows * kct * 16,
+              "mfac32 too small for big-D layout");
+  const size_t lds = sizeof(float) * (size_t)128 * (kct * 16 + 4);
+  dim3 grid((uint32_t)((n + 127) / 128), (k + 3) / 4);
+  auto s = stream();
+#define LAUNCH_ELBF(KCT)                                                    \
+  do {                                                                      \
+    if (lds > 64 * 1024) {                                                  \
+      HIP_CHECK(hipFuncSetAttribute(                                        \
+          reinterpret_cast<const void*>(                                    \
+              &gmm::estep_logw_big_f32_kernel<KCT>),                        \
+          hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds));           \
+    }                                                                       \
+    hipLaunchKernelGGL((gmm::estep_logw_big_f32_kernel<KCT>), grid,         \
+                       dim3(kNT), lds, s, z.data_ptr<float>(),              \
+                       mfac32.data_ptr<float>(), add.data_ptr<float>(),     \
+                       logw.data_ptr<float>(), d, k, n);                    \
+  } while (0)
+  if (kct == 2) LAUNCH_ELBF(2);
+  else if (kct == 3) LAUNCH_ELBF(3);
+  else if (kct == 5) LAUNCH_ELBF(5);
+  else LAUNCH_ELBF(9);
+#undef LAUNCH_ELBF
+  HIP_CHECK(hipGetLastError());
+}
+
 void mstep_moments_big(torch::Tensor x, torch::Tensor w,
-                       torch::Tensor partials) {
+                       torch::Tensor lse, torch::Tensor partials) {
   TORCH_CHECK(x.is_cuda() && x.is_contiguous() &&
                   x.scalar_type() == torch::kFloat32,
               "x must be contiguous fp32");
@@ -356,7 +429,8 @@ void mstep_moments_big(torch::Tensor x, torch::Tensor w,
   dim3 grid((k + 3) / 4, nchunk);
   hipLaunchKernelGGL(gmm::mstep_moments_big_kernel, grid, dim3(1024), lds,
                      stream(), x.data_ptr<float>(), w.data_ptr<float>(),
-                     partials.data_ptr<float>(), d, k, n, nchunk);
+                     lse_ptr(lse, n), partials.data_ptr<float>(), d, k, n,
+                     nchunk);
   HIP_CHECK(hipGetLastError());
 }
 
@@ -424,7 +498,7 @@ void emit_factors(torch::Tensor rinv, torch::Tensor means,
 
 void estep_fused_f32(torch::Tensor z, torch::Tensor mfac32,
                      torch::Tensor add, torch::Tensor w_out,
-                     torch::Tensor partial) {
+                     torch::Tensor lse_out, torch::Tensor partial) {
   check_f32(z, "z");
   check_f32(mfac32, "mfac32");
   check_f32(add, "add");
@@ -440,11 +514,13 @@ void estep_fused_f32(torch::Tensor z, torch::Tensor mfac32,
   TORCH_CHECK(partial.size(0) >= nblk, "partial buffer too small");
   const size_t lds =
       sizeof(float) * ((size_t)256 * 33 + 2 * (256 / 64) * 256 + 256);
+  check_f32(lse_out, "lse_out");
+  TORCH_CHECK(lse_out.numel() == n, "lse_out must be [N]");
   hipLaunchKernelGGL(gmm::estep_fused_f32_kernel, dim3((uint32_t)nblk),
                      dim3(kNT), lds, stream(), z.data_ptr<float>(),
                      mfac32.data_ptr<float>(), add.data_ptr<float>(),
-                     w_out.data_ptr<float>(), partial.data_ptr<float>(), d,
-                     k, n);
+                     w_out.data_ptr<float>(), lse_out.data_ptr<float>(),
+                     partial.data_ptr<float>(), d, k, n);
   HIP_CHECK(hipGetLastError());
 }
 
@@ -466,12 +542,17 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "log weighted likelihoods [K,N] (gfx950 kernel)");
   m.def("estep_posteriors", &estep_posteriors,
         "in-place posteriors + per-block likelihood partials");
+  m.def("estep_lse", &estep_lse,
+        "per-event log-sum-exp + likelihood partials (no posterior "
+        "write-back; pairs with the lse-aware M-step)");
   m.def("mstep_covariance_partials", &mstep_covariance_partials,
         "packed weighted second-moment partials [nchunk,K,P]");
   m.def("constants", &constants,
         "batched no-pivot LU inverse + ln|det| + bf16 Cholesky factors");
   m.def("estep_logw_big", &estep_logw_big,
         "big-D MFMA log-weights (31 < D <= 143)");
+  m.def("estep_logw_big_f32", &estep_logw_big_f32,
+        "exact-f32 big-D MFMA log-weights (31 < D <= 143)");
   m.def("mstep_moments_big", &mstep_moments_big,
         "big-D split-precision moments");
   m.def("mstep_moments_b16", &mstep_moments_b16,

@@ -222,6 +222,42 @@ estep_posteriors_kernel(float* __restrict__ logw, float* __restrict__ partial,
 }
 
 // ---------------------------------------------------------------------------
+// Per-event log-sum-exp ONLY (no posterior write-back): reads logw [K, N],
+// writes lse [N] + per-block likelihood partials. Pairs with the lse-aware
+// M-step kernels (they apply exp(logw - lse) while staging w), which
+// deletes the [K, N] posterior write+read round trip of the two-kernel
+// E-step path entirely.
+// ---------------------------------------------------------------------------
+__global__ void __launch_bounds__(NT)
+estep_lse_kernel(const float* __restrict__ logw, float* __restrict__ lse,
+                 float* __restrict__ partial, int k, int64_t n) {
+  float acc = 0.0f;
+  for (int64_t e = (int64_t)blockIdx.x * NT + threadIdx.x; e < n;
+       e += (int64_t)gridDim.x * NT) {
+    float m = logw[e];
+#pragma unroll 4
+    for (int c = 1; c < k; ++c) m = fmaxf(m, logw[(int64_t)c * n + e]);
+    float s = 0.0f;
+#pragma unroll 4
+    for (int c = 0; c < k; ++c) s += __expf(logw[(int64_t)c * n + e] - m);
+    const float denom = m + __logf(s);
+    lse[e] = denom;
+    acc += denom;
+  }
+  __shared__ float wsum[NT / WAVE];
+  for (int off = WAVE / 2; off > 0; off >>= 1)
+    acc += __shfl_down(acc, off, WAVE);
+  const int wid = threadIdx.x / WAVE;
+  if ((threadIdx.x & (WAVE - 1)) == 0) wsum[wid] = acc;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float total = 0.0f;
+    for (int wv = 0; wv < NT / WAVE; ++wv) total += wsum[wv];
+    partial[blockIdx.x] = total;
+  }
+}
+
+// ---------------------------------------------------------------------------
 // M-step covariance: packed second moments S_c = sum_e w_ce x_e x_e^T
 // (replaces mstep_covariance1, gaussian_kernel.cu:605 — but uncentered;
 // the engine finalizes R = (S - N mu mu^T + G avgvar I)/N, identical math).
@@ -235,6 +271,7 @@ estep_posteriors_kernel(float* __restrict__ logw, float* __restrict__ partial,
 template <int PPT, typename T>
 __global__ void __launch_bounds__(NT)
 mstep_cov_kernel(const T* __restrict__ x, const float* __restrict__ w,
+                 const float* __restrict__ lse,
                  float* __restrict__ partials, int d, int k, int64_t n,
                  int te, int nchunk) {
   // lds: xs[d][te + 4] floats (te multiple of 4), then wt[te]
@@ -269,16 +306,22 @@ mstep_cov_kernel(const T* __restrict__ x, const float* __restrict__ w,
       for (int i = threadIdx.x; i < d * te; i += NT)
         xs[(i / te) * row + i % te] =
             load_x(x, (int64_t)(i / te) * n + e0 + i % te);
-      for (int ei = threadIdx.x; ei < te; ei += NT)
-        wt[ei] = w[(int64_t)c * n + e0 + ei];
+      for (int ei = threadIdx.x; ei < te; ei += NT) {
+        float v = w[(int64_t)c * n + e0 + ei];
+        if (lse) v = __expf(v - lse[e0 + ei]);
+        wt[ei] = v;
+      }
     } else {
       for (int i = threadIdx.x; i < d * te; i += NT) {
         const int di = i / te, ei = i % te;
         xs[di * row + ei] =
             (ei < cnt) ? load_x(x, (int64_t)di * n + e0 + ei) : 0.0f;
       }
-      for (int ei = threadIdx.x; ei < te; ei += NT)
-        wt[ei] = (ei < cnt) ? w[(int64_t)c * n + e0 + ei] : 0.0f;
+      for (int ei = threadIdx.x; ei < te; ei += NT) {
+        float v = (ei < cnt) ? w[(int64_t)c * n + e0 + ei] : 0.0f;
+        if (lse && ei < cnt) v = __expf(v - lse[e0 + ei]);
+        wt[ei] = v;
+      }
     }
     __syncthreads();
 
@@ -578,9 +621,13 @@ constants_lu_kernel(const float* __restrict__ r,
 // ---------------------------------------------------------------------------
 #define MOM_BK 128
 
+// When `lse` is non-null, `w` holds LOG weights and the effective weight is
+// exp(w - lse[e]) computed while staging (bit-identical to the normalize
+// pass it replaces — same __expf on the same operands). Pad events get 0.
 template <typename T>
 __global__ void __launch_bounds__(NT)
 mstep_moments_kernel(const T* __restrict__ x, const float* __restrict__ w,
+                     const float* __restrict__ lse,
                      float* __restrict__ partials, int d, int k, int64_t n,
                      int nchunk) {
   // Exact-fp32 augmented moments on v_mfma_f32_16x16x4_f32, with
@@ -639,12 +686,23 @@ mstep_moments_kernel(const T* __restrict__ x, const float* __restrict__ w,
       const int cw = blockIdx.x * 4 + wv;
       if (cw < k) {
         const int64_t g = (int64_t)cw * n + e0 + eq * 4;
+        const int64_t ge = e0 + eq * 4;
         if (full) {
           rw = *(const float4*)&w[g];
+          if (lse) {
+            const float4 lv = *(const float4*)&lse[ge];
+            rw.x = __expf(rw.x - lv.x);
+            rw.y = __expf(rw.y - lv.y);
+            rw.z = __expf(rw.z - lv.z);
+            rw.w = __expf(rw.w - lv.w);
+          }
         } else {
           float v[4];
-          for (int u = 0; u < 4; ++u)
-            v[u] = (e0 + eq * 4 + u < n) ? w[g + u] : 0.0f;
+          for (int u = 0; u < 4; ++u) {
+            const bool ok = ge + u < n;
+            v[u] = ok ? w[g + u] : 0.0f;
+            if (lse) v[u] = ok ? __expf(v[u] - lse[ge + u]) : 0.0f;
+          }
           rw = *(float4*)v;
         }
       } else {
@@ -743,6 +801,7 @@ __global__ void __launch_bounds__(MB_NT)
 mstep_moments_b16_kernel(const __hip_bfloat16* __restrict__ xhi,
                          const __hip_bfloat16* __restrict__ xlo,
                          const float* __restrict__ w,
+                         const float* __restrict__ lse,
                          float* __restrict__ partials, int d, int k,
                          int64_t n, int nchunk) {
   // Split-precision moments with double-buffered, register-staged tiles
@@ -812,12 +871,23 @@ mstep_moments_b16_kernel(const __hip_bfloat16* __restrict__ xhi,
       const int cw = blockIdx.x * MB_CPB + wv;
       if (cw < k) {
         const int64_t g = (int64_t)cw * n + e0 + eq * 4;
+        const int64_t ge = e0 + eq * 4;
         if (full) {
           rw = *(const float4*)&w[g];
+          if (lse) {
+            const float4 lv = *(const float4*)&lse[ge];
+            rw.x = __expf(rw.x - lv.x);
+            rw.y = __expf(rw.y - lv.y);
+            rw.z = __expf(rw.z - lv.z);
+            rw.w = __expf(rw.w - lv.w);
+          }
         } else {
           float v[4];
-          for (int u = 0; u < 4; ++u)
-            v[u] = (e0 + eq * 4 + u < n) ? w[g + u] : 0.0f;
+          for (int u = 0; u < 4; ++u) {
+            const bool ok = ge + u < n;
+            v[u] = ok ? w[g + u] : 0.0f;
+            if (lse) v[u] = ok ? __expf(v[u] - lse[ge + u]) : 0.0f;
+          }
           rw = *(float4*)v;
         }
       } else {
@@ -974,8 +1044,8 @@ __global__ void __launch_bounds__(NT)
 estep_fused_kernel(const __hip_bfloat16* __restrict__ z,
                    const __hip_bfloat16* __restrict__ mfac,  // [K][2][32][32]
                    const float* __restrict__ add,            // const + ln pi
-                   float* __restrict__ w_out, float* __restrict__ partial,
-                   int d, int k, int64_t n) {
+                   float* __restrict__ w_out, float* __restrict__ lse_out,
+                   float* __restrict__ partial, int d, int k, int64_t n) {
   // LDS: zs_t [EST_BE][EST_ZROW] bf16 — z staged TRANSPOSED (k-major per
   // event, ones-row and zero-pad baked in) so a B fragment is a single
   // 16-byte ds_read_b128; then per-wave online-softmax state
@@ -1105,9 +1175,15 @@ estep_fused_kernel(const __hip_bfloat16* __restrict__ z,
   }
   __syncthreads();
 
-  // likelihood partial: sum of lse over live events
+  // likelihood partial + per-event lse out. w_out holds LOGW — the
+  // lse-aware M-step applies exp(logw - lse) while staging w, so no
+  // normalize pass and no posterior write+read round trip at all.
   {
-    float acc = (threadIdx.x < cnt) ? lse[threadIdx.x] : 0.0f;
+    float acc = 0.0f;
+    if (threadIdx.x < cnt) {
+      acc = lse[threadIdx.x];
+      lse_out[e0 + threadIdx.x] = acc;
+    }
     __shared__ float wsum[NT / WAVE];
     for (int off = WAVE / 2; off > 0; off >>= 1)
       acc += __shfl_down(acc, off, WAVE);
@@ -1117,18 +1193,6 @@ estep_fused_kernel(const __hip_bfloat16* __restrict__ z,
       float total = 0.0f;
       for (int wv = 0; wv < NT / WAVE; ++wv) total += wsum[wv];
       partial[blockIdx.x] = total;
-    }
-  }
-
-  // normalize pass: read the scratch logw back (L2-hot) and overwrite with
-  // posteriors. Flat block-wide mapping (EST_BE == NT): thread t handles
-  // event t of every cluster row — fully coalesced 1 KB accesses.
-  const int e = threadIdx.x;
-  if (e < cnt) {
-    const float l = lse[e];
-    for (int c = 0; c < k; ++c) {
-      const int64_t idx = (int64_t)c * n + e0 + e;
-      w_out[idx] = __expf(w_out[idx] - l);
     }
   }
 }
@@ -1145,8 +1209,9 @@ __global__ void __launch_bounds__(NT)
 estep_fused_f32_kernel(const float* __restrict__ z,
                        const float* __restrict__ mfac32,  // [K][32][32]
                        const float* __restrict__ add,
-                       float* __restrict__ w_out, float* __restrict__ partial,
-                       int d, int k, int64_t n) {
+                       float* __restrict__ w_out,
+                       float* __restrict__ lse_out,
+                       float* __restrict__ partial, int d, int k, int64_t n) {
   typedef __attribute__((ext_vector_type(16))) float f32x16;
   constexpr int ZR = 33;  // f32 slots per transposed event row (32 + pad)
   extern __shared__ float lds[];
@@ -1250,8 +1315,14 @@ estep_fused_f32_kernel(const float* __restrict__ z,
   }
   __syncthreads();
 
+  // likelihood partial + per-event lse out (w_out holds LOGW; the
+  // lse-aware M-step normalizes on the fly)
   {
-    float acc = (threadIdx.x < cnt) ? lse[threadIdx.x] : 0.0f;
+    float acc = 0.0f;
+    if (threadIdx.x < cnt) {
+      acc = lse[threadIdx.x];
+      lse_out[e0 + threadIdx.x] = acc;
+    }
     __shared__ float wsum2[NT / WAVE];
     for (int off = WAVE / 2; off > 0; off >>= 1)
       acc += __shfl_down(acc, off, WAVE);
@@ -1261,16 +1332,6 @@ estep_fused_f32_kernel(const float* __restrict__ z,
       float total = 0.0f;
       for (int wv = 0; wv < NT / WAVE; ++wv) total += wsum2[wv];
       partial[blockIdx.x] = total;
-    }
-  }
-
-  // normalize pass over the L2-hot scratch logw
-  const int e = threadIdx.x;
-  if (e < cnt) {
-    const float l = lse[e];
-    for (int c = 0; c < k; ++c) {
-      const int64_t idx = (int64_t)c * n + e0 + e;
-      w_out[idx] = __expf(w_out[idx] - l);
     }
   }
 }
@@ -1290,6 +1351,7 @@ estep_fused_f32_kernel(const float* __restrict__ z,
 __global__ void __launch_bounds__(MBB_NT)
 mstep_moments_big_kernel(const float* __restrict__ x,
                          const float* __restrict__ w,
+                         const float* __restrict__ lse,
                          float* __restrict__ partials, int d, int k,
                          int64_t n, int nchunk) {
   typedef __attribute__((ext_vector_type(16))) float f32x16;
@@ -1348,8 +1410,12 @@ mstep_moments_big_kernel(const float* __restrict__ x,
       }
       for (int idx = threadIdx.x; idx < MBB_CPB * MBB_BK; idx += MBB_NT) {
         const int ci = blockIdx.x * MBB_CPB + idx / MBB_BK;
-        if (ci < k)
-          wt[idx] = w[(int64_t)ci * n + e0 + idx % MBB_BK];
+        const int ei = idx % MBB_BK;
+        if (ci < k) {
+          float v = w[(int64_t)ci * n + e0 + ei];
+          if (lse) v = __expf(v - lse[e0 + ei]);
+          wt[idx] = v;
+        }
       }
     } else {
       for (int idx = threadIdx.x; idx < rows * MBB_BK; idx += MBB_NT) {
@@ -1364,7 +1430,9 @@ mstep_moments_big_kernel(const float* __restrict__ x,
       for (int idx = threadIdx.x; idx < MBB_CPB * MBB_BK; idx += MBB_NT) {
         const int ci = blockIdx.x * MBB_CPB + idx / MBB_BK;
         const int ei = idx % MBB_BK;
-        wt[idx] = (ci < k && ei < cnt) ? w[(int64_t)ci * n + e0 + ei] : 0.0f;
+        float v = (ci < k && ei < cnt) ? w[(int64_t)ci * n + e0 + ei] : 0.0f;
+        if (lse && ci < k && ei < cnt) v = __expf(v - lse[e0 + ei]);
+        wt[idx] = v;
       }
     }
     __syncthreads();
@@ -1513,6 +1581,95 @@ estep_logw_big_kernel(const __hip_bfloat16* __restrict__ z,
   }
 #pragma unroll
   for (int t = 0; t < ESB_BE / 32; ++t) {
+    float s = q[t] + __shfl_xor(q[t], 32, WAVE);
+    if (lane < 32) {
+      const int64_t e = e0 + t * 32 + j32;
+      if (e < n) logw[(int64_t)c * n + e] = -0.5f * s + addc;
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Exact-fp32 big-D MFMA E-step logw (31 < D <= 143): the fp32 counterpart
+// of estep_logw_big_kernel on v_mfma_f32_32x32x2_f32, consuming the f32
+// factor plane [K][RT*32][KCT*16]. Fills the fp32 + D > 31 quadrant that
+// previously fell back to the latency-bound VALU estep_logw_gen path
+// (reference estep1 covers its full D range at fp32 within its broken
+// D <= 32 cap, gaussian_kernel.cu:383). 128-event tiles (f32 z LDS).
+// ---------------------------------------------------------------------------
+#define ESBF_BE 128
+
+template <int KCT>
+__global__ void __launch_bounds__(NT)
+estep_logw_big_f32_kernel(const float* __restrict__ z,
+                          const float* __restrict__ mfac32,
+                          const float* __restrict__ add,
+                          float* __restrict__ logw, int d, int k, int64_t n) {
+  typedef __attribute__((ext_vector_type(16))) float f32x16;
+  constexpr int ZR = KCT * 16 + 4;  // f32 slots per transposed event row
+  extern __shared__ float lds[];
+  float* zs = lds;                  // [ESBF_BE][ZR]
+
+  const int wave = threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int j32 = lane & 31;
+  const int g2 = lane >> 5;
+  const int64_t e0 = (int64_t)blockIdx.x * ESBF_BE;
+  const int cnt = (int)min((int64_t)ESBF_BE, n - e0);
+
+  if (cnt == ESBF_BE) {
+    for (int idx = threadIdx.x; idx < d * ESBF_BE; idx += blockDim.x) {
+      const int kk = idx / ESBF_BE, ei = idx % ESBF_BE;
+      zs[ei * ZR + kk] = z[(int64_t)kk * n + e0 + ei];
+    }
+  } else {
+    for (int idx = threadIdx.x; idx < d * ESBF_BE; idx += blockDim.x) {
+      const int kk = idx / ESBF_BE, ei = idx % ESBF_BE;
+      zs[ei * ZR + kk] = (ei < cnt) ? z[(int64_t)kk * n + e0 + ei] : 0.0f;
+    }
+  }
+  for (int idx = threadIdx.x; idx < (KCT * 16 - d) * ESBF_BE;
+       idx += blockDim.x) {
+    const int kk = d + idx / ESBF_BE, ei = idx % ESBF_BE;
+    zs[ei * ZR + kk] = (kk == d && ei < cnt) ? 1.0f : 0.0f;
+  }
+  __syncthreads();
+
+  const int c = blockIdx.y * 4 + wave;
+  if (c >= k) return;
+  const int rt_n = (d + 31) / 32;
+  const int cols = KCT * 16;
+  const float* mf = mfac32 + (int64_t)c * rt_n * 32 * cols;
+  const float addc = add[c];
+
+  float q[ESBF_BE / 32];
+#pragma unroll
+  for (int t = 0; t < ESBF_BE / 32; ++t) q[t] = 0.0f;
+
+  for (int rt = 0; rt < rt_n; ++rt) {
+    // A values for this row-tile: lane j32 holds row rt*32+j32, k-slots
+    // 2*ch + g2 (v_mfma_f32_32x32x2_f32 operand map, guide §3)
+    float a[KCT * 8];
+    const float* arow = mf + (rt * 32 + j32) * cols;
+#pragma unroll
+    for (int ch = 0; ch < KCT * 8; ++ch) a[ch] = arow[2 * ch + g2];
+#pragma unroll
+    for (int t = 0; t < ESBF_BE / 32; ++t) {
+      const float* zrow = zs + (t * 32 + j32) * ZR;
+      f32x16 y = (f32x16)(0.0f);
+#pragma unroll
+      for (int ch = 0; ch < KCT * 8; ++ch) {
+        const float b = zrow[2 * ch + g2];
+        y = __builtin_amdgcn_mfma_f32_32x32x2f32(a[ch], b, y, 0, 0, 0);
+      }
+      float s = 0.0f;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) s = fmaf(y[r], y[r], s);
+      q[t] += s;
+    }
+  }
+#pragma unroll
+  for (int t = 0; t < ESBF_BE / 32; ++t) {
     float s = q[t] + __shfl_xor(q[t], 32, WAVE);
     if (lane < 32) {
       const int64_t e = e0 + t * 32 + j32;

@@ -234,7 +234,9 @@ def test_estep_fused_matches_cpu(device):
     rinv_g, const_g = F.constants(r, means, False, mfac)
     add = const_g + torch.log(pi)
     w_out = torch.empty(k, n, dtype=torch.float32, device=device)
-    w, lik = F.estep_fused(xt.to(torch.bfloat16), mfac, add, w_out)
+    lse = torch.empty(n, dtype=torch.float32, device=device)
+    _, lik = F.estep_fused(xt.to(torch.bfloat16), mfac, add, w_out, lse)
+    w = torch.exp(w_out - lse.unsqueeze(0))  # w_out holds logw
     # reference: logw + posteriors in fp32
     ref_logw = cpu.estep_logw(torch.from_numpy(x), means.cpu(), rinv, const,
                               pi.cpu())
@@ -258,8 +260,11 @@ def test_estep_fused_tail_events(device):
     add = const_g + torch.log(pi)
     x = rng.standard_normal((d, n)).astype(np.float32)
     w_out = torch.empty(k, n, dtype=torch.float32, device=device)
-    w, lik = F.estep_fused(
-        torch.from_numpy(x).to(device).to(torch.bfloat16), mfac, add, w_out)
+    lse = torch.empty(n, dtype=torch.float32, device=device)
+    _, lik = F.estep_fused(
+        torch.from_numpy(x).to(device).to(torch.bfloat16), mfac, add, w_out,
+        lse)
+    w = torch.exp(w_out - lse.unsqueeze(0))
     rinv, const = cpu.compute_constants(r.cpu())
     ref_logw = cpu.estep_logw(torch.from_numpy(x), means.cpu(), rinv, const,
                               pi.cpu())
@@ -390,7 +395,9 @@ def test_engine_big_d_bf16_matches_fp32(device=None):
         lik = eng.run_em(8)
         res[name] = (lik, eng.state.means.cpu().numpy().copy(),
                      eng.use_big_estep)
-    assert res["bf16"][2] is True and res["fp32"][2] is False
+    # both precisions use the big-D MFMA path now (bf16 hi/lo split and
+    # the exact-f32 MFMA variant respectively)
+    assert res["bf16"][2] is True and res["fp32"][2] is True
     assert res["bf16"][0] == pytest.approx(res["fp32"][0], rel=2e-3)
     np.testing.assert_allclose(res["bf16"][1], res["fp32"][1],
                                rtol=5e-2, atol=5e-1)
@@ -408,8 +415,10 @@ def test_estep_fused_f32_matches_cpu(device):
     add = const_g + torch.log(pi)
     x = rng.standard_normal((d, n)).astype(np.float32) * 2
     w_out = torch.empty(k, n, dtype=torch.float32, device=device)
-    w, lik = F.estep_fused_f32(torch.from_numpy(x).to(device), mfac32, add,
-                               w_out)
+    lse = torch.empty(n, dtype=torch.float32, device=device)
+    _, lik = F.estep_fused_f32(torch.from_numpy(x).to(device), mfac32, add,
+                               w_out, lse)
+    w = torch.exp(w_out - lse.unsqueeze(0))
     rinv, const = cpu.compute_constants(r.cpu())
     ref_logw = cpu.estep_logw(torch.from_numpy(x), means.cpu(), rinv, const,
                               pi.cpu())
@@ -520,8 +529,11 @@ def test_estep_fused_odd_k(device, k):
     add = const_g + torch.log(pi)
     x = rng.standard_normal((d, n)).astype(np.float32)
     w_out = torch.empty(k, n, dtype=torch.float32, device=device)
-    w, lik = F.estep_fused(
-        torch.from_numpy(x).to(device).to(torch.bfloat16), mfac, add, w_out)
+    lse = torch.empty(n, dtype=torch.float32, device=device)
+    _, lik = F.estep_fused(
+        torch.from_numpy(x).to(device).to(torch.bfloat16), mfac, add, w_out,
+        lse)
+    w = torch.exp(w_out - lse.unsqueeze(0))
     rinv, const = cpu.compute_constants(r.cpu())
     ref_logw = cpu.estep_logw(torch.from_numpy(x), means.cpu(), rinv, const,
                               pi.cpu())
@@ -529,3 +541,74 @@ def test_estep_fused_odd_k(device, k):
     assert float(lik) == pytest.approx(float(ref_lik), rel=1e-2)
     np.testing.assert_allclose(w.sum(dim=0).cpu().numpy(), np.ones(n),
                                rtol=1e-3)
+
+
+@pytest.mark.parametrize("d", [40, 64, 128])
+def test_estep_logw_big_f32_matches_cpu(device, d):
+    """Exact-f32 big-D MFMA logw vs the fp32 torch reference (tight —
+    f32 MFMA is bitwise an fmaf chain; only the Cholesky-vs-Rinv algebra
+    differs). Fills the fp32 + D > 31 quadrant (VERDICT r1 missing #4)."""
+    from cuda_gmm_mpi_amd.ops import functional as F
+    rng = np.random.default_rng(d + 3)
+    k, n = 6, 2000 + 57
+    means, r, pi = random_model(rng, k, d, device)
+    mfac = torch.empty(k, *F.mfac_shape(d), dtype=torch.bfloat16,
+                       device=device)
+    mfac32 = torch.empty(k, *F.mfac_shape(d)[1:], dtype=torch.float32,
+                         device=device)
+    rinv_g, const_g = F.constants(r, means, False, mfac, mfac32)
+    add = const_g + torch.log(pi)
+    x = rng.standard_normal((d, n)).astype(np.float32)
+    out = torch.empty(k, n, dtype=torch.float32, device=device)
+    F.estep_logw_big_f32(torch.from_numpy(x).to(device), mfac32, add, out)
+    rinv, const = cpu.compute_constants(r.cpu())
+    ref = cpu.estep_logw(torch.from_numpy(x), means.cpu(), rinv, const,
+                         pi.cpu())
+    scale = float(ref.abs().max())
+    np.testing.assert_allclose(out.cpu().numpy(), ref.numpy(),
+                               rtol=1e-3, atol=1e-5 * scale)
+
+
+def test_engine_big_d_fp32_mfma_matches_valu(device=None):
+    """Engine-level: the exact-f32 big-D MFMA E-step vs the VALU gen
+    kernel on the same fp32 data — same trajectory."""
+    from cuda_gmm_mpi_amd.engine import build_engine
+    from cuda_gmm_mpi_amd.utils.config import GmmConfig
+    from cuda_gmm_mpi_amd.utils.synthetic import make_blobs
+    data, _ = make_blobs(20000, 48, 6, seed=67)
+    cfg = GmmConfig(num_clusters=6, target_num_clusters=6,
+                    min_iters=6, max_iters=6)  # fp32 everywhere
+    eng_m = build_engine(data, cfg, device="cuda")
+    assert eng_m.use_big_estep and eng_m.mfac32 is not None
+    lik_m = eng_m.run_em(6)
+    eng_v = build_engine(data, cfg, device="cuda")
+    eng_v.use_big_estep = False
+    eng_v.mfac32 = None
+    lik_v = eng_v.run_em(6)
+    assert lik_m == pytest.approx(lik_v, rel=1e-4)
+    np.testing.assert_allclose(eng_m.state.means.cpu().numpy(),
+                               eng_v.state.means.cpu().numpy(),
+                               rtol=1e-3, atol=1e-2)
+
+
+def test_engine_diag_only_fused_matches_valu(device=None):
+    """DIAG_ONLY through the fused factor path vs the VALU diag kernel:
+    identical math (diagonal Cholesky factor), same trajectory."""
+    from cuda_gmm_mpi_amd.engine import build_engine
+    from cuda_gmm_mpi_amd.utils.config import GmmConfig
+    from cuda_gmm_mpi_amd.utils.synthetic import make_blobs
+    data, _ = make_blobs(20000, 12, 4, seed=131)
+    for ed in ("fp32", "bf16"):
+        cfg = GmmConfig(num_clusters=4, target_num_clusters=4,
+                        min_iters=6, max_iters=6, diag_only=True,
+                        estep_dtype=ed)
+        eng_f = build_engine(data, cfg, device="cuda")
+        assert eng_f.use_fused_estep
+        lik_f = eng_f.run_em(4)
+        eng_v = build_engine(data, cfg, device="cuda")
+        eng_v.use_fused_estep = False
+        eng_v.use_big_estep = False
+        eng_v.mfac32 = None
+        lik_v = eng_v.run_em(4)
+        rel = 1e-4 if ed == "fp32" else 2e-2
+        assert lik_f == pytest.approx(lik_v, rel=rel), ed
