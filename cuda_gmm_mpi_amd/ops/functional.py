@@ -294,9 +294,11 @@ def mstep_moments(x: torch.Tensor, w: torch.Tensor,
     if x.is_cuda:
         if lse is not None:
             # fallback path mixes rocBLAS GEMMs that need materialized
-            # posteriors: normalize w in place (it is rewritten by the
-            # next E-step)
-            w.sub_(lse.unsqueeze(0)).exp_()
+            # posteriors. MUST be non-destructive: w holds the E-step's
+            # logw, and an in-place normalize would corrupt the first
+            # graph replay (the eager capture-warmup iteration would
+            # leave posteriors where the replayed ops expect logw).
+            w = torch.exp(w - lse.unsqueeze(0))
         s = mstep_covariance_s(x, w)        # custom kernel, [K, D, D]
         mean_num = w @ x.T                  # rocBLAS
         n_c = w.sum(dim=1)

@@ -1,0 +1,313 @@
+#!/usr/bin/env python3
+"""Generate the golden-output fixture: an INDEPENDENT, direct numpy
+transcription of the reference binary's single-rank/single-GPU execution
+(gaussianMPI), pinned to the quirk ledger, producing the
+exact `.summary`/`.results` byte formats (gaussian.cu:1180-1197,
+:1042-1059).
+
+This is a test oracle, deliberately written WITHOUT the package: no
+imports from cuda_gmm_mpi_amd, no shared helpers — a from-scratch
+walk of the reference's order of operations so the committed fixture
+cross-checks the framework end to end (no CUDA toolchain exists in this
+environment to run the real binary; VERDICT r1 task 6 sanctions a
+carefully validated CPU transcription).
+
+Transcribed reference semantics, line-cited:
+  - seeding: host seed_clusters strided means + N=num_events/K
+    (gaussian.cu:108-123); GPU seed kernel survivors R=I, pi=1/K,
+    avgvar=(mean per-dim variance)/COVARIANCE_DYNAMIC_RANGE
+    (gaussian_kernel.cu:316-326)
+  - constants: ln determinant via LU (gaussian_kernel.cu:139),
+    constant = -D/2 ln(2pi) - 0.5 ln|R| (:241); pi floor 1e-10 for
+    N < 0.5 (:184-190)
+  - E-step: logw = -0.5 q + constant + ln pi (:442); posteriors via
+    max + log-sum-exp (:483-502); likelihood = sum log P(x) (:505-511)
+  - M-step: N = sum w; means = sum w x / N if N > 0.5 else 0
+    (gaussian.cu:610-622); covariance = centered per-event sums, zeroed
+    in-kernel when N < 1 (gaussian_kernel.cu:658), + avgvar on the
+    diagonal (:673-675), / N if N > 0.5 else identity (gaussian.cu:663-679)
+  - fixed 100 iterations (MIN_ITERS == MAX_ITERS, gaussian.h:26-27)
+  - rissanen = -L + 0.5 (K (1 + D + 0.5(D+1)D) - 1) ln(N D)
+    (gaussian.cu:826); target-K model saved (:839)
+  - merge: eliminate N < 0.5 (:866-874); exhaustive pair scan with
+    cluster_distance (:1203-1208) on add_clusters output whose constant
+    consumes the LOG10 determinant of invert_cpu (invert_matrix.cpp:61
+    vs gaussian.cu:1249 — quirk #2); min pair merged + compacted
+    (:899-907)
+  - output: writeCluster '%.3f'-formatted blocks; .results rows
+    '%f'-formatted data CSV \t membership CSV
+"""
+import os
+import struct
+import sys
+
+import numpy as np
+
+OUT = os.path.join(os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+                   "tests", "fixtures")
+
+F32 = np.float32
+LOG2PI = float(np.log(2.0 * np.pi))
+
+
+def lu_ln_det_inv(a):
+    """No-pivot LU inversion + NATURAL log determinant — the GPU invert
+    (gaussian_kernel.cu:107-169) used by constants_kernel."""
+    n = a.shape[0]
+    d = a.astype(np.float64).copy()
+    # Doolittle, matching invert_matrix.cpp:42-58 elimination order
+    d[0, 1:] /= d[0, 0]
+    for i in range(1, n):
+        for j in range(i, n):
+            d[j, i] -= d[j, :i] @ d[:i, i]
+        if i < n - 1:
+            for j in range(i + 1, n):
+                d[i, j] = (d[i, j] - d[i, :i] @ d[:i, j]) / d[i, i]
+    ln_det = float(np.sum(np.log(np.abs(np.diag(d)))))
+    return np.linalg.inv(a.astype(np.float64)), ln_det
+
+
+def lu_log10_det(a):
+    """log10 determinant via the same no-pivot LU — the CPU invert_cpu
+    consumed by the merge path (invert_matrix.cpp:61, quirk #2)."""
+    n = a.shape[0]
+    if n == 1:
+        return float(np.log(a[0, 0]))  # 1x1 case uses logf (:39)
+    d = a.astype(np.float64).copy()
+    d[0, 1:] /= d[0, 0]
+    for i in range(1, n):
+        for j in range(i, n):
+            d[j, i] -= d[j, :i] @ d[:i, i]
+        if i < n - 1:
+            for j in range(i + 1, n):
+                d[i, j] = (d[i, j] - d[i, :i] @ d[:i, j]) / d[i, i]
+    return float(np.sum(np.log10(np.abs(np.diag(d)))))
+
+
+class Clusters:
+    def __init__(self, k, d):
+        self.N = np.zeros(k, F32)
+        self.pi = np.zeros(k, F32)
+        self.constant = np.zeros(k, F32)
+        self.avgvar = np.zeros(k, F32)
+        self.means = np.zeros((k, d), F32)
+        self.R = np.zeros((k, d, d), F32)
+        self.Rinv = np.zeros((k, d, d), F32)
+
+    @property
+    def k(self):
+        return len(self.N)
+
+    def compact(self, keep):
+        out = Clusters(len(keep), self.means.shape[1])
+        for f in ("N", "pi", "constant", "avgvar", "means", "R", "Rinv"):
+            setattr(out, f, getattr(self, f)[keep].copy())
+        return out
+
+
+def seed(data, k, cdr=1e3):
+    n, d = data.shape
+    c = Clusters(k, d)
+    # GPU seed kernel survivors (gaussian_kernel.cu:316-326)
+    var = data.astype(np.float64).var(axis=0)  # E[x^2] - mean^2 per dim
+    c.avgvar[:] = float(var.mean() / cdr)
+    c.pi[:] = 1.0 / k
+    for i in range(k):
+        c.R[i] = np.eye(d, dtype=F32)
+    # host seed_clusters overwrite (gaussian.cu:108-123)
+    fraction = (n - 1.0) / (k - 1.0) if k > 1 else 0.0
+    for i in range(k):
+        c.N[i] = n / float(k)
+        c.means[i] = data[int(i * fraction)]
+    return c
+
+
+def constants(c):
+    """constants_kernel (gaussian_kernel.cu:196-243) + compute_pi."""
+    d = c.means.shape[1]
+    for i in range(c.k):
+        inv, ln_det = lu_ln_det_inv(c.R[i])
+        c.Rinv[i] = inv.astype(F32)
+        c.constant[i] = F32(-d * 0.5 * LOG2PI - 0.5 * ln_det)
+    total = float(c.N.sum())
+    for i in range(c.k):
+        c.pi[i] = 1e-10 if c.N[i] < 0.5 else c.N[i] / total
+
+
+def estep(data_t, c):
+    """estep1 + estep2: returns (posteriors [K, N], likelihood)."""
+    d, n = data_t.shape
+    logw = np.empty((c.k, n), np.float64)
+    for i in range(c.k):
+        dx = data_t - c.means[i][:, None].astype(np.float64)
+        q = np.einsum("in,ij,jn->n", dx, c.Rinv[i].astype(np.float64), dx)
+        logw[i] = -0.5 * q + float(c.constant[i]) + np.log(float(c.pi[i]))
+    m = logw.max(axis=0)
+    denom = m + np.log(np.exp(logw - m).sum(axis=0))
+    w = np.exp(logw - denom)
+    return w, float(denom.sum())
+
+
+def mstep(data_t, c, w):
+    d, n = data_t.shape
+    for i in range(c.k):
+        n_i = float(w[i].sum())
+        c.N[i] = n_i
+        if n_i > 0.5:
+            c.means[i] = ((data_t * w[i]).sum(axis=1) / n_i).astype(F32)
+        else:
+            c.means[i] = 0.0
+        if n_i >= 1.0:  # kernel-side zero check (gaussian_kernel.cu:658)
+            dx = data_t - c.means[i][:, None].astype(np.float64)
+            cov = (w[i] * dx) @ dx.T
+        else:
+            cov = np.zeros((d, d))
+        cov += np.eye(d) * float(c.avgvar[i])  # G=1 GPU partial (:673-675)
+        if n_i > 0.5:
+            c.R[i] = (cov / n_i).astype(F32)
+        else:
+            c.R[i] = np.eye(d, dtype=F32)
+    constants(c)
+
+
+def run_em(data_t, c, iters=100):
+    w, lik = estep(data_t, c)
+    for _ in range(iters):
+        mstep(data_t, c, w)
+        w, lik = estep(data_t, c)
+    return w, lik
+
+
+class _CView:
+    """Single-cluster view for the merge math."""
+    def __init__(self, c, i):
+        self.N = float(c.N[i])
+        self.means = c.means[i].astype(np.float64)
+        self.R = c.R[i].astype(np.float64)
+        self.constant = float(c.constant[i])
+
+
+def merge_step(c):
+    """Empty elimination + exhaustive min-distance pair merge
+    (gaussian.cu:866-907)."""
+    keep = [i for i in range(c.k) if c.N[i] >= 0.5]
+    c = c.compact(np.array(keep, dtype=int))
+    if c.k < 2:
+        return c
+    d = c.means.shape[1]
+    best = None
+    for i in range(c.k):
+        for j in range(i + 1, c.k):
+            a, b = _CView(c, i), _CView(c, j)
+            means, r, const = merge_pair(a, b, d)
+            # cluster_distance (gaussian.cu:1203-1208)
+            dist = (a.N * a.constant + b.N * b.constant
+                    - (a.N + b.N) * const)
+            if best is None or dist < best[0]:
+                best = (dist, i, j, means, r, const)
+    _, i, j, means, r, const = best
+    # merge into i (gaussian.cu:899-907)
+    c.N[i] = c.N[i] + c.N[j]
+    c.means[i] = means.astype(F32)
+    c.R[i] = r.astype(F32)
+    c.constant[i] = F32(const)
+    c.Rinv[i] = np.linalg.inv(r).astype(F32)
+    keep = [x for x in range(c.k) if x != j]
+    return c.compact(np.array(keep, dtype=int))
+
+
+def merge_pair(a, b, d):
+    wt1 = a.N / (a.N + b.N)
+    wt2 = 1.0 - wt1
+    means = wt1 * a.means + wt2 * b.means
+    r = np.zeros((d, d))
+    for i in range(d):
+        for j in range(d):
+            r[i, j] = (wt1 * (a.R[i, j] + (a.means[i] - means[i]) *
+                              (a.means[j] - means[j]))
+                       + wt2 * (b.R[i, j] + (b.means[i] - means[i]) *
+                                (b.means[j] - means[j])))
+    log10_det = lu_log10_det(r)
+    const = -d * 0.5 * LOG2PI - 0.5 * log10_det
+    return means, r, const
+
+
+def rissanen(lik, k, d, n):
+    nparams = 1 + d + 0.5 * (d + 1) * d
+    return -lik + 0.5 * (k * nparams - 1) * np.log(float(n * d))
+
+
+def write_summary(path, c):
+    """writeCluster per cluster (gaussian.cu:1180-1197): '%.3f' fields."""
+    d = c.means.shape[1]
+    with open(path, "w") as f:
+        for i in range(c.k):
+            f.write(f"Cluster #{i}\n")
+            f.write(f"Probability: {c.pi[i]:.3f}\n")
+            f.write(f"N: {c.N[i]:.3f}\n")
+            f.write("Means: " + ", ".join(f"{v:.3f}" for v in c.means[i])
+                    + "\n\n")
+            f.write("R Matrix:\n")
+            for r in range(d):
+                f.write(", ".join(f"{c.R[i, r, cc]:.3f}" for cc in range(d))
+                        + "\n")
+            f.write("\n\n")
+
+
+def write_results(path, data, w):
+    """Per-event 'data CSV \\t membership CSV' rows ('%f' = 6 decimals,
+    gaussian.cu:1042-1059)."""
+    with open(path, "w") as f:
+        for e in range(data.shape[0]):
+            f.write(",".join(f"{v:f}" for v in data[e]))
+            f.write("\t")
+            f.write(",".join(f"{w[i, e]:f}" for i in range(w.shape[0])))
+            f.write("\n")
+
+
+def main():
+    os.makedirs(OUT, exist_ok=True)
+    rng = np.random.default_rng(424242)
+    # 3 well-separated blobs, FCS-like scale, N divisible by nothing fancy
+    n, d, k0, ktarget = 600, 3, 3, 2
+    centers = np.array([[200.0, 300.0, 150.0],
+                        [500.0, 180.0, 420.0],
+                        [320.0, 520.0, 260.0]])
+    counts = [220, 190, 190]
+    rows = []
+    for c, cnt in zip(centers, counts):
+        rows.append(rng.normal(c, 18.0, size=(cnt, d)))
+    data = np.vstack(rows).astype(F32)
+    data = data[rng.permutation(n)]
+
+    with open(os.path.join(OUT, "golden_small.bin"), "wb") as f:
+        f.write(struct.pack("<ii", n, d))
+        f.write(data.tobytes())
+
+    data_t = data.T.astype(np.float64)
+    c = seed(data, k0)
+    constants(c)
+
+    best = None
+    kk = k0
+    while kk >= 1:
+        w, lik = run_em(data_t, c, iters=100)
+        riss = rissanen(lik, kk, d, n)
+        if kk == k0 or kk == ktarget:
+            best = (c, w, riss, kk)
+        if kk <= ktarget:
+            break
+        c = merge_step(c)
+        kk = c.k
+        # post-merge: merged cluster carries the host constant/Rinv into
+        # the next E-step (quirk #8) — constants NOT recomputed here
+
+    c, w, riss, kk = best
+    write_summary(os.path.join(OUT, "golden_small.summary"), c)
+    write_results(os.path.join(OUT, "golden_small.results"), data, w)
+    print(f"fixture written: K={kk} rissanen={riss:.4f} "
+          f"likelihood={w.shape} -> {OUT}")
+
+
+if __name__ == "__main__":
+    sys.exit(main())
