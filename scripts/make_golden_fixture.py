@@ -238,18 +238,20 @@ def rissanen(lik, k, d, n):
 
 
 def write_summary(path, c):
-    """writeCluster per cluster (gaussian.cu:1180-1197): '%.3f' fields."""
+    """writeCluster per cluster (gaussian.cu:1180-1197). Byte layout:
+    "Probability: %f\n", "N: %f\n", "Means: " + D*"%.3f " + "\n",
+    "\nR Matrix:\n", D rows of D*"%.3f " each ending "\n"."""
     d = c.means.shape[1]
     with open(path, "w") as f:
         for i in range(c.k):
             f.write(f"Cluster #{i}\n")
-            f.write(f"Probability: {c.pi[i]:.3f}\n")
-            f.write(f"N: {c.N[i]:.3f}\n")
-            f.write("Means: " + ", ".join(f"{v:.3f}" for v in c.means[i])
-                    + "\n\n")
-            f.write("R Matrix:\n")
+            f.write(f"Probability: {c.pi[i]:f}\n")
+            f.write(f"N: {c.N[i]:f}\n")
+            f.write("Means: " + "".join(f"{v:.3f} " for v in c.means[i])
+                    + "\n")
+            f.write("\nR Matrix:\n")
             for r in range(d):
-                f.write(", ".join(f"{c.R[i, r, cc]:.3f}" for cc in range(d))
+                f.write("".join(f"{c.R[i, r, cc]:.3f} " for cc in range(d))
                         + "\n")
             f.write("\n\n")
 
