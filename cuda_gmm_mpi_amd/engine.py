@@ -144,11 +144,12 @@ class EmEngine:
 
 
     def _refresh_mfac(self, k: int) -> None:
-        """Re-emit the fused-E-step factors from the CURRENT Rinv without
-        touching the constants — after an MDL merge or a checkpoint resume
-        the merged cluster's constant/Rinv come from the host path and must
-        feed the next E-step unchanged, but the factor
-        tables must match the compacted Rinv."""
+        """Re-emit the fused-E-step factors from the CURRENT covariance R
+        (Cholesky of R, stable for any PD covariance) without touching
+        the constants — after an MDL merge or a checkpoint resume the
+        merged cluster's constant comes from the host path and must feed
+        the next E-step unchanged, but the factor tables
+        must match the compacted state."""
         if self.mfac is None and self.mfac32 is None:
             return
         from .ops.backend import hip_ext
@@ -156,7 +157,7 @@ class EmEngine:
         empty_b = torch.empty(0, dtype=torch.bfloat16, device=self.device)
         empty_f = torch.empty(0, dtype=torch.float32, device=self.device)
         hip_ext().emit_factors(
-            st.Rinv.contiguous(), st.means.contiguous(),
+            st.R.contiguous(), st.means.contiguous(),
             self.mfac[:k] if self.mfac is not None else empty_b,
             self.mfac32[:k] if self.mfac32 is not None else empty_f,
         )

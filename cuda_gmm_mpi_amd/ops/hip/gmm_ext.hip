@@ -229,11 +229,11 @@ void constants(torch::Tensor r, torch::Tensor means, torch::Tensor pi,
                        r.data_ptr<float>(), rinv.data_ptr<float>(),
                        logdet.data_ptr<float>(), d);
     if (make_mfac) {
-      // diag Rinv is a valid input to the generic factor path: run the LU
-      // kernel's emit on the diagonal inverse via a tiny dedicated kernel
-      hipLaunchKernelGGL(gmm::emit_mfac_from_rinv_kernel, dim3(k), dim3(kNT),
+      // diag R is a valid input to the generic factor path (diagonal
+      // Cholesky); emit from R via the standalone kernel
+      hipLaunchKernelGGL(gmm::emit_mfac_from_r_kernel, dim3(k), dim3(kNT),
                          sizeof(float) * (2 * (size_t)d * d + d), s,
-                         rinv.data_ptr<float>(), means.data_ptr<float>(), mp,
+                         r.data_ptr<float>(), means.data_ptr<float>(), mp,
                          mp32, d);
     }
   } else {
@@ -468,12 +468,12 @@ void mstep_finalize(torch::Tensor packed, torch::Tensor avgvar,
   HIP_CHECK(hipGetLastError());
 }
 
-void emit_factors(torch::Tensor rinv, torch::Tensor means,
+void emit_factors(torch::Tensor r, torch::Tensor means,
                   torch::Tensor mfac, torch::Tensor mfac32) {
-  check_f32(rinv, "rinv");
+  check_f32(r, "r");
   check_f32(means, "means");
-  const int k = (int)rinv.size(0);
-  const int d = (int)rinv.size(1);
+  const int k = (int)r.size(0);
+  const int d = (int)r.size(1);
   __hip_bfloat16* mp = nullptr;
   float* mp32 = nullptr;
   if (mfac.numel() > 0) {
@@ -487,11 +487,11 @@ void emit_factors(torch::Tensor rinv, torch::Tensor means,
   const size_t lds = sizeof(float) * (2 * (size_t)d * d + d);
   if (lds > 64 * 1024) {
     HIP_CHECK(hipFuncSetAttribute(
-        reinterpret_cast<const void*>(&gmm::emit_mfac_from_rinv_kernel),
+        reinterpret_cast<const void*>(&gmm::emit_mfac_from_r_kernel),
         hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds));
   }
-  hipLaunchKernelGGL(gmm::emit_mfac_from_rinv_kernel, dim3(k), dim3(kNT),
-                     lds, stream(), rinv.data_ptr<float>(),
+  hipLaunchKernelGGL(gmm::emit_mfac_from_r_kernel, dim3(k), dim3(kNT),
+                     lds, stream(), r.data_ptr<float>(),
                      means.data_ptr<float>(), mp, mp32, d);
   HIP_CHECK(hipGetLastError());
 }
@@ -562,7 +562,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mstep_finalize", &mstep_finalize,
         "finalize N/means/R/pi from all-reduced packed moments");
   m.def("emit_factors", &emit_factors,
-        "re-emit E-step factors from an existing Rinv (post-merge/resume)");
+        "re-emit E-step factors from the covariance R (post-merge/resume)");
   m.def("estep_fused_f32", &estep_fused_f32,
         "exact-f32 MFMA fused E-step (D <= 31)");
   m.def("estep_fused", &estep_fused,

@@ -302,14 +302,19 @@ mstep_cov_kernel(const T* __restrict__ x, const float* __restrict__ w,
     __syncthreads();
     // stage x tile (coalesced per dimension row) and w tile; branchless
     // when the tile is full (guide §5 trap 4c)
+    // per-element runtime selects inside staging loops serialize every
+    // load behind vmcnt(0) (guide §5 trap 4c) — the lse test is hoisted
+    // into duplicated loops instead
     if (cnt == te) {
       for (int i = threadIdx.x; i < d * te; i += NT)
         xs[(i / te) * row + i % te] =
             load_x(x, (int64_t)(i / te) * n + e0 + i % te);
-      for (int ei = threadIdx.x; ei < te; ei += NT) {
-        float v = w[(int64_t)c * n + e0 + ei];
-        if (lse) v = __expf(v - lse[e0 + ei]);
-        wt[ei] = v;
+      if (lse) {
+        for (int ei = threadIdx.x; ei < te; ei += NT)
+          wt[ei] = __expf(w[(int64_t)c * n + e0 + ei] - lse[e0 + ei]);
+      } else {
+        for (int ei = threadIdx.x; ei < te; ei += NT)
+          wt[ei] = w[(int64_t)c * n + e0 + ei];
       }
     } else {
       for (int i = threadIdx.x; i < d * te; i += NT) {
@@ -317,10 +322,13 @@ mstep_cov_kernel(const T* __restrict__ x, const float* __restrict__ w,
         xs[di * row + ei] =
             (ei < cnt) ? load_x(x, (int64_t)di * n + e0 + ei) : 0.0f;
       }
-      for (int ei = threadIdx.x; ei < te; ei += NT) {
-        float v = (ei < cnt) ? w[(int64_t)c * n + e0 + ei] : 0.0f;
-        if (lse && ei < cnt) v = __expf(v - lse[e0 + ei]);
-        wt[ei] = v;
+      if (lse) {
+        for (int ei = threadIdx.x; ei < te; ei += NT)
+          wt[ei] = (ei < cnt)
+              ? __expf(w[(int64_t)c * n + e0 + ei] - lse[e0 + ei]) : 0.0f;
+      } else {
+        for (int ei = threadIdx.x; ei < te; ei += NT)
+          wt[ei] = (ei < cnt) ? w[(int64_t)c * n + e0 + ei] : 0.0f;
       }
     }
     __syncthreads();
@@ -409,37 +417,68 @@ mstep_finalize_kernel(const float* __restrict__ packed,
   }
 }
 
-// Emit the fused-E-step factor M = [U | -U mu] (U^T U = Rinv, upper
-// Cholesky) as bf16 hi/lo pairs into mfac[c][2][32][32], stored in MFMA
-// A-fragment k-order (mfma_b16_k). `u` is scratch LDS holding Rinv [d*d]
-// (upper triangle becomes U in place) + u0 [d].
-__device__ inline void emit_mfac(float* u, const float* __restrict__ means,
+// Emit the fused-E-step factor M = [F | -F mu] with F = L^-1 where
+// L L^T = R (lower Cholesky of the COVARIANCE, not of Rinv): then
+// q = ||F z||^2 = z^T R^-1 z exactly. Factoring R — which is positive
+// definite by construction (covariance + avgvar ridge; identity resets)
+// — is stable where the previous chol(Rinv) exploded: a no-pivot fp32
+// LU inverse of an ill-conditioned R loses positive-definiteness and
+// clamped pivots cascade into ~1e15-scale factor rows whose quadratic
+// forms overflow fp32 (K >> N/D regimes).
+// `a` is scratch LDS holding R [d*d] (becomes L, then F in place);
+// `o` [d*d] is the snapshot for the cross-thread triangular inversion;
+// u0 [d] follows o.
+__device__ inline void emit_mfac(float* a, float* o,
+                                 const float* __restrict__ r_global,
+                                 const float* __restrict__ means,
                                  __hip_bfloat16* __restrict__ mfac,
                                  float* __restrict__ mfac32, int c, int d) {
   const int tid = threadIdx.x;
-  float* u0 = u + d * d;
-  // in-place upper Cholesky of Rinv (SPD up to rounding; clamped pivots)
+  float* u0 = o + d * d;
+  for (int t = tid; t < d * d; t += blockDim.x)
+    a[t] = r_global[(int64_t)c * d * d + t];
+  __syncthreads();
+  // in-place lower Cholesky of R (column j: pivot, then rows below)
   for (int j = 0; j < d; ++j) {
     if (tid == 0) {
-      float s = u[j * d + j];
-      for (int kk = 0; kk < j; ++kk) s -= u[kk * d + j] * u[kk * d + j];
-      u[j * d + j] = sqrtf(fmaxf(s, 1e-30f));
+      const float diag0 = fabsf(a[j * d + j]);
+      float s = a[j * d + j];
+      for (int kk = 0; kk < j; ++kk) s -= a[j * d + kk] * a[j * d + kk];
+      a[j * d + j] = sqrtf(fmaxf(s, 1e-8f * diag0 + 1e-30f));
     }
     __syncthreads();
-    const float piv = u[j * d + j];
+    const float piv = a[j * d + j];
     for (int i = j + 1 + tid; i < d; i += blockDim.x) {
-      float s = u[j * d + i];
+      float s = a[i * d + j];
 #pragma unroll 8
-      for (int kk = 0; kk < j; ++kk) s -= u[kk * d + j] * u[kk * d + i];
-      u[j * d + i] = s / piv;
+      for (int kk = 0; kk < j; ++kk) s -= a[i * d + kk] * a[j * d + kk];
+      a[i * d + j] = s / piv;
     }
     __syncthreads();
   }
-  // u0 = -U mu (mu = centered cluster means)
+  // snapshot L, then invert in place: F = L^-1 (lower). Thread i owns
+  // column i (serial down rows); cross-column L reads come from the
+  // snapshot, this thread's own F values from a.
+  for (int t = tid; t < d * d; t += blockDim.x) o[t] = a[t];
+  __syncthreads();
+  for (int i = tid; i < d; i += blockDim.x) {
+    for (int j = i; j < d; ++j) {
+      float xv = 1.0f;
+      if (i != j) {
+        xv = 0.0f;
+#pragma unroll 8
+        for (int kk = i; kk < j; ++kk)
+          xv -= o[j * d + kk] * a[kk * d + i];
+      }
+      a[j * d + i] = xv / o[j * d + j];
+    }
+  }
+  __syncthreads();
+  // u0 = -F mu (mu = centered cluster means; F lower: j <= i)
   for (int i = tid; i < d; i += blockDim.x) {
     float s = 0.0f;
 #pragma unroll 8
-    for (int j = i; j < d; ++j) s += u[i * d + j] * means[c * d + j];
+    for (int j = 0; j <= i; ++j) s += a[i * d + j] * means[c * d + j];
     u0[i] = -s;
   }
   __syncthreads();
@@ -456,7 +495,7 @@ __device__ inline void emit_mfac(float* u, const float* __restrict__ means,
     const int i = t / cols, kx = t % cols;
     float v = 0.0f;
     if (i < d) {
-      if (kx < d) v = (kx >= i) ? u[i * d + kx] : 0.0f;
+      if (kx < d) v = (kx <= i) ? a[i * d + kx] : 0.0f;
       else if (kx == d) v = u0[i];
     }
     const __hip_bfloat16 hi = __float2bfloat16(v);
@@ -500,7 +539,7 @@ constants_lu_kernel(const float* __restrict__ r,
       o[0] = oc[0];
     }
     __syncthreads();
-    if (mfac != nullptr) emit_mfac(o, means, mfac, mfac32, c, 1);
+    if (mfac != nullptr) emit_mfac(a, o, r, means, mfac, mfac32, c, 1);
     return;
   }
 
@@ -595,11 +634,11 @@ constants_lu_kernel(const float* __restrict__ r,
     for (int kk = (i > j ? i : j); kk < d; ++kk)
       s = fmaf((j == kk) ? 1.0f : a[j * d + kk], a[kk * d + i], s);
     oc[j * d + i] = s;
-    o[j * d + i] = s;  // LDS copy for the Cholesky factor below
   }
   if (mfac != nullptr) {
     __syncthreads();
-    emit_mfac(o, means, mfac, mfac32, c, d);
+    // factor emission re-reads R from global and factors it directly
+    emit_mfac(a, o, r, means, mfac, mfac32, c, d);
   }
 }
 
@@ -687,15 +726,15 @@ mstep_moments_kernel(const T* __restrict__ x, const float* __restrict__ w,
       if (cw < k) {
         const int64_t g = (int64_t)cw * n + e0 + eq * 4;
         const int64_t ge = e0 + eq * 4;
-        if (full) {
+        if (full && lse) {
           rw = *(const float4*)&w[g];
-          if (lse) {
-            const float4 lv = *(const float4*)&lse[ge];
-            rw.x = __expf(rw.x - lv.x);
-            rw.y = __expf(rw.y - lv.y);
-            rw.z = __expf(rw.z - lv.z);
-            rw.w = __expf(rw.w - lv.w);
-          }
+          const float4 lv = *(const float4*)&lse[ge];
+          rw.x = __expf(rw.x - lv.x);
+          rw.y = __expf(rw.y - lv.y);
+          rw.z = __expf(rw.z - lv.z);
+          rw.w = __expf(rw.w - lv.w);
+        } else if (full) {
+          rw = *(const float4*)&w[g];
         } else {
           float v[4];
           for (int u = 0; u < 4; ++u) {
@@ -872,15 +911,15 @@ mstep_moments_b16_kernel(const __hip_bfloat16* __restrict__ xhi,
       if (cw < k) {
         const int64_t g = (int64_t)cw * n + e0 + eq * 4;
         const int64_t ge = e0 + eq * 4;
-        if (full) {
+        if (full && lse) {
           rw = *(const float4*)&w[g];
-          if (lse) {
-            const float4 lv = *(const float4*)&lse[ge];
-            rw.x = __expf(rw.x - lv.x);
-            rw.y = __expf(rw.y - lv.y);
-            rw.z = __expf(rw.z - lv.z);
-            rw.w = __expf(rw.w - lv.w);
-          }
+          const float4 lv = *(const float4*)&lse[ge];
+          rw.x = __expf(rw.x - lv.x);
+          rw.y = __expf(rw.y - lv.y);
+          rw.z = __expf(rw.z - lv.z);
+          rw.w = __expf(rw.w - lv.w);
+        } else if (full) {
+          rw = *(const float4*)&w[g];
         } else {
           float v[4];
           for (int u = 0; u < 4; ++u) {
@@ -1408,13 +1447,19 @@ mstep_moments_big_kernel(const float* __restrict__ x,
           put(idx / MBB_BK, idx % MBB_BK,
               (idx / MBB_BK == d) ? 1.0f : 0.0f);
       }
-      for (int idx = threadIdx.x; idx < MBB_CPB * MBB_BK; idx += MBB_NT) {
-        const int ci = blockIdx.x * MBB_CPB + idx / MBB_BK;
-        const int ei = idx % MBB_BK;
-        if (ci < k) {
-          float v = w[(int64_t)ci * n + e0 + ei];
-          if (lse) v = __expf(v - lse[e0 + ei]);
-          wt[idx] = v;
+      if (lse) {
+        for (int idx = threadIdx.x; idx < MBB_CPB * MBB_BK; idx += MBB_NT) {
+          const int ci = blockIdx.x * MBB_CPB + idx / MBB_BK;
+          const int ei = idx % MBB_BK;
+          if (ci < k)
+            wt[idx] = __expf(w[(int64_t)ci * n + e0 + ei] - lse[e0 + ei]);
+        }
+      } else {
+        for (int idx = threadIdx.x; idx < MBB_CPB * MBB_BK; idx += MBB_NT) {
+          const int ci = blockIdx.x * MBB_CPB + idx / MBB_BK;
+          const int ei = idx % MBB_BK;
+          if (ci < k)
+            wt[idx] = w[(int64_t)ci * n + e0 + ei];
         }
       }
     } else {
@@ -1427,12 +1472,20 @@ mstep_moments_big_kernel(const float* __restrict__ x,
         }
         put(di, ei, v);
       }
-      for (int idx = threadIdx.x; idx < MBB_CPB * MBB_BK; idx += MBB_NT) {
-        const int ci = blockIdx.x * MBB_CPB + idx / MBB_BK;
-        const int ei = idx % MBB_BK;
-        float v = (ci < k && ei < cnt) ? w[(int64_t)ci * n + e0 + ei] : 0.0f;
-        if (lse && ci < k && ei < cnt) v = __expf(v - lse[e0 + ei]);
-        wt[idx] = v;
+      if (lse) {
+        for (int idx = threadIdx.x; idx < MBB_CPB * MBB_BK; idx += MBB_NT) {
+          const int ci = blockIdx.x * MBB_CPB + idx / MBB_BK;
+          const int ei = idx % MBB_BK;
+          wt[idx] = (ci < k && ei < cnt)
+              ? __expf(w[(int64_t)ci * n + e0 + ei] - lse[e0 + ei]) : 0.0f;
+        }
+      } else {
+        for (int idx = threadIdx.x; idx < MBB_CPB * MBB_BK; idx += MBB_NT) {
+          const int ci = blockIdx.x * MBB_CPB + idx / MBB_BK;
+          const int ei = idx % MBB_BK;
+          wt[idx] = (ci < k && ei < cnt)
+              ? w[(int64_t)ci * n + e0 + ei] : 0.0f;
+        }
       }
     }
     __syncthreads();
@@ -1678,19 +1731,15 @@ estep_logw_big_f32_kernel(const float* __restrict__ z,
   }
 }
 
-// Factor emission from an existing Rinv (diag-only path): stage Rinv into
-// LDS and run the same Cholesky/emit as the LU kernel.
+// Standalone factor emission from the covariance R (diag-only constants
+// and the post-merge/resume refresh path).
 __global__ void __launch_bounds__(NT)
-emit_mfac_from_rinv_kernel(const float* __restrict__ rinv,
-                           const float* __restrict__ means,
-                           __hip_bfloat16* __restrict__ mfac,
-                           float* __restrict__ mfac32, int d) {
+emit_mfac_from_r_kernel(const float* __restrict__ r,
+                        const float* __restrict__ means,
+                        __hip_bfloat16* __restrict__ mfac,
+                        float* __restrict__ mfac32, int d) {
   extern __shared__ float buf[];
-  const int c = blockIdx.x;
-  for (int t = threadIdx.x; t < d * d; t += NT)
-    buf[t] = rinv[(int64_t)c * d * d + t];
-  __syncthreads();
-  emit_mfac(buf, means, mfac, mfac32, c, d);
+  emit_mfac(buf, buf + d * d, r, means, mfac, mfac32, blockIdx.x, d);
 }
 
 // DIAG_ONLY constants (gaussian_kernel.cu:215-223)

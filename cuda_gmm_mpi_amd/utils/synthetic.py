@@ -88,4 +88,23 @@ def make_supported_blobs(num_events: int, num_dims: int, num_clusters: int,
         labels[pos:pos + n_c] = c
         pos += n_c
     perm = rng.permutation(num_events)
-    return data[perm], labels[perm]
+    data, labels = data[perm], labels[perm]
+    # The reference seeds cluster c's mean from the event at row
+    # c*(N-1)/(K-1) (gaussian.cu:108-123). Arrange the row order so each
+    # seed position holds an event from a DISTINCT cluster — every blob
+    # starts seeded, so EM at K0=K keeps all clusters supported and the
+    # MDL sweep cannot mass-eliminate its way past the target.
+    if k > 1:
+        seed_rows = np.round(np.arange(k) * (num_events - 1.0)
+                             / (k - 1.0)).astype(np.int64)
+        used = set(seed_rows.tolist())
+        for c, p in enumerate(seed_rows):
+            if labels[p] == c:
+                continue
+            cand = np.nonzero(labels == c)[0]
+            swap = next(int(s) for s in cand
+                        if int(s) not in used or int(s) == int(p))
+            data[[p, swap]] = data[[swap, p]]
+            labels[[p, swap]] = labels[[swap, p]]
+            used.add(swap)
+    return data, labels
