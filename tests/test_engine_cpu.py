@@ -524,7 +524,11 @@ def test_sweep_lands_on_target_with_supported_blobs():
     LANDS on the target K (the reference's save-target path,
     gaussian.cu:839) instead of jumping past it."""
     from cuda_gmm_mpi_amd.utils.synthetic import make_supported_blobs
-    data, _ = make_supported_blobs(6000, 5, 30, seed=11)
+    # unit-spread data keeps |det R| ~ 1, so the bug-compat log10/ln
+    # merge-constant shift (quirk #2, ~0.28*ln|det| nats) cannot inflate
+    # the merged cluster's first post-merge E-step and starve others
+    data, _ = make_supported_blobs(6000, 5, 30, seed=11,
+                                   scale=10.0, spread=1.0)
     cfg = GmmConfig(num_clusters=30, target_num_clusters=10,
                     min_iters=4, max_iters=4)
     res = build_engine(data, cfg, device="cpu").sweep()
