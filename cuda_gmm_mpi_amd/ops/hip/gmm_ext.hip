@@ -232,13 +232,13 @@ void constants(torch::Tensor r, torch::Tensor means, torch::Tensor pi,
       // diag R is a valid input to the generic factor path (diagonal
       // Cholesky); emit from R via the standalone kernel
       hipLaunchKernelGGL(gmm::emit_mfac_from_r_kernel, dim3(k), dim3(kNT),
-                         sizeof(float) * (2 * (size_t)d * d + d), s,
+                         sizeof(float) * (2 * (size_t)d * (d | 1) + d), s,
                          r.data_ptr<float>(), means.data_ptr<float>(), mp,
                          mp32, d);
     }
   } else {
     // working buffer + read-only LU snapshot (+ u0 scratch for the factor)
-    const size_t lds = sizeof(float) * (2 * (size_t)d * d + d);
+    const size_t lds = sizeof(float) * (2 * (size_t)d * (d | 1) + d);
     if (lds > 64 * 1024) {  // gfx950: 160 KiB LDS/CU; opt in past 64 KiB
       HIP_CHECK(hipFuncSetAttribute(
           reinterpret_cast<const void*>(&gmm::constants_lu_kernel),
@@ -484,7 +484,7 @@ void emit_factors(torch::Tensor r, torch::Tensor means,
     check_f32(mfac32, "mfac32");
     mp32 = mfac32.data_ptr<float>();
   }
-  const size_t lds = sizeof(float) * (2 * (size_t)d * d + d);
+  const size_t lds = sizeof(float) * (2 * (size_t)d * (d | 1) + d);
   if (lds > 64 * 1024) {
     HIP_CHECK(hipFuncSetAttribute(
         reinterpret_cast<const void*>(&gmm::emit_mfac_from_r_kernel),

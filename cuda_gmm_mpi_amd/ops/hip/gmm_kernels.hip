@@ -434,32 +434,35 @@ __device__ inline void emit_mfac(float* a, float* o,
                                  __hip_bfloat16* __restrict__ mfac,
                                  float* __restrict__ mfac32, int c, int d) {
   const int tid = threadIdx.x;
-  float* u0 = o + d * d;
+  // odd LDS row stride: stride-d column walks with gcd(d, 32) > 1 put
+  // whole lane groups on one bank (8-way at d = 24)
+  const int ldp = d | 1;
+  float* u0 = o + d * ldp;
   for (int t = tid; t < d * d; t += blockDim.x)
-    a[t] = r_global[(int64_t)c * d * d + t];
+    a[(t / d) * ldp + t % d] = r_global[(int64_t)c * d * d + t];
   __syncthreads();
   // in-place lower Cholesky of R (column j: pivot, then rows below)
   for (int j = 0; j < d; ++j) {
     if (tid == 0) {
-      const float diag0 = fabsf(a[j * d + j]);
-      float s = a[j * d + j];
-      for (int kk = 0; kk < j; ++kk) s -= a[j * d + kk] * a[j * d + kk];
-      a[j * d + j] = sqrtf(fmaxf(s, 1e-8f * diag0 + 1e-30f));
+      const float diag0 = fabsf(a[j * ldp + j]);
+      float s = a[j * ldp + j];
+      for (int kk = 0; kk < j; ++kk) s -= a[j * ldp + kk] * a[j * ldp + kk];
+      a[j * ldp + j] = sqrtf(fmaxf(s, 1e-8f * diag0 + 1e-30f));
     }
     __syncthreads();
-    const float piv = a[j * d + j];
+    const float piv = a[j * ldp + j];
     for (int i = j + 1 + tid; i < d; i += blockDim.x) {
-      float s = a[i * d + j];
+      float s = a[i * ldp + j];
 #pragma unroll 8
-      for (int kk = 0; kk < j; ++kk) s -= a[i * d + kk] * a[j * d + kk];
-      a[i * d + j] = s / piv;
+      for (int kk = 0; kk < j; ++kk) s -= a[i * ldp + kk] * a[j * ldp + kk];
+      a[i * ldp + j] = s / piv;
     }
     __syncthreads();
   }
   // snapshot L, then invert in place: F = L^-1 (lower). Thread i owns
   // column i (serial down rows); cross-column L reads come from the
   // snapshot, this thread's own F values from a.
-  for (int t = tid; t < d * d; t += blockDim.x) o[t] = a[t];
+  for (int t = tid; t < d * ldp; t += blockDim.x) o[t] = a[t];
   __syncthreads();
   for (int i = tid; i < d; i += blockDim.x) {
     for (int j = i; j < d; ++j) {
@@ -468,9 +471,9 @@ __device__ inline void emit_mfac(float* a, float* o,
         xv = 0.0f;
 #pragma unroll 8
         for (int kk = i; kk < j; ++kk)
-          xv -= o[j * d + kk] * a[kk * d + i];
+          xv -= o[j * ldp + kk] * a[kk * ldp + i];
       }
-      a[j * d + i] = xv / o[j * d + j];
+      a[j * ldp + i] = xv / o[j * ldp + j];
     }
   }
   __syncthreads();
@@ -478,7 +481,7 @@ __device__ inline void emit_mfac(float* a, float* o,
   for (int i = tid; i < d; i += blockDim.x) {
     float s = 0.0f;
 #pragma unroll 8
-    for (int j = 0; j <= i; ++j) s += a[i * d + j] * means[c * d + j];
+    for (int j = 0; j <= i; ++j) s += a[i * ldp + j] * means[c * d + j];
     u0[i] = -s;
   }
   __syncthreads();
@@ -495,7 +498,7 @@ __device__ inline void emit_mfac(float* a, float* o,
     const int i = t / cols, kx = t % cols;
     float v = 0.0f;
     if (i < d) {
-      if (kx < d) v = (kx <= i) ? a[i * d + kx] : 0.0f;
+      if (kx < d) v = (kx <= i) ? a[i * ldp + kx] : 0.0f;
       else if (kx == d) v = u0[i];
     }
     const __hip_bfloat16 hi = __float2bfloat16(v);
@@ -521,18 +524,22 @@ constants_lu_kernel(const float* __restrict__ r,
   extern __shared__ float a[];
   const int c = blockIdx.x;
   const int tid = threadIdx.x;
-  float* o = a + d * d;
+  // odd LDS row stride (see emit_mfac): kills the gcd(d, 32)-way bank
+  // conflicts of stride-d column walks
+  const int ldp = d | 1;
+  float* o = a + d * ldp;
   const float* rc = r + (int64_t)c * d * d;
   float* oc = rinv + (int64_t)c * d * d;
 
-  for (int t = tid; t < d * d; t += blockDim.x) a[t] = rc[t];
+  for (int t = tid; t < d * d; t += blockDim.x)
+    a[(t / d) * ldp + t % d] = rc[t];
   __syncthreads();
 
   if (d == 1) {
     if (tid == 0) {
-      const float ld = __logf(a[0]);
-      logdet[c] = ld;
-      const float cst = -0.5f * 1.8378770664093453f - 0.5f * ld;  // ln(2pi)
+      const float lg = __logf(a[0]);
+      logdet[c] = lg;
+      const float cst = -0.5f * 1.8378770664093453f - 0.5f * lg;  // ln(2pi)
       if (constant) constant[c] = cst;
       if (add) add[c] = cst + __logf(pi[c]);
       oc[0] = 1.0f / a[0];
@@ -552,18 +559,20 @@ constants_lu_kernel(const float* __restrict__ r,
     for (int j = i + tid; j < d; j += blockDim.x) {
       float s = 0.0f;
 #pragma unroll 8
-      for (int kk = 0; kk < i; ++kk) s = fmaf(a[j * d + kk], a[kk * d + i], s);
-      a[j * d + i] -= s;
+      for (int kk = 0; kk < i; ++kk)
+        s = fmaf(a[j * ldp + kk], a[kk * ldp + i], s);
+      a[j * ldp + i] -= s;
     }
     __syncthreads();
     if (i == d - 1) break;
     // row i of U: cols j > i in parallel
-    const float pivot = a[i * d + i];
+    const float pivot = a[i * ldp + i];
     for (int j = i + 1 + tid; j < d; j += blockDim.x) {
       float s = 0.0f;
 #pragma unroll 8
-      for (int kk = 0; kk < i; ++kk) s = fmaf(a[i * d + kk], a[kk * d + j], s);
-      a[i * d + j] = (a[i * d + j] - s) / pivot;
+      for (int kk = 0; kk < i; ++kk)
+        s = fmaf(a[i * ldp + kk], a[kk * ldp + j], s);
+      a[i * ldp + j] = (a[i * ldp + j] - s) / pivot;
     }
     __syncthreads();
   }
@@ -573,7 +582,8 @@ constants_lu_kernel(const float* __restrict__ r,
     __shared__ float wsum[NT / WAVE];
     const int nw = (blockDim.x + WAVE - 1) / WAVE;
     float acc = 0.0f;
-    for (int i = tid; i < d; i += blockDim.x) acc += __logf(fabsf(a[i * d + i]));
+    for (int i = tid; i < d; i += blockDim.x)
+      acc += __logf(fabsf(a[i * ldp + i]));
     for (int off = WAVE / 2; off > 0; off >>= 1)
       acc += __shfl_down(acc, off, WAVE);
     if ((tid & (WAVE - 1)) == 0) wsum[tid / WAVE] = acc;
@@ -591,7 +601,7 @@ constants_lu_kernel(const float* __restrict__ r,
   __syncthreads();
   // snapshot the LU factor: the inversion below reads original L/U values
   // that the in-place writes would otherwise clobber across threads
-  for (int t = tid; t < d * d; t += blockDim.x) o[t] = a[t];
+  for (int t = tid; t < d * ldp; t += blockDim.x) o[t] = a[t];
   __syncthreads();
 
   // invert L: column i per thread, serial down rows; cross-column reads and
@@ -604,9 +614,9 @@ constants_lu_kernel(const float* __restrict__ r,
         xv = 0.0f;
 #pragma unroll 8
         for (int kk = i; kk < j; ++kk)
-          xv -= o[j * d + kk] * a[kk * d + i];
+          xv -= o[j * ldp + kk] * a[kk * ldp + i];
       }
-      a[j * d + i] = xv / o[j * d + j];
+      a[j * ldp + i] = xv / o[j * ldp + j];
     }
   }
   // invert U: row i per thread, serial across cols; column reads from the
@@ -618,8 +628,8 @@ constants_lu_kernel(const float* __restrict__ r,
       float s = 0.0f;
 #pragma unroll 8
       for (int kk = i; kk < j; ++kk)
-        s += o[kk * d + j] * ((i == kk) ? 1.0f : a[i * d + kk]);
-      a[i * d + j] = -s;
+        s += o[kk * ldp + j] * ((i == kk) ? 1.0f : a[i * ldp + kk]);
+      a[i * ldp + j] = -s;
     }
   }
   __syncthreads();
@@ -632,7 +642,7 @@ constants_lu_kernel(const float* __restrict__ r,
     float s = 0.0f;
 #pragma unroll 8
     for (int kk = (i > j ? i : j); kk < d; ++kk)
-      s = fmaf((j == kk) ? 1.0f : a[j * d + kk], a[kk * d + i], s);
+      s = fmaf((j == kk) ? 1.0f : a[j * ldp + kk], a[kk * ldp + i], s);
     oc[j * d + i] = s;
   }
   if (mfac != nullptr) {
@@ -1739,7 +1749,8 @@ emit_mfac_from_r_kernel(const float* __restrict__ r,
                         __hip_bfloat16* __restrict__ mfac,
                         float* __restrict__ mfac32, int d) {
   extern __shared__ float buf[];
-  emit_mfac(buf, buf + d * d, r, means, mfac, mfac32, blockIdx.x, d);
+  emit_mfac(buf, buf + d * (d | 1), r, means, mfac, mfac32,
+            blockIdx.x, d);
 }
 
 // DIAG_ONLY constants (gaussian_kernel.cu:215-223)
