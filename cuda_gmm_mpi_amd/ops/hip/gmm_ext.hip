@@ -236,6 +236,21 @@ void constants(torch::Tensor r, torch::Tensor means, torch::Tensor pi,
                          r.data_ptr<float>(), means.data_ptr<float>(), mp,
                          mp32, d);
     }
+  } else if (d <= 48) {
+    // wave-per-cluster variant: 4 independent waves per block, zero
+    // workgroup barriers (the block variant pays ~100 __syncthreads per
+    // cluster, each draining the LDS pipeline)
+    const size_t lds =
+        4 * sizeof(float) * (2 * (size_t)d * (d | 1) + d);
+    if (lds > 64 * 1024) {
+      HIP_CHECK(hipFuncSetAttribute(
+          reinterpret_cast<const void*>(&gmm::constants_wave_kernel),
+          hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds));
+    }
+    hipLaunchKernelGGL(gmm::constants_wave_kernel, dim3((k + 3) / 4),
+                       dim3(kNT), lds, s, r.data_ptr<float>(),
+                       means.data_ptr<float>(), pip, rinv.data_ptr<float>(),
+                       logdet.data_ptr<float>(), cst, addp, mp, mp32, d, k);
   } else {
     // working buffer + read-only LU snapshot (+ u0 scratch for the factor)
     const size_t lds = sizeof(float) * (2 * (size_t)d * (d | 1) + d);
@@ -356,10 +371,34 @@ void estep_logw_big(torch::Tensor z, torch::Tensor mfac, torch::Tensor add,
   TORCH_CHECK(logw.size(0) == k && logw.size(1) == n, "logw shape");
   const int kct =
       d + 1 <= 32 ? 2 : d + 1 <= 48 ? 3 : d + 1 <= 80 ? 5 : 9;
-  const size_t lds = (size_t)256 * (kct * 16 + 8) * 2;
-  dim3 grid((uint32_t)((n + 255) / 256), (k + 3) / 4);
+  const int rows = ((d + 31) / 32) * 32;
+  const int rt_n = rows / 32;
+  const int arow = kct * 16 + 8;
+  const size_t lds = (size_t)(2 * rows + 128) * arow * 2 +
+                     (size_t)rt_n * 128 * 4;
+  const int64_t tiles = (n + 127) / 128;
+  // enough blocks to fill the chip; each block amortizes one staged
+  // factor table over tiles/nchunk z tiles
+  const int nchunk =
+      (int)std::min<int64_t>(tiles, std::max<int64_t>(1, 2048 / k));
+  dim3 grid((uint32_t)nchunk, k);
   auto s = stream();
-#define LAUNCH_ELB(KCT)                                                        do {                                                                           if (lds > 64 * 1024) {                                                         HIP_CHECK(hipFuncSetAttribute(                                                   reinterpret_cast<const void*>(&gmm::estep_logw_big_kernel<KCT>),             hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds));                }                                                                            hipLaunchKernelGGL((gmm::estep_logw_big_kernel<KCT>), grid, dim3(kNT),                          lds, s,                                                                      reinterpret_cast<const __hip_bfloat16*>(z.data_ptr()),                       reinterpret_cast<const __hip_bfloat16*>(                                         mfac.data_ptr()),                                                        add.data_ptr<float>(), logw.data_ptr<float>(), d, k,                         n);                                                     } while (0)
+#define LAUNCH_ELB(KCT)                                                     \
+  do {                                                                      \
+    if (lds > 64 * 1024) {                                                  \
+      HIP_CHECK(hipFuncSetAttribute(                                        \
+          reinterpret_cast<const void*>(&gmm::estep_logw_big2_kernel<KCT>), \
+          hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds));           \
+    }                                                                       \
+    hipLaunchKernelGGL((gmm::estep_logw_big2_kernel<KCT>), grid,            \
+                       dim3(1024), lds, s,                                  \
+                       reinterpret_cast<const __hip_bfloat16*>(             \
+                           z.data_ptr()),                                   \
+                       reinterpret_cast<const __hip_bfloat16*>(             \
+                           mfac.data_ptr()),                                \
+                       add.data_ptr<float>(), logw.data_ptr<float>(), d, k, \
+                       n, nchunk);                                          \
+  } while (0)
   if (kct == 2) LAUNCH_ELB(2);
   else if (kct == 3) LAUNCH_ELB(3);
   else if (kct == 5) LAUNCH_ELB(5);

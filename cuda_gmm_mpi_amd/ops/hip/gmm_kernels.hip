@@ -508,6 +508,191 @@ __device__ inline void emit_mfac(float* a, float* o,
   }
 }
 
+// Wave-level LDS ordering: all 64 lanes of a CDNA wave share one program
+// counter, so cross-lane LDS visibility needs only completion of this
+// wave's own LDS ops (lgkmcnt) plus compiler fences — no workgroup
+// barrier. This is what makes the wave-per-cluster constants variant
+// ~barrier-free (the block variant pays ~100 __syncthreads per cluster,
+// each draining the LDS pipeline).
+__device__ inline void wsync() {
+  __builtin_amdgcn_wave_barrier();
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  __builtin_amdgcn_wave_barrier();
+}
+
+// Wave-per-cluster constants (d <= 48): cluster c = blockIdx.x*4 + wave.
+// Same reference-faithful math as constants_lu_kernel (no-pivot LU,
+// ln|det|, snapshot triangular inversion, chol(R) factor emission), with
+// every loop parallelized over the 64 lanes of one wave and zero
+// workgroup barriers.
+__global__ void __launch_bounds__(NT)
+constants_wave_kernel(const float* __restrict__ r,
+                      const float* __restrict__ means,
+                      const float* __restrict__ pi,
+                      float* __restrict__ rinv, float* __restrict__ logdet,
+                      float* __restrict__ constant, float* __restrict__ add,
+                      __hip_bfloat16* __restrict__ mfac,
+                      float* __restrict__ mfac32, int d, int k) {
+  extern __shared__ float lds[];
+  const int wave = threadIdx.x / WAVE;
+  const int l = threadIdx.x & (WAVE - 1);
+  const int c = blockIdx.x * (NT / WAVE) + wave;
+  if (c >= k) return;
+  const int ldp = d | 1;
+  float* a = lds + (size_t)wave * (2 * d * ldp + d);
+  float* o = a + d * ldp;
+  float* u0 = o + d * ldp;
+  const float* rc = r + (int64_t)c * d * d;
+  float* oc = rinv + (int64_t)c * d * d;
+
+  for (int t = l; t < d * d; t += WAVE) a[(t / d) * ldp + t % d] = rc[t];
+  wsync();
+
+  if (d == 1) {
+    if (l == 0) {
+      const float lg = __logf(a[0]);
+      logdet[c] = lg;
+      const float cst = -0.5f * 1.8378770664093453f - 0.5f * lg;
+      if (constant) constant[c] = cst;
+      if (add) add[c] = cst + __logf(pi[c]);
+      oc[0] = 1.0f / a[0];
+    }
+    wsync();
+  } else {
+    for (int j = 1 + l; j < d; j += WAVE) a[j] /= a[0];
+    wsync();
+    for (int i = 1; i < d; ++i) {
+      for (int j = i + l; j < d; j += WAVE) {
+        float s = 0.0f;
+#pragma unroll 8
+        for (int kk = 0; kk < i; ++kk)
+          s = fmaf(a[j * ldp + kk], a[kk * ldp + i], s);
+        a[j * ldp + i] -= s;
+      }
+      wsync();
+      if (i == d - 1) break;
+      const float pivot = a[i * ldp + i];
+      for (int j = i + 1 + l; j < d; j += WAVE) {
+        float s = 0.0f;
+#pragma unroll 8
+        for (int kk = 0; kk < i; ++kk)
+          s = fmaf(a[i * ldp + kk], a[kk * ldp + j], s);
+        a[i * ldp + j] = (a[i * ldp + j] - s) / pivot;
+      }
+      wsync();
+    }
+    {
+      float acc = 0.0f;
+      for (int i = l; i < d; i += WAVE) acc += __logf(fabsf(a[i * ldp + i]));
+      for (int off = WAVE / 2; off > 0; off >>= 1)
+        acc += __shfl_down(acc, off, WAVE);
+      if (l == 0) {
+        logdet[c] = acc;
+        const float cst = -d * 0.5f * 1.8378770664093453f - 0.5f * acc;
+        if (constant) constant[c] = cst;
+        if (add) add[c] = cst + __logf(pi[c]);
+      }
+    }
+    for (int t = l; t < d * ldp; t += WAVE) o[t] = a[t];
+    wsync();
+    // invert L (lane = column; d <= 48 < WAVE: one lane each)
+    if (l < d) {
+      const int i = l;
+      for (int j = i; j < d; ++j) {
+        float xv = 1.0f;
+        if (i != j) {
+          xv = 0.0f;
+#pragma unroll 8
+          for (int kk = i; kk < j; ++kk)
+            xv -= o[j * ldp + kk] * a[kk * ldp + i];
+        }
+        a[j * ldp + i] = xv / o[j * ldp + j];
+      }
+      // invert U (lane = row); disjoint writes from the L columns
+      const int i2 = l;
+      for (int j = i2 + 1; j < d; ++j) {
+        float s = 0.0f;
+#pragma unroll 8
+        for (int kk = i2; kk < j; ++kk)
+          s += o[kk * ldp + j] * ((i2 == kk) ? 1.0f : a[i2 * ldp + kk]);
+        a[i2 * ldp + j] = -s;
+      }
+    }
+    wsync();
+    for (int t = l; t < d * d; t += WAVE) {
+      const int j = t / d, i = t % d;
+      float s = 0.0f;
+#pragma unroll 8
+      for (int kk = (i > j ? i : j); kk < d; ++kk)
+        s = fmaf((j == kk) ? 1.0f : a[j * ldp + kk], a[kk * ldp + i], s);
+      oc[j * d + i] = s;
+    }
+  }
+  if (mfac == nullptr) return;
+
+  // ---- factor emission: lower Cholesky of R, F = L^-1, M = [F | -F mu]
+  for (int t = l; t < d * d; t += WAVE)
+    a[(t / d) * ldp + t % d] = rc[t];
+  wsync();
+  for (int j = 0; j < d; ++j) {
+    if (l == 0) {
+      const float diag0 = fabsf(a[j * ldp + j]);
+      float s = a[j * ldp + j];
+      for (int kk = 0; kk < j; ++kk) s -= a[j * ldp + kk] * a[j * ldp + kk];
+      a[j * ldp + j] = sqrtf(fmaxf(s, 1e-8f * diag0 + 1e-30f));
+    }
+    wsync();
+    const float piv = a[j * ldp + j];
+    for (int i = j + 1 + l; i < d; i += WAVE) {
+      float s = a[i * ldp + j];
+#pragma unroll 8
+      for (int kk = 0; kk < j; ++kk) s -= a[i * ldp + kk] * a[j * ldp + kk];
+      a[i * ldp + j] = s / piv;
+    }
+    wsync();
+  }
+  for (int t = l; t < d * ldp; t += WAVE) o[t] = a[t];
+  wsync();
+  if (l < d) {
+    const int i = l;
+    for (int j = i; j < d; ++j) {
+      float xv = 1.0f;
+      if (i != j) {
+        xv = 0.0f;
+#pragma unroll 8
+        for (int kk = i; kk < j; ++kk)
+          xv -= o[j * ldp + kk] * a[kk * ldp + i];
+      }
+      a[j * ldp + i] = xv / o[j * ldp + j];
+    }
+  }
+  wsync();
+  for (int i = l; i < d; i += WAVE) {
+    float s = 0.0f;
+#pragma unroll 8
+    for (int j = 0; j <= i; ++j) s += a[i * ldp + j] * means[c * d + j];
+    u0[i] = -s;
+  }
+  wsync();
+  const int rows = ((d + 31) / 32) * 32;
+  const int cols = kc_tier((d + 1 + 15) / 16) * 16;
+  const int cells = rows * cols;
+  __hip_bfloat16* out = mfac + (int64_t)c * 2 * cells;
+  float* out32 = mfac32 ? mfac32 + (int64_t)c * cells : nullptr;
+  for (int t = l; t < cells; t += WAVE) {
+    const int i = t / cols, kx = t % cols;
+    float v = 0.0f;
+    if (i < d) {
+      if (kx < d) v = (kx <= i) ? a[i * ldp + kx] : 0.0f;
+      else if (kx == d) v = u0[i];
+    }
+    const __hip_bfloat16 hi = __float2bfloat16(v);
+    out[t] = hi;
+    out[cells + t] = __float2bfloat16(v - __bfloat162float(hi));
+    if (out32) out32[t] = v;
+  }
+}
+
 __global__ void __launch_bounds__(NT)
 constants_lu_kernel(const float* __restrict__ r,
                     const float* __restrict__ means,
@@ -1648,6 +1833,121 @@ estep_logw_big_kernel(const __hip_bfloat16* __restrict__ z,
     if (lane < 32) {
       const int64_t e = e0 + t * 32 + j32;
       if (e < n) logw[(int64_t)c * n + e] = -0.5f * s + addc;
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Big-D MFMA E-step v2 (round 2): ONE CLUSTER PER BLOCK with the whole
+// factor table [2][RT*32][KCT*16] staged in LDS once and reused across
+// MANY 128-event z tiles (tile-strided chunks). v1 re-read every
+// cluster's fragments from L2/HBM per 256-event block (~295 GB/iter at
+// config 4 by the round-1 ledger); v2's factor traffic is
+// K * nchunk * |M_c| ~ 0.4 GB. 16 waves split the (row-tile, event-32)
+// work; per-tile partials combine through LDS (fixed order:
+// deterministic).
+// ---------------------------------------------------------------------------
+#define ESB2_BE 128
+#define ESB2_NT 1024
+
+template <int KCT>
+__global__ void __launch_bounds__(ESB2_NT)
+estep_logw_big2_kernel(const __hip_bfloat16* __restrict__ z,
+                       const __hip_bfloat16* __restrict__ mfac,
+                       const float* __restrict__ add,
+                       float* __restrict__ logw, int d, int k, int64_t n,
+                       int nchunk) {
+  typedef __attribute__((ext_vector_type(16))) float f32x16;
+  constexpr int COLS = KCT * 16;
+  constexpr int AROW = COLS + 8;   // bf16 row stride (16-lane b128 groups
+                                   // land on distinct banks: 76 = 12 mod 64)
+  extern __shared__ float lds[];
+  const int rt_n = (d + 31) / 32;
+  const int rows = rt_n * 32;
+  __bf16* ah = (__bf16*)lds;             // [rows][AROW]
+  __bf16* al = ah + rows * AROW;
+  __bf16* zs = al + rows * AROW;         // [ESB2_BE][AROW]
+  float* qpart = (float*)(zs + ESB2_BE * AROW);  // [rt_n][ESB2_BE]
+
+  const int wave = threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int j32 = lane & 31;
+  const int g2 = lane >> 5;
+  const int c = blockIdx.y;
+  const int chunk = blockIdx.x;
+  const float addc = add[c];
+
+  // stage this cluster's whole factor table (hi+lo planes) into padded
+  // LDS rows; vectorized 8-bf16 copies, coalesced over the row-major
+  // global layout [c][2][rows][cols]
+  {
+    const uint4* src_h =
+        (const uint4*)(mfac + (int64_t)c * 2 * rows * COLS);
+    const uint4* src_l = src_h + rows * COLS / 8;
+    for (int q8 = threadIdx.x; q8 < rows * COLS / 8; q8 += ESB2_NT) {
+      const int row = q8 / (COLS / 8), col8 = q8 % (COLS / 8);
+      *(uint4*)(ah + row * AROW + col8 * 8) = src_h[q8];
+      *(uint4*)(al + row * AROW + col8 * 8) = src_l[q8];
+    }
+  }
+
+  const int64_t tiles = (n + ESB2_BE - 1) / ESB2_BE;
+  for (int64_t tile = chunk; tile < tiles; tile += nchunk) {
+    const int64_t e0 = tile * ESB2_BE;
+    const int cnt = (int)min((int64_t)ESB2_BE, n - e0);
+    __syncthreads();
+    // transposed z staging, ones-row and zero-pad baked in
+    if (cnt == ESB2_BE) {
+      for (int idx = threadIdx.x; idx < d * ESB2_BE; idx += ESB2_NT) {
+        const int kk = idx / ESB2_BE, ei = idx % ESB2_BE;
+        zs[ei * AROW + kk] =
+            (__bf16)__bfloat162float(z[(int64_t)kk * n + e0 + ei]);
+      }
+    } else {
+      for (int idx = threadIdx.x; idx < d * ESB2_BE; idx += ESB2_NT) {
+        const int kk = idx / ESB2_BE, ei = idx % ESB2_BE;
+        zs[ei * AROW + kk] = (__bf16)(
+            (ei < cnt) ? __bfloat162float(z[(int64_t)kk * n + e0 + ei])
+                       : 0.0f);
+      }
+    }
+    for (int idx = threadIdx.x; idx < (COLS - d) * ESB2_BE;
+         idx += ESB2_NT) {
+      const int kk = d + idx / ESB2_BE, ei = idx % ESB2_BE;
+      zs[ei * AROW + kk] = (__bf16)((kk == d && ei < cnt) ? 1.0f : 0.0f);
+    }
+    __syncthreads();
+
+    // (row-tile, event-32-tile) pairs split across the 16 waves
+    for (int pair = wave; pair < rt_n * (ESB2_BE / 32);
+         pair += ESB2_NT / WAVE) {
+      const int rt = pair / (ESB2_BE / 32);
+      const int t = pair % (ESB2_BE / 32);
+      const __bf16* arow = ah + (rt * 32 + j32) * AROW;
+      const __bf16* lrow = al + (rt * 32 + j32) * AROW;
+      const __bf16* zrow = zs + (t * 32 + j32) * AROW;
+      f32x16 y = (f32x16)(0.0f);
+#pragma unroll
+      for (int kc = 0; kc < KCT; ++kc) {
+        const bf16x8 b = *(const bf16x8*)(zrow + kc * 16 + 8 * g2);
+        const bf16x8 fa = *(const bf16x8*)(arow + kc * 16 + 8 * g2);
+        const bf16x8 fl = *(const bf16x8*)(lrow + kc * 16 + 8 * g2);
+        y = __builtin_amdgcn_mfma_f32_32x32x16_bf16(fa, b, y, 0, 0, 0);
+        y = __builtin_amdgcn_mfma_f32_32x32x16_bf16(fl, b, y, 0, 0, 0);
+      }
+      float s = 0.0f;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) s = fmaf(y[r], y[r], s);
+      s += __shfl_xor(s, 32, WAVE);
+      if (lane < 32) qpart[rt * ESB2_BE + t * 32 + j32] = s;
+    }
+    __syncthreads();
+    // combine row-tile partials (fixed order) and write logw
+    if (threadIdx.x < ESB2_BE && threadIdx.x < cnt) {
+      const int e = threadIdx.x;
+      float q = qpart[e];
+      for (int rt = 1; rt < rt_n; ++rt) q += qpart[rt * ESB2_BE + e];
+      logw[(int64_t)c * n + e0 + e] = -0.5f * q + addc;
     }
   }
 }
