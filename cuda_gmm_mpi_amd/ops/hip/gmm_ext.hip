@@ -236,21 +236,6 @@ void constants(torch::Tensor r, torch::Tensor means, torch::Tensor pi,
                          r.data_ptr<float>(), means.data_ptr<float>(), mp,
                          mp32, d);
     }
-  } else if (d <= 48) {
-    // wave-per-cluster variant: 4 independent waves per block, zero
-    // workgroup barriers (the block variant pays ~100 __syncthreads per
-    // cluster, each draining the LDS pipeline)
-    const size_t lds =
-        4 * sizeof(float) * (2 * (size_t)d * (d | 1) + d);
-    if (lds > 64 * 1024) {
-      HIP_CHECK(hipFuncSetAttribute(
-          reinterpret_cast<const void*>(&gmm::constants_wave_kernel),
-          hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds));
-    }
-    hipLaunchKernelGGL(gmm::constants_wave_kernel, dim3((k + 3) / 4),
-                       dim3(kNT), lds, s, r.data_ptr<float>(),
-                       means.data_ptr<float>(), pip, rinv.data_ptr<float>(),
-                       logdet.data_ptr<float>(), cst, addp, mp, mp32, d, k);
   } else {
     // working buffer + read-only LU snapshot (+ u0 scratch for the factor)
     const size_t lds = sizeof(float) * (2 * (size_t)d * (d | 1) + d);

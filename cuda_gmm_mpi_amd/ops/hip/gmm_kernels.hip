@@ -508,191 +508,15 @@ __device__ inline void emit_mfac(float* a, float* o,
   }
 }
 
-// Wave-level LDS ordering: all 64 lanes of a CDNA wave share one program
-// counter, so cross-lane LDS visibility needs only completion of this
-// wave's own LDS ops (lgkmcnt) plus compiler fences — no workgroup
-// barrier. This is what makes the wave-per-cluster constants variant
-// ~barrier-free (the block variant pays ~100 __syncthreads per cluster,
-// each draining the LDS pipeline).
-__device__ inline void wsync() {
-  __builtin_amdgcn_wave_barrier();
-  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-  __builtin_amdgcn_wave_barrier();
-}
-
-// Wave-per-cluster constants (d <= 48): cluster c = blockIdx.x*4 + wave.
-// Same reference-faithful math as constants_lu_kernel (no-pivot LU,
-// ln|det|, snapshot triangular inversion, chol(R) factor emission), with
-// every loop parallelized over the 64 lanes of one wave and zero
-// workgroup barriers.
-__global__ void __launch_bounds__(NT)
-constants_wave_kernel(const float* __restrict__ r,
-                      const float* __restrict__ means,
-                      const float* __restrict__ pi,
-                      float* __restrict__ rinv, float* __restrict__ logdet,
-                      float* __restrict__ constant, float* __restrict__ add,
-                      __hip_bfloat16* __restrict__ mfac,
-                      float* __restrict__ mfac32, int d, int k) {
-  extern __shared__ float lds[];
-  const int wave = threadIdx.x / WAVE;
-  const int l = threadIdx.x & (WAVE - 1);
-  const int c = blockIdx.x * (NT / WAVE) + wave;
-  if (c >= k) return;
-  const int ldp = d | 1;
-  float* a = lds + (size_t)wave * (2 * d * ldp + d);
-  float* o = a + d * ldp;
-  float* u0 = o + d * ldp;
-  const float* rc = r + (int64_t)c * d * d;
-  float* oc = rinv + (int64_t)c * d * d;
-
-  for (int t = l; t < d * d; t += WAVE) a[(t / d) * ldp + t % d] = rc[t];
-  wsync();
-
-  if (d == 1) {
-    if (l == 0) {
-      const float lg = __logf(a[0]);
-      logdet[c] = lg;
-      const float cst = -0.5f * 1.8378770664093453f - 0.5f * lg;
-      if (constant) constant[c] = cst;
-      if (add) add[c] = cst + __logf(pi[c]);
-      oc[0] = 1.0f / a[0];
-    }
-    wsync();
-  } else {
-    for (int j = 1 + l; j < d; j += WAVE) a[j] /= a[0];
-    wsync();
-    for (int i = 1; i < d; ++i) {
-      for (int j = i + l; j < d; j += WAVE) {
-        float s = 0.0f;
-#pragma unroll 8
-        for (int kk = 0; kk < i; ++kk)
-          s = fmaf(a[j * ldp + kk], a[kk * ldp + i], s);
-        a[j * ldp + i] -= s;
-      }
-      wsync();
-      if (i == d - 1) break;
-      const float pivot = a[i * ldp + i];
-      for (int j = i + 1 + l; j < d; j += WAVE) {
-        float s = 0.0f;
-#pragma unroll 8
-        for (int kk = 0; kk < i; ++kk)
-          s = fmaf(a[i * ldp + kk], a[kk * ldp + j], s);
-        a[i * ldp + j] = (a[i * ldp + j] - s) / pivot;
-      }
-      wsync();
-    }
-    {
-      float acc = 0.0f;
-      for (int i = l; i < d; i += WAVE) acc += __logf(fabsf(a[i * ldp + i]));
-      for (int off = WAVE / 2; off > 0; off >>= 1)
-        acc += __shfl_down(acc, off, WAVE);
-      if (l == 0) {
-        logdet[c] = acc;
-        const float cst = -d * 0.5f * 1.8378770664093453f - 0.5f * acc;
-        if (constant) constant[c] = cst;
-        if (add) add[c] = cst + __logf(pi[c]);
-      }
-    }
-    for (int t = l; t < d * ldp; t += WAVE) o[t] = a[t];
-    wsync();
-    // invert L (lane = column; d <= 48 < WAVE: one lane each)
-    if (l < d) {
-      const int i = l;
-      for (int j = i; j < d; ++j) {
-        float xv = 1.0f;
-        if (i != j) {
-          xv = 0.0f;
-#pragma unroll 8
-          for (int kk = i; kk < j; ++kk)
-            xv -= o[j * ldp + kk] * a[kk * ldp + i];
-        }
-        a[j * ldp + i] = xv / o[j * ldp + j];
-      }
-      // invert U (lane = row); disjoint writes from the L columns
-      const int i2 = l;
-      for (int j = i2 + 1; j < d; ++j) {
-        float s = 0.0f;
-#pragma unroll 8
-        for (int kk = i2; kk < j; ++kk)
-          s += o[kk * ldp + j] * ((i2 == kk) ? 1.0f : a[i2 * ldp + kk]);
-        a[i2 * ldp + j] = -s;
-      }
-    }
-    wsync();
-    for (int t = l; t < d * d; t += WAVE) {
-      const int j = t / d, i = t % d;
-      float s = 0.0f;
-#pragma unroll 8
-      for (int kk = (i > j ? i : j); kk < d; ++kk)
-        s = fmaf((j == kk) ? 1.0f : a[j * ldp + kk], a[kk * ldp + i], s);
-      oc[j * d + i] = s;
-    }
-  }
-  if (mfac == nullptr) return;
-
-  // ---- factor emission: lower Cholesky of R, F = L^-1, M = [F | -F mu]
-  for (int t = l; t < d * d; t += WAVE)
-    a[(t / d) * ldp + t % d] = rc[t];
-  wsync();
-  for (int j = 0; j < d; ++j) {
-    if (l == 0) {
-      const float diag0 = fabsf(a[j * ldp + j]);
-      float s = a[j * ldp + j];
-      for (int kk = 0; kk < j; ++kk) s -= a[j * ldp + kk] * a[j * ldp + kk];
-      a[j * ldp + j] = sqrtf(fmaxf(s, 1e-8f * diag0 + 1e-30f));
-    }
-    wsync();
-    const float piv = a[j * ldp + j];
-    for (int i = j + 1 + l; i < d; i += WAVE) {
-      float s = a[i * ldp + j];
-#pragma unroll 8
-      for (int kk = 0; kk < j; ++kk) s -= a[i * ldp + kk] * a[j * ldp + kk];
-      a[i * ldp + j] = s / piv;
-    }
-    wsync();
-  }
-  for (int t = l; t < d * ldp; t += WAVE) o[t] = a[t];
-  wsync();
-  if (l < d) {
-    const int i = l;
-    for (int j = i; j < d; ++j) {
-      float xv = 1.0f;
-      if (i != j) {
-        xv = 0.0f;
-#pragma unroll 8
-        for (int kk = i; kk < j; ++kk)
-          xv -= o[j * ldp + kk] * a[kk * ldp + i];
-      }
-      a[j * ldp + i] = xv / o[j * ldp + j];
-    }
-  }
-  wsync();
-  for (int i = l; i < d; i += WAVE) {
-    float s = 0.0f;
-#pragma unroll 8
-    for (int j = 0; j <= i; ++j) s += a[i * ldp + j] * means[c * d + j];
-    u0[i] = -s;
-  }
-  wsync();
-  const int rows = ((d + 31) / 32) * 32;
-  const int cols = kc_tier((d + 1 + 15) / 16) * 16;
-  const int cells = rows * cols;
-  __hip_bfloat16* out = mfac + (int64_t)c * 2 * cells;
-  float* out32 = mfac32 ? mfac32 + (int64_t)c * cells : nullptr;
-  for (int t = l; t < cells; t += WAVE) {
-    const int i = t / cols, kx = t % cols;
-    float v = 0.0f;
-    if (i < d) {
-      if (kx < d) v = (kx <= i) ? a[i * ldp + kx] : 0.0f;
-      else if (kx == d) v = u0[i];
-    }
-    const __hip_bfloat16 hi = __float2bfloat16(v);
-    out[t] = hi;
-    out[cells + t] = __float2bfloat16(v - __bfloat162float(hi));
-    if (out32) out32[t] = v;
-  }
-}
-
+// NOTE (negative result, round 2): a wave-per-cluster constants variant
+// (4 independent waves per block, zero workgroup barriers, cross-lane
+// LDS ordering via the wave's own lgkmcnt) measured SLOWER than this
+// block-per-cluster kernel: 64/103 us vs 47/77 us (LU-only / LU+emit,
+// K=64 D=24). With one cluster per wave only K/4 blocks run (16 CUs at
+// K=64 vs 64) and each serial chain has a quarter of the lanes and no
+// co-resident waves to hide LDS latency; the ~100 __syncthreads of the
+// block variant are cheaper than that. Kept here as a ledger entry —
+// do not rediscover (full details in profiles/LADDER.md).
 __global__ void __launch_bounds__(NT)
 constants_lu_kernel(const float* __restrict__ r,
                     const float* __restrict__ means,
@@ -1693,20 +1517,29 @@ mstep_moments_big_kernel(const float* __restrict__ x,
       const float4 wv1 = *(const float4*)(wt + cw * MBB_BK + eb + 4);
       const float wv[8] = {wv0.x, wv0.y, wv0.z, wv0.w,
                            wv1.x, wv1.y, wv1.z, wv1.w};
+      // the weighted A-side hi/lo split is per (row-tile, chunk) only:
+      // hoist it across tile-pairs sharing tr (the quarter assignment
+      // keeps same-tr pairs adjacent — ~2x less split VALU at RT2=5)
+      int prev_tr = -1;
+      bf16x8 a_hi, a_lo;
 #pragma unroll
       for (int pp = 0; pp < MBB_PMAX; ++pp) {
         if (p_lo + pp >= p_hi) break;
         const int tr = ptr[pp], tc = ptc[pp];
-        const bf16x8 zah = *(const bf16x8*)(zhi + (tr * 32 + j32) * zbr + eb);
-        const bf16x8 zal = *(const bf16x8*)(zlo + (tr * 32 + j32) * zbr + eb);
-        bf16x8 a_hi, a_lo;
+        if (tr != prev_tr) {
+          const bf16x8 zah =
+              *(const bf16x8*)(zhi + (tr * 32 + j32) * zbr + eb);
+          const bf16x8 zal =
+              *(const bf16x8*)(zlo + (tr * 32 + j32) * zbr + eb);
 #pragma unroll
-        for (int u = 0; u < 8; ++u) {
-          const float zf = (float)zah[u] + (float)zal[u];
-          const float av = wv[u] * zf;
-          const __bf16 hi = (__bf16)av;
-          a_hi[u] = hi;
-          a_lo[u] = (__bf16)(av - (float)hi);
+          for (int u = 0; u < 8; ++u) {
+            const float zf = (float)zah[u] + (float)zal[u];
+            const float av = wv[u] * zf;
+            const __bf16 hi = (__bf16)av;
+            a_hi[u] = hi;
+            a_lo[u] = (__bf16)(av - (float)hi);
+          }
+          prev_tr = tr;
         }
         const bf16x8 b_hi = *(const bf16x8*)(zhi + (tc * 32 + j32) * zbr + eb);
         const bf16x8 b_lo = *(const bf16x8*)(zlo + (tc * 32 + j32) * zbr + eb);
