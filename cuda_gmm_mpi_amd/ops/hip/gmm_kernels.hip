@@ -46,6 +46,8 @@ __device__ inline void tri_row_col(int t, int* i, int* j) {
 __device__ inline int mfma_b16_k(int group, int u) { return 8 * group + u; }
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(2))) __bf16 bf16x2;
+typedef __attribute__((ext_vector_type(2))) float f32x2;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 typedef __attribute__((ext_vector_type(16))) float f32x16_t;
 
@@ -1002,14 +1004,24 @@ mstep_moments_b16_kernel(const __hip_bfloat16* __restrict__ xhi,
       const float4 wv1 = *(const float4*)(wt + wave * MB_BK + eb + 4);
       const float wv[8] = {wv0.x, wv0.y, wv0.z, wv0.w,
                            wv1.x, wv1.y, wv1.z, wv1.w};
+      // packed A-side split (v_pk_add/mul/fma on f32 pairs): the scalar
+      // per-element form was the kernel's VALU bottleneck (70% VALUBusy)
       bf16x8 a_hi, a_lo;
+      const bf16x2* h2 = (const bf16x2*)&b_hi;
+      const bf16x2* l2 = (const bf16x2*)&b_lo;
+      bf16x2* ah2 = (bf16x2*)&a_hi;
+      bf16x2* al2 = (bf16x2*)&a_lo;
+      const f32x2* wv2 = (const f32x2*)wv;
 #pragma unroll
-      for (int u = 0; u < 8; ++u) {
-        const float zf = (float)b_hi[u] + (float)b_lo[u];
-        const float av = wv[u] * zf;
-        const __bf16 hi = (__bf16)av;
-        a_hi[u] = hi;
-        a_lo[u] = (__bf16)(av - (float)hi);
+      for (int u2 = 0; u2 < 4; ++u2) {
+        const f32x2 hf = {(float)h2[u2][0], (float)h2[u2][1]};
+        const f32x2 lf = {(float)l2[u2][0], (float)l2[u2][1]};
+        const f32x2 av = wv2[u2] * (hf + lf);
+        const bf16x2 hi = {(__bf16)av.x, (__bf16)av.y};
+        const f32x2 hif = {(float)hi[0], (float)hi[1]};
+        const f32x2 lo = av - hif;
+        ah2[u2] = hi;
+        al2[u2] = (bf16x2){(__bf16)lo.x, (__bf16)lo.y};
       }
       accA = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a_hi, b_hi, accA, 0, 0, 0);
       accB = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a_hi, b_lo, accB, 0, 0, 0);
@@ -1193,9 +1205,15 @@ estep_fused_kernel(const __hip_bfloat16* __restrict__ z,
       y = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a_h1, b1, y, 0, 0, 0);
       y = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a_l0, b0, y, 0, 0, 0);
       y = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a_l1, b1, y, 0, 0, 0);
-      float s = 0.0f;
+      // packed dot (v_pk_fma_f32: 2 fp32 FMAs/instr — the VALU-bound
+      // epilogue was 128 accvgpr reads + 122 scalar fmac per c-iteration)
+      f32x2 s2 = {0.0f, 0.0f};
 #pragma unroll
-      for (int r = 0; r < 16; ++r) s = fmaf(y[r], y[r], s);
+      for (int r = 0; r < 8; ++r) {
+        const f32x2 yp = {y[2 * r], y[2 * r + 1]};
+        s2 = yp * yp + s2;
+      }
+      float s = s2.x + s2.y;
       // the 32 Y rows live across the two lane halves: one cross-half sum
       s += __shfl_xor(s, 32, WAVE);
       if (lane < 32) {
@@ -1338,9 +1356,13 @@ estep_fused_f32_kernel(const float* __restrict__ z,
         const float b = zrow[2 * ch + g2];
         y = __builtin_amdgcn_mfma_f32_32x32x2f32(a[ch], b, y, 0, 0, 0);
       }
-      float s = 0.0f;
+      f32x2 s2 = {0.0f, 0.0f};
 #pragma unroll
-      for (int r = 0; r < 16; ++r) s = fmaf(y[r], y[r], s);
+      for (int r = 0; r < 8; ++r) {
+        const f32x2 yp = {y[2 * r], y[2 * r + 1]};
+        s2 = yp * yp + s2;
+      }
+      float s = s2.x + s2.y;
       s += __shfl_xor(s, 32, WAVE);
       if (lane < 32) {
         const int e = t * 32 + j32;
@@ -1531,13 +1553,21 @@ mstep_moments_big_kernel(const float* __restrict__ x,
               *(const bf16x8*)(zhi + (tr * 32 + j32) * zbr + eb);
           const bf16x8 zal =
               *(const bf16x8*)(zlo + (tr * 32 + j32) * zbr + eb);
+          const bf16x2* h2 = (const bf16x2*)&zah;
+          const bf16x2* l2 = (const bf16x2*)&zal;
+          bf16x2* ah2 = (bf16x2*)&a_hi;
+          bf16x2* al2 = (bf16x2*)&a_lo;
+          const f32x2* wv2 = (const f32x2*)wv;
 #pragma unroll
-          for (int u = 0; u < 8; ++u) {
-            const float zf = (float)zah[u] + (float)zal[u];
-            const float av = wv[u] * zf;
-            const __bf16 hi = (__bf16)av;
-            a_hi[u] = hi;
-            a_lo[u] = (__bf16)(av - (float)hi);
+          for (int u2 = 0; u2 < 4; ++u2) {
+            const f32x2 hf = {(float)h2[u2][0], (float)h2[u2][1]};
+            const f32x2 lf = {(float)l2[u2][0], (float)l2[u2][1]};
+            const f32x2 av = wv2[u2] * (hf + lf);
+            const bf16x2 hi = {(__bf16)av.x, (__bf16)av.y};
+            const f32x2 hif = {(float)hi[0], (float)hi[1]};
+            const f32x2 lo = av - hif;
+            ah2[u2] = hi;
+            al2[u2] = (bf16x2){(__bf16)lo.x, (__bf16)lo.y};
           }
           prev_tr = tr;
         }
@@ -1729,25 +1759,36 @@ estep_logw_big2_kernel(const __hip_bfloat16* __restrict__ z,
     const int64_t e0 = tile * ESB2_BE;
     const int cnt = (int)min((int64_t)ESB2_BE, n - e0);
     __syncthreads();
-    // transposed z staging, ones-row and zero-pad baked in
-    if (cnt == ESB2_BE) {
-      for (int idx = threadIdx.x; idx < d * ESB2_BE; idx += ESB2_NT) {
-        const int kk = idx / ESB2_BE, ei = idx % ESB2_BE;
+    // transposed z staging: each thread gathers 4 consecutive k-slots of
+    // one event (4 coalesced-by-wave global reads) and writes ONE 8-byte
+    // ds_write_b64 — scalar b16 transpose writes were a 4-way bank
+    // conflict (stride-AROW lanes 8 apart collide mod 32 for any 16B
+    // row stride; 20% LDSBankConflict measured)
+    {
+      const int kq_total = ((d + 3) / 4) * ESB2_BE;
+      for (int idx = threadIdx.x; idx < kq_total; idx += ESB2_NT) {
+        const int kk0 = (idx / ESB2_BE) * 4, ei = idx % ESB2_BE;
+        __bf16 v[4];
+#pragma unroll
+        for (int u = 0; u < 4; ++u) {
+          const int kk = kk0 + u;
+          float vf = 0.0f;
+          if (ei < cnt) {
+            if (kk < d) vf = __bfloat162float(z[(int64_t)kk * n + e0 + ei]);
+            else if (kk == d) vf = 1.0f;  // ones row inside the quad
+          }
+          v[u] = (__bf16)vf;
+        }
+        *(uint2*)(zs + ei * AROW + kk0) = *(uint2*)v;
+      }
+      const int kpad0 = ((d + 3) / 4) * 4;
+      for (int idx = threadIdx.x; idx < (COLS - kpad0) * ESB2_BE;
+           idx += ESB2_NT) {
+        const int kk = kpad0 + idx / ESB2_BE, ei = idx % ESB2_BE;
+        // d is only >= kpad0 when d % 4 == 0: the ones row then lives here
         zs[ei * AROW + kk] =
-            (__bf16)__bfloat162float(z[(int64_t)kk * n + e0 + ei]);
+            (__bf16)((kk == d && ei < cnt) ? 1.0f : 0.0f);
       }
-    } else {
-      for (int idx = threadIdx.x; idx < d * ESB2_BE; idx += ESB2_NT) {
-        const int kk = idx / ESB2_BE, ei = idx % ESB2_BE;
-        zs[ei * AROW + kk] = (__bf16)(
-            (ei < cnt) ? __bfloat162float(z[(int64_t)kk * n + e0 + ei])
-                       : 0.0f);
-      }
-    }
-    for (int idx = threadIdx.x; idx < (COLS - d) * ESB2_BE;
-         idx += ESB2_NT) {
-      const int kk = d + idx / ESB2_BE, ei = idx % ESB2_BE;
-      zs[ei * AROW + kk] = (__bf16)((kk == d && ei < cnt) ? 1.0f : 0.0f);
     }
     __syncthreads();
 
@@ -1768,9 +1809,13 @@ estep_logw_big2_kernel(const __hip_bfloat16* __restrict__ z,
         y = __builtin_amdgcn_mfma_f32_32x32x16_bf16(fa, b, y, 0, 0, 0);
         y = __builtin_amdgcn_mfma_f32_32x32x16_bf16(fl, b, y, 0, 0, 0);
       }
-      float s = 0.0f;
+      f32x2 s2 = {0.0f, 0.0f};
 #pragma unroll
-      for (int r = 0; r < 16; ++r) s = fmaf(y[r], y[r], s);
+      for (int r = 0; r < 8; ++r) {
+        const f32x2 yp = {y[2 * r], y[2 * r + 1]};
+        s2 = yp * yp + s2;
+      }
+      float s = s2.x + s2.y;
       s += __shfl_xor(s, 32, WAVE);
       if (lane < 32) qpart[rt * ESB2_BE + t * 32 + j32] = s;
     }
