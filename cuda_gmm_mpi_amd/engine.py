@@ -170,10 +170,10 @@ class EmEngine:
             pi_add = (
                 (st.pi, self._add[:k]) if self._add is not None else None
             )
-            rinv, const = F.constants(st.R, st.means, self.cfg.diag_only,
-                                      mfac, mfac32, pi_add=pi_add)
-            st.Rinv.copy_(rinv)
-            st.constant.copy_(const)
+            # written in place into the state (no copy-back kernels)
+            F.constants(st.R, st.means, self.cfg.diag_only,
+                        mfac, mfac32, pi_add=pi_add,
+                        out=(st.Rinv, st.constant))
         self.profile.count("constants")
 
     def _sync_add(self, k: int) -> None:

@@ -57,7 +57,9 @@ def estep_lse(logw: torch.Tensor, lse: torch.Tensor) -> torch.Tensor:
     nblocks = 1024
     partial = torch.zeros(nblocks, dtype=torch.float32, device=logw.device)
     hip_ext().estep_lse(logw, lse, partial)
-    return partial.sum()
+    lik = torch.empty(1, dtype=torch.float32, device=logw.device)
+    hip_ext().reduce_scalar(partial, lik)
+    return lik
 
 
 def mstep_n_means(x_aug_t: torch.Tensor, w: torch.Tensor
@@ -113,7 +115,8 @@ def mstep_covariance_s(x: torch.Tensor, w: torch.Tensor,
         if lse is None:
             lse = torch.empty(0, dtype=torch.float32, device=x.device)
         hip_ext().mstep_covariance_partials(x, w, lse, partials)
-        packed = partials.sum(dim=0)                      # [K, P]
+        packed = torch.empty((k, p), dtype=torch.float32, device=x.device)
+        hip_ext().reduce_chunks(partials, packed)         # [K, P]
         idx = _tri_unpack_index(d, x.device)
         s = packed[:, idx].view(k, d, d)
         if out is not None:
@@ -132,7 +135,8 @@ def mstep_covariance_s(x: torch.Tensor, w: torch.Tensor,
 def constants(r: torch.Tensor, means: torch.Tensor | None = None,
               diag_only: bool = False, mfac: torch.Tensor | None = None,
               mfac32: torch.Tensor | None = None,
-              pi_add: tuple[torch.Tensor, torch.Tensor] | None = None
+              pi_add: tuple[torch.Tensor, torch.Tensor] | None = None,
+              out: tuple[torch.Tensor, torch.Tensor] | None = None
               ) -> tuple[torch.Tensor, torch.Tensor]:
     """(Rinv [K,D,D], constant [K]) via no-pivot LU + ln|det|.
 
@@ -142,7 +146,7 @@ def constants(r: torch.Tensor, means: torch.Tensor | None = None,
     """
     if r.is_cuda:
         k, d, _ = r.shape
-        rinv = torch.empty_like(r)
+        rinv = out[0] if out is not None else torch.empty_like(r)
         logdet = torch.empty(k, dtype=torch.float32, device=r.device)
         if means is None:
             means = torch.zeros(k, d, dtype=torch.float32, device=r.device)
@@ -153,18 +157,32 @@ def constants(r: torch.Tensor, means: torch.Tensor | None = None,
         empty = torch.empty(0, dtype=torch.float32, device=r.device)
         if pi_add is not None:
             pi_t, add_t = pi_add
-            const_t = torch.empty(k, dtype=torch.float32, device=r.device)
+            const_t = (out[1] if out is not None
+                       else torch.empty(k, dtype=torch.float32,
+                                        device=r.device))
             hip_ext().constants(r, means, pi_t, rinv, logdet, const_t,
                                 add_t, mfac, mfac32, bool(diag_only))
             if diag_only:  # diag kernel does not emit constant/add
-                const_t = -d * 0.5 * cpu.LOG_2PI - 0.5 * logdet
-                add_t.copy_(const_t + torch.log(pi_t))
+                cval = -d * 0.5 * cpu.LOG_2PI - 0.5 * logdet
+                if out is not None:
+                    const_t.copy_(cval)
+                else:
+                    const_t = cval
+                add_t.copy_(cval + torch.log(pi_t))
             return rinv, const_t
         hip_ext().constants(r, means, empty, rinv, logdet, empty, empty,
                             mfac, mfac32, bool(diag_only))
         const = -d * 0.5 * cpu.LOG_2PI - 0.5 * logdet
+        if out is not None:
+            out[1].copy_(const)
+            const = out[1]
         return rinv, const
-    return cpu.compute_constants(r, diag_only)
+    rinv_c, const_c = cpu.compute_constants(r, diag_only)
+    if out is not None:
+        out[0].copy_(rinv_c)
+        out[1].copy_(const_c)
+        return out[0], out[1]
+    return rinv_c, const_c
 
 
 def estep_fused(z: torch.Tensor, mfac: torch.Tensor, add: torch.Tensor,
@@ -179,7 +197,9 @@ def estep_fused(z: torch.Tensor, mfac: torch.Tensor, add: torch.Tensor,
     # every launched block writes its partial slot: no zero-fill needed
     partial = torch.empty(nblk, dtype=torch.float32, device=z.device)
     hip_ext().estep_fused(z, mfac, add, w_out, lse, partial)
-    return w_out, partial.sum()
+    lik = torch.empty(1, dtype=torch.float32, device=z.device)
+    hip_ext().reduce_scalar(partial, lik)
+    return w_out, lik
 
 
 def estep_fused_available(device: torch.device, dtype: str, d: int,
@@ -198,7 +218,9 @@ def estep_fused_f32(z: torch.Tensor, mfac32: torch.Tensor, add: torch.Tensor,
     nblk = (n + 255) // 256
     partial = torch.empty(nblk, dtype=torch.float32, device=z.device)
     hip_ext().estep_fused_f32(z, mfac32, add, w_out, lse, partial)
-    return w_out, partial.sum()
+    lik = torch.empty(1, dtype=torch.float32, device=z.device)
+    hip_ext().reduce_scalar(partial, lik)
+    return w_out, lik
 
 
 def estep_big_available(device: torch.device, dtype: str, d: int) -> bool:
@@ -274,7 +296,9 @@ def mstep_moments(x: torch.Tensor, w: torch.Tensor,
                                         partials)
         else:
             hip_ext().mstep_moments(x, w, lse_t, partials)
-        return partials.sum(dim=0)
+        out = torch.empty((k, pp), dtype=torch.float32, device=x.device)
+        hip_ext().reduce_chunks(partials, out)
+        return out
     if x.is_cuda and precision == "bf16x3" and d <= 159:
         tiles = (n + 63) // 64  # MBB_BK
         if nchunk is None:
@@ -289,7 +313,9 @@ def mstep_moments(x: torch.Tensor, w: torch.Tensor,
         lse_t = (lse if lse is not None
                  else torch.empty(0, dtype=torch.float32, device=x.device))
         hip_ext().mstep_moments_big(x, w, lse_t, partials)
-        return partials.sum(dim=0)
+        out = torch.empty((k, pp), dtype=torch.float32, device=x.device)
+        hip_ext().reduce_chunks(partials, out)
+        return out
     packed = torch.empty((k, pp), dtype=torch.float32, device=x.device)
     if x.is_cuda:
         if lse is not None:

@@ -469,6 +469,29 @@ void mfma_probe32(torch::Tensor a, torch::Tensor b, torch::Tensor c) {
   HIP_CHECK(hipGetLastError());
 }
 
+void reduce_chunks(torch::Tensor partials, torch::Tensor out) {
+  check_f32(partials, "partials");
+  check_f32(out, "out");
+  const int c = (int)partials.size(0);
+  const int64_t m = partials.numel() / c;
+  TORCH_CHECK(out.numel() == m, "out shape");
+  const int grid = (int)std::min<int64_t>((m + kNT - 1) / kNT, 4096);
+  hipLaunchKernelGGL(gmm::reduce_chunks_kernel, dim3(grid), dim3(kNT), 0,
+                     stream(), partials.data_ptr<float>(),
+                     out.data_ptr<float>(), c, m);
+  HIP_CHECK(hipGetLastError());
+}
+
+void reduce_scalar(torch::Tensor in, torch::Tensor out) {
+  check_f32(in, "in");
+  check_f32(out, "out");
+  TORCH_CHECK(out.numel() >= 1);
+  hipLaunchKernelGGL(gmm::reduce_scalar_kernel, dim3(1), dim3(kNT), 0,
+                     stream(), in.data_ptr<float>(), out.data_ptr<float>(),
+                     in.numel());
+  HIP_CHECK(hipGetLastError());
+}
+
 void mstep_finalize(torch::Tensor packed, torch::Tensor avgvar,
                     int64_t world, torch::Tensor n_out, torch::Tensor means,
                     torch::Tensor r_out, torch::Tensor pi, bool diag_only) {
@@ -583,6 +606,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "split-precision bf16x3 augmented moments");
   m.def("mstep_moments", &mstep_moments,
         "fused augmented moments [S|mean_num|N] via f32 MFMA");
+  m.def("reduce_chunks", &reduce_chunks,
+        "deterministic chunk-partials reduction out[j] = sum_i in[i][j]");
+  m.def("reduce_scalar", &reduce_scalar,
+        "deterministic scalar sum out[0] = sum(in)");
   m.def("mstep_finalize", &mstep_finalize,
         "finalize N/means/R/pi from all-reduced packed moments");
   m.def("emit_factors", &emit_factors,

@@ -259,6 +259,39 @@ estep_lse_kernel(const float* __restrict__ logw, float* __restrict__ lse,
   }
 }
 
+// Deterministic chunk reduction: out[j] = sum_i in[i*m + j] (fixed i
+// order). Replaces torch's generic reduce_kernel (19.6 us for the
+// [256, K*Pp] moments partials; this is bandwidth-bound ~4 us).
+__global__ void __launch_bounds__(NT)
+reduce_chunks_kernel(const float* __restrict__ in, float* __restrict__ out,
+                     int c, int64_t m) {
+  for (int64_t j = (int64_t)blockIdx.x * NT + threadIdx.x; j < m;
+       j += (int64_t)gridDim.x * NT) {
+    float s = 0.0f;
+    for (int i = 0; i < c; ++i) s += in[(int64_t)i * m + j];
+    out[j] = s;
+  }
+}
+
+// Scalar sum: out[0] = sum in[0..n) (single block, fixed order per lane
+// then wave-order combine — deterministic).
+__global__ void __launch_bounds__(NT)
+reduce_scalar_kernel(const float* __restrict__ in, float* __restrict__ out,
+                     int64_t n) {
+  float acc = 0.0f;
+  for (int64_t i = threadIdx.x; i < n; i += NT) acc += in[i];
+  __shared__ float wsum[NT / WAVE];
+  for (int off = WAVE / 2; off > 0; off >>= 1)
+    acc += __shfl_down(acc, off, WAVE);
+  if ((threadIdx.x & (WAVE - 1)) == 0) wsum[threadIdx.x / WAVE] = acc;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float total = 0.0f;
+    for (int wv = 0; wv < NT / WAVE; ++wv) total += wsum[wv];
+    out[0] = total;
+  }
+}
+
 // ---------------------------------------------------------------------------
 // M-step covariance: packed second moments S_c = sum_e w_ce x_e x_e^T
 // (replaces mstep_covariance1, gaussian_kernel.cu:605 — but uncentered;
@@ -1769,15 +1802,24 @@ estep_logw_big2_kernel(const __hip_bfloat16* __restrict__ z,
       for (int idx = threadIdx.x; idx < kq_total; idx += ESB2_NT) {
         const int kk0 = (idx / ESB2_BE) * 4, ei = idx % ESB2_BE;
         __bf16 v[4];
+        // kk0 is wave-uniform (ESB2_BE >= WAVE): full quads take the
+        // branchless path — per-element ternaries around loads would
+        // serialize them behind vmcnt(0) (guide trap 4c, measured -15%)
+        if (cnt == ESB2_BE && kk0 + 3 < d) {
+          const __hip_bfloat16* zp = z + (int64_t)kk0 * n + e0 + ei;
 #pragma unroll
-        for (int u = 0; u < 4; ++u) {
-          const int kk = kk0 + u;
-          float vf = 0.0f;
-          if (ei < cnt) {
-            if (kk < d) vf = __bfloat162float(z[(int64_t)kk * n + e0 + ei]);
-            else if (kk == d) vf = 1.0f;  // ones row inside the quad
+          for (int u = 0; u < 4; ++u) v[u] = (__bf16)(zp[(int64_t)u * n]);
+        } else {
+#pragma unroll
+          for (int u = 0; u < 4; ++u) {
+            const int kk = kk0 + u;
+            float vf = 0.0f;
+            if (ei < cnt) {
+              if (kk < d) vf = __bfloat162float(z[(int64_t)kk * n + e0 + ei]);
+              else if (kk == d) vf = 1.0f;  // ones row inside the quad
+            }
+            v[u] = (__bf16)vf;
           }
-          v[u] = (__bf16)vf;
         }
         *(uint2*)(zs + ei * AROW + kk0) = *(uint2*)v;
       }

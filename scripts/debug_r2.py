@@ -47,6 +47,42 @@ def nan_isolation():
         print(f"[nan] {variant}: {['%.6e' % v for v in liks]}", flush=True)
 
 
+def flagship_timing():
+    rng = np.random.default_rng(2)
+    d, n, k = 24, 1_000_000, 64
+    x = torch.from_numpy(
+        rng.standard_normal((d, n)).astype(np.float32)).cuda()
+    xb = x.to(torch.bfloat16)
+    xs = F.split_bf16_planes(x)
+    w = torch.rand(k, n, device="cuda")
+    lse = torch.zeros(n, device="cuda")
+    means = torch.randn(k, d, device="cuda")
+    r = torch.eye(d, device="cuda").expand(k, d, d).contiguous() * 3
+    mfac = torch.empty(k, 2, 32, 32, dtype=torch.bfloat16, device="cuda")
+    rinv, const = F.constants(r, means, False, mfac)
+    add = const + float(np.log(1.0 / k))
+    wo = torch.empty(k, n, device="cuda")
+    print("[f] estep_fused bf16  : %.3f ms"
+          % timeit(lambda: F.estep_fused(xb, mfac, add, wo, lse), 50),
+          flush=True)
+    lw = torch.randn(k, n, device="cuda") * 3
+    print("[f] moments_b16 +lse  : %.3f ms"
+          % timeit(lambda: F.mstep_moments(x, lw, precision="bf16x3",
+                                           x_split=xs, lse=lse), 30),
+          flush=True)
+    print("[f] moments_b16 plain : %.3f ms"
+          % timeit(lambda: F.mstep_moments(x, w, precision="bf16x3",
+                                           x_split=xs), 30), flush=True)
+    mfac32 = torch.empty(k, 32, 32, dtype=torch.float32, device="cuda")
+    rinv, const = F.constants(r, means, False, mfac, mfac32)
+    print("[f] estep_fused_f32   : %.3f ms"
+          % timeit(lambda: F.estep_fused_f32(x, mfac32, add, wo, lse), 30),
+          flush=True)
+    print("[f] moments_f32 +lse  : %.3f ms"
+          % timeit(lambda: F.mstep_moments(x, lw, precision="fp32",
+                                           lse=lse), 20), flush=True)
+
+
 def big_path_timing():
     rng = np.random.default_rng(1)
     d, n, k = 128, 500_000, 256
@@ -141,5 +177,7 @@ if __name__ == "__main__":
         nan_isolation()
     if which in ("all", "timing"):
         big_path_timing()
+    if which in ("all", "flagship"):
+        flagship_timing()
     if which in ("all", "constants"):
         constants_decomposition()

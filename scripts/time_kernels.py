@@ -22,5 +22,6 @@ mfac = torch.empty(k,2,32,32,dtype=torch.bfloat16,device="cuda")
 rinv, const = F.constants(r, means, False, mfac)
 add = const + torch.log(torch.full((k,),1.0/k,device="cuda"))
 xb = x.to(torch.bfloat16); wo = torch.empty(k,n,device="cuda")
-def es(): return F.estep_fused(xb, mfac, add, wo)
+lse = torch.empty(n, device="cuda")
+def es(): return F.estep_fused(xb, mfac, add, wo, lse)
 print("estep_fused  : %.3f ms" % t(es))
