@@ -1037,24 +1037,14 @@ mstep_moments_b16_kernel(const __hip_bfloat16* __restrict__ xhi,
       const float4 wv1 = *(const float4*)(wt + wave * MB_BK + eb + 4);
       const float wv[8] = {wv0.x, wv0.y, wv0.z, wv0.w,
                            wv1.x, wv1.y, wv1.z, wv1.w};
-      // packed A-side split (v_pk_add/mul/fma on f32 pairs): the scalar
-      // per-element form was the kernel's VALU bottleneck (70% VALUBusy)
       bf16x8 a_hi, a_lo;
-      const bf16x2* h2 = (const bf16x2*)&b_hi;
-      const bf16x2* l2 = (const bf16x2*)&b_lo;
-      bf16x2* ah2 = (bf16x2*)&a_hi;
-      bf16x2* al2 = (bf16x2*)&a_lo;
-      const f32x2* wv2 = (const f32x2*)wv;
 #pragma unroll
-      for (int u2 = 0; u2 < 4; ++u2) {
-        const f32x2 hf = {(float)h2[u2][0], (float)h2[u2][1]};
-        const f32x2 lf = {(float)l2[u2][0], (float)l2[u2][1]};
-        const f32x2 av = wv2[u2] * (hf + lf);
-        const bf16x2 hi = {(__bf16)av.x, (__bf16)av.y};
-        const f32x2 hif = {(float)hi[0], (float)hi[1]};
-        const f32x2 lo = av - hif;
-        ah2[u2] = hi;
-        al2[u2] = (bf16x2){(__bf16)lo.x, (__bf16)lo.y};
+      for (int u = 0; u < 8; ++u) {
+        const float zf = (float)b_hi[u] + (float)b_lo[u];
+        const float av = wv[u] * zf;
+        const __bf16 hi = (__bf16)av;
+        a_hi[u] = hi;
+        a_lo[u] = (__bf16)(av - (float)hi);
       }
       accA = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a_hi, b_hi, accA, 0, 0, 0);
       accB = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a_hi, b_lo, accB, 0, 0, 0);
@@ -1238,15 +1228,9 @@ estep_fused_kernel(const __hip_bfloat16* __restrict__ z,
       y = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a_h1, b1, y, 0, 0, 0);
       y = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a_l0, b0, y, 0, 0, 0);
       y = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a_l1, b1, y, 0, 0, 0);
-      // packed dot (v_pk_fma_f32: 2 fp32 FMAs/instr — the VALU-bound
-      // epilogue was 128 accvgpr reads + 122 scalar fmac per c-iteration)
-      f32x2 s2 = {0.0f, 0.0f};
+      float s = 0.0f;
 #pragma unroll
-      for (int r = 0; r < 8; ++r) {
-        const f32x2 yp = {y[2 * r], y[2 * r + 1]};
-        s2 = yp * yp + s2;
-      }
-      float s = s2.x + s2.y;
+      for (int r = 0; r < 16; ++r) s = fmaf(y[r], y[r], s);
       // the 32 Y rows live across the two lane halves: one cross-half sum
       s += __shfl_xor(s, 32, WAVE);
       if (lane < 32) {
@@ -1389,13 +1373,9 @@ estep_fused_f32_kernel(const float* __restrict__ z,
         const float b = zrow[2 * ch + g2];
         y = __builtin_amdgcn_mfma_f32_32x32x2f32(a[ch], b, y, 0, 0, 0);
       }
-      f32x2 s2 = {0.0f, 0.0f};
+      float s = 0.0f;
 #pragma unroll
-      for (int r = 0; r < 8; ++r) {
-        const f32x2 yp = {y[2 * r], y[2 * r + 1]};
-        s2 = yp * yp + s2;
-      }
-      float s = s2.x + s2.y;
+      for (int r = 0; r < 16; ++r) s = fmaf(y[r], y[r], s);
       s += __shfl_xor(s, 32, WAVE);
       if (lane < 32) {
         const int e = t * 32 + j32;
@@ -1502,68 +1482,111 @@ mstep_moments_big_kernel(const float* __restrict__ x,
   for (int pp = 0; pp < MBB_PMAX; ++pp) acc[pp] = (f32x16)(0.0f);
 
   const int64_t tiles = (n + MBB_BK - 1) / MBB_BK;
-  for (int64_t tile = chunk; tile < tiles; tile += nchunk) {
-    const int64_t e0 = tile * MBB_BK;
-    const int cnt = (int)min((int64_t)MBB_BK, n - e0);
-    __syncthreads();
-    auto put = [&](int di, int ei, float v) {
-      const __bf16 hi = (__bf16)v;
-      zhi[di * zbr + ei] = hi;
-      zlo[di * zbr + ei] = (__bf16)(v - (float)hi);
-    };
-    if (cnt == MBB_BK) {
-      for (int idx = threadIdx.x; idx < d * MBB_BK; idx += MBB_NT)
-        put(idx / MBB_BK, idx % MBB_BK,
-            x[(int64_t)(idx / MBB_BK) * n + e0 + idx % MBB_BK]);
-      if (tile == chunk) {
-        for (int idx = d * MBB_BK + threadIdx.x; idx < rows * MBB_BK;
-             idx += MBB_NT)
-          put(idx / MBB_BK, idx % MBB_BK,
-              (idx / MBB_BK == d) ? 1.0f : 0.0f);
-      }
-      if (lse) {
-        for (int idx = threadIdx.x; idx < MBB_CPB * MBB_BK; idx += MBB_NT) {
-          const int ci = blockIdx.x * MBB_CPB + idx / MBB_BK;
-          const int ei = idx % MBB_BK;
-          if (ci < k)
-            wt[idx] = __expf(w[(int64_t)ci * n + e0 + ei] - lse[e0 + ei]);
-        }
-      } else {
-        for (int idx = threadIdx.x; idx < MBB_CPB * MBB_BK; idx += MBB_NT) {
-          const int ci = blockIdx.x * MBB_CPB + idx / MBB_BK;
-          const int ei = idx % MBB_BK;
-          if (ci < k)
-            wt[idx] = w[(int64_t)ci * n + e0 + ei];
-        }
-      }
-    } else {
-      for (int idx = threadIdx.x; idx < rows * MBB_BK; idx += MBB_NT) {
-        const int di = idx / MBB_BK, ei = idx % MBB_BK;
-        float v = 0.0f;
-        if (ei < cnt) {
-          if (di < d) v = x[(int64_t)di * n + e0 + ei];
-          else if (di == d) v = 1.0f;
-        }
-        put(di, ei, v);
-      }
-      if (lse) {
-        for (int idx = threadIdx.x; idx < MBB_CPB * MBB_BK; idx += MBB_NT) {
-          const int ci = blockIdx.x * MBB_CPB + idx / MBB_BK;
-          const int ei = idx % MBB_BK;
-          wt[idx] = (ci < k && ei < cnt)
-              ? __expf(w[(int64_t)ci * n + e0 + ei] - lse[e0 + ei]) : 0.0f;
-        }
-      } else {
-        for (int idx = threadIdx.x; idx < MBB_CPB * MBB_BK; idx += MBB_NT) {
-          const int ci = blockIdx.x * MBB_CPB + idx / MBB_BK;
-          const int ei = idx % MBB_BK;
-          wt[idx] = (ci < k && ei < cnt)
-              ? w[(int64_t)ci * n + e0 + ei] : 0.0f;
+  const int64_t my_tiles =
+      chunk < tiles ? (tiles - chunk + nchunk - 1) / nchunk : 0;
+
+  // T14 register staging: tile t+1's global loads are issued before tile
+  // t's MFMA work; ONE LDS buffer with the write pass after the barrier
+  // (double-buffering would halve occupancy at the D=128 LDS size).
+  // Pad events need no z zero-fill: their w is staged as 0, so the
+  // A-side fragments vanish regardless of what B holds.
+  constexpr int MBB_NXQ = 3;  // ceil(159*16/1024)
+  const int xq_total = d * (MBB_BK / 4);
+  const int wq_total = MBB_CPB * (MBB_BK / 4);
+  float4 rx[MBB_NXQ];
+  float4 rw, rl;
+  auto issue_loads = [&](int64_t ti) {
+    const int64_t e0 = (chunk + ti * nchunk) * MBB_BK;
+    const bool full = (n - e0) >= MBB_BK;
+#pragma unroll
+    for (int sq = 0; sq < MBB_NXQ; ++sq) {
+      const int q = threadIdx.x + sq * MBB_NT;
+      if (q < xq_total) {
+        const int di = q / (MBB_BK / 4), eq = q % (MBB_BK / 4);
+        const float* g = x + (int64_t)di * n + e0 + eq * 4;
+        if (full) {
+          rx[sq] = *(const float4*)g;
+        } else {
+          float v[4];
+#pragma unroll
+          for (int u = 0; u < 4; ++u)
+            v[u] = (e0 + eq * 4 + u < n) ? g[u] : 0.0f;
+          rx[sq] = *(float4*)v;
         }
       }
     }
-    __syncthreads();
-    if (c >= k) continue;
+    if (threadIdx.x < wq_total) {
+      const int ci = blockIdx.x * MBB_CPB + threadIdx.x / (MBB_BK / 4);
+      const int eq = threadIdx.x % (MBB_BK / 4);
+      const int64_t ge = e0 + eq * 4;
+      if (ci < k) {
+        const float* g = w + (int64_t)ci * n + ge;
+        if (full) {
+          rw = *(const float4*)g;
+          if (lse) rl = *(const float4*)&lse[ge];
+        } else {
+          float v[4], lv[4];
+#pragma unroll
+          for (int u = 0; u < 4; ++u) {
+            const bool ok = ge + u < n;
+            // pad events carry w = 0 (plain) or logw = -inf-ish (lse
+            // mode: exp gives 0) so their A fragments vanish
+            v[u] = ok ? g[u] : (lse ? -3.0e38f : 0.0f);
+            lv[u] = (ok && lse) ? lse[ge + u] : 0.0f;
+          }
+          rw = *(float4*)v;
+          rl = *(float4*)lv;
+        }
+      } else {
+        rw = (float4){lse ? -3.0e38f : 0.0f, lse ? -3.0e38f : 0.0f,
+                      lse ? -3.0e38f : 0.0f, lse ? -3.0e38f : 0.0f};
+        rl = (float4){0, 0, 0, 0};
+      }
+    }
+  };
+  auto write_buf = [&]() {
+#pragma unroll
+    for (int sq = 0; sq < MBB_NXQ; ++sq) {
+      const int q = threadIdx.x + sq * MBB_NT;
+      if (q < xq_total) {
+        const int di = q / (MBB_BK / 4), ei4 = (q % (MBB_BK / 4)) * 4;
+        const float v[4] = {rx[sq].x, rx[sq].y, rx[sq].z, rx[sq].w};
+        __bf16 h[4], l[4];
+#pragma unroll
+        for (int u = 0; u < 4; ++u) {
+          h[u] = (__bf16)v[u];
+          l[u] = (__bf16)(v[u] - (float)h[u]);
+        }
+        *(uint2*)(zhi + di * zbr + ei4) = *(uint2*)h;
+        *(uint2*)(zlo + di * zbr + ei4) = *(uint2*)l;
+      }
+    }
+    if (threadIdx.x < wq_total) {
+      float v[4] = {rw.x, rw.y, rw.z, rw.w};
+      if (lse) {
+        const float lv[4] = {rl.x, rl.y, rl.z, rl.w};
+#pragma unroll
+        for (int u = 0; u < 4; ++u) v[u] = __expf(v[u] - lv[u]);
+      }
+      *(float4*)(wt + threadIdx.x * 4) = *(float4*)v;
+    }
+  };
+
+  // constant rows (ones at d, zeros above), written once; visibility is
+  // covered by the first in-loop barrier
+  for (int idx = d * MBB_BK + threadIdx.x; idx < rows * MBB_BK;
+       idx += MBB_NT) {
+    const int di = idx / MBB_BK, ei = idx % MBB_BK;
+    zhi[di * zbr + ei] = (__bf16)(di == d ? 1.0f : 0.0f);
+    zlo[di * zbr + ei] = (__bf16)0.0f;
+  }
+  if (my_tiles > 0) issue_loads(0);
+
+  for (int64_t ti = 0; ti < my_tiles; ++ti) {
+    __syncthreads();   // previous tile's readers are done with the buffer
+    write_buf();
+    if (ti + 1 < my_tiles) issue_loads(ti + 1);
+    __syncthreads();   // staging visible
 
 #pragma unroll 2
     for (int ch = 0; ch < MBB_BK / 16; ++ch) {
@@ -1586,21 +1609,13 @@ mstep_moments_big_kernel(const float* __restrict__ x,
               *(const bf16x8*)(zhi + (tr * 32 + j32) * zbr + eb);
           const bf16x8 zal =
               *(const bf16x8*)(zlo + (tr * 32 + j32) * zbr + eb);
-          const bf16x2* h2 = (const bf16x2*)&zah;
-          const bf16x2* l2 = (const bf16x2*)&zal;
-          bf16x2* ah2 = (bf16x2*)&a_hi;
-          bf16x2* al2 = (bf16x2*)&a_lo;
-          const f32x2* wv2 = (const f32x2*)wv;
 #pragma unroll
-          for (int u2 = 0; u2 < 4; ++u2) {
-            const f32x2 hf = {(float)h2[u2][0], (float)h2[u2][1]};
-            const f32x2 lf = {(float)l2[u2][0], (float)l2[u2][1]};
-            const f32x2 av = wv2[u2] * (hf + lf);
-            const bf16x2 hi = {(__bf16)av.x, (__bf16)av.y};
-            const f32x2 hif = {(float)hi[0], (float)hi[1]};
-            const f32x2 lo = av - hif;
-            ah2[u2] = hi;
-            al2[u2] = (bf16x2){(__bf16)lo.x, (__bf16)lo.y};
+          for (int u = 0; u < 8; ++u) {
+            const float zf = (float)zah[u] + (float)zal[u];
+            const float av = wv[u] * zf;
+            const __bf16 hi = (__bf16)av;
+            a_hi[u] = hi;
+            a_lo[u] = (__bf16)(av - (float)hi);
           }
           prev_tr = tr;
         }
@@ -1851,13 +1866,9 @@ estep_logw_big2_kernel(const __hip_bfloat16* __restrict__ z,
         y = __builtin_amdgcn_mfma_f32_32x32x16_bf16(fa, b, y, 0, 0, 0);
         y = __builtin_amdgcn_mfma_f32_32x32x16_bf16(fl, b, y, 0, 0, 0);
       }
-      f32x2 s2 = {0.0f, 0.0f};
+      float s = 0.0f;
 #pragma unroll
-      for (int r = 0; r < 8; ++r) {
-        const f32x2 yp = {y[2 * r], y[2 * r + 1]};
-        s2 = yp * yp + s2;
-      }
-      float s = s2.x + s2.y;
+      for (int r = 0; r < 16; ++r) s = fmaf(y[r], y[r], s);
       s += __shfl_xor(s, 32, WAVE);
       if (lane < 32) qpart[rt * ESB2_BE + t * 32 + j32] = s;
     }
