@@ -296,7 +296,7 @@ void mstep_moments_b16(torch::Tensor xhi, torch::Tensor xlo,
                   partials.size(2) == dp * (dp + 1) / 2,
               "partials must be [nchunk, K, Dp*(Dp+1)/2]");
   // two buffers x (zhi+zlo planes + w tiles for 8 clusters)
-  const size_t lds = 2 * (2 * 32 * 136 * 2 + (8 * 128 + 128) * 4);
+  const size_t lds = 2 * (2 * 32 * 136 * 2 + 8 * 128 * 4);
   dim3 grid((k + 7) / 8, nchunk);
   hipLaunchKernelGGL(gmm::mstep_moments_b16_kernel, grid, dim3(512), lds,
                      stream(),

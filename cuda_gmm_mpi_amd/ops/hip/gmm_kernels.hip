@@ -917,9 +917,8 @@ mstep_moments_b16_kernel(const __hip_bfloat16* __restrict__ xhi,
   constexpr int ZBR = MB_BK + 8;
   constexpr int PLANE = 32 * ZBR;          // bf16 elements per z plane
   constexpr int BUFB = 2 * PLANE;          // zhi+zlo per buffer (bf16)
-  constexpr int WROW = MB_CPB * MB_BK + MB_BK;  // w rows + one lse row
   __bf16* zbuf = (__bf16*)lds;             // [2][2*PLANE]
-  float* wbuf = (float*)(zbuf + 2 * BUFB); // [2][WROW]
+  float* wbuf = (float*)(zbuf + 2 * BUFB); // [2][MB_CPB*MB_BK]
 
   const int wave = threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
@@ -943,7 +942,7 @@ mstep_moments_b16_kernel(const __hip_bfloat16* __restrict__ xhi,
 
   // ---- staging helpers -------------------------------------------------
   uint2 rxh[2], rxl[2];
-  float4 rw, rls;
+  float4 rw;
   auto issue_loads = [&](int64_t tile) {
     const int64_t e0 = tile * MB_BK;
     const bool full = (n - e0) >= MB_BK;
@@ -975,31 +974,27 @@ mstep_moments_b16_kernel(const __hip_bfloat16* __restrict__ xhi,
       const int cw = blockIdx.x * MB_CPB + wv;
       if (cw < k) {
         const int64_t g = (int64_t)cw * n + e0 + eq * 4;
-        if (full) {
+        const int64_t ge = e0 + eq * 4;
+        if (full && lse) {
+          rw = *(const float4*)&w[g];
+          const float4 lv = *(const float4*)&lse[ge];
+          rw.x = __expf(rw.x - lv.x);
+          rw.y = __expf(rw.y - lv.y);
+          rw.z = __expf(rw.z - lv.z);
+          rw.w = __expf(rw.w - lv.w);
+        } else if (full) {
           rw = *(const float4*)&w[g];
         } else {
           float v[4];
-          for (int u = 0; u < 4; ++u)
-            v[u] = (e0 + eq * 4 + u < n) ? w[g + u] : 0.0f;
+          for (int u = 0; u < 4; ++u) {
+            const bool ok = ge + u < n;
+            v[u] = ok ? w[g + u] : 0.0f;
+            if (lse) v[u] = ok ? __expf(v[u] - lse[ge + u]) : 0.0f;
+          }
           rw = *(float4*)v;
         }
       } else {
         rw = (float4){0, 0, 0, 0};
-      }
-    }
-    // lse staged ONCE per block (shared by the 8 clusters; per-cluster
-    // loads were 8x redundant L2 traffic, ~19 us/iter). w stays RAW
-    // logw in LDS; exp happens at A-build. Pad events get lse = +3e38
-    // so exp(logw - lse) -> 0.
-    if (lse && threadIdx.x < MB_BK / 4) {
-      const int64_t ge = e0 + threadIdx.x * 4;
-      if (full) {
-        rls = *(const float4*)&lse[ge];
-      } else {
-        float lv[4];
-        for (int u = 0; u < 4; ++u)
-          lv[u] = (ge + u < n) ? lse[ge + u] : 3.0e38f;
-        rls = *(float4*)lv;
       }
     }
   };
@@ -1016,9 +1011,7 @@ mstep_moments_b16_kernel(const __hip_bfloat16* __restrict__ xhi,
       }
     }
     if (threadIdx.x < MB_CPB * (MB_BK / 4))
-      *(float4*)(wbuf + buf * WROW + threadIdx.x * 4) = rw;
-    if (lse && threadIdx.x < MB_BK / 4)
-      *(float4*)(wbuf + buf * WROW + MB_CPB * MB_BK + threadIdx.x * 4) = rls;
+      *(float4*)(wbuf + buf * MB_CPB * MB_BK + threadIdx.x * 4) = rw;
   };
   // constant rows (ones at d, zeros above) in BOTH buffers, written once;
   // a tail tile is always the globally last so no re-fix is needed
@@ -1043,8 +1036,7 @@ mstep_moments_b16_kernel(const __hip_bfloat16* __restrict__ xhi,
     if (ti + 1 < my_tiles) issue_loads(chunk + (ti + 1) * nchunk);
     const __bf16* zh = zbuf + cur * BUFB;
     const __bf16* zl = zh + PLANE;
-    const float* wt = wbuf + cur * WROW;
-    const float* lrow = wt + MB_CPB * MB_BK;
+    const float* wt = wbuf + cur * MB_CPB * MB_BK;
 
 #pragma unroll 2
     for (int ch = 0; ch < MB_BK / 16; ++ch) {
@@ -1053,16 +1045,8 @@ mstep_moments_b16_kernel(const __hip_bfloat16* __restrict__ xhi,
       const bf16x8 b_lo = *(const bf16x8*)(zl + j32 * ZBR + eb);
       const float4 wv0 = *(const float4*)(wt + wave * MB_BK + eb);
       const float4 wv1 = *(const float4*)(wt + wave * MB_BK + eb + 4);
-      float wv[8] = {wv0.x, wv0.y, wv0.z, wv0.w,
-                     wv1.x, wv1.y, wv1.z, wv1.w};
-      if (lse) {  // w holds RAW logw; normalize with the block-shared lse
-        const float4 l0 = *(const float4*)(lrow + eb);
-        const float4 l1 = *(const float4*)(lrow + eb + 4);
-        const float lv[8] = {l0.x, l0.y, l0.z, l0.w,
-                             l1.x, l1.y, l1.z, l1.w};
-#pragma unroll
-        for (int u = 0; u < 8; ++u) wv[u] = __expf(wv[u] - lv[u]);
-      }
+      const float wv[8] = {wv0.x, wv0.y, wv0.z, wv0.w,
+                           wv1.x, wv1.y, wv1.z, wv1.w};
       bf16x8 a_hi, a_lo;
 #pragma unroll
       for (int u = 0; u < 8; ++u) {
