@@ -87,6 +87,12 @@ class EmEngine:
         self.use_fused_estep = F.estep_fused_available(
             self.device, config.estep_dtype, self.d, k0,
         )
+        # v1 lw-in-LDS variant writes posteriors directly — preferred
+        # when K fits its LDS budget (the lse-aware M-step's staging exp
+        # is on its VALU-bound critical path; measured ~19 us/iter)
+        self.use_lds_estep = F.estep_fused_lds_available(
+            self.device, config.estep_dtype, self.d, k0,
+        )
         self.use_big_estep = F.estep_big_available(
             self.device, config.estep_dtype, self.d,
         ) and not self.use_fused_estep
@@ -189,7 +195,18 @@ class EmEngine:
         """E-step into self.w[:k]; returns shard-partial likelihood tensor."""
         st = self.state.shrink(k)
         with self.profile.time("e_step"):
-            if self.use_fused_estep:
+            self._w_is_logw = False
+            if self.use_fused_estep and self.use_lds_estep:
+                add = self._add[:k]
+                if self.mfac32 is not None:
+                    w, lik = F.estep_fused_f32_lds(self.x_estep,
+                                                   self.mfac32[:k], add,
+                                                   self.w[:k])
+                else:
+                    w, lik = F.estep_fused_lds(self.x_estep, self.mfac[:k],
+                                               add, self.w[:k])
+            elif self.use_fused_estep:
+                self._w_is_logw = True
                 add = self._add[:k]
                 if self.mfac32 is not None:
                     w, lik = F.estep_fused_f32(self.x_estep, self.mfac32[:k],
@@ -203,16 +220,19 @@ class EmEngine:
                     logw = F.estep_logw_big_f32(self.x_estep,
                                                 self.mfac32[:k], add,
                                                 self.w[:k])
+                    self._w_is_logw = True
                 else:
                     logw = F.estep_logw_big(self.x_estep, self.mfac[:k],
                                             add, self.w[:k])
                 lik = F.estep_lse(logw, self._lse)
+                self._w_is_logw = True
             elif self.device.type == "cuda":
                 logw = F.estep_logw(
                     self.x_estep, st.means, st.Rinv, st.constant, st.pi,
                     self.cfg.diag_only, out=self.w[:k],
                 )
                 lik = F.estep_lse(logw, self._lse)
+                self._w_is_logw = True
             else:
                 logw = F.estep_logw(
                     self.x_estep, st.means, st.Rinv, st.constant, st.pi,
@@ -262,9 +282,10 @@ class EmEngine:
         """
         st = self.state.shrink(k)
         with self.profile.time("m_step"):
-            packed = F.mstep_moments(self.x, self.w[:k],
-                                     precision=self.cfg.mstep_precision,
-                                     x_split=self.x_split, lse=self._lse)
+            packed = F.mstep_moments(
+                self.x, self.w[:k], precision=self.cfg.mstep_precision,
+                x_split=self.x_split,
+                lse=self._lse if getattr(self, "_w_is_logw", False) else None)
         with self.profile.time("comm"):
             pdist.all_reduce_(packed)
         with self.profile.time("m_step"):
@@ -576,8 +597,8 @@ class EmEngine:
 
     def posteriors(self, k: int) -> torch.Tensor:
         """Normalized posteriors [k, n_shard] for the current E-step state
-        (on CUDA the w buffer holds log weights; normalize with the lse)."""
-        if self._lse is None:
+        (paths that emit log weights normalize with the lse)."""
+        if self._lse is None or not getattr(self, "_w_is_logw", False):
             return self.w[:k]
         return torch.exp(self.w[:k] - self._lse.unsqueeze(0))
 

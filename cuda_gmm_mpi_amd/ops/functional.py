@@ -209,6 +209,44 @@ def estep_fused_available(device: torch.device, dtype: str, d: int,
     return device.type == "cuda" and d <= 31
 
 
+def estep_fused_lds_available(device: torch.device, dtype: str, d: int,
+                              k: int) -> bool:
+    """v1 lw-in-LDS fast path: writes posteriors directly (the M-step
+    then skips the exp at staging, which sits on its VALU-bound critical
+    path — measured ~19 us/iter at K=64). LDS-bounded in K."""
+    if device.type != "cuda" or d > 31:
+        return False
+    if dtype == "bf16":
+        return 128 * 40 * 2 + 4 * k * 132 <= 64 * 1024   # K <= 104
+    return 4 * (128 * 33 + k * 132) <= 64 * 1024         # K <= 85
+
+
+def estep_fused_lds(z: torch.Tensor, mfac: torch.Tensor, add: torch.Tensor,
+                    w_out: torch.Tensor
+                    ) -> tuple[torch.Tensor, torch.Tensor]:
+    """v1 fused bf16 E-step: POSTERIORS into w_out, likelihood scalar."""
+    n = z.shape[1]
+    nblk = (n + 127) // 128
+    partial = torch.empty(nblk, dtype=torch.float32, device=z.device)
+    hip_ext().estep_fused_lds(z, mfac, add, w_out, partial)
+    lik = torch.empty(1, dtype=torch.float32, device=z.device)
+    hip_ext().reduce_scalar(partial, lik)
+    return w_out, lik
+
+
+def estep_fused_f32_lds(z: torch.Tensor, mfac32: torch.Tensor,
+                        add: torch.Tensor, w_out: torch.Tensor
+                        ) -> tuple[torch.Tensor, torch.Tensor]:
+    """v1 exact-f32 fused E-step: POSTERIORS into w_out."""
+    n = z.shape[1]
+    nblk = (n + 127) // 128
+    partial = torch.empty(nblk, dtype=torch.float32, device=z.device)
+    hip_ext().estep_fused_f32_lds(z, mfac32, add, w_out, partial)
+    lik = torch.empty(1, dtype=torch.float32, device=z.device)
+    hip_ext().reduce_scalar(partial, lik)
+    return w_out, lik
+
+
 def estep_fused_f32(z: torch.Tensor, mfac32: torch.Tensor, add: torch.Tensor,
                     w_out: torch.Tensor, lse: torch.Tensor
                     ) -> tuple[torch.Tensor, torch.Tensor]:

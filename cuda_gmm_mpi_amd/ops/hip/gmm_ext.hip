@@ -341,6 +341,64 @@ void estep_fused(torch::Tensor z, torch::Tensor mfac, torch::Tensor add,
   HIP_CHECK(hipGetLastError());
 }
 
+void estep_fused_lds(torch::Tensor z, torch::Tensor mfac, torch::Tensor add,
+                     torch::Tensor w_out, torch::Tensor partial) {
+  TORCH_CHECK(z.is_cuda() && z.is_contiguous() &&
+                  z.scalar_type() == torch::kBFloat16,
+              "z must be contiguous bf16 [D,N]");
+  TORCH_CHECK(mfac.is_contiguous() && mfac.scalar_type() == torch::kBFloat16,
+              "mfac must be bf16");
+  check_f32(add, "add");
+  check_f32(w_out, "w_out");
+  check_f32(partial, "partial");
+  const int d = (int)z.size(0);
+  const int64_t n = z.size(1);
+  const int k = (int)add.size(0);
+  TORCH_CHECK(d <= 31, "estep_fused_lds needs D <= 31");
+  TORCH_CHECK(mfac.numel() >= (int64_t)k * 2 * 32 * 32, "mfac too small");
+  TORCH_CHECK(w_out.size(0) == k && w_out.size(1) == n, "w_out shape");
+  const int64_t nblk = (n + 128 - 1) / 128;
+  TORCH_CHECK(partial.size(0) >= nblk, "partial buffer too small");
+  const size_t zbytes = (size_t)128 * 40 * 2;  // transposed z tile
+  const size_t lds = zbytes + sizeof(float) * (size_t)k * (128 + 4);
+  TORCH_CHECK(lds <= 64 * 1024,
+              "estep_fused_lds LDS budget exceeded (K too big)");
+  hipLaunchKernelGGL(gmm::estep_fused_lds_kernel, dim3((uint32_t)nblk),
+                     dim3(kNT), lds, stream(),
+                     reinterpret_cast<const __hip_bfloat16*>(z.data_ptr()),
+                     reinterpret_cast<const __hip_bfloat16*>(mfac.data_ptr()),
+                     add.data_ptr<float>(), w_out.data_ptr<float>(),
+                     partial.data_ptr<float>(), d, k, n);
+  HIP_CHECK(hipGetLastError());
+}
+
+void estep_fused_f32_lds(torch::Tensor z, torch::Tensor mfac32,
+                         torch::Tensor add, torch::Tensor w_out,
+                         torch::Tensor partial) {
+  check_f32(z, "z");
+  check_f32(mfac32, "mfac32");
+  check_f32(add, "add");
+  check_f32(w_out, "w_out");
+  check_f32(partial, "partial");
+  const int d = (int)z.size(0);
+  const int64_t n = z.size(1);
+  const int k = (int)add.size(0);
+  TORCH_CHECK(d <= 31, "estep_fused_f32_lds needs D <= 31");
+  TORCH_CHECK(mfac32.numel() >= (int64_t)k * 32 * 32, "mfac32 too small");
+  TORCH_CHECK(w_out.size(0) == k && w_out.size(1) == n, "w_out shape");
+  const int64_t nblk = (n + 128 - 1) / 128;
+  TORCH_CHECK(partial.size(0) >= nblk, "partial buffer too small");
+  const size_t lds =
+      sizeof(float) * ((size_t)128 * 33 + (size_t)k * (128 + 4));
+  TORCH_CHECK(lds <= 64 * 1024, "estep_fused_f32_lds LDS budget exceeded");
+  hipLaunchKernelGGL(gmm::estep_fused_f32_lds_kernel, dim3((uint32_t)nblk),
+                     dim3(kNT), lds, stream(), z.data_ptr<float>(),
+                     mfac32.data_ptr<float>(), add.data_ptr<float>(),
+                     w_out.data_ptr<float>(), partial.data_ptr<float>(), d,
+                     k, n);
+  HIP_CHECK(hipGetLastError());
+}
+
 void estep_logw_big(torch::Tensor z, torch::Tensor mfac, torch::Tensor add,
                     torch::Tensor logw) {
   TORCH_CHECK(z.is_cuda() && z.is_contiguous() &&
@@ -616,6 +674,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "re-emit E-step factors from the covariance R (post-merge/resume)");
   m.def("estep_fused_f32", &estep_fused_f32,
         "exact-f32 MFMA fused E-step (D <= 31)");
+  m.def("estep_fused_lds", &estep_fused_lds,
+        "small-K fused E-step (v1, lw in LDS, posteriors written)");
+  m.def("estep_fused_f32_lds", &estep_fused_f32_lds,
+        "small-K exact-f32 fused E-step (v1)");
   m.def("estep_fused", &estep_fused,
         "fused bf16-MFMA E-step: posteriors + likelihood partials");
   m.def("mfma_probe", &mfma_probe, "bf16 MFMA fragment-layout probe");

@@ -1440,6 +1440,286 @@ estep_fused_f32_kernel(const float* __restrict__ z,
 }
 
 // ---------------------------------------------------------------------------
+// Small-K fused E-step (v1, lw-in-LDS): 128-event blocks, log-weights
+// kept in LDS, posteriors + likelihood fused in-block — logw never
+// touches HBM and the M-step reads plain posteriors (no exp at staging).
+// LDS-bounded: K <= ~104 bf16 / ~85 f32. Same-box A/B showed this beats
+// the online-softmax variant by ~19 us/iter at the flagship K=64 (the
+// M-step's staging exp is on its VALU-bound critical path), so it is the
+// preferred path when K fits; the any-K online-softmax kernels cover the
+// rest.
+// ---------------------------------------------------------------------------
+#define ESTL_ZROW 40
+#define ESTL_BE 128  // v1 block size (lw-in-LDS variant)  // bf16 per transposed-z row (32 k-slots + pad)
+
+__global__ void __launch_bounds__(NT)
+estep_fused_lds_kernel(const __hip_bfloat16* __restrict__ z,
+                   const __hip_bfloat16* __restrict__ mfac,  // [K][2][32][32]
+                   const float* __restrict__ add,            // const + ln pi
+                   float* __restrict__ w_out, float* __restrict__ partial,
+                   int d, int k, int64_t n) {
+  // LDS: zs_t [ESTL_BE][ESTL_ZROW] bf16 — z staged TRANSPOSED (k-major per
+  // event, ones-row and zero-pad baked in) so a B fragment is a single
+  // 16-byte ds_read_b128; then lw [k][ESTL_BE+4] f32.
+  extern __shared__ float lds[];
+  const int lrow = ESTL_BE + 4;
+  __hip_bfloat16* zs = (__hip_bfloat16*)lds;
+  float* lw = lds + (ESTL_BE * ESTL_ZROW) / 2;
+
+  const int wave = threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int64_t e0 = (int64_t)blockIdx.x * ESTL_BE;
+  const int cnt = (int)min((int64_t)ESTL_BE, n - e0);
+
+  __bf16* zsb = (__bf16*)zs;
+  if (cnt == ESTL_BE) {
+    // branchless staging (guide §5 trap 4c): coalesced reads of the d data
+    // rows, transposed scatter into LDS; then the constant rows
+    for (int idx = threadIdx.x; idx < d * ESTL_BE; idx += blockDim.x) {
+      const int kk = idx / ESTL_BE, ei = idx % ESTL_BE;
+      zsb[ei * ESTL_ZROW + kk] =
+          (__bf16)__bfloat162float(z[(int64_t)kk * n + e0 + ei]);
+    }
+  } else {
+    for (int idx = threadIdx.x; idx < d * ESTL_BE; idx += blockDim.x) {
+      const int kk = idx / ESTL_BE, ei = idx % ESTL_BE;
+      zsb[ei * ESTL_ZROW + kk] = (__bf16)(
+          (ei < cnt) ? __bfloat162float(z[(int64_t)kk * n + e0 + ei]) : 0.0f);
+    }
+  }
+  for (int idx = threadIdx.x; idx < (32 - d) * ESTL_BE; idx += blockDim.x) {
+    const int kk = d + idx / ESTL_BE, ei = idx % ESTL_BE;
+    zsb[ei * ESTL_ZROW + kk] =
+        (__bf16)((kk == d && ei < cnt) ? 1.0f : 0.0f);
+  }
+  __syncthreads();
+
+  // 32x32x16 bf16 MFMA: one 32-row tile covers all of M (Dp <= 32) and 32
+  // events; WAVES SPLIT THE CLUSTER LOOP (c = wave, wave+4, ...) so each
+  // cluster's factor fragments are fetched once per block, not once per
+  // wave — the A-fragment L2 traffic was the previous bottleneck.
+  // A lane l -> A[i=l&31][kk=8*(l>>5)+u] per 16-deep chunk; B lane l ->
+  // B[kk][j=l&31]; C/D col=l&31, row=(reg&3)+8*(reg>>2)+4*(l>>5) (guide §3).
+  const int j32 = lane & 31;
+  const int g2 = lane >> 5;
+  typedef __attribute__((ext_vector_type(16))) float f32x16;
+  const bf16x8* mf = (const bf16x8*)mfac;  // rows of 32 bf16 = 4 frags each
+  const int fq0 = g2;      // chunk 0 covers k [0,16): slots {0,1}
+  const int fq1 = 2 + g2;  // chunk 1 covers k [16,32): slots {2,3}
+  const int nwaves = NT / WAVE;
+
+  bf16x8 nx_h0, nx_l0, nx_h1, nx_l1;
+  float nx_add;
+  auto load_a = [&](int c) {
+    const int64_t base = ((int64_t)c * 2) * 32 * 4;  // in bf16x8 units
+    nx_h0 = mf[base + j32 * 4 + fq0];
+    nx_l0 = mf[base + 32 * 4 + j32 * 4 + fq0];
+    nx_h1 = mf[base + j32 * 4 + fq1];
+    nx_l1 = mf[base + 32 * 4 + j32 * 4 + fq1];
+    nx_add = add[c];
+  };
+  if (wave < k) load_a(wave);
+
+  for (int c = wave; c < k; c += nwaves) {
+    const bf16x8 a_h0 = nx_h0, a_l0 = nx_l0, a_h1 = nx_h1, a_l1 = nx_l1;
+    const float addc = nx_add;
+    if (c + nwaves < k) load_a(c + nwaves);
+#pragma unroll
+    for (int t = 0; t < ESTL_BE / 32; ++t) {
+      // B fragments: contiguous 16 B of the transposed z row
+      const bf16x8 b0 =
+          *(const bf16x8*)(zs + (t * 32 + j32) * ESTL_ZROW + 8 * g2);
+      const bf16x8 b1 =
+          *(const bf16x8*)(zs + (t * 32 + j32) * ESTL_ZROW + 16 + 8 * g2);
+      // one accumulator chain: Y = (Mhi+Mlo)(chunk0+chunk1) summed by the
+      // MFMAs themselves (the VALU epilogue was the measured bottleneck)
+      f32x16 y = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+          a_h0, b0, (f32x16)(0.0f), 0, 0, 0);
+      y = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a_h1, b1, y, 0, 0, 0);
+      y = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a_l0, b0, y, 0, 0, 0);
+      y = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a_l1, b1, y, 0, 0, 0);
+      float s = 0.0f;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) s = fmaf(y[r], y[r], s);
+      // the 32 Y rows live across the two lane halves: one cross-half sum
+      s += __shfl_xor(s, 32, WAVE);
+      if (lane < 32) lw[c * lrow + t * 32 + j32] = -0.5f * s + addc;
+    }
+  }
+  __syncthreads();
+
+  // pass 2: posteriors + likelihood, cluster loop split across BOTH
+  // thread halves (the serial 64-deep store loop was 24% of the kernel —
+  // store-issue-bound): threads t and t+ESTL_BE each handle half the
+  // clusters of event t, combining max/sum through LDS.
+  __shared__ float pmax[2][ESTL_BE];
+  __shared__ float psum[2][ESTL_BE];
+  float acc = 0.0f;
+  {
+    const int t = threadIdx.x & (ESTL_BE - 1);
+    const int half = threadIdx.x >> 7;       // ESTL_BE == 128
+    const int mid = (k + 1) / 2;   // half 0 never empty (K=1, odd K)
+    const int c_lo = half * mid;
+    const int c_hi = half ? k : mid;
+    if (t < cnt && c_lo < c_hi) {
+      float m = lw[c_lo * lrow + t];
+#pragma unroll 4
+      for (int c = c_lo + 1; c < c_hi; ++c)
+        m = fmaxf(m, lw[c * lrow + t]);
+      pmax[half][t] = m;
+    } else if (t < ESTL_BE) {
+      pmax[half][t] = -3.0e38f;
+    }
+    __syncthreads();
+    const float m = fmaxf(pmax[0][t], pmax[1][t]);
+    float s = 0.0f;
+    if (t < cnt) {
+#pragma unroll 4
+      for (int c = c_lo; c < c_hi; ++c) {
+        const float e = __expf(lw[c * lrow + t] - m);
+        lw[c * lrow + t] = e;
+        s += e;
+      }
+    }
+    psum[half][t] = s;
+    __syncthreads();
+    const float total = psum[0][t] + psum[1][t];
+    if (t < cnt && c_lo < c_hi) {
+      const float inv = 1.0f / total;
+#pragma unroll 4
+      for (int c = c_lo; c < c_hi; ++c)
+        w_out[(int64_t)c * n + e0 + t] = lw[c * lrow + t] * inv;
+      if (half == 0) acc = m + __logf(total);
+    }
+  }
+  __shared__ float wsum[NT / WAVE];
+  for (int off = WAVE / 2; off > 0; off >>= 1)
+    acc += __shfl_down(acc, off, WAVE);
+  if (lane == 0) wsum[wave] = acc;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float total = 0.0f;
+    for (int wv = 0; wv < NT / WAVE; ++wv) total += wsum[wv];
+    partial[blockIdx.x] = total;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Exact-fp32 fused E-step (D <= 31): same structure as estep_fused_lds_kernel
+// but on v_mfma_f32_32x32x2_f32 — f32 in / f32 accumulate, bitwise an fmaf
+// chain (guide §3), consuming the fp32 factor plane. 16 dependent MFMAs
+// per (cluster, 32-event tile); issue interval == dependent latency (64),
+// so the chain runs at the f32 matrix rate. This makes the CLI's default
+// exact mode ~4x faster than the VALU quadratic-form path.
+// ---------------------------------------------------------------------------
+__global__ void __launch_bounds__(NT)
+estep_fused_f32_lds_kernel(const float* __restrict__ z,
+                       const float* __restrict__ mfac32,  // [K][32][32]
+                       const float* __restrict__ add,
+                       float* __restrict__ w_out, float* __restrict__ partial,
+                       int d, int k, int64_t n) {
+  typedef __attribute__((ext_vector_type(16))) float f32x16;
+  constexpr int ZR = 33;  // f32 slots per transposed event row (32 + pad)
+  extern __shared__ float lds[];
+  float* zs = lds;                         // [ESTL_BE][ZR]
+  const int lrow = ESTL_BE + 4;
+  float* lw = lds + ESTL_BE * ZR;           // [k][lrow]
+
+  const int wave = threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int j32 = lane & 31;
+  const int g2 = lane >> 5;
+  const int64_t e0 = (int64_t)blockIdx.x * ESTL_BE;
+  const int cnt = (int)min((int64_t)ESTL_BE, n - e0);
+
+  if (cnt == ESTL_BE) {
+    for (int idx = threadIdx.x; idx < d * ESTL_BE; idx += blockDim.x) {
+      const int kk = idx / ESTL_BE, ei = idx % ESTL_BE;
+      zs[ei * ZR + kk] = z[(int64_t)kk * n + e0 + ei];
+    }
+  } else {
+    for (int idx = threadIdx.x; idx < d * ESTL_BE; idx += blockDim.x) {
+      const int kk = idx / ESTL_BE, ei = idx % ESTL_BE;
+      zs[ei * ZR + kk] =
+          (ei < cnt) ? z[(int64_t)kk * n + e0 + ei] : 0.0f;
+    }
+  }
+  for (int idx = threadIdx.x; idx < (32 - d) * ESTL_BE; idx += blockDim.x) {
+    const int kk = d + idx / ESTL_BE, ei = idx % ESTL_BE;
+    zs[ei * ZR + kk] = (kk == d && ei < cnt) ? 1.0f : 0.0f;
+  }
+  __syncthreads();
+
+  const int nwaves = NT / WAVE;
+
+  // prefetched A rows: 16 f32 per lane (row j32, k-slots 2*ch + g2)
+  float nx_a[16];
+  float nx_add;
+  auto load_a = [&](int c) {
+    const float* row = mfac32 + ((int64_t)c * 32 + j32) * 32;
+#pragma unroll
+    for (int ch = 0; ch < 16; ++ch) nx_a[ch] = row[2 * ch + g2];
+    nx_add = add[c];
+  };
+  if (wave < k) load_a(wave);
+
+  for (int c = wave; c < k; c += nwaves) {
+    float a[16];
+#pragma unroll
+    for (int ch = 0; ch < 16; ++ch) a[ch] = nx_a[ch];
+    const float addc = nx_add;
+    if (c + nwaves < k) load_a(c + nwaves);
+#pragma unroll 2
+    for (int t = 0; t < ESTL_BE / 32; ++t) {
+      const float* zrow = zs + (t * 32 + j32) * ZR;
+      f32x16 y = (f32x16)(0.0f);
+#pragma unroll
+      for (int ch = 0; ch < 16; ++ch) {
+        const float b = zrow[2 * ch + g2];
+        y = __builtin_amdgcn_mfma_f32_32x32x2f32(a[ch], b, y, 0, 0, 0);
+      }
+      float s = 0.0f;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) s = fmaf(y[r], y[r], s);
+      s += __shfl_xor(s, 32, WAVE);
+      if (lane < 32) lw[c * lrow + t * 32 + j32] = -0.5f * s + addc;
+    }
+  }
+  __syncthreads();
+
+  // pass 2: identical to the bf16 kernel
+  float acc = 0.0f;
+  if (threadIdx.x < ESTL_BE && threadIdx.x < cnt) {
+    const int t = threadIdx.x;
+    float m = lw[t];
+#pragma unroll 4
+    for (int c = 1; c < k; ++c) m = fmaxf(m, lw[c * lrow + t]);
+    float s = 0.0f;
+#pragma unroll 4
+    for (int c = 0; c < k; ++c) {
+      const float e = __expf(lw[c * lrow + t] - m);
+      lw[c * lrow + t] = e;
+      s += e;
+    }
+    const float inv = 1.0f / s;
+#pragma unroll 4
+    for (int c = 0; c < k; ++c)
+      w_out[(int64_t)c * n + e0 + t] = lw[c * lrow + t] * inv;
+    acc = m + __logf(s);
+  }
+  __shared__ float wsum2[NT / WAVE];
+  for (int off = WAVE / 2; off > 0; off >>= 1)
+    acc += __shfl_down(acc, off, WAVE);
+  if (lane == 0) wsum2[wave] = acc;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float total = 0.0f;
+    for (int wv = 0; wv < NT / WAVE; ++wv) total += wsum2[wv];
+    partial[blockIdx.x] = total;
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Big-D split-precision moments (31 < D <= 159): same packed output as the
 // small-D kernels. Row-tiles of 32 cover the padded (D+1) dims; the
 // RT2*(RT2+1)/2 tile-pairs of the symmetric output are split across two

@@ -612,3 +612,54 @@ def test_engine_diag_only_fused_matches_valu(device=None):
         lik_v = eng_v.run_em(4)
         rel = 1e-4 if ed == "fp32" else 2e-2
         assert lik_f == pytest.approx(lik_v, rel=rel), ed
+
+
+def test_estep_fused_lds_matches_cpu(device):
+    """v1 lw-in-LDS fused E-step (the small-K fast path: posteriors
+    written directly) vs the fp32 torch reference."""
+    from cuda_gmm_mpi_amd.ops import functional as F
+    rng = np.random.default_rng(311)
+    k, d, n = 8, 24, 20000 + 57
+    means, r, pi = random_model(rng, k, d, device)
+    mfac = torch.empty(k, 2, 32, 32, dtype=torch.bfloat16, device=device)
+    mfac32 = torch.empty(k, 32, 32, dtype=torch.float32, device=device)
+    rinv_g, const_g = F.constants(r, means, False, mfac, mfac32)
+    add = const_g + torch.log(pi)
+    x = rng.standard_normal((d, n)).astype(np.float32) * 2
+    rinv, const = cpu.compute_constants(r.cpu())
+    ref_logw = cpu.estep_logw(torch.from_numpy(x), means.cpu(), rinv, const,
+                              pi.cpu())
+    ref_w, ref_lik = cpu.estep_posteriors(ref_logw)
+    w_out = torch.empty(k, n, dtype=torch.float32, device=device)
+    w, lik = F.estep_fused_lds(
+        torch.from_numpy(x).to(device).to(torch.bfloat16), mfac, add, w_out)
+    np.testing.assert_allclose(w.cpu().numpy(), ref_w.numpy(),
+                               rtol=5e-2, atol=2e-2)
+    assert float(lik) == pytest.approx(float(ref_lik), rel=2e-3)
+    w32, lik32 = F.estep_fused_f32_lds(torch.from_numpy(x).to(device),
+                                       mfac32, add, w_out)
+    np.testing.assert_allclose(w32.cpu().numpy(), ref_w.numpy(),
+                               rtol=2e-3, atol=2e-4)
+    assert float(lik32) == pytest.approx(float(ref_lik), rel=1e-4)
+
+
+def test_engine_lds_vs_lse_estep_equivalent(device=None):
+    """The small-K lds fast path and the any-K online-softmax path give
+    the same trajectory (engine-level)."""
+    from cuda_gmm_mpi_amd.engine import build_engine
+    from cuda_gmm_mpi_amd.utils.config import GmmConfig
+    from cuda_gmm_mpi_amd.utils.synthetic import make_blobs
+    data, _ = make_blobs(30000, 24, 8, seed=211)
+    cfg = GmmConfig(num_clusters=8, target_num_clusters=8,
+                    min_iters=8, max_iters=8, estep_dtype="bf16",
+                    mstep_precision="bf16x3")
+    eng_a = build_engine(data, cfg, device="cuda")
+    assert eng_a.use_lds_estep
+    lik_a = eng_a.run_em(8)
+    eng_b = build_engine(data, cfg, device="cuda")
+    eng_b.use_lds_estep = False  # force the online-softmax variant
+    lik_b = eng_b.run_em(8)
+    assert lik_a == pytest.approx(lik_b, rel=1e-3)
+    np.testing.assert_allclose(eng_a.state.means.cpu().numpy(),
+                               eng_b.state.means.cpu().numpy(),
+                               rtol=1e-2, atol=1e-1)
