@@ -1727,9 +1727,11 @@ estep_fused_f32_lds_kernel(const float* __restrict__ z,
 // Grid (ceil(K/2), nchunk); BK = 64 events per tile.
 // ---------------------------------------------------------------------------
 #define MBB_BK 64
-#define MBB_NT 1024  // 16 waves: 4 clusters x 4 pair-quarters
-#define MBB_CPB 4
-#define MBB_PMAX 4  // tile-pairs per wave (RT2 <= 5 -> TP <= 15 -> 4/quarter)
+#define MBB_NT 1024  // 16 waves: 2 clusters x 8 pair-groups
+#define MBB_CPB 2
+#define MBB_PMAX 2  // tile-pairs per wave: acc[4] x f32x16 = 64 AGPRs
+                    // spilled 72 B/lane in the chunk loop (measured);
+                    // 2 pairs (32 acc regs) fits
 
 __global__ void __launch_bounds__(MBB_NT)
 mstep_moments_big_kernel(const float* __restrict__ x,
@@ -1752,11 +1754,11 @@ mstep_moments_big_kernel(const float* __restrict__ x,
   const int lane = threadIdx.x & (WAVE - 1);
   const int j32 = lane & 31;
   const int g2 = lane >> 5;
-  const int cw = wave >> 2;              // which of the block's 4 clusters
-  const int quarter = wave & 3;          // which quarter of the tile-pairs
+  const int cw = wave >> 3;              // which of the block's 2 clusters
+  const int group = wave & 7;            // which eighth of the tile-pairs
   const int c = blockIdx.x * MBB_CPB + cw;
   const int chunk = blockIdx.y;
-  const int p_lo = quarter * MBB_PMAX;
+  const int p_lo = group * MBB_PMAX;
   const int p_hi = min(tp, p_lo + MBB_PMAX);
   // hoisted pair->tile mapping (tri_row_col has a sqrtf: keep it out of
   // the chunk loop)
