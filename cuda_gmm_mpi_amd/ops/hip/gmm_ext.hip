@@ -425,6 +425,10 @@ void estep_logw_big(torch::Tensor z, torch::Tensor mfac, torch::Tensor add,
   const int nchunk =
       (int)std::min<int64_t>(tiles, std::max<int64_t>(1, 2048 / k));
   dim3 grid((uint32_t)nchunk, k);
+  // block size matched to the (row-tile x event-tile) pair count so no
+  // waves idle at small rt_n (D <= 79 has rt_n <= 3 -> <= 12 pairs)
+  const uint32_t nthreads =
+      std::min<uint32_t>(1024, std::max<uint32_t>(256, rt_n * 4 * 64));
   auto s = stream();
 #define LAUNCH_ELB(KCT)                                                     \
   do {                                                                      \
@@ -434,7 +438,7 @@ void estep_logw_big(torch::Tensor z, torch::Tensor mfac, torch::Tensor add,
           hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds));           \
     }                                                                       \
     hipLaunchKernelGGL((gmm::estep_logw_big2_kernel<KCT>), grid,            \
-                       dim3(1024), lds, s,                                  \
+                       dim3(nthreads), lds, s,                                  \
                        reinterpret_cast<const __hip_bfloat16*>(             \
                            z.data_ptr()),                                   \
                        reinterpret_cast<const __hip_bfloat16*>(             \

@@ -2087,7 +2087,8 @@ estep_logw_big2_kernel(const __hip_bfloat16* __restrict__ z,
     const uint4* src_h =
         (const uint4*)(mfac + (int64_t)c * 2 * rows * COLS);
     const uint4* src_l = src_h + rows * COLS / 8;
-    for (int q8 = threadIdx.x; q8 < rows * COLS / 8; q8 += ESB2_NT) {
+    for (int q8 = threadIdx.x; q8 < rows * COLS / 8;
+         q8 += (int)blockDim.x) {
       const int row = q8 / (COLS / 8), col8 = q8 % (COLS / 8);
       *(uint4*)(ah + row * AROW + col8 * 8) = src_h[q8];
       *(uint4*)(al + row * AROW + col8 * 8) = src_l[q8];
@@ -2106,7 +2107,8 @@ estep_logw_big2_kernel(const __hip_bfloat16* __restrict__ z,
     // row stride; 20% LDSBankConflict measured)
     {
       const int kq_total = ((d + 3) / 4) * ESB2_BE;
-      for (int idx = threadIdx.x; idx < kq_total; idx += ESB2_NT) {
+      for (int idx = threadIdx.x; idx < kq_total;
+           idx += (int)blockDim.x) {
         const int kk0 = (idx / ESB2_BE) * 4, ei = idx % ESB2_BE;
         __bf16 v[4];
         // kk0 is wave-uniform (ESB2_BE >= WAVE): full quads take the
@@ -2132,7 +2134,7 @@ estep_logw_big2_kernel(const __hip_bfloat16* __restrict__ z,
       }
       const int kpad0 = ((d + 3) / 4) * 4;
       for (int idx = threadIdx.x; idx < (COLS - kpad0) * ESB2_BE;
-           idx += ESB2_NT) {
+           idx += (int)blockDim.x) {
         const int kk = kpad0 + idx / ESB2_BE, ei = idx % ESB2_BE;
         // d is only >= kpad0 when d % 4 == 0: the ones row then lives here
         zs[ei * AROW + kk] =
@@ -2143,7 +2145,7 @@ estep_logw_big2_kernel(const __hip_bfloat16* __restrict__ z,
 
     // (row-tile, event-32-tile) pairs split across the 16 waves
     for (int pair = wave; pair < rt_n * (ESB2_BE / 32);
-         pair += ESB2_NT / WAVE) {
+         pair += (int)(blockDim.x / WAVE)) {
       const int rt = pair / (ESB2_BE / 32);
       const int t = pair % (ESB2_BE / 32);
       const __bf16* arow = ah + (rt * 32 + j32) * AROW;
