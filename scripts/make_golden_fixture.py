@@ -267,6 +267,69 @@ def write_results(path, data, w):
             f.write("\n")
 
 
+def constants_diag(c):
+    """constants_kernel DIAG_ONLY (gaussian_kernel.cu:215-223): invert the
+    diagonal only; det = product of diagonal entries, then log."""
+    d = c.means.shape[1]
+    for i in range(c.k):
+        dd = 1.0
+        for j in range(d):
+            dd *= float(c.R[i][j, j])
+        ln_det = float(np.log(dd))
+        c.Rinv[i] = np.diag(1.0 / np.diag(c.R[i])).astype(F32)
+        c.constant[i] = F32(-d * 0.5 * LOG2PI - 0.5 * ln_det)
+    total = float(c.N.sum())
+    for i in range(c.k):
+        c.pi[i] = 1e-10 if c.N[i] < 0.5 else c.N[i] / total
+
+
+def estep_diag(data_t, c):
+    """estep1 DIAG_ONLY (gaussian_kernel.cu:430-433): diagonal quadratic
+    form only."""
+    d, n = data_t.shape
+    logw = np.empty((c.k, n), np.float64)
+    for i in range(c.k):
+        dx = data_t - c.means[i][:, None].astype(np.float64)
+        rd = np.diag(c.Rinv[i]).astype(np.float64)
+        q = (dx * dx * rd[:, None]).sum(axis=0)
+        logw[i] = -0.5 * q + float(c.constant[i]) + np.log(float(c.pi[i]))
+    m = logw.max(axis=0)
+    denom = m + np.log(np.exp(logw - m).sum(axis=0))
+    w = np.exp(logw - denom)
+    return w, float(denom.sum())
+
+
+def mstep_diag(data_t, c, w):
+    """M-step with off-diagonal covariance zeroed (engine diag rule)."""
+    d, n = data_t.shape
+    for i in range(c.k):
+        n_i = float(w[i].sum())
+        c.N[i] = n_i
+        if n_i > 0.5:
+            c.means[i] = ((data_t * w[i]).sum(axis=1) / n_i).astype(F32)
+        else:
+            c.means[i] = 0.0
+        if n_i >= 1.0:
+            dx = data_t - c.means[i][:, None].astype(np.float64)
+            var = (w[i] * dx * dx).sum(axis=1)
+        else:
+            var = np.zeros(d)
+        var = var + float(c.avgvar[i])
+        if n_i > 0.5:
+            c.R[i] = np.diag(var / n_i).astype(F32)
+        else:
+            c.R[i] = np.eye(d, dtype=F32)
+    constants_diag(c)
+
+
+def run_em_diag(data_t, c, iters=100):
+    w, lik = estep_diag(data_t, c)
+    for _ in range(iters):
+        mstep_diag(data_t, c, w)
+        w, lik = estep_diag(data_t, c)
+    return w, lik
+
+
 def main():
     os.makedirs(OUT, exist_ok=True)
     rng = np.random.default_rng(424242)
@@ -309,6 +372,23 @@ def main():
     write_results(os.path.join(OUT, "golden_small.results"), data, w)
     print(f"fixture written: K={kk} rissanen={riss:.4f} "
           f"likelihood={w.shape} -> {OUT}")
+
+    # ---- DIAG_ONLY fixture: K=2, no merge, diagonal covariance path ----
+    rng = np.random.default_rng(777)
+    n2, d2 = 500, 3
+    a = rng.normal([100.0, 220.0, 160.0], [6.0, 9.0, 5.0], size=(260, d2))
+    b = rng.normal([320.0, 120.0, 300.0], [8.0, 5.0, 7.0], size=(240, d2))
+    data2 = np.vstack([a, b]).astype(F32)[rng.permutation(n2)]
+    with open(os.path.join(OUT, "golden_diag.bin"), "wb") as f:
+        f.write(struct.pack("<ii", n2, d2))
+        f.write(data2.tobytes())
+    dt2 = data2.T.astype(np.float64)
+    c2 = seed(data2, 2)
+    constants_diag(c2)
+    w2, lik2 = run_em_diag(dt2, c2, iters=100)
+    write_summary(os.path.join(OUT, "golden_diag.summary"), c2)
+    write_results(os.path.join(OUT, "golden_diag.results"), data2, w2)
+    print(f"diag fixture written: likelihood={lik2:.4f}")
 
 
 if __name__ == "__main__":

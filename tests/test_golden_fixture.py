@@ -100,3 +100,54 @@ def test_results_match_fixture(cli_output):
     np.testing.assert_allclose(gw, ww, atol=5e-3)
     # every event assigned to the same cluster
     assert (gw.argmax(axis=1) == ww.argmax(axis=1)).all()
+
+
+@pytest.fixture(scope="module")
+def cli_output_diag(tmp_path_factory):
+    out = str(tmp_path_factory.mktemp("golden_diag") / "o")
+    rc = main(["2", os.path.join(FIX, "golden_diag.bin"), out, "2",
+               "--device", "cpu", "--no-center", "--diag-only"])
+    assert rc == 0
+    return out
+
+
+def test_diag_fixture_summary(cli_output_diag):
+    got = parse_summary(open(cli_output_diag + ".summary").read())
+    want = parse_summary(open(os.path.join(FIX, "golden_diag.summary")).read())
+    assert len(got) == len(want) == 2
+    order = [int(np.argmin([np.linalg.norm(w["means"] - g["means"])
+                            for g in got])) for w in want]
+    assert sorted(order) == [0, 1]
+    for wcl, gi in zip(want, order):
+        gcl = got[gi]
+        assert gcl["pi"] == pytest.approx(wcl["pi"], abs=2e-3)
+        assert gcl["N"] == pytest.approx(wcl["N"], rel=2e-3)
+        np.testing.assert_allclose(gcl["means"], wcl["means"],
+                                   rtol=2e-3, atol=0.2)
+        # diagonal covariance: off-diagonals must be exactly zero in BOTH
+        for m in (gcl["R"], wcl["R"]):
+            off = m - np.diag(np.diag(m))
+            assert np.abs(off).max() == 0.0
+        np.testing.assert_allclose(np.diag(gcl["R"]), np.diag(wcl["R"]),
+                                   rtol=5e-2, atol=0.5)
+
+
+def test_diag_fixture_results(cli_output_diag):
+    got_lines = open(cli_output_diag + ".results").read().splitlines()
+    want_lines = open(os.path.join(FIX,
+                                   "golden_diag.results")).read().splitlines()
+    assert len(got_lines) == len(want_lines) == 500
+    for g, w in zip(got_lines, want_lines):
+        assert g.split("\t")[0] == w.split("\t")[0]
+    gw = np.array([[float(v) for v in ln.split("\t")[1].split(",")]
+                   for ln in got_lines])
+    ww = np.array([[float(v) for v in ln.split("\t")[1].split(",")]
+                   for ln in want_lines])
+    got_sum = parse_summary(open(cli_output_diag + ".summary").read())
+    want_sum = parse_summary(
+        open(os.path.join(FIX, "golden_diag.summary")).read())
+    order = [int(np.argmin([np.linalg.norm(w["means"] - g["means"])
+                            for g in got_sum])) for w in want_sum]
+    gw = gw[:, order]
+    np.testing.assert_allclose(gw, ww, atol=5e-3)
+    assert (gw.argmax(axis=1) == ww.argmax(axis=1)).all()
