@@ -2151,24 +2151,18 @@ estep_logw_big2_kernel(const __hip_bfloat16* __restrict__ z,
       const __bf16* arow = ah + (rt * 32 + j32) * AROW;
       const __bf16* lrow = al + (rt * 32 + j32) * AROW;
       const __bf16* zrow = zs + (t * 32 + j32) * AROW;
-      // two INDEPENDENT accumulator chains (hi and lo planes): the
-      // single-chain form was 2*KCT = 18 dependent MFMAs at D=128
-      f32x16 y0 = (f32x16)(0.0f);
-      f32x16 y1 = (f32x16)(0.0f);
+      f32x16 y = (f32x16)(0.0f);
 #pragma unroll
       for (int kc = 0; kc < KCT; ++kc) {
         const bf16x8 b = *(const bf16x8*)(zrow + kc * 16 + 8 * g2);
         const bf16x8 fa = *(const bf16x8*)(arow + kc * 16 + 8 * g2);
         const bf16x8 fl = *(const bf16x8*)(lrow + kc * 16 + 8 * g2);
-        y0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(fa, b, y0, 0, 0, 0);
-        y1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(fl, b, y1, 0, 0, 0);
+        y = __builtin_amdgcn_mfma_f32_32x32x16_bf16(fa, b, y, 0, 0, 0);
+        y = __builtin_amdgcn_mfma_f32_32x32x16_bf16(fl, b, y, 0, 0, 0);
       }
       float s = 0.0f;
 #pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        const float yr = y0[r] + y1[r];
-        s = fmaf(yr, yr, s);
-      }
+      for (int r = 0; r < 16; ++r) s = fmaf(y[r], y[r], s);
       s += __shfl_xor(s, 32, WAVE);
       if (lane < 32) qpart[rt * ESB2_BE + t * 32 + j32] = s;
     }
