@@ -191,8 +191,11 @@ class EmEngine:
 
     # ------------------------------------------------------------------ EM
 
-    def _estep(self, k: int) -> torch.Tensor:
-        """E-step into self.w[:k]; returns shard-partial likelihood tensor."""
+    def _estep(self, k: int, need_lik: bool = True) -> torch.Tensor | None:
+        """E-step into self.w[:k]; returns the shard-partial likelihood
+        tensor, or None with ``need_lik=False`` (the dead-epsilon loop
+        only reads the likelihood after its FINAL iteration — the
+        reference computes-and-discards it every iteration)."""
         st = self.state.shrink(k)
         with self.profile.time("e_step"):
             self._w_is_logw = False
@@ -201,19 +204,20 @@ class EmEngine:
                 if self.mfac32 is not None:
                     w, lik = F.estep_fused_f32_lds(self.x_estep,
                                                    self.mfac32[:k], add,
-                                                   self.w[:k])
+                                                   self.w[:k], need_lik)
                 else:
                     w, lik = F.estep_fused_lds(self.x_estep, self.mfac[:k],
-                                               add, self.w[:k])
+                                               add, self.w[:k], need_lik)
             elif self.use_fused_estep:
                 self._w_is_logw = True
                 add = self._add[:k]
                 if self.mfac32 is not None:
                     w, lik = F.estep_fused_f32(self.x_estep, self.mfac32[:k],
-                                               add, self.w[:k], self._lse)
+                                               add, self.w[:k], self._lse,
+                                               need_lik)
                 else:
                     w, lik = F.estep_fused(self.x_estep, self.mfac[:k], add,
-                                           self.w[:k], self._lse)
+                                           self.w[:k], self._lse, need_lik)
             elif self.use_big_estep:
                 add = self._add[:k]
                 if self.mfac32 is not None:
@@ -224,14 +228,14 @@ class EmEngine:
                 else:
                     logw = F.estep_logw_big(self.x_estep, self.mfac[:k],
                                             add, self.w[:k])
-                lik = F.estep_lse(logw, self._lse)
+                lik = F.estep_lse(logw, self._lse, need_lik)
                 self._w_is_logw = True
             elif self.device.type == "cuda":
                 logw = F.estep_logw(
                     self.x_estep, st.means, st.Rinv, st.constant, st.pi,
                     self.cfg.diag_only, out=self.w[:k],
                 )
-                lik = F.estep_lse(logw, self._lse)
+                lik = F.estep_lse(logw, self._lse, need_lik)
                 self._w_is_logw = True
             else:
                 logw = F.estep_logw(
@@ -254,10 +258,13 @@ class EmEngine:
         self._finish_likelihood(lik_part)
         return float(self._lik_dev.item())
 
-    def _iteration_body(self, k: int) -> None:
-        """One EM iteration ending with the reduced likelihood on device."""
+    def _iteration_body(self, k: int, need_lik: bool = True) -> None:
+        """One EM iteration; with need_lik the reduced likelihood lands in
+        the persistent device scalar."""
         self._mstep(k)
-        self._finish_likelihood(self._estep(k))
+        lik = self._estep(k, need_lik)
+        if need_lik and lik is not None:
+            self._finish_likelihood(lik)
 
     def _can_graph(self) -> bool:
         import os
@@ -319,12 +326,17 @@ class EmEngine:
         if cfg.min_iters >= cfg.max_iters and not cfg.verbose:
             # epsilon is dead (reference default MIN_ITERS == MAX_ITERS,
             # gaussian.h:26-27): the loop runs exactly min_iters times no
-            # matter what the likelihood does, so keep it on device and
-            # read the scalar ONCE per K instead of syncing the host every
-            # iteration (config 5's 22-K sweep paid ~2,200 syncs).
-            self._finish_likelihood(self._estep(k))
-            for _ in range(cfg.min_iters):
-                self.em_iteration(k)
+            # matter what the likelihood does. Keep the scalar on device,
+            # read it ONCE per K, and skip the likelihood reduce chain on
+            # every iteration but the last (the reference computes and
+            # discards it; the final model and likelihood are identical).
+            self._estep(k, need_lik=False)
+            for _ in range(max(0, cfg.min_iters - 1)):
+                self.em_iteration(k, need_lik=False)
+            if cfg.min_iters >= 1:
+                self.em_iteration(k, need_lik=True)
+            else:
+                self._finish_likelihood(self._estep(k))
             lik = float(self._lik_dev.item())
             self.likelihood = lik
             self.total_em_iterations += cfg.min_iters
@@ -347,7 +359,7 @@ class EmEngine:
         self.total_em_iterations += iters
         return lik
 
-    def em_iteration(self, k: int) -> None:
+    def em_iteration(self, k: int, need_lik: bool = True) -> None:
         """One EM iteration: M-step + all-reduce + constants + E-step +
         likelihood reduce. On GPU the whole sequence replays as one
         hipGraph (captured lazily per K; the capturing call runs the
@@ -356,27 +368,28 @@ class EmEngine:
         (Requires a prior _estep so self.w holds posteriors.)
         """
         if self._can_graph():
-            g = self._graphs.get(k)
+            key = (k, need_lik)
+            g = self._graphs.get(key)
             if g is None:
-                self._capture_iteration(k)
+                self._capture_iteration(k, need_lik)
                 return  # the capture's eager warmup WAS this iteration
             if g:
                 g.replay()
                 return
-        self._iteration_body(k)
+        self._iteration_body(k, need_lik)
 
-    def _capture_iteration(self, k: int):
+    def _capture_iteration(self, k: int, need_lik: bool = True):
         """Run one eager iteration (counts), then capture the graph."""
-        self._iteration_body(k)
+        self._iteration_body(k, need_lik)
         try:
             self.profile.paused = True
             torch.cuda.synchronize()
             g = torch.cuda.CUDAGraph()
             with torch.cuda.graph(g):
-                self._iteration_body(k)
-            self._graphs[k] = g
+                self._iteration_body(k, need_lik)
+            self._graphs[(k, need_lik)] = g
         except Exception:  # noqa: BLE001 — graphs are an optimization only
-            self._graphs[k] = False
+            self._graphs[(k, need_lik)] = False
         finally:
             self.profile.paused = False
         return None

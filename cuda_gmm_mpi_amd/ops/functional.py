@@ -48,7 +48,8 @@ def estep_posteriors(logw: torch.Tensor) -> tuple[torch.Tensor, torch.Tensor]:
     return logw, lik
 
 
-def estep_lse(logw: torch.Tensor, lse: torch.Tensor) -> torch.Tensor:
+def estep_lse(logw: torch.Tensor, lse: torch.Tensor,
+              need_lik: bool = True) -> torch.Tensor | None:
     """Per-event log-sum-exp into lse [N] + likelihood scalar (CUDA).
 
     logw is NOT normalized — the lse-aware M-step kernels apply
@@ -57,6 +58,8 @@ def estep_lse(logw: torch.Tensor, lse: torch.Tensor) -> torch.Tensor:
     nblocks = 1024
     partial = torch.zeros(nblocks, dtype=torch.float32, device=logw.device)
     hip_ext().estep_lse(logw, lse, partial)
+    if not need_lik:
+        return None
     lik = torch.empty(1, dtype=torch.float32, device=logw.device)
     hip_ext().reduce_scalar(partial, lik)
     return lik
@@ -186,8 +189,9 @@ def constants(r: torch.Tensor, means: torch.Tensor | None = None,
 
 
 def estep_fused(z: torch.Tensor, mfac: torch.Tensor, add: torch.Tensor,
-                w_out: torch.Tensor, lse: torch.Tensor
-                ) -> tuple[torch.Tensor, torch.Tensor]:
+                w_out: torch.Tensor, lse: torch.Tensor,
+                need_lik: bool = True
+                ) -> tuple[torch.Tensor, torch.Tensor | None]:
     """Fused bf16-MFMA E-step (CUDA only): LOG weights into w_out [K,N],
     per-event log-sum-exp into lse [N], returns (w_out, likelihood).
     Online-softmax over 256-event blocks; the lse-aware M-step applies
@@ -197,6 +201,8 @@ def estep_fused(z: torch.Tensor, mfac: torch.Tensor, add: torch.Tensor,
     # every launched block writes its partial slot: no zero-fill needed
     partial = torch.empty(nblk, dtype=torch.float32, device=z.device)
     hip_ext().estep_fused(z, mfac, add, w_out, lse, partial)
+    if not need_lik:
+        return w_out, None
     lik = torch.empty(1, dtype=torch.float32, device=z.device)
     hip_ext().reduce_scalar(partial, lik)
     return w_out, lik
@@ -222,40 +228,48 @@ def estep_fused_lds_available(device: torch.device, dtype: str, d: int,
 
 
 def estep_fused_lds(z: torch.Tensor, mfac: torch.Tensor, add: torch.Tensor,
-                    w_out: torch.Tensor
-                    ) -> tuple[torch.Tensor, torch.Tensor]:
+                    w_out: torch.Tensor, need_lik: bool = True
+                    ) -> tuple[torch.Tensor, torch.Tensor | None]:
     """v1 fused bf16 E-step: POSTERIORS into w_out, likelihood scalar."""
     n = z.shape[1]
     nblk = (n + 127) // 128
     partial = torch.empty(nblk, dtype=torch.float32, device=z.device)
     hip_ext().estep_fused_lds(z, mfac, add, w_out, partial)
+    if not need_lik:
+        return w_out, None
     lik = torch.empty(1, dtype=torch.float32, device=z.device)
     hip_ext().reduce_scalar(partial, lik)
     return w_out, lik
 
 
 def estep_fused_f32_lds(z: torch.Tensor, mfac32: torch.Tensor,
-                        add: torch.Tensor, w_out: torch.Tensor
-                        ) -> tuple[torch.Tensor, torch.Tensor]:
+                        add: torch.Tensor, w_out: torch.Tensor,
+                        need_lik: bool = True
+                        ) -> tuple[torch.Tensor, torch.Tensor | None]:
     """v1 exact-f32 fused E-step: POSTERIORS into w_out."""
     n = z.shape[1]
     nblk = (n + 127) // 128
     partial = torch.empty(nblk, dtype=torch.float32, device=z.device)
     hip_ext().estep_fused_f32_lds(z, mfac32, add, w_out, partial)
+    if not need_lik:
+        return w_out, None
     lik = torch.empty(1, dtype=torch.float32, device=z.device)
     hip_ext().reduce_scalar(partial, lik)
     return w_out, lik
 
 
 def estep_fused_f32(z: torch.Tensor, mfac32: torch.Tensor, add: torch.Tensor,
-                    w_out: torch.Tensor, lse: torch.Tensor
-                    ) -> tuple[torch.Tensor, torch.Tensor]:
+                    w_out: torch.Tensor, lse: torch.Tensor,
+                    need_lik: bool = True
+                    ) -> tuple[torch.Tensor, torch.Tensor | None]:
     """Exact-f32 MFMA fused E-step (CUDA, D <= 31, any K): log weights
     into w_out, per-event log-sum-exp into lse."""
     n = z.shape[1]
     nblk = (n + 255) // 256
     partial = torch.empty(nblk, dtype=torch.float32, device=z.device)
     hip_ext().estep_fused_f32(z, mfac32, add, w_out, lse, partial)
+    if not need_lik:
+        return w_out, None
     lik = torch.empty(1, dtype=torch.float32, device=z.device)
     hip_ext().reduce_scalar(partial, lik)
     return w_out, lik
