@@ -151,3 +151,25 @@ def test_diag_fixture_results(cli_output_diag):
     gw = gw[:, order]
     np.testing.assert_allclose(gw, ww, atol=5e-3)
     assert (gw.argmax(axis=1) == ww.argmax(axis=1)).all()
+
+
+def test_fixture_generator_reproduces_committed_bytes(tmp_path):
+    """The committed fixtures are exactly what the (deterministic)
+    transcription generator produces — guards both against accidental
+    generator edits and against stale fixtures after deliberate ones."""
+    import importlib.util
+    import sys
+    gen = os.path.join(os.path.dirname(FIX), "..", "scripts",
+                       "make_golden_fixture.py")
+    spec = importlib.util.spec_from_file_location("mgf", gen)
+    mod = importlib.util.module_from_spec(spec)
+    sys.modules["mgf"] = mod
+    spec.loader.exec_module(mod)
+    mod.OUT = str(tmp_path)
+    mod.main()
+    for name in ("golden_small.bin", "golden_small.summary",
+                 "golden_small.results", "golden_diag.bin",
+                 "golden_diag.summary", "golden_diag.results"):
+        got = open(os.path.join(tmp_path, name), "rb").read()
+        want = open(os.path.join(FIX, name), "rb").read()
+        assert got == want, f"fixture drift: {name}"
