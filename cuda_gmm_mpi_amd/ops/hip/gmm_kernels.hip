@@ -1880,7 +1880,7 @@ mstep_moments_big_kernel(const float* __restrict__ x,
     if (ti + 1 < my_tiles) issue_loads(ti + 1);
     __syncthreads();   // staging visible
 
-#pragma unroll 2
+#pragma unroll
     for (int ch = 0; ch < MBB_BK / 16; ++ch) {
       const int eb = ch * 16 + 8 * g2;
       const float4 wv0 = *(const float4*)(wt + cw * MBB_BK + eb);
