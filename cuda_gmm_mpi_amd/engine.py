@@ -118,6 +118,12 @@ class EmEngine:
             if self.device.type == "cuda" else None
         )
 
+        # factor-path iterations skip the LU kernel; Rinv refreshed per K
+        self._lazy_rinv = (
+            self.device.type == "cuda"
+            and (self.use_fused_estep or self.use_big_estep)
+            and not config.diag_only
+        )
         self.state = GmmState.empty(k0, self.d, self.device)
         seed_state(
             self.state, seed_means - center.unsqueeze(0), var_per_dim,
@@ -126,6 +132,7 @@ class EmEngine:
         # constants for the seeded R=I state (constants_kernel after seeding,
         # gaussian.cu:404); also fills the add buffer
         self._update_constants(self.state)
+        self._refresh_rinv(k0)  # state.Rinv valid from the start
 
         # membership / logw buffer, cluster-major [K, n_shard]. On CUDA
         # this holds LOG weights after the E-step, with the per-event
@@ -166,6 +173,7 @@ class EmEngine:
             st.R.contiguous(), st.means.contiguous(),
             self.mfac[:k] if self.mfac is not None else empty_b,
             self.mfac32[:k] if self.mfac32 is not None else empty_f,
+            empty_f, empty_f, empty_f,  # constants NOT recomputed (quirk #8)
         )
 
     def _update_constants(self, st: GmmState) -> None:
@@ -176,11 +184,41 @@ class EmEngine:
             pi_add = (
                 (st.pi, self._add[:k]) if self._add is not None else None
             )
-            # written in place into the state (no copy-back kernels)
-            F.constants(st.R, st.means, self.cfg.diag_only,
-                        mfac, mfac32, pi_add=pi_add,
-                        out=(st.Rinv, st.constant))
+            if self._lazy_rinv and mfac is not None and pi_add is not None:
+                # factor-path iterations never read Rinv: one emission
+                # kernel produces the Cholesky factors AND the constants
+                # (ln|R| = 2 sum ln diag L, fp-equivalent to the LU det);
+                # the reference-faithful LU Rinv is refreshed once per K
+                # (_refresh_rinv) for the merge/output path. Saves the
+                # ~47 us LU kernel on every EM iteration.
+                from .ops.backend import hip_ext
+                empty_f = torch.empty(0, dtype=torch.float32,
+                                      device=self.device)
+                hip_ext().emit_factors(
+                    st.R, st.means, mfac,
+                    mfac32 if mfac32 is not None else empty_f,
+                    st.pi, st.constant, self._add[:k],
+                )
+            else:
+                # written in place into the state (no copy-back kernels)
+                F.constants(st.R, st.means, self.cfg.diag_only,
+                            mfac, mfac32, pi_add=pi_add,
+                            out=(st.Rinv, st.constant))
         self.profile.count("constants")
+
+    def _refresh_rinv(self, k: int) -> None:
+        """Reference-faithful no-pivot-LU Rinv for the merge/output/
+        checkpoint consumers (the factor-path iterations keep it lazy)."""
+        if not self._lazy_rinv:
+            return
+        from .ops.backend import hip_ext
+        st = self.state.shrink(k)
+        empty_f = torch.empty(0, dtype=torch.float32, device=self.device)
+        empty_b = torch.empty(0, dtype=torch.bfloat16, device=self.device)
+        logdet = torch.empty(k, dtype=torch.float32, device=self.device)
+        hip_ext().constants(st.R.contiguous(), st.means.contiguous(),
+                            empty_f, st.Rinv, logdet, empty_f, empty_f,
+                            empty_b, empty_f, False)
 
     def _sync_add(self, k: int) -> None:
         """Recompute constant+ln(pi) after host-side param loads (merge /
@@ -531,6 +569,9 @@ class EmEngine:
 
         while k >= stop:
             lik = self.run_em(k)
+            # lazy-Rinv paths: make state.Rinv reference-faithful before
+            # it is saved, merged, checkpointed or broadcast
+            self._refresh_rinv(k)
             riss = rissanen_score(lik, k, self.d, self.n_total)
             riss_by_k[k] = riss
             if cfg.enable_print and self.rank == 0:

@@ -234,7 +234,7 @@ void constants(torch::Tensor r, torch::Tensor means, torch::Tensor pi,
       hipLaunchKernelGGL(gmm::emit_mfac_from_r_kernel, dim3(k), dim3(kNT),
                          sizeof(float) * (2 * (size_t)d * (d | 1) + d), s,
                          r.data_ptr<float>(), means.data_ptr<float>(), mp,
-                         mp32, d);
+                         mp32, d, nullptr, nullptr, nullptr);
     }
   } else {
     // working buffer + read-only LU snapshot (+ u0 scratch for the factor)
@@ -578,7 +578,9 @@ void mstep_finalize(torch::Tensor packed, torch::Tensor avgvar,
 }
 
 void emit_factors(torch::Tensor r, torch::Tensor means,
-                  torch::Tensor mfac, torch::Tensor mfac32) {
+                  torch::Tensor mfac, torch::Tensor mfac32,
+                  torch::Tensor pi, torch::Tensor constant,
+                  torch::Tensor add) {
   check_f32(r, "r");
   check_f32(means, "means");
   const int k = (int)r.size(0);
@@ -599,9 +601,13 @@ void emit_factors(torch::Tensor r, torch::Tensor means,
         reinterpret_cast<const void*>(&gmm::emit_mfac_from_r_kernel),
         hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds));
   }
+  const float* pip = pi.numel() > 0 ? pi.data_ptr<float>() : nullptr;
+  float* cst = constant.numel() > 0 ? constant.data_ptr<float>() : nullptr;
+  float* addp = add.numel() > 0 ? add.data_ptr<float>() : nullptr;
+  TORCH_CHECK(addp == nullptr || pip != nullptr, "add output requires pi");
   hipLaunchKernelGGL(gmm::emit_mfac_from_r_kernel, dim3(k), dim3(kNT),
                      lds, stream(), r.data_ptr<float>(),
-                     means.data_ptr<float>(), mp, mp32, d);
+                     means.data_ptr<float>(), mp, mp32, d, pip, cst, addp);
   HIP_CHECK(hipGetLastError());
 }
 

@@ -477,7 +477,10 @@ __device__ inline void emit_mfac(float* a, float* o,
                                  const float* __restrict__ r_global,
                                  const float* __restrict__ means,
                                  __hip_bfloat16* __restrict__ mfac,
-                                 float* __restrict__ mfac32, int c, int d) {
+                                 float* __restrict__ mfac32, int c, int d,
+                                 const float* __restrict__ pi = nullptr,
+                                 float* __restrict__ constant = nullptr,
+                                 float* __restrict__ add = nullptr) {
   const int tid = threadIdx.x;
   // odd LDS row stride: stride-d column walks with gcd(d, 32) > 1 put
   // whole lane groups on one bank (8-way at d = 24)
@@ -501,6 +504,30 @@ __device__ inline void emit_mfac(float* a, float* o,
 #pragma unroll 8
       for (int kk = 0; kk < j; ++kk) s -= a[i * ldp + kk] * a[j * ldp + kk];
       a[i * ldp + j] = s / piv;
+    }
+    __syncthreads();
+  }
+  // ln|R| = 2 sum ln diag(L): lets the EM iteration skip the LU kernel
+  // entirely on the factor paths (constant = -D/2 ln 2pi - 0.5 ln|R|,
+  // fp-equivalent to the LU determinant; the reference-faithful LU Rinv
+  // is refreshed once per K for the merge/output path)
+  if (constant != nullptr) {
+    __shared__ float wsum_e[NT / WAVE];
+    const int nw = (blockDim.x + WAVE - 1) / WAVE;
+    float acc = 0.0f;
+    for (int i = tid; i < d; i += blockDim.x)
+      acc += __logf(a[i * ldp + i]);
+    for (int off = WAVE / 2; off > 0; off >>= 1)
+      acc += __shfl_down(acc, off, WAVE);
+    if ((tid & (WAVE - 1)) == 0) wsum_e[tid / WAVE] = acc;
+    __syncthreads();
+    if (tid == 0) {
+      const float ld2 = 2.0f * (wsum_e[0] + ((nw > 1) ? wsum_e[1] : 0.0f) +
+                                ((nw > 2) ? wsum_e[2] : 0.0f) +
+                                ((nw > 3) ? wsum_e[3] : 0.0f));
+      const float cst = -d * 0.5f * 1.8378770664093453f - 0.5f * ld2;
+      constant[c] = cst;
+      if (add) add[c] = cst + __logf(pi[c]);
     }
     __syncthreads();
   }
@@ -2272,10 +2299,13 @@ __global__ void __launch_bounds__(NT)
 emit_mfac_from_r_kernel(const float* __restrict__ r,
                         const float* __restrict__ means,
                         __hip_bfloat16* __restrict__ mfac,
-                        float* __restrict__ mfac32, int d) {
+                        float* __restrict__ mfac32, int d,
+                        const float* __restrict__ pi,
+                        float* __restrict__ constant,
+                        float* __restrict__ add) {
   extern __shared__ float buf[];
   emit_mfac(buf, buf + d * (d | 1), r, means, mfac, mfac32,
-            blockIdx.x, d);
+            blockIdx.x, d, pi, constant, add);
 }
 
 // DIAG_ONLY constants (gaussian_kernel.cu:215-223)
