@@ -118,12 +118,6 @@ class EmEngine:
             if self.device.type == "cuda" else None
         )
 
-        # factor-path iterations skip the LU kernel; Rinv refreshed per K
-        self._lazy_rinv = (
-            self.device.type == "cuda"
-            and (self.use_fused_estep or self.use_big_estep)
-            and not config.diag_only
-        )
         self.state = GmmState.empty(k0, self.d, self.device)
         seed_state(
             self.state, seed_means - center.unsqueeze(0), var_per_dim,
@@ -205,6 +199,16 @@ class EmEngine:
                             mfac, mfac32, pi_add=pi_add,
                             out=(st.Rinv, st.constant))
         self.profile.count("constants")
+
+    @property
+    def _lazy_rinv(self) -> bool:
+        """Factor-path iterations skip the LU kernel (Rinv refreshed per
+        K). A PROPERTY of the live dispatch flags: tests (and debugging)
+        flip use_fused_estep/use_big_estep after construction, and the
+        VALU E-step they force DOES read Rinv every iteration."""
+        return (self.device.type == "cuda"
+                and (self.use_fused_estep or self.use_big_estep)
+                and not self.cfg.diag_only)
 
     def _refresh_rinv(self, k: int) -> None:
         """Reference-faithful no-pivot-LU Rinv for the merge/output/
