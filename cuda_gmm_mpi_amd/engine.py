@@ -337,6 +337,24 @@ class EmEngine:
                 lse=self._lse if getattr(self, "_w_is_logw", False) else None)
         with self.profile.time("comm"):
             pdist.all_reduce_(packed)
+        lazy = (self._lazy_rinv and self._add is not None
+                and self.mfac is not None)
+        if lazy:
+            # ONE kernel: finalize (N/means/R/pi, reference rules) fused
+            # with the factor + constants emission — R stays L1-hot and
+            # the iteration drops a launch; Rinv stays lazy (per K)
+            from .ops.backend import hip_ext
+            with self.profile.time("m_step"):
+                empty_f = torch.empty(0, dtype=torch.float32,
+                                      device=self.device)
+                hip_ext().mstep_finalize_emit(
+                    packed, st.avgvar, self.world, st.N, st.means, st.R,
+                    st.pi, st.constant, self._add[:k], self.mfac[:k],
+                    self.mfac32[:k] if self.mfac32 is not None else empty_f,
+                )
+            self.profile.count("params")
+            self.profile.count("constants")
+            return
         with self.profile.time("m_step"):
             if self.device.type == "cuda":
                 # one kernel: N, means, R, pi from the packed moments with

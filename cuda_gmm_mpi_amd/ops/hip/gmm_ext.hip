@@ -577,6 +577,50 @@ void mstep_finalize(torch::Tensor packed, torch::Tensor avgvar,
   HIP_CHECK(hipGetLastError());
 }
 
+void mstep_finalize_emit(torch::Tensor packed, torch::Tensor avgvar,
+                         int64_t world, torch::Tensor n_out,
+                         torch::Tensor means, torch::Tensor r_out,
+                         torch::Tensor pi, torch::Tensor constant,
+                         torch::Tensor add, torch::Tensor mfac,
+                         torch::Tensor mfac32) {
+  check_f32(packed, "packed");
+  check_f32(avgvar, "avgvar");
+  check_f32(n_out, "n_out");
+  check_f32(means, "means");
+  check_f32(r_out, "r_out");
+  check_f32(pi, "pi");
+  check_f32(constant, "constant");
+  check_f32(add, "add");
+  const int k = (int)means.size(0);
+  const int d = (int)means.size(1);
+  TORCH_CHECK(packed.size(0) == k &&
+                  packed.size(1) == (d + 1) * (d + 2) / 2,
+              "packed shape mismatch");
+  TORCH_CHECK(mfac.numel() > 0 && mfac.scalar_type() == torch::kBFloat16 &&
+              mfac.is_contiguous());
+  __hip_bfloat16* mp =
+      reinterpret_cast<__hip_bfloat16*>(mfac.data_ptr());
+  float* mp32 = nullptr;
+  if (mfac32.numel() > 0) {
+    check_f32(mfac32, "mfac32");
+    mp32 = mfac32.data_ptr<float>();
+  }
+  const size_t lds = sizeof(float) * (2 * (size_t)d * (d | 1) + d);
+  if (lds > 64 * 1024) {
+    HIP_CHECK(hipFuncSetAttribute(
+        reinterpret_cast<const void*>(&gmm::mstep_finalize_emit_kernel),
+        hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds));
+  }
+  hipLaunchKernelGGL(gmm::mstep_finalize_emit_kernel, dim3(k), dim3(kNT),
+                     lds, stream(), packed.data_ptr<float>(),
+                     avgvar.data_ptr<float>(), (int)world,
+                     n_out.data_ptr<float>(), means.data_ptr<float>(),
+                     r_out.data_ptr<float>(), pi.data_ptr<float>(),
+                     constant.data_ptr<float>(), add.data_ptr<float>(), mp,
+                     mp32, d, k);
+  HIP_CHECK(hipGetLastError());
+}
+
 void emit_factors(torch::Tensor r, torch::Tensor means,
                   torch::Tensor mfac, torch::Tensor mfac32,
                   torch::Tensor pi, torch::Tensor constant,
@@ -678,6 +722,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "deterministic chunk-partials reduction out[j] = sum_i in[i][j]");
   m.def("reduce_scalar", &reduce_scalar,
         "deterministic scalar sum out[0] = sum(in)");
+  m.def("mstep_finalize_emit", &mstep_finalize_emit,
+        "fused M-step finalize + factor/constants emission (fast path)");
   m.def("mstep_finalize", &mstep_finalize,
         "finalize N/means/R/pi from all-reduced packed moments");
   m.def("emit_factors", &emit_factors,

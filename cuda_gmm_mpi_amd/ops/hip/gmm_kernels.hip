@@ -580,6 +580,66 @@ __device__ inline void emit_mfac(float* a, float* o,
   }
 }
 
+// Fused M-step finalize + factor/constants emission (one block per
+// cluster): the finalize's R lands in this block's L1, the emission
+// reloads it without a kernel boundary, and the EM iteration drops one
+// launch. Fast-path only (lazy-Rinv engines); the split kernels remain
+// for the VALU/diag paths.
+__global__ void __launch_bounds__(NT)
+mstep_finalize_emit_kernel(const float* __restrict__ packed,
+                           const float* __restrict__ avgvar, int world,
+                           float* __restrict__ n_out,
+                           float* __restrict__ means,
+                           float* __restrict__ r_out,
+                           float* __restrict__ pi,
+                           float* __restrict__ constant,
+                           float* __restrict__ add,
+                           __hip_bfloat16* __restrict__ mfac,
+                           float* __restrict__ mfac32, int d, int k) {
+  extern __shared__ float lds[];
+  const int ldp = d | 1;
+  float* a = lds;                 // emit working buffer [d*ldp]
+  float* o = a + d * ldp;         // emit snapshot [d*ldp] (+u0 [d])
+  float* mu = o + d * ldp;        // borrow u0 space for mu during finalize
+  const int c = blockIdx.x;
+  const int dp = d + 1;
+  const int pp = dp * (dp + 1) / 2;
+  const int p = d * (d + 1) / 2;
+  const float* row = packed + (int64_t)c * pp;
+  const float n_c = row[pp - 1];
+  const bool ge1 = n_c >= 1.0f;
+  const bool gt05 = n_c > 0.5f;
+
+  if (threadIdx.x == 0) {
+    n_out[c] = n_c;
+    float total = 0.0f;
+    for (int cc = 0; cc < k; ++cc) total += packed[(int64_t)cc * pp + pp - 1];
+    pi[c] = (n_c < 0.5f) ? 1e-10f : n_c / total;
+  }
+  for (int i = threadIdx.x; i < d; i += NT) {
+    const float m = gt05 ? row[p + i] / n_c : 0.0f;
+    mu[i] = m;
+    means[(int64_t)c * d + i] = m;
+  }
+  __syncthreads();
+  const float reg = world * avgvar[c];
+  for (int t = threadIdx.x; t < p; t += NT) {
+    int i, j;
+    tri_row_col(t, &i, &j);
+    float cov = ge1 ? row[t] - n_c * mu[i] * mu[j] : 0.0f;
+    if (i == j) cov += reg;
+    float rv;
+    if (gt05) rv = cov / n_c;
+    else rv = (i == j) ? 1.0f : 0.0f;
+    r_out[((int64_t)c * d + i) * d + j] = rv;
+    if (i != j) r_out[((int64_t)c * d + j) * d + i] = rv;
+  }
+  __syncthreads();
+  // R written by THIS block is L1-visible after the barrier; the
+  // emission path reloads it (and means) through the same pointers
+  emit_mfac(a, o, r_out, means, mfac, mfac32, c, d, pi, constant, add);
+}
+
 // NOTE (negative result, round 2): a wave-per-cluster constants variant
 // (4 independent waves per block, zero workgroup barriers, cross-lane
 // LDS ordering via the wave's own lgkmcnt) measured SLOWER than this
