@@ -511,7 +511,14 @@ void mstep_moments_big(torch::Tensor x, torch::Tensor w,
                   partials.size(2) == dp * (dp + 1) / 2,
               "partials must be [nchunk, K, Dp*(Dp+1)/2]");
   const int rows = ((dp + 31) / 32) * 32;
-  const size_t lds = (size_t)2 * rows * (64 + 8) * 2 + 2 * 64 * 4;
+  // double-buffered: 2 x (zhi+zlo planes + w tiles)
+  const size_t lds =
+      2 * ((size_t)2 * rows * (64 + 8) * 2 + 2 * 64 * 4);
+  if (lds > 64 * 1024) {
+    HIP_CHECK(hipFuncSetAttribute(
+        reinterpret_cast<const void*>(&gmm::mstep_moments_big_kernel),
+        hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds));
+  }
   dim3 grid((k + 1) / 2, nchunk);
   hipLaunchKernelGGL(gmm::mstep_moments_big_kernel, grid, dim3(1024), lds,
                      stream(), x.data_ptr<float>(), w.data_ptr<float>(),

@@ -1832,10 +1832,16 @@ mstep_moments_big_kernel(const float* __restrict__ x,
   const int rows = rt2 * 32;
   const int tp = rt2 * (rt2 + 1) / 2;
   const int zbr = MBB_BK + 8;  // bf16 row stride
+  // DOUBLE-BUFFERED LDS (ablation: single-buffer T14 still exposed
+  // ~37% of the kernel as staging+barrier time — the write pass sat
+  // between two barriers on the critical path): per buffer
+  // zhi/zlo [rows][zbr] bf16 + wt [MBB_CPB][MBB_BK] f32; one barrier
+  // per tile. 93 KB at D=128 -> 1 block/CU of 16 waves (4 waves/SIMD).
   extern __shared__ float lds[];
-  __bf16* zhi = (__bf16*)lds;            // [rows][zbr]
-  __bf16* zlo = zhi + rows * zbr;
-  float* wt = (float*)(zlo + rows * zbr);  // [MBB_CPB][MBB_BK]
+  const int zplane = rows * zbr;               // bf16 elems per z plane
+  const int bufsz_b = 2 * zplane;              // bf16 elems per buffer
+  __bf16* zbuf = (__bf16*)lds;                 // [2][2*zplane]
+  float* wbuf = (float*)(zbuf + 2 * bufsz_b);  // [2][MBB_CPB*MBB_BK]
 
   const int wave = threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
@@ -1923,7 +1929,10 @@ mstep_moments_big_kernel(const float* __restrict__ x,
       }
     }
   };
-  auto write_buf = [&]() {
+  auto write_buf = [&](int buf) {
+    __bf16* zhi = zbuf + buf * bufsz_b;
+    __bf16* zlo = zhi + zplane;
+    float* wt = wbuf + buf * MBB_CPB * MBB_BK;
 #pragma unroll
     for (int sq = 0; sq < MBB_NXQ; ++sq) {
       const int q = threadIdx.x + sq * MBB_NT;
@@ -1951,21 +1960,30 @@ mstep_moments_big_kernel(const float* __restrict__ x,
     }
   };
 
-  // constant rows (ones at d, zeros above), written once; visibility is
-  // covered by the first in-loop barrier
-  for (int idx = d * MBB_BK + threadIdx.x; idx < rows * MBB_BK;
-       idx += MBB_NT) {
-    const int di = idx / MBB_BK, ei = idx % MBB_BK;
-    zhi[di * zbr + ei] = (__bf16)(di == d ? 1.0f : 0.0f);
-    zlo[di * zbr + ei] = (__bf16)0.0f;
+  // constant rows (ones at d, zeros above) in BOTH buffers, once;
+  // visibility covered by the first barrier
+  for (int bset = 0; bset < 2; ++bset) {
+    __bf16* zhi = zbuf + bset * bufsz_b;
+    __bf16* zlo = zhi + zplane;
+    for (int idx = d * MBB_BK + threadIdx.x; idx < rows * MBB_BK;
+         idx += MBB_NT) {
+      const int di = idx / MBB_BK, ei = idx % MBB_BK;
+      zhi[di * zbr + ei] = (__bf16)(di == d ? 1.0f : 0.0f);
+      zlo[di * zbr + ei] = (__bf16)0.0f;
+    }
   }
-  if (my_tiles > 0) issue_loads(0);
+  if (my_tiles > 0) {
+    issue_loads(0);
+    write_buf(0);
+  }
+  __syncthreads();
 
+  int cur = 0;
   for (int64_t ti = 0; ti < my_tiles; ++ti) {
-    __syncthreads();   // previous tile's readers are done with the buffer
-    write_buf();
     if (ti + 1 < my_tiles) issue_loads(ti + 1);
-    __syncthreads();   // staging visible
+    const __bf16* zhi = zbuf + cur * bufsz_b;
+    const __bf16* zlo = zhi + zplane;
+    const float* wt = wbuf + cur * MBB_CPB * MBB_BK;
 
 #pragma unroll
     for (int ch = 0; ch < MBB_BK / 16; ++ch) {
@@ -2008,6 +2026,11 @@ mstep_moments_big_kernel(const float* __restrict__ x,
             __builtin_amdgcn_mfma_f32_32x32x16_bf16(a_lo, b_hi, acc[pp], 0, 0, 0);
       }
     }
+    // write the NEXT tile into the other buffer; single barrier per
+    // tile (nobody reads that buffer this iteration)
+    if (ti + 1 < my_tiles) write_buf(cur ^ 1);
+    __syncthreads();
+    cur ^= 1;
   }
 
   if (c >= k) return;
