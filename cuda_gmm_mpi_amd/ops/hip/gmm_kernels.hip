@@ -1853,31 +1853,12 @@ mstep_moments_big_kernel(const float* __restrict__ x,
   const int chunk = blockIdx.y;
   const int p_lo = group * MBB_PMAX;
   const int p_hi = min(tp, p_lo + MBB_PMAX);
-  // Pair ordering: tri order at group stride 2 almost never puts two
-  // same-row pairs in one wave, killing the A-split hoist (the ablation
-  // shows the split is ~40% of the kernel). Reorder so each group of 2
-  // holds same-row pairs where possible (6 of 8 groups at RT2=5);
-  // leftovers (odd rows) are appended. Built once per block.
-  __shared__ int porder[16];
-  if (threadIdx.x == 0) {
-    int idx = 0, leftovers[8], nl = 0;
-    for (int r = 0; r < rt2; ++r) {
-      const int cnt = r + 1, cbase = r * (r + 1) / 2;
-      for (int f = 0; f < cnt / 2; ++f) {
-        porder[idx++] = cbase + 2 * f;
-        porder[idx++] = cbase + 2 * f + 1;
-      }
-      if (cnt & 1) leftovers[nl++] = cbase + cnt - 1;
-    }
-    for (int i = 0; i < nl; ++i) porder[idx++] = leftovers[i];
-  }
-  __syncthreads();
   // hoisted pair->tile mapping (tri_row_col has a sqrtf: keep it out of
   // the chunk loop)
   int ptr[MBB_PMAX], ptc[MBB_PMAX];
 #pragma unroll
   for (int pp = 0; pp < MBB_PMAX; ++pp) {
-    if (p_lo + pp < tp) tri_row_col(porder[p_lo + pp], &ptr[pp], &ptc[pp]);
+    if (p_lo + pp < tp) tri_row_col(p_lo + pp, &ptr[pp], &ptc[pp]);
     else { ptr[pp] = 0; ptc[pp] = 0; }
   }
 
