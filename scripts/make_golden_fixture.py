@@ -390,6 +390,23 @@ def main():
     write_results(os.path.join(OUT, "golden_diag.results"), data2, w2)
     print(f"diag fixture written: likelihood={lik2:.4f}")
 
+    # ---- K=1 fixture: single-cluster path (seed fraction=0, pi=1,
+    # no merge loop; gaussian.cu:117 k==1 branch) ----
+    rng = np.random.default_rng(31337)
+    n3, d3 = 400, 2
+    data3 = rng.normal([250.0, 180.0], [20.0, 35.0],
+                       size=(n3, d3)).astype(F32)
+    with open(os.path.join(OUT, "golden_k1.bin"), "wb") as f:
+        f.write(struct.pack("<ii", n3, d3))
+        f.write(data3.tobytes())
+    dt3 = data3.T.astype(np.float64)
+    c3 = seed(data3, 1)
+    constants(c3)
+    w3, lik3 = run_em(dt3, c3, iters=100)
+    write_summary(os.path.join(OUT, "golden_k1.summary"), c3)
+    write_results(os.path.join(OUT, "golden_k1.results"), data3, w3)
+    print(f"k1 fixture written: likelihood={lik3:.4f}")
+
 
 if __name__ == "__main__":
     sys.exit(main())

@@ -153,6 +153,38 @@ def test_diag_fixture_results(cli_output_diag):
     assert (gw.argmax(axis=1) == ww.argmax(axis=1)).all()
 
 
+@pytest.fixture(scope="module")
+def cli_output_k1(tmp_path_factory):
+    out = str(tmp_path_factory.mktemp("golden_k1") / "o")
+    rc = main(["1", os.path.join(FIX, "golden_k1.bin"), out, "1",
+               "--device", "cpu", "--no-center"])
+    assert rc == 0
+    return out
+
+
+def test_k1_fixture_summary(cli_output_k1):
+    got = parse_summary(open(cli_output_k1 + ".summary").read())
+    want = parse_summary(open(os.path.join(FIX, "golden_k1.summary")).read())
+    assert len(got) == len(want) == 1
+    g, w = got[0], want[0]
+    assert g["pi"] == pytest.approx(1.0) and w["pi"] == pytest.approx(1.0)
+    assert g["N"] == pytest.approx(w["N"], rel=1e-3)
+    np.testing.assert_allclose(g["means"], w["means"], rtol=2e-3, atol=0.2)
+    scale = np.abs(w["R"]).max()
+    np.testing.assert_allclose(g["R"], w["R"], rtol=5e-2, atol=5e-2 * scale)
+
+
+def test_k1_fixture_results(cli_output_k1):
+    got_lines = open(cli_output_k1 + ".results").read().splitlines()
+    want_lines = open(os.path.join(FIX,
+                                   "golden_k1.results")).read().splitlines()
+    assert len(got_lines) == len(want_lines) == 400
+    for g, w in zip(got_lines, want_lines):
+        assert g.split("\t")[0] == w.split("\t")[0]
+        # single cluster: membership column is identically 1.000000
+        assert g.split("\t")[1] == w.split("\t")[1] == "1.000000"
+
+
 def test_fixture_generator_reproduces_committed_bytes(tmp_path):
     """The committed fixtures are exactly what the (deterministic)
     transcription generator produces — guards both against accidental
@@ -169,7 +201,9 @@ def test_fixture_generator_reproduces_committed_bytes(tmp_path):
     mod.main()
     for name in ("golden_small.bin", "golden_small.summary",
                  "golden_small.results", "golden_diag.bin",
-                 "golden_diag.summary", "golden_diag.results"):
+                 "golden_diag.summary", "golden_diag.results",
+                 "golden_k1.bin", "golden_k1.summary",
+                 "golden_k1.results"):
         got = open(os.path.join(tmp_path, name), "rb").read()
         want = open(os.path.join(FIX, name), "rb").read()
         assert got == want, f"fixture drift: {name}"
