@@ -417,28 +417,35 @@ void estep_logw_big(torch::Tensor z, torch::Tensor mfac, torch::Tensor add,
   const int rows = ((d + 31) / 32) * 32;
   const int rt_n = rows / 32;
   const int arow = kct * 16 + 8;
-  const size_t lds = (size_t)(2 * rows + 128) * arow * 2 +
-                     (size_t)rt_n * 128 * 4;
-  const int64_t tiles = (n + 127) / 128;
+  // 256-event tiles where the LDS fits (gfx950: 160 KB/workgroup max),
+  // else 128 (only D > 128, rows = 160, needs the smaller tile)
+  const size_t lds256 = (size_t)(2 * rows + 256) * arow * 2 +
+                        (size_t)rt_n * 256 * 4;
+  const size_t lds128 = (size_t)(2 * rows + 128) * arow * 2 +
+                        (size_t)rt_n * 128 * 4;
+  const int be = lds256 <= 160 * 1024 ? 256 : 128;
+  const size_t lds = be == 256 ? lds256 : lds128;
+  const int64_t tiles = (n + be - 1) / be;
   // enough blocks to fill the chip; each block amortizes one staged
   // factor table over tiles/nchunk z tiles
   const int nchunk =
       (int)std::min<int64_t>(tiles, std::max<int64_t>(1, 2048 / k));
   dim3 grid((uint32_t)nchunk, k);
-  // block size matched to the (row-tile x event-tile) pair count so no
-  // waves idle at small rt_n (D <= 79 has rt_n <= 3 -> <= 12 pairs)
-  const uint32_t nthreads =
-      std::min<uint32_t>(1024, std::max<uint32_t>(256, rt_n * 4 * 64));
+  // block size matched to the (row-tile x event-tile) task count so no
+  // waves idle at small rt_n
+  const uint32_t nthreads = std::min<uint32_t>(
+      1024, std::max<uint32_t>(256, rt_n * (be / 32) * 64));
   auto s = stream();
-#define LAUNCH_ELB(KCT)                                                     \
+#define LAUNCH_ELB(KCT, BE)                                                 \
   do {                                                                      \
     if (lds > 64 * 1024) {                                                  \
       HIP_CHECK(hipFuncSetAttribute(                                        \
-          reinterpret_cast<const void*>(&gmm::estep_logw_big2_kernel<KCT>), \
+          reinterpret_cast<const void*>(                                    \
+              &gmm::estep_logw_big2_kernel<KCT, BE>),                       \
           hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds));           \
     }                                                                       \
-    hipLaunchKernelGGL((gmm::estep_logw_big2_kernel<KCT>), grid,            \
-                       dim3(nthreads), lds, s,                                  \
+    hipLaunchKernelGGL((gmm::estep_logw_big2_kernel<KCT, BE>), grid,        \
+                       dim3(nthreads), lds, s,                              \
                        reinterpret_cast<const __hip_bfloat16*>(             \
                            z.data_ptr()),                                   \
                        reinterpret_cast<const __hip_bfloat16*>(             \
@@ -446,10 +453,11 @@ void estep_logw_big(torch::Tensor z, torch::Tensor mfac, torch::Tensor add,
                        add.data_ptr<float>(), logw.data_ptr<float>(), d, k, \
                        n, nchunk);                                          \
   } while (0)
-  if (kct == 2) LAUNCH_ELB(2);
-  else if (kct == 3) LAUNCH_ELB(3);
-  else if (kct == 5) LAUNCH_ELB(5);
-  else LAUNCH_ELB(9);
+  if (kct == 2) LAUNCH_ELB(2, 256);
+  else if (kct == 3) LAUNCH_ELB(3, 256);
+  else if (kct == 5) LAUNCH_ELB(5, 256);
+  else if (be == 256) LAUNCH_ELB(9, 256);
+  else LAUNCH_ELB(9, 128);
 #undef LAUNCH_ELB
   HIP_CHECK(hipGetLastError());
 }

@@ -2160,10 +2160,9 @@ estep_logw_big_kernel(const __hip_bfloat16* __restrict__ z,
 // work; per-tile partials combine through LDS (fixed order:
 // deterministic).
 // ---------------------------------------------------------------------------
-#define ESB2_BE 128
 #define ESB2_NT 1024
 
-template <int KCT>
+template <int KCT, int BE>
 __global__ void __launch_bounds__(ESB2_NT)
 estep_logw_big2_kernel(const __hip_bfloat16* __restrict__ z,
                        const __hip_bfloat16* __restrict__ mfac,
@@ -2179,8 +2178,8 @@ estep_logw_big2_kernel(const __hip_bfloat16* __restrict__ z,
   const int rows = rt_n * 32;
   __bf16* ah = (__bf16*)lds;             // [rows][AROW]
   __bf16* al = ah + rows * AROW;
-  __bf16* zs = al + rows * AROW;         // [ESB2_BE][AROW]
-  float* qpart = (float*)(zs + ESB2_BE * AROW);  // [rt_n][ESB2_BE]
+  __bf16* zs = al + rows * AROW;         // [BE][AROW]
+  float* qpart = (float*)(zs + BE * AROW);  // [rt_n][BE]
 
   const int wave = threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
@@ -2205,10 +2204,10 @@ estep_logw_big2_kernel(const __hip_bfloat16* __restrict__ z,
     }
   }
 
-  const int64_t tiles = (n + ESB2_BE - 1) / ESB2_BE;
+  const int64_t tiles = (n + BE - 1) / BE;
   for (int64_t tile = chunk; tile < tiles; tile += nchunk) {
-    const int64_t e0 = tile * ESB2_BE;
-    const int cnt = (int)min((int64_t)ESB2_BE, n - e0);
+    const int64_t e0 = tile * BE;
+    const int cnt = (int)min((int64_t)BE, n - e0);
     __syncthreads();
     // transposed z staging: each thread gathers 4 consecutive k-slots of
     // one event (4 coalesced-by-wave global reads) and writes ONE 8-byte
@@ -2216,15 +2215,15 @@ estep_logw_big2_kernel(const __hip_bfloat16* __restrict__ z,
     // conflict (stride-AROW lanes 8 apart collide mod 32 for any 16B
     // row stride; 20% LDSBankConflict measured)
     {
-      const int kq_total = ((d + 3) / 4) * ESB2_BE;
+      const int kq_total = ((d + 3) / 4) * BE;
       for (int idx = threadIdx.x; idx < kq_total;
            idx += (int)blockDim.x) {
-        const int kk0 = (idx / ESB2_BE) * 4, ei = idx % ESB2_BE;
+        const int kk0 = (idx / BE) * 4, ei = idx % BE;
         __bf16 v[4];
-        // kk0 is wave-uniform (ESB2_BE >= WAVE): full quads take the
+        // kk0 is wave-uniform (BE >= WAVE): full quads take the
         // branchless path — per-element ternaries around loads would
         // serialize them behind vmcnt(0) (guide trap 4c, measured -15%)
-        if (cnt == ESB2_BE && kk0 + 3 < d) {
+        if (cnt == BE && kk0 + 3 < d) {
           const __hip_bfloat16* zp = z + (int64_t)kk0 * n + e0 + ei;
 #pragma unroll
           for (int u = 0; u < 4; ++u) v[u] = (__bf16)(zp[(int64_t)u * n]);
@@ -2243,9 +2242,9 @@ estep_logw_big2_kernel(const __hip_bfloat16* __restrict__ z,
         *(uint2*)(zs + ei * AROW + kk0) = *(uint2*)v;
       }
       const int kpad0 = ((d + 3) / 4) * 4;
-      for (int idx = threadIdx.x; idx < (COLS - kpad0) * ESB2_BE;
+      for (int idx = threadIdx.x; idx < (COLS - kpad0) * BE;
            idx += (int)blockDim.x) {
-        const int kk = kpad0 + idx / ESB2_BE, ei = idx % ESB2_BE;
+        const int kk = kpad0 + idx / BE, ei = idx % BE;
         // d is only >= kpad0 when d % 4 == 0: the ones row then lives here
         zs[ei * AROW + kk] =
             (__bf16)((kk == d && ei < cnt) ? 1.0f : 0.0f);
@@ -2253,17 +2252,33 @@ estep_logw_big2_kernel(const __hip_bfloat16* __restrict__ z,
     }
     __syncthreads();
 
-    // (row-tile, event-32-tile) pairs split across the 16 waves
-    for (int pair = wave; pair < rt_n * (ESB2_BE / 32);
-         pair += (int)(blockDim.x / WAVE)) {
-      const int rt = pair / (ESB2_BE / 32);
-      const int t = pair % (ESB2_BE / 32);
+    // (row-tile, event-32-tile) tasks. Task cost falls with rt (the
+    // zero-tile skip below), so enumerate tasks rt-DESCENDING and deal
+    // them to waves in a zigzag: each wave's summed cost is near-uniform
+    // and the post-loop barrier waits for the least-idle wave.
+    const int etiles = BE / 32;
+    const int T = rt_n * etiles;
+    const int W = (int)(blockDim.x / WAVE);
+    const int kaug = d >> 4;  // column-tile holding the augmented column
+    for (int s2 = 0;; ++s2) {
+      const int idx =
+          (s2 & 1) ? ((s2 + 1) * W - 1 - wave) : (s2 * W + wave);
+      if (idx >= T) break;
+      const int rt = rt_n - 1 - idx / etiles;
+      const int t = idx % etiles;
+      // F = [chol(R)^-1 | -F mu] is lower-triangular + one augmented
+      // column: row-tile rt's live column-tiles are 0..(64rt+31)/16 and
+      // the augmented tile; the rest of the padded table is zeros
+      // (emit_mfac) and MFMA accumulation of zeros is bitwise a no-op,
+      // so skipping those tiles is exact
+      const int kcmax = min(KCT - 1, 2 * rt + 1);
       const __bf16* arow = ah + (rt * 32 + j32) * AROW;
       const __bf16* lrow = al + (rt * 32 + j32) * AROW;
       const __bf16* zrow = zs + (t * 32 + j32) * AROW;
       f32x16 y = (f32x16)(0.0f);
 #pragma unroll
       for (int kc = 0; kc < KCT; ++kc) {
+        if (kc > kcmax && kc != kaug) continue;
         const bf16x8 b = *(const bf16x8*)(zrow + kc * 16 + 8 * g2);
         const bf16x8 fa = *(const bf16x8*)(arow + kc * 16 + 8 * g2);
         const bf16x8 fl = *(const bf16x8*)(lrow + kc * 16 + 8 * g2);
@@ -2274,14 +2289,14 @@ estep_logw_big2_kernel(const __hip_bfloat16* __restrict__ z,
 #pragma unroll
       for (int r = 0; r < 16; ++r) s = fmaf(y[r], y[r], s);
       s += __shfl_xor(s, 32, WAVE);
-      if (lane < 32) qpart[rt * ESB2_BE + t * 32 + j32] = s;
+      if (lane < 32) qpart[rt * BE + t * 32 + j32] = s;
     }
     __syncthreads();
     // combine row-tile partials (fixed order) and write logw
-    if (threadIdx.x < ESB2_BE && threadIdx.x < cnt) {
+    if (threadIdx.x < BE && threadIdx.x < cnt) {
       const int e = threadIdx.x;
       float q = qpart[e];
-      for (int rt = 1; rt < rt_n; ++rt) q += qpart[rt * ESB2_BE + e];
+      for (int rt = 1; rt < rt_n; ++rt) q += qpart[rt * BE + e];
       logw[(int64_t)c * n + e0 + e] = -0.5f * q + addc;
     }
   }
@@ -2346,17 +2361,26 @@ estep_logw_big_f32_kernel(const float* __restrict__ z,
 
   for (int rt = 0; rt < rt_n; ++rt) {
     // A values for this row-tile: lane j32 holds row rt*32+j32, k-slots
-    // 2*ch + g2 (v_mfma_f32_32x32x2_f32 operand map, guide §3)
+    // 2*ch + g2 (v_mfma_f32_32x32x2_f32 operand map, guide §3).
+    // F is lower-triangular + one augmented column: k-chunks past
+    // chmax hold only zeros (emit_mfac pads), except the augmented
+    // chunk — skipping them is exact (acc + 0*b is a bitwise no-op)
+    // and cuts ~43% of the rate-limiting f32 MFMAs at D=128. rt is
+    // wave-uniform, so the branches do not diverge.
+    const int chmax = min(KCT * 8 - 1, 16 * rt + 15);
+    const int chaug = d >> 1;  // k-chunk holding the augmented column
     float a[KCT * 8];
     const float* arow = mf + (rt * 32 + j32) * cols;
 #pragma unroll
-    for (int ch = 0; ch < KCT * 8; ++ch) a[ch] = arow[2 * ch + g2];
+    for (int ch = 0; ch < KCT * 8; ++ch)
+      a[ch] = (ch <= chmax || ch == chaug) ? arow[2 * ch + g2] : 0.0f;
 #pragma unroll
     for (int t = 0; t < ESBF_BE / 32; ++t) {
       const float* zrow = zs + (t * 32 + j32) * ZR;
       f32x16 y = (f32x16)(0.0f);
 #pragma unroll
       for (int ch = 0; ch < KCT * 8; ++ch) {
+        if (ch > chmax && ch != chaug) continue;
         const float b = zrow[2 * ch + g2];
         y = __builtin_amdgcn_mfma_f32_32x32x2f32(a[ch], b, y, 0, 0, 0);
       }
