@@ -1831,6 +1831,17 @@ mstep_moments_big_kernel(const float* __restrict__ x,
   const int rt2 = (dp + 31) / 32;
   const int rows = rt2 * 32;
   const int tp = rt2 * (rt2 + 1) / 2;
+  // Augmented-row split: when D % 32 == 0 the ones row (index d) is the
+  // ONLY live row of the last row-tile, so the rt2-1 tile-pairs touching
+  // it run full MFMA just to produce T[d, :] = [sum w z | sum w] (the
+  // mean numerators and N). Drop those pairs from the MFMA set (at
+  // D=128: 15 -> 10, one third of the MFMA + A-build work) and let the
+  // waves they would have occupied accumulate that row directly from
+  // the staged LDS planes with plain FMA (lane = event slot:
+  // conflict-free, and far off the pair-waves' critical path).
+  const bool aug_split = (d & 31) == 0;
+  const int tp_mfma = aug_split ? (rt2 - 1) * rt2 / 2 : tp;
+  const int pair_groups = (tp_mfma + MBB_PMAX - 1) / MBB_PMAX;
   const int zbr = MBB_BK + 8;  // bf16 row stride
   // DOUBLE-BUFFERED LDS (ablation: single-buffer T14 still exposed
   // ~37% of the kernel as staging+barrier time — the write pass sat
@@ -1852,19 +1863,26 @@ mstep_moments_big_kernel(const float* __restrict__ x,
   const int c = blockIdx.x * MBB_CPB + cw;
   const int chunk = blockIdx.y;
   const int p_lo = group * MBB_PMAX;
-  const int p_hi = min(tp, p_lo + MBB_PMAX);
+  const int p_hi = min(tp_mfma, p_lo + MBB_PMAX);
+  const bool is_aug = aug_split && group >= pair_groups;
+  // lane <-> dim mapping: each aug lane owns ONE dim of T[d, :] and
+  // accumulates over every event serially (one scalar accumulator —
+  // a per-lane dim ARRAY spilled 7 VGPRs and taxed every wave)
+  const int augdim = (group - pair_groups) * WAVE + lane;
   // hoisted pair->tile mapping (tri_row_col has a sqrtf: keep it out of
   // the chunk loop)
   int ptr[MBB_PMAX], ptc[MBB_PMAX];
 #pragma unroll
   for (int pp = 0; pp < MBB_PMAX; ++pp) {
-    if (p_lo + pp < tp) tri_row_col(p_lo + pp, &ptr[pp], &ptc[pp]);
+    if (p_lo + pp < tp_mfma) tri_row_col(p_lo + pp, &ptr[pp], &ptc[pp]);
     else { ptr[pp] = 0; ptc[pp] = 0; }
   }
 
   f32x16 acc[MBB_PMAX];
 #pragma unroll
   for (int pp = 0; pp < MBB_PMAX; ++pp) acc[pp] = (f32x16)(0.0f);
+
+  float acc_aug = 0.0f;  // this lane's running sum for T[d, augdim]
 
   const int64_t tiles = (n + MBB_BK - 1) / MBB_BK;
   const int64_t my_tiles =
@@ -1985,6 +2003,7 @@ mstep_moments_big_kernel(const float* __restrict__ x,
     const __bf16* zlo = zhi + zplane;
     const float* wt = wbuf + cur * MBB_CPB * MBB_BK;
 
+    if (p_lo < p_hi) {
 #pragma unroll
     for (int ch = 0; ch < MBB_BK / 16; ++ch) {
       const int eb = ch * 16 + 8 * g2;
@@ -2026,6 +2045,21 @@ mstep_moments_big_kernel(const float* __restrict__ x,
             __builtin_amdgcn_mfma_f32_32x32x16_bf16(a_lo, b_hi, acc[pp], 0, 0, 0);
       }
     }
+    } else if (is_aug && augdim < dp) {
+      // T[d, augdim] = sum_e w_e z[augdim, e]: serial event loop from
+      // the staged planes (2-way ILP; fixed order -> deterministic).
+      // These waves sit well off the pair-waves' critical path.
+      const __bf16* zh = zhi + augdim * zbr;
+      const __bf16* zl = zlo + augdim * zbr;
+      const float* wrow = wt + cw * MBB_BK;
+      float a0 = 0.0f, a1 = 0.0f;
+#pragma unroll 8
+      for (int e = 0; e < MBB_BK; e += 2) {
+        a0 = fmaf(wrow[e], (float)zh[e] + (float)zl[e], a0);
+        a1 = fmaf(wrow[e + 1], (float)zh[e + 1] + (float)zl[e + 1], a1);
+      }
+      acc_aug += a0 + a1;
+    }
     // write the NEXT tile into the other buffer; single barrier per
     // tile (nobody reads that buffer this iteration)
     if (ti + 1 < my_tiles) write_buf(cur ^ 1);
@@ -2048,6 +2082,8 @@ mstep_moments_big_kernel(const float* __restrict__ x,
         out[gi * (gi + 1) / 2 + gj] = acc[pp][r];
     }
   }
+  if (is_aug && augdim < dp)
+    out[d * (d + 1) / 2 + augdim] = acc_aug;  // packed row d = T[d, :]
 }
 
 // ---------------------------------------------------------------------------
