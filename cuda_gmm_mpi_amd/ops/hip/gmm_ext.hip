@@ -428,8 +428,10 @@ void estep_logw_big(torch::Tensor z, torch::Tensor mfac, torch::Tensor add,
   const int64_t tiles = (n + be - 1) / be;
   // enough blocks to fill the chip; each block amortizes one staged
   // factor table over tiles/nchunk z tiles
-  const int nchunk =
+  int nchunk =
       (int)std::min<int64_t>(tiles, std::max<int64_t>(1, 2048 / k));
+  if (const char* e = std::getenv("GMM_BIG2_NCHUNK"))
+    nchunk = (int)std::min<int64_t>(tiles, std::max(1, atoi(e)));
   dim3 grid((uint32_t)nchunk, k);
   // block size matched to the (row-tile x event-tile) task count so no
   // waves idle at small rt_n
