@@ -103,6 +103,44 @@ def test_results_match_fixture(cli_output):
 
 
 @pytest.fixture(scope="module")
+def cli_output_centered(tmp_path_factory):
+    """Same run WITH the default internal centering: the oracle never
+    centers, so matching it pins the centering redesign as exactly
+    translation-invariant end to end (engine.py deliberate-redesign #1)."""
+    out = str(tmp_path_factory.mktemp("golden_c") / "o")
+    rc = main(["3", os.path.join(FIX, "golden_small.bin"), out, "2",
+               "--device", "cpu"])
+    assert rc == 0
+    return out
+
+
+def test_centered_run_matches_uncentered_fixture(cli_output_centered):
+    got = parse_summary(open(cli_output_centered + ".summary").read())
+    want = parse_summary(open(os.path.join(FIX, "golden_small.summary")).read())
+    assert len(got) == len(want) == 2
+    order = []
+    for wcl in want:
+        d = [np.linalg.norm(wcl["means"] - g["means"]) for g in got]
+        order.append(int(np.argmin(d)))
+    assert sorted(order) == [0, 1]
+    for wcl, gi in zip(want, order):
+        gcl = got[gi]
+        assert gcl["pi"] == pytest.approx(wcl["pi"], abs=2e-3)
+        assert gcl["N"] == pytest.approx(wcl["N"], rel=2e-3)
+        np.testing.assert_allclose(gcl["means"], wcl["means"],
+                                   rtol=2e-3, atol=0.5)
+        scale = np.abs(wcl["R"]).max()
+        np.testing.assert_allclose(gcl["R"], wcl["R"],
+                                   rtol=5e-2, atol=5e-2 * scale)
+    # .results data halves byte-identical (de-centering restores input)
+    got_lines = open(cli_output_centered + ".results").read().splitlines()
+    want_lines = open(os.path.join(FIX,
+                                   "golden_small.results")).read().splitlines()
+    for g, w in zip(got_lines, want_lines):
+        assert g.split("\t")[0] == w.split("\t")[0]
+
+
+@pytest.fixture(scope="module")
 def cli_output_diag(tmp_path_factory):
     out = str(tmp_path_factory.mktemp("golden_diag") / "o")
     rc = main(["2", os.path.join(FIX, "golden_diag.bin"), out, "2",
