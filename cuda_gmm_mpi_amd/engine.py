@@ -80,7 +80,7 @@ class EmEngine:
 
         k0 = config.num_clusters
         # MFMA E-step paths (D <= 31: fused online-softmax, bf16 or
-        # exact-f32, any K; 31 < D <= 143: big-D logw, bf16 or exact-f32);
+        # exact-f32, any K; 31 < D <= 142: big-D logw, bf16 or exact-f32);
         # otherwise the VALU kernels. DIAG_ONLY routes through the same
         # factor path: a diagonal Rinv yields a diagonal Cholesky factor,
         # so q = ||Uz+u0||^2 equals the diagonal quadratic form exactly.

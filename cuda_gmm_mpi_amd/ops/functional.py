@@ -147,8 +147,33 @@ def constants(r: torch.Tensor, means: torch.Tensor | None = None,
     emits the fused-E-step factor M = [U | -U mu] (U^T U = Rinv) as bf16
     hi/lo fragment pairs — requires ``means`` and D <= 31.
     """
+    k, d, _ = r.shape
+    if r.is_cuda and d > 142:
+        # the constants/emission kernels stage TWO d x (d|1) fp32 planes
+        # + a d-float mu buffer in LDS (race-free LU snapshot /
+        # triangular-inverse working set); that crosses gfx950's 160 KB
+        # workgroup limit at exactly D = 143. Fall back to the CPU LU
+        # (bit-faithful invert_cpu math) — the MFMA factor paths are
+        # gated to D <= 142, so no caller needs mfac here.
+        if mfac is not None and mfac.numel() > 0:
+            raise ValueError("factor emission needs D <= 142 (LDS bound)")
+        rinv_c, const_c = cpu.compute_constants(r.detach().cpu(), diag_only)
+        rinv = out[0] if out is not None else torch.empty_like(r)
+        rinv.copy_(rinv_c.to(r.device))
+        const = const_c.to(r.device)
+        if pi_add is not None:
+            pi_t, add_t = pi_add
+            const_t = (out[1] if out is not None
+                       else torch.empty(k, dtype=torch.float32,
+                                        device=r.device))
+            const_t.copy_(const)
+            add_t.copy_(const + torch.log(pi_t))
+            return rinv, const_t
+        if out is not None:
+            out[1].copy_(const)
+            return rinv, out[1]
+        return rinv, const
     if r.is_cuda:
-        k, d, _ = r.shape
         rinv = out[0] if out is not None else torch.empty_like(r)
         logdet = torch.empty(k, dtype=torch.float32, device=r.device)
         if means is None:
@@ -276,9 +301,10 @@ def estep_fused_f32(z: torch.Tensor, mfac32: torch.Tensor, add: torch.Tensor,
 
 
 def estep_big_available(device: torch.device, dtype: str, d: int) -> bool:
-    """Big-D MFMA logw path gate (D <= 143, bf16 or exact-f32 MFMA). The
+    """Big-D MFMA logw path gate (D <= 142, bf16 or exact-f32 MFMA; the
+    factor-emission LDS working set crosses 160 KB at D = 143). The
     engine prefers the fused kernel for D <= 31."""
-    return device.type == "cuda" and dtype in ("bf16", "fp32") and 1 <= d <= 143
+    return device.type == "cuda" and dtype in ("bf16", "fp32") and 1 <= d <= 142
 
 
 def mfac_shape(d: int) -> tuple[int, int, int]:

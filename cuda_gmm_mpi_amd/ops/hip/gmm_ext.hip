@@ -215,8 +215,8 @@ void constants(torch::Tensor r, torch::Tensor means, torch::Tensor pi,
     TORCH_CHECK(mfac.is_cuda() && mfac.is_contiguous() &&
                     mfac.scalar_type() == torch::kBFloat16 &&
                     mfac.numel() == (int64_t)k * 2 * rows * kct * 16 &&
-                    d <= 143,
-                "mfac must be bf16 [K,2,RT*32,KCT*16] with D <= 143");
+                    d <= 142,
+                "mfac must be bf16 [K,2,RT*32,KCT*16] with D <= 142");
     mp = reinterpret_cast<__hip_bfloat16*>(mfac.data_ptr());
   }
   auto s = stream();
@@ -410,7 +410,7 @@ void estep_logw_big(torch::Tensor z, torch::Tensor mfac, torch::Tensor add,
   const int d = (int)z.size(0);
   const int64_t n = z.size(1);
   const int k = (int)add.size(0);
-  TORCH_CHECK(d >= 1 && d <= 143, "estep_logw_big supports D <= 143");
+  TORCH_CHECK(d >= 1 && d <= 142, "estep_logw_big supports D <= 143");
   TORCH_CHECK(logw.size(0) == k && logw.size(1) == n, "logw shape");
   const int kct =
       d + 1 <= 32 ? 2 : d + 1 <= 48 ? 3 : d + 1 <= 80 ? 5 : 9;
@@ -473,7 +473,7 @@ void estep_logw_big_f32(torch::Tensor z, torch::Tensor mfac32,
   const int d = (int)z.size(0);
   const int64_t n = z.size(1);
   const int k = (int)add.size(0);
-  TORCH_CHECK(d >= 1 && d <= 143, "estep_logw_big_f32 supports D <= 143");
+  TORCH_CHECK(d >= 1 && d <= 142, "estep_logw_big_f32 supports D <= 143");
   TORCH_CHECK(logw.size(0) == k && logw.size(1) == n, "logw shape");
   const int kct =
       d + 1 <= 32 ? 2 : d + 1 <= 48 ? 3 : d + 1 <= 80 ? 5 : 9;
@@ -726,9 +726,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("constants", &constants,
         "batched no-pivot LU inverse + ln|det| + bf16 Cholesky factors");
   m.def("estep_logw_big", &estep_logw_big,
-        "big-D MFMA log-weights (31 < D <= 143)");
+        "big-D MFMA log-weights (31 < D <= 142)");
   m.def("estep_logw_big_f32", &estep_logw_big_f32,
-        "exact-f32 big-D MFMA log-weights (31 < D <= 143)");
+        "exact-f32 big-D MFMA log-weights (31 < D <= 142)");
   m.def("mstep_moments_big", &mstep_moments_big,
         "big-D split-precision moments");
   m.def("mstep_moments_b16", &mstep_moments_b16,

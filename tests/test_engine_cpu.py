@@ -264,8 +264,10 @@ def test_fused_gate_boundaries():
     assert not F.estep_fused_available(dev, "bf16", 32, 8)   # D > 31
     assert F.estep_big_available(dev, "bf16", 32)
     assert F.estep_big_available(dev, "bf16", 24)
-    assert F.estep_big_available(dev, "bf16", 143)
-    assert not F.estep_big_available(dev, "bf16", 144)
+    assert F.estep_big_available(dev, "bf16", 142)
+    # D >= 143: the factor-emission LDS working set crosses gfx950's
+    # 160 KB limit -> constants() falls back to CPU, E-step goes VALU
+    assert not F.estep_big_available(dev, "bf16", 143)
     assert F.estep_big_available(dev, "fp32", 64)  # exact-f32 MFMA tier
     assert F.estep_fused_available(dev, "fp32", 24, 512)
 

@@ -334,7 +334,7 @@ def test_mstep_moments_b16_matches_cpu(device, d, n):
                                rtol=3e-4, atol=3e-4 * scale)
 
 
-@pytest.mark.parametrize("d", [40, 64, 100, 128, 143])
+@pytest.mark.parametrize("d", [40, 64, 100, 128, 142])
 def test_estep_logw_big_matches_cpu(device, d):
     """Big-D MFMA logw vs fp32 torch reference (bf16-class tolerance)."""
     from cuda_gmm_mpi_amd.ops import functional as F
