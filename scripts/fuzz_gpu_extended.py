@@ -39,10 +39,11 @@ for trial in range(60):
 print(f"moments fuzz: 60 trials, {fails} failures", flush=True)
 
 efails = 0
+BIG_D = {20: 96, 22: 128, 24: 145}  # dedicated big-D engine trials
 for trial in range(25):
-    d = int(rng.integers(2, 40))
+    d = BIG_D.get(trial, int(rng.integers(2, 40)))
     k = int(rng.integers(2, 14))
-    n = int(rng.integers(600, 20000))
+    n = int(rng.integers(600, 20000)) if trial not in BIG_D else 4000
     data, _ = make_blobs(n, d, max(2, k // 2), seed=int(rng.integers(1e6)))
     iters = int(rng.integers(1, 7))
     ed = "bf16" if trial % 2 else "fp32"
