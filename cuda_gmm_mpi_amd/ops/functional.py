@@ -105,6 +105,18 @@ def mstep_covariance_s(x: torch.Tensor, w: torch.Tensor,
     deterministic per-chunk packed partials, summed + unpacked here; CPU:
     batched torch matmuls.
     """
+    if x.is_cuda and x.shape[0] > 128:
+        # past the covariance kernel's D cap: per-cluster rocBLAS GEMMs
+        # (the exact-fp32 D>128 quadrant; ~K GEMMs of D x N x D)
+        k, n = w.shape
+        d = x.shape[0]
+        if lse is not None and lse.numel() > 0:
+            w = torch.exp(w - lse.unsqueeze(0))
+        s = out if out is not None else torch.empty(
+            (k, d, d), dtype=torch.float32, device=x.device)
+        for c in range(k):
+            s[c] = (x * w[c].unsqueeze(0)) @ x.T
+        return s
     if x.is_cuda:
         k = w.shape[0]
         d = x.shape[0]

@@ -278,3 +278,23 @@ def test_gpu_em_likelihood_monotone_fp32():
         liks.append(float(eng._lik_dev.item()))
     diffs = np.diff(np.array(liks))
     assert np.all(diffs >= -1e-6 * abs(liks[-1])), diffs
+
+
+@pytest.mark.parametrize("d,ed", [(140, "fp32"), (145, "fp32"), (145, "bf16")])
+def test_engine_past_factor_dcap_matches_cpu(d, ed):
+    """D in (128, 142]: fp32 MFMA E-step + rocBLAS covariance GEMMs;
+    D > 142: VALU E-step + CPU-LU constants (the factor-emission LDS
+    working set crosses gfx950's 160 KB at D=143) + GEMM covariance.
+    Both quadrants must still match the CPU golden path."""
+    data, _ = make_blobs(3000, d, 3, seed=d)
+    cfg = GmmConfig(num_clusters=3, target_num_clusters=3,
+                    min_iters=4, max_iters=4, estep_dtype=ed,
+                    mstep_precision="bf16x3" if ed == "bf16" else "fp32")
+    eng_g, lik_g = run("cuda", cfg, data)
+    cfg_c = GmmConfig(num_clusters=3, target_num_clusters=3,
+                      min_iters=4, max_iters=4)
+    eng_c, lik_c = run("cpu", cfg_c, data)
+    assert np.isfinite(lik_g)
+    assert lik_g == pytest.approx(lik_c, rel=5e-3 if ed == "bf16" else 1e-4)
+    np.testing.assert_allclose(eng_g.state.N.cpu().numpy(),
+                               eng_c.state.N.numpy(), rtol=2e-2)
