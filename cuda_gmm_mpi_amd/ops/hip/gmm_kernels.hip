@@ -1461,12 +1461,18 @@ estep_fused_f32_kernel(const float* __restrict__ z,
     for (int ch = 0; ch < 16; ++ch) a[ch] = nx_a[ch];
     const float addc = nx_add;
     if (c + nwaves < k) load_a(c + nwaves);
+    // columns past the augmented one (index d) are zero padding and
+    // f32 MFMA accumulation of zeros is bitwise a no-op: skipping them
+    // is exact and cuts the 16-deep dependent MFMA chain to d/2+1
+    // (13 at D=24). d is kernel-uniform, so no divergence.
+    const int chmax = d >> 1;
 #pragma unroll 2
     for (int t = 0; t < EST_BE / 32; ++t) {
       const float* zrow = zs + (t * 32 + j32) * ZR;
       f32x16 y = (f32x16)(0.0f);
 #pragma unroll
       for (int ch = 0; ch < 16; ++ch) {
+        if (ch > chmax) continue;
         const float b = zrow[2 * ch + g2];
         y = __builtin_amdgcn_mfma_f32_32x32x2f32(a[ch], b, y, 0, 0, 0);
       }
@@ -1762,6 +1768,7 @@ estep_fused_f32_lds_kernel(const float* __restrict__ z,
       f32x16 y = (f32x16)(0.0f);
 #pragma unroll
       for (int ch = 0; ch < 16; ++ch) {
+        if (ch > (d >> 1)) continue;  // zero padding past the aug column: exact skip
         const float b = zrow[2 * ch + g2];
         y = __builtin_amdgcn_mfma_f32_32x32x2f32(a[ch], b, y, 0, 0, 0);
       }
