@@ -334,7 +334,7 @@ def test_mstep_moments_b16_matches_cpu(device, d, n):
                                rtol=3e-4, atol=3e-4 * scale)
 
 
-@pytest.mark.parametrize("d", [40, 64, 100, 128, 142])
+@pytest.mark.parametrize("d", [32, 33, 40, 64, 100, 128, 142])
 def test_estep_logw_big_matches_cpu(device, d):
     """Big-D MFMA logw vs fp32 torch reference (bf16-class tolerance)."""
     from cuda_gmm_mpi_amd.ops import functional as F
@@ -357,8 +357,8 @@ def test_estep_logw_big_matches_cpu(device, d):
                                rtol=5e-2, atol=2.0)
 
 
-@pytest.mark.parametrize("d,n", [(40, 5000), (64, 3001), (128, 2000),
-                                 (143, 1500)])
+@pytest.mark.parametrize("d,n", [(32, 4000), (40, 5000), (64, 3001),
+                                 (128, 2000), (143, 1500)])
 def test_mstep_moments_big_matches_cpu(device, d, n):
     from cuda_gmm_mpi_amd.ops import functional as F
     rng = np.random.default_rng(d * 7 + 1)
