@@ -161,10 +161,21 @@ class EmEngine:
             return
         from .ops.backend import hip_ext
         st = self.state.shrink(k)
+        r_src = st.R
+        if self.cfg.diag_only:
+            # DIAG parity through a merge: the reference's diag E-step
+            # reads diag(inv(R_merged_full)) — the host merge computed
+            # the FULL-matrix inverse and quirk #8 carries it into the
+            # next E-step — which is NOT inv(diag(R_full)). Emit the
+            # factors from the diagonal matrix with exactly that
+            # inverse; for an already-diagonal R (seed, resume, every
+            # iteration after the first M-step) this is identical to R.
+            d_inv = st.Rinv.diagonal(dim1=-2, dim2=-1)
+            r_src = torch.diag_embed(1.0 / d_inv)
         empty_b = torch.empty(0, dtype=torch.bfloat16, device=self.device)
         empty_f = torch.empty(0, dtype=torch.float32, device=self.device)
         hip_ext().emit_factors(
-            st.R.contiguous(), st.means.contiguous(),
+            r_src.contiguous(), st.means.contiguous(),
             self.mfac[:k] if self.mfac is not None else empty_b,
             self.mfac32[:k] if self.mfac32 is not None else empty_f,
             empty_f, empty_f, empty_f,  # constants NOT recomputed (quirk #8)

@@ -298,3 +298,28 @@ def test_engine_past_factor_dcap_matches_cpu(d, ed):
     assert lik_g == pytest.approx(lik_c, rel=5e-3 if ed == "bf16" else 1e-4)
     np.testing.assert_allclose(eng_g.state.N.cpu().numpy(),
                                eng_c.state.N.numpy(), rtol=2e-2)
+
+
+def test_diag_sweep_with_merge_matches_cpu():
+    """DIAG_ONLY through an MDL merge: the first post-merge E-step must
+    use diag(inv(R_merged_full)) — the host merge's full inverse, carried
+    by quirk #8 — not the full quadratic form of chol(R_merged). The GPU
+    factor path and the CPU golden path must take the same trajectory."""
+    data, _ = make_blobs(6000, 4, 3, seed=77)
+    for ed in ("fp32", "bf16"):
+        cfg = GmmConfig(num_clusters=5, target_num_clusters=2,
+                        min_iters=6, max_iters=6, diag_only=True,
+                        estep_dtype=ed)
+        eng_g = build_engine(data, cfg, device="cuda")
+        res_g = eng_g.sweep()
+        cfg_c = GmmConfig(num_clusters=5, target_num_clusters=2,
+                          min_iters=6, max_iters=6, diag_only=True)
+        eng_c = build_engine(data, cfg_c, device="cpu")
+        res_c = eng_c.sweep()
+        assert res_g.num_clusters == res_c.num_clusters
+        assert res_g.min_rissanen == pytest.approx(
+            res_c.min_rissanen, rel=5e-3 if ed == "bf16" else 1e-4)
+        assert res_g.rissanen_by_k.keys() == res_c.rissanen_by_k.keys()
+        for kk in res_g.rissanen_by_k:
+            assert res_g.rissanen_by_k[kk] == pytest.approx(
+                res_c.rissanen_by_k[kk], rel=1e-2 if ed == "bf16" else 1e-3)
