@@ -407,6 +407,31 @@ def main():
     write_results(os.path.join(OUT, "golden_k1.results"), data3, w3)
     print(f"k1 fixture written: likelihood={lik3:.4f}")
 
+    # ---- multi-merge fixture: K0=6 -> target 2 (four passes through
+    # empty-elimination + pair merge + the quirk-#8 constant carry) ----
+    rng = np.random.default_rng(90210)
+    n4, d4 = 800, 2
+    centers4 = np.array([[150.0, 400.0], [420.0, 150.0],
+                         [500.0, 480.0], [220.0, 200.0]])
+    rows4 = [rng.normal(c, 14.0, size=(200, d4)) for c in centers4]
+    data4 = np.vstack(rows4).astype(F32)[rng.permutation(n4)]
+    with open(os.path.join(OUT, "golden_multi.bin"), "wb") as f:
+        f.write(struct.pack("<ii", n4, d4))
+        f.write(data4.tobytes())
+    dt4 = data4.T.astype(np.float64)
+    c4 = seed(data4, 6)
+    constants(c4)
+    kk = 6
+    while True:
+        w4, lik4 = run_em(dt4, c4, iters=100)
+        if kk <= 2:
+            break
+        c4 = merge_step(c4)
+        kk = c4.k
+    write_summary(os.path.join(OUT, "golden_multi.summary"), c4)
+    write_results(os.path.join(OUT, "golden_multi.results"), data4, w4)
+    print(f"multi-merge fixture written: K={kk} likelihood={lik4:.4f}")
+
 
 if __name__ == "__main__":
     sys.exit(main())

@@ -223,6 +223,59 @@ def test_k1_fixture_results(cli_output_k1):
         assert g.split("\t")[1] == w.split("\t")[1] == "1.000000"
 
 
+@pytest.fixture(scope="module")
+def cli_output_multi(tmp_path_factory):
+    out = str(tmp_path_factory.mktemp("golden_multi") / "o")
+    rc = main(["6", os.path.join(FIX, "golden_multi.bin"), out, "2",
+               "--device", "cpu", "--no-center"])
+    assert rc == 0
+    return out
+
+
+def test_multi_merge_fixture_summary(cli_output_multi):
+    """K0=6 -> 2: four merge/elimination passes with the quirk-#8
+    constant carry between each; the CLI must land on the same final
+    two clusters as the float64 oracle."""
+    got = parse_summary(open(cli_output_multi + ".summary").read())
+    want = parse_summary(open(os.path.join(FIX,
+                                           "golden_multi.summary")).read())
+    assert len(got) == len(want) == 2
+    order = []
+    for wcl in want:
+        d = [np.linalg.norm(wcl["means"] - g["means"]) for g in got]
+        order.append(int(np.argmin(d)))
+    assert sorted(order) == [0, 1]
+    for wcl, gi in zip(want, order):
+        gcl = got[gi]
+        assert gcl["pi"] == pytest.approx(wcl["pi"], abs=5e-3)
+        assert gcl["N"] == pytest.approx(wcl["N"], rel=5e-3)
+        np.testing.assert_allclose(gcl["means"], wcl["means"],
+                                   rtol=5e-3, atol=1.0)
+        scale = np.abs(wcl["R"]).max()
+        np.testing.assert_allclose(gcl["R"], wcl["R"],
+                                   rtol=1e-1, atol=1e-1 * scale)
+
+
+def test_multi_merge_fixture_results(cli_output_multi):
+    got_lines = open(cli_output_multi + ".results").read().splitlines()
+    want_lines = open(os.path.join(FIX,
+                                   "golden_multi.results")).read().splitlines()
+    assert len(got_lines) == len(want_lines) == 800
+    for g, w in zip(got_lines, want_lines):
+        assert g.split("\t")[0] == w.split("\t")[0]
+    gw = np.array([[float(v) for v in ln.split("\t")[1].split(",")]
+                   for ln in got_lines])
+    ww = np.array([[float(v) for v in ln.split("\t")[1].split(",")]
+                   for ln in want_lines])
+    got_sum = parse_summary(open(cli_output_multi + ".summary").read())
+    want_sum = parse_summary(
+        open(os.path.join(FIX, "golden_multi.summary")).read())
+    order = [int(np.argmin([np.linalg.norm(w["means"] - g["means"])
+                            for g in got_sum])) for w in want_sum]
+    gw = gw[:, order]
+    assert (gw.argmax(axis=1) == ww.argmax(axis=1)).all()
+
+
 def test_fixture_generator_reproduces_committed_bytes(tmp_path):
     """The committed fixtures are exactly what the (deterministic)
     transcription generator produces — guards both against accidental
@@ -241,7 +294,8 @@ def test_fixture_generator_reproduces_committed_bytes(tmp_path):
                  "golden_small.results", "golden_diag.bin",
                  "golden_diag.summary", "golden_diag.results",
                  "golden_k1.bin", "golden_k1.summary",
-                 "golden_k1.results"):
+                 "golden_k1.results", "golden_multi.bin",
+                 "golden_multi.summary", "golden_multi.results"):
         got = open(os.path.join(tmp_path, name), "rb").read()
         want = open(os.path.join(FIX, name), "rb").read()
         assert got == want, f"fixture drift: {name}"
