@@ -276,6 +276,31 @@ def test_multi_merge_fixture_results(cli_output_multi):
     assert (gw.argmax(axis=1) == ww.argmax(axis=1)).all()
 
 
+def test_csv_input_byte_matches_bin_run(tmp_path):
+    """The same dataset fed as CSV (header line + comma rows, the
+    readData.cpp:49-129 format) must produce byte-identical .summary and
+    .results to the .bin run — 9-significant-digit text round-trips
+    fp32 exactly, so the parsed events are bitwise the same."""
+    import struct
+    raw = open(os.path.join(FIX, "golden_small.bin"), "rb").read()
+    n, d = struct.unpack("<ii", raw[:8])
+    data = np.frombuffer(raw[8:], dtype=np.float32).reshape(n, d)
+    csv = tmp_path / "g.csv"
+    with open(csv, "w") as f:
+        f.write(",".join(f"c{i}" for i in range(d)) + "\n")  # header row
+        for row in data:
+            f.write(",".join(f"{v:.9g}" for v in row) + "\n")
+    out_bin = str(tmp_path / "ob")
+    out_csv = str(tmp_path / "oc")
+    assert main(["3", os.path.join(FIX, "golden_small.bin"), out_bin, "2",
+                 "--device", "cpu", "--no-center"]) == 0
+    assert main(["3", str(csv), out_csv, "2",
+                 "--device", "cpu", "--no-center"]) == 0
+    for suf in (".summary", ".results"):
+        assert open(out_csv + suf, "rb").read() == \
+            open(out_bin + suf, "rb").read(), f"CSV vs BIN drift in {suf}"
+
+
 def test_fixture_generator_reproduces_committed_bytes(tmp_path):
     """The committed fixtures are exactly what the (deterministic)
     transcription generator produces — guards both against accidental
