@@ -10,6 +10,8 @@ custom kernels for the fused hot ops).
 """
 from __future__ import annotations
 
+import os
+
 import torch
 
 from . import cpu_reference as cpu
@@ -391,6 +393,8 @@ def mstep_moments(x: torch.Tensor, w: torch.Tensor,
         return out
     if x.is_cuda and precision == "bf16x3" and d <= 159:
         tiles = (n + 63) // 64  # MBB_BK
+        if nchunk is None and os.environ.get("GMM_MOMENTS_NCHUNK"):
+            nchunk = int(os.environ["GMM_MOMENTS_NCHUNK"])
         if nchunk is None:
             # big-D packed rows are large (Pp ~ 8k at D=128): a small byte
             # cap starves chunk-parallelism (7 chunks = 448 blocks at
