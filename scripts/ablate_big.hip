@@ -1,5 +1,10 @@
 // Standalone ablation harness for mstep_moments_big_kernel (round 2,
 // T14 form): attributes the kernel's cycles by removing pieces.
+// NOTE: this models the SINGLE-buffer T14 form the shipped kernel had
+// when the ablation ran (it located the ~37% staging+barrier exposure
+// that motivated R22's double-buffering); the shipped kernel has since
+// gained double-buffered LDS and the augmented-row split (R26), so
+// re-run conclusions only after porting those here.
 // VARIANT: 0=full 1=no-MFMA 2=no-A-build 3=no-staging(stale LDS)
 //          4=no-A-build (same as 2; kept for symmetry) 5=constant-B
 // Build: hipcc -O3 --offload-arch=gfx950 -o ablate_big ablate_big.hip
