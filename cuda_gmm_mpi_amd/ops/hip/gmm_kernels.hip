@@ -2053,17 +2053,26 @@ mstep_moments_big_kernel(const float* __restrict__ x,
       }
     }
     } else if (is_aug && augdim < dp) {
-      // T[d, augdim] = sum_e w_e z[augdim, e]: serial event loop from
-      // the staged planes (2-way ILP; fixed order -> deterministic).
-      // These waves sit well off the pair-waves' critical path.
+      // T[d, augdim] = sum_e w_e z[augdim, e] from the staged planes
+      // (fixed order -> deterministic). b128 row reads: scalar b16
+      // reads at the zbr=72 lane stride were 8-way bank-conflicted
+      // (36 dwords = 4 mod 32; 12.5% LDSBankConflict on the final PMC);
+      // one bf16x8 read covers 8 events at a 2-way worst case. These
+      // waves sit well off the pair-waves' critical path either way.
       const __bf16* zh = zhi + augdim * zbr;
       const __bf16* zl = zlo + augdim * zbr;
       const float* wrow = wt + cw * MBB_BK;
       float a0 = 0.0f, a1 = 0.0f;
-#pragma unroll 8
-      for (int e = 0; e < MBB_BK; e += 2) {
-        a0 = fmaf(wrow[e], (float)zh[e] + (float)zl[e], a0);
-        a1 = fmaf(wrow[e + 1], (float)zh[e + 1] + (float)zl[e + 1], a1);
+#pragma unroll
+      for (int e8 = 0; e8 < MBB_BK; e8 += 8) {
+        const bf16x8 vh = *(const bf16x8*)(zh + e8);
+        const bf16x8 vl = *(const bf16x8*)(zl + e8);
+#pragma unroll
+        for (int u = 0; u < 8; u += 2) {
+          a0 = fmaf(wrow[e8 + u], (float)vh[u] + (float)vl[u], a0);
+          a1 = fmaf(wrow[e8 + u + 1],
+                    (float)vh[u + 1] + (float)vl[u + 1], a1);
+        }
       }
       acc_aug += a0 + a1;
     }
