@@ -15,7 +15,8 @@ from cuda_gmm_mpi_amd.ops import functional as F
 from cuda_gmm_mpi_amd.utils.config import GmmConfig
 from cuda_gmm_mpi_amd.utils.synthetic import make_blobs
 
-rng = np.random.default_rng(20240914)
+rng = np.random.default_rng(int(sys.argv[1]) if len(sys.argv) > 1
+                            else 20240914)
 dev = "cuda"
 fails = 0
 
