@@ -154,3 +154,30 @@ def test_reduce_order_after_mass_elimination(rng):
         assert k == 1 and (c1, c2) == (0, 1)
     k, _, _ = reduce_order(hc)
     assert hc.N[0] == pytest.approx(n_live, rel=1e-6)
+
+
+def test_merge_identical_clusters_is_idempotent(rng):
+    """Merging a cluster with an exact copy of itself must reproduce it:
+    means and R unchanged (the cross terms vanish), N doubled."""
+    hc = make_clusters(rng, 2, 3)
+    for f in ("means", "R", "pi", "N", "avgvar", "constant"):
+        getattr(hc, f)[1] = getattr(hc, f)[0]
+    m = add_clusters(hc, 0, 1, bug_compat=False)
+    assert m.N == pytest.approx(float(2 * hc.N[0]))
+    np.testing.assert_allclose(m.means, hc.means[0], rtol=1e-6)
+    np.testing.assert_allclose(m.R, hc.R[0], rtol=1e-5, atol=1e-6)
+
+
+def test_cluster_distance_self_merge_is_zero(rng):
+    """distance(a, a-copy) = N1·c1 + N2·c2 − (N1+N2)·c12 with c12 = c1
+    when the merged cluster equals its parents — the scan can never
+    prefer a worse pair over an exact duplicate."""
+    from cuda_gmm_mpi_amd.models.merge import cluster_distance
+    hc = make_clusters(rng, 3, 3)
+    for f in ("means", "R", "pi", "N", "avgvar", "constant"):
+        getattr(hc, f)[1] = getattr(hc, f)[0]
+    dist, m = cluster_distance(hc, 0, 1, bug_compat=False)
+    assert m.N == pytest.approx(float(2 * hc.N[0]))
+    # c12 is recomputed from the merged R (fp32): allow fp noise only
+    scale = abs(float(hc.N[0] * hc.constant[0])) + 1.0
+    assert abs(dist) < 1e-4 * scale
