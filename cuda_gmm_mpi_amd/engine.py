@@ -579,6 +579,19 @@ class EmEngine:
                   if self.rank == 0 else None)
             if self.world > 1:
                 ck = self._bcast_checkpoint(ck)
+            if ck and ck["state"] is not None and ck["k"] < stop:
+                # the checkpoint's sweep position is already BELOW this
+                # run's target: it was written under a different target
+                # (checkpoints lag one merge, and best-state snapshots
+                # follow the OLD target's save rule), so the model this
+                # run asks for may not exist in it. Resuming would
+                # silently return a stale snapshot — start fresh instead.
+                if self.rank == 0:
+                    import warnings
+                    warnings.warn(
+                        f"checkpoint at K={ck['k']} is below the requested "
+                        f"target K={stop}; ignoring it and sweeping fresh")
+                ck = None
             if ck and ck["state"] is not None and ck["k"] <= k:
                 k = ck["k"]
                 st = ck["state"].to(self.device)

@@ -535,3 +535,25 @@ def test_sweep_lands_on_target_with_supported_blobs():
                     min_iters=4, max_iters=4)
     res = build_engine(data, cfg, device="cpu").sweep()
     assert res.num_clusters == 10
+
+
+def test_resume_with_raised_target_starts_fresh(tmp_path):
+    """A checkpoint whose sweep position is below the new run's target
+    was written under a DIFFERENT target (snapshots follow the old
+    target's save rule), so resuming from it would silently return a
+    stale model. The engine must warn, ignore it, and sweep fresh to
+    the requested K."""
+    import warnings
+    data, _ = make_blobs(2000, 2, 3, seed=41)
+    ckdir = str(tmp_path / "ck")
+    cfg1 = GmmConfig(num_clusters=6, target_num_clusters=2,
+                     min_iters=3, max_iters=3, checkpoint_dir=ckdir)
+    r1 = build_engine(data, cfg1, device="cpu").sweep()
+    assert r1.num_clusters == 2
+    cfg2 = GmmConfig(num_clusters=6, target_num_clusters=4,
+                     min_iters=3, max_iters=3, checkpoint_dir=ckdir)
+    with warnings.catch_warnings(record=True) as w:
+        warnings.simplefilter("always")
+        r2 = build_engine(data, cfg2, device="cpu").sweep()
+    assert r2.num_clusters == 4
+    assert any("below the requested" in str(x.message) for x in w)
