@@ -579,6 +579,18 @@ class EmEngine:
                   if self.rank == 0 else None)
             if self.world > 1:
                 ck = self._bcast_checkpoint(ck)
+            if (ck and ck["state"] is not None
+                    and ck["state"].means.shape[1] != self.d):
+                # a checkpoint for a different dataset shape: loading it
+                # would crash mid-copy with an opaque shape error
+                if self.rank == 0:
+                    import warnings
+                    warnings.warn(
+                        f"checkpoint dimensionality "
+                        f"D={int(ck['state'].means.shape[1])} does not "
+                        f"match the dataset D={self.d}; ignoring it and "
+                        f"sweeping fresh")
+                ck = None
             if ck and ck["state"] is not None and ck["k"] < stop:
                 # the checkpoint's sweep position is already BELOW this
                 # run's target: it was written under a different target

@@ -557,3 +557,21 @@ def test_resume_with_raised_target_starts_fresh(tmp_path):
         r2 = build_engine(data, cfg2, device="cpu").sweep()
     assert r2.num_clusters == 4
     assert any("below the requested" in str(x.message) for x in w)
+
+
+def test_resume_with_mismatched_dims_starts_fresh(tmp_path):
+    """A checkpoint from a different dataset shape must be ignored with
+    a warning, not crash mid-load with a shape error."""
+    import warnings
+    ckdir = str(tmp_path / "ck")
+    data2, _ = make_blobs(2400, 2, 3, seed=41)
+    cfg = GmmConfig(num_clusters=5, target_num_clusters=3,
+                    min_iters=2, max_iters=2, checkpoint_dir=ckdir)
+    build_engine(data2, cfg, device="cpu").sweep()
+    data3, _ = make_blobs(2400, 3, 3, seed=42)
+    with warnings.catch_warnings(record=True) as w:
+        warnings.simplefilter("always")
+        r = build_engine(data3, cfg, device="cpu").sweep()
+    assert r.num_clusters == 3
+    assert r.state.means.shape[1] == 3
+    assert any("dimensionality" in str(x.message) for x in w)
